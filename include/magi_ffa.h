@@ -1,0 +1,152 @@
+/* magi_ffa.h — C-ABI boundary of the MI355X-native flex-flash-attention engine.
+ *
+ * These entry points are the drop-in replacement for the reference's JIT'd FFA
+ * kernel module surface (reference: magi_attention/functional/flex_flash_attn.py:261-290
+ * `mod.fwd`, :527-554 `mod.bwd`, resolved by _flex_flash_attn_jit.py:456
+ * `get_ffa_jit_mod`). The host side (magi_attention/functional/flex_flash_attn.py
+ * in THIS repo) binds them via ctypes; INTEGRATION.md shows the binding a
+ * maintainer would add.
+ *
+ * Conventions: all device pointers; q/k/v bf16 row-major [tokens, heads, dim];
+ * ranges int32 [n,2] half-open; attn_type_map int32 (0=full, 1=causal,
+ * 2=inv_causal, 3=bi_causal; semantics = reference flex_flash_attn.py:1247-1341);
+ * stream is a hipStream_t. All functions return 0 on success, nonzero hipError_t
+ * or negative validation error otherwise.
+ */
+#ifndef MAGI_FFA_H
+#define MAGI_FFA_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef struct magi_ffa_fwd_args {
+  const void* q;            /* bf16 [total_q, hq, d] */
+  const void* k;            /* bf16 [total_k, hk, d] */
+  const void* v;            /* bf16 [total_k, hk, d] */
+  void* out;                /* f32 (accumulate/merge mode) or bf16 [total_q, hq, d] */
+  float* lse;               /* f32 [total_q, hq], caller-initialised to -inf */
+  const int32_t* q_ranges;  /* [n, 2] */
+  const int32_t* k_ranges;  /* [n, 2] */
+  const int32_t* attn_type_map; /* [n] or NULL (= all full) */
+  int32_t* locks;           /* zeroed int32 [ceil(total_q/128) * hq]; NULL iff
+                               disable_atomic_reduction */
+  int64_t n_ranges;
+  int64_t total_q;
+  int64_t total_k;
+  int32_t hq;
+  int32_t hk;
+  int32_t d;                /* head dim: 64 or 128 */
+  int32_t max_seqlen_q;     /* upper bound on max q-range length */
+  float softmax_scale;
+  float softcap;            /* 0 = disabled; else score=softcap*tanh(s*scale/softcap),
+                               softmax scale becomes softcap (reference
+                               mainloop_fwd...hpp:466-489) */
+  int32_t out_is_fp32;      /* 1: out is f32 accumulator (merge into it);
+                               0: bf16 direct store (requires disable_atomic_reduction) */
+  int32_t disable_atomic_reduction; /* 1: q_ranges disjoint, skip lock/merge */
+  int32_t cu_margin;        /* CUs left free for comm kernels (sm_margin analogue) */
+  void* stream;             /* hipStream_t */
+} magi_ffa_fwd_args;
+
+int magi_ffa_fwd(const magi_ffa_fwd_args* args);
+
+typedef struct magi_ffa_bwd_args {
+  const void* dout;         /* bf16 [total_q, hq, d] */
+  const void* q;            /* bf16 [total_q, hq, d] */
+  const void* k;            /* bf16 [total_k, hk, d] */
+  const void* v;            /* bf16 [total_k, hk, d] */
+  const void* out;          /* bf16 or f32 [total_q, hq, d] (forward output) */
+  const float* lse;         /* f32 [total_q, hq] */
+  float* dq;                /* f32 [total_q, hq, d], zeroed, atomic-accumulated */
+  float* dk;                /* f32 [total_k, hk, d], zeroed, atomic-accumulated */
+  float* dv;                /* f32 [total_k, hk, d], zeroed, atomic-accumulated */
+  float* dpsum;             /* f32 [total_q, hq] workspace = rowsum(dO*O) */
+  const int32_t* q_ranges;  /* [n, 2] */
+  const int32_t* k_ranges;  /* [n, 2] */
+  const int32_t* attn_type_map; /* [n] or NULL */
+  int64_t n_ranges;
+  int64_t total_q;
+  int64_t total_k;
+  int32_t hq;
+  int32_t hk;
+  int32_t d;
+  int32_t max_seqlen_k;     /* upper bound on max k-range length */
+  int32_t out_is_fp32;      /* dtype of `out` above */
+  float softmax_scale;
+  float softcap;
+  int32_t cu_margin;
+  void* stream;
+} magi_ffa_bwd_args;
+
+/* dPsum preprocess: dpsum[t,h] = sum_d dout[t,h,d] * out[t,h,d]
+ * (reference flash_bwd_preprocess_kernel.h:42). */
+int magi_ffa_bwd_preprocess(const magi_ffa_bwd_args* args);
+
+/* 5-matmul backward mainloop (reference flash_bwd_kernel_sm90.h:41):
+ * recompute P, dV += P^T dO, dP = dO V^T, dS = P*(dP-dpsum),
+ * dK += dS^T Q, dQ += dS K (fp32 atomics). Requires dpsum filled. */
+int magi_ffa_bwd(const magi_ffa_bwd_args* args);
+
+/* ------------------------------------------------------------------ *
+ * Range row ops (reference common/range_op/_range_*.py — Triton there,
+ * hand-written HIP here; pack/unpack engine of the group collectives). *
+ * ------------------------------------------------------------------ */
+
+typedef struct magi_range_op_args {
+  const void* input;        /* [T_in, row_elems] */
+  void* output;             /* [T_out, row_elems] */
+  const int32_t* in_ranges;   /* [n,2] rows of input  */
+  const int32_t* out_starts;  /* [n]   start row in output for each range */
+  const float* in_lse;      /* optional [T_in, h] (lse-weighted reduce) */
+  float* out_lse;           /* optional [T_out, h] */
+  int64_t n_ranges;
+  int64_t row_elems;        /* elements per row (h*d) */
+  int64_t total_rows;       /* total rows moved (sum of range lengths) */
+  int32_t elem_size;        /* bytes per element: 2 (bf16) or 4 (f32) */
+  int32_t n_heads;          /* for lse variant: row_elems = n_heads*d */
+  int32_t reduce_op;        /* 0=copy(gather/scatter), 1=sum, 2=lse-weighted */
+  void* stream;
+} magi_range_op_args;
+
+/* output[out_starts[i] + j] = input[in_ranges[i].start + j] (copy), or
+ * output rows += / lse-merge input rows (reduce). */
+int magi_range_gather(const magi_range_op_args* args);
+int magi_range_reduce(const magi_range_op_args* args);
+
+/* Fused merge of two partial (out, lse) sets (reference functional/utils.py:371
+ * correct_out_lse_kernel): out1/lse1 updated in place with out2/lse2 merged in. */
+typedef struct magi_correct_args {
+  void* out1;               /* f32 [T, h, d], in/out */
+  float* lse1;              /* f32 [T, h], in/out */
+  const void* out2;         /* f32 [T, h, d] */
+  const float* lse2;        /* f32 [T, h] */
+  int64_t total_rows;       /* T */
+  int32_t n_heads;
+  int32_t d;
+  void* stream;
+} magi_correct_args;
+
+int magi_correct_out_lse(const magi_correct_args* args);
+
+/* ------------------------------------------------------------------ *
+ * magi_attn_ext surface helpers (reference csrc/extensions/)           *
+ * ------------------------------------------------------------------ */
+
+/* GPU spin barrier (reference extensions/kernel_barrier.cu:103): counter in
+ * device memory; produce() increments from one stream, synchronize() launches
+ * a 1-thread spin kernel on another stream that waits counter >= target. */
+int magi_kernel_barrier_produce(int32_t* counter, void* stream);
+int magi_kernel_barrier_synchronize(const int32_t* counter, int32_t target,
+                                    void* stream);
+
+/* Version/identity probe so tests can verify the native library is loaded. */
+int magi_ffa_abi_version(void);
+
+#ifdef __cplusplus
+} /* extern "C" */
+#endif
+
+#endif /* MAGI_FFA_H */

@@ -1,0 +1,8 @@
+# magi_attention — MI355X-native distributed flex-flash-attention engine.
+#
+# A from-scratch CDNA4/gfx950 rebuild with the API surface of
+# SandAI-org/MagiAttention (the reference). Compute path: hand-written HIP
+# kernels behind a C-ABI (include/magi_ffa.h) + RCCL-over-xGMI collectives.
+__version__ = "0.1.0"
+
+from . import common  # noqa: F401

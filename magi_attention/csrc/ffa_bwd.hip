@@ -1,0 +1,385 @@
+// ffa_bwd.hip — MI355X-native flex-flash-attention BACKWARD (gfx950).
+//
+// Re-designed for CDNA4 (reference behaviour: flash_bwd_kernel_sm90.h:41 /
+// mainloop_bwd_sm90_tma_gmma_ws.hpp — 5-matmul recompute backward with fp32
+// atomic dQ (and dK/dV) accumulation; preprocess dPsum = rowsum(dO*O),
+// flash_bwd_preprocess_kernel.h:42).
+//
+// Structure: ONE wave (64 threads) per (k-tile of 32 rows, range, q-head).
+// For its K/V tile the wave loops over the slice's q tiles of 32 rows:
+//   S^T  = K Q^T                        (swapped MFMA, like the fwd)
+//   P^T  = exp2(t - lse*log2e)          (recompute, base-2 like the fwd)
+//   dP^T = V dO^T
+//   dS^T = P^T (dP^T - dPsum) * dscale
+//   dV  += P^T dO     (P  transposed through a small LDS tile)
+//   dK  += dS^T Q     (dS transposed through the same LDS tile)
+//   dQ  += unsafeAtomicAdd(dS K)   (dS^T -> A-fragment in-register via permlane)
+// dK/dV accumulate in AGPRs across the q loop and are atomicAdd'd once.
+
+#include <hip/hip_runtime.h>
+#include <math.h>
+
+#include "../../include/magi_ffa.h"
+
+#define BWD_BN 32  // k rows per wave
+#define BWD_BM 32  // q rows per inner tile
+
+using f32x16 = __attribute__((ext_vector_type(16))) float;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using bf16_t = __bf16;
+
+#define DEV_INLINE __device__ __forceinline__
+
+DEV_INLINE int crow(int r, int hi) { return (r & 3) + 8 * (r >> 2) + 4 * hi; }
+
+DEV_INLINE unsigned pack_bf16_pair(float lo, float hi) {
+  union { __bf16 b; unsigned short u; } a, b;
+  a.b = (__bf16)lo;
+  b.b = (__bf16)hi;
+  return ((unsigned)b.u << 16) | a.u;
+}
+
+// Rebuild an MFMA A-fragment (A[i=lo32][k=8*hi+e]) from values held in the
+// C/D layout (val[r] at row crow(r,hi), col lo32) — same cvt_pk +
+// permlane32_swap path as the forward kernel. tt selects k in [16*tt,16*tt+16).
+DEV_INLINE bf16x8 cframe_to_afrag(const float* val, int tt) {
+  unsigned c0 = pack_bf16_pair(val[8 * tt + 0], val[8 * tt + 1]);
+  unsigned c1 = pack_bf16_pair(val[8 * tt + 2], val[8 * tt + 3]);
+  unsigned c2 = pack_bf16_pair(val[8 * tt + 4], val[8 * tt + 5]);
+  unsigned c3 = pack_bf16_pair(val[8 * tt + 6], val[8 * tt + 7]);
+  {
+    auto r2 = __builtin_amdgcn_permlane32_swap(c0, c2, false, false);
+    c0 = r2[0];
+    c2 = r2[1];
+  }
+  {
+    auto r2 = __builtin_amdgcn_permlane32_swap(c1, c3, false, false);
+    c1 = r2[0];
+    c3 = r2[1];
+  }
+  union {
+    unsigned u[4];
+    bf16x8 v;
+  } cvt;
+  cvt.u[0] = c0;
+  cvt.u[1] = c1;
+  cvt.u[2] = c2;
+  cvt.u[3] = c3;
+  return cvt.v;
+}
+
+struct BwdParams {
+  const bf16_t* dout;
+  const bf16_t* q;
+  const bf16_t* k;
+  const bf16_t* v;
+  const void* out;  // bf16 or f32 per OUT_F32
+  const float* lse;
+  float* dq;
+  float* dk;
+  float* dv;
+  float* dpsum;
+  const int* q_ranges;
+  const int* k_ranges;
+  const int* attn_type_map;
+  int hq, hk, gqa;
+  float scale;
+  float softcap;
+  long long total_q, total_k;
+};
+
+// ---------------- preprocess: dpsum = rowsum(dO * O) ----------------
+template <bool OUT_F32>
+__global__ __launch_bounds__(256) void bwd_preprocess_kernel(BwdParams p, int d,
+                                                             long long n_rows) {
+  // one wave per (token, head) row
+  const long long row = (long long)blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (row >= n_rows) return;
+  const int lane = threadIdx.x & 63;
+  const int per = d / 64;  // elements per lane (1 for d=64, 2 for d=128)
+  const bf16_t* do_p = p.dout + row * d;
+  float acc = 0.f;
+  for (int e = 0; e < per; ++e) {
+    const int idx = lane * per + e;
+    const float dov = (float)do_p[idx];
+    float ov;
+    if (OUT_F32)
+      ov = ((const float*)p.out)[row * d + idx];
+    else
+      ov = (float)((const bf16_t*)p.out)[row * d + idx];
+    acc += dov * ov;
+  }
+#pragma unroll
+  for (int s = 32; s > 0; s >>= 1) acc += __shfl_xor(acc, s, 64);
+  if (lane == 0) p.dpsum[row] = acc;
+}
+
+// ---------------- mainloop ----------------
+template <int D, bool HAS_SOFTCAP>
+__global__ __launch_bounds__(64) void ffa_bwd_kernel(BwdParams p) {
+  constexpr int DF = D / 16;
+  constexpr int DT = D / 32;
+  const int ri = blockIdx.y;
+  const int h = blockIdx.z;
+  const int ks = p.k_ranges[2 * ri], ke = p.k_ranges[2 * ri + 1];
+  const int n0 = ks + blockIdx.x * BWD_BN;
+  if (n0 >= ke) return;
+  const int qs = p.q_ranges[2 * ri], qe = p.q_ranges[2 * ri + 1];
+  if (qe <= qs) return;
+  const int atype = p.attn_type_map ? p.attn_type_map[ri] : 0;
+
+  const int lane = threadIdx.x & 63;
+  const int lo32 = lane & 31;
+  const int hi = lane >> 5;
+
+  const float sl2 = HAS_SOFTCAP ? p.softcap * 1.4426950408889634f
+                                : p.scale * 1.4426950408889634f;
+  const float cap_pre = HAS_SOFTCAP ? p.scale / p.softcap : 0.f;
+  const float log2e = 1.4426950408889634f;
+
+  const int kh = h / p.gqa;
+  const size_t k_pitch = (size_t)p.hk * D;
+  const size_t q_pitch = (size_t)p.hq * D;
+
+  const int krow = n0 + lo32;           // this lane's k row (A-layout row)
+  const bool kvalid = krow < ke;
+  const int kcl = kvalid ? krow : (ke - 1);
+
+  // K/V fragments, A-layout (row lo32, d-slices by hi): 16B loads
+  bf16x8 kfA[DF], vfA[DF];
+  {
+    const bf16_t* kp = p.k + (size_t)kcl * k_pitch + (size_t)kh * D;
+    const bf16_t* vp = p.v + (size_t)kcl * k_pitch + (size_t)kh * D;
+#pragma unroll
+    for (int dd = 0; dd < DF; ++dd) {
+      kfA[dd] = *(const bf16x8*)(kp + dd * 16 + hi * 8);
+      vfA[dd] = *(const bf16x8*)(vp + dd * 16 + hi * 8);
+    }
+  }
+  // K as B-fragments for dQ += dS K: B[kk=8hi+e][j=d0+lo32] (column loads)
+  union Bf {
+    unsigned short u[8];
+    bf16x8 v;
+  };
+  bf16x8 kfB[DT][2];
+#pragma unroll
+  for (int dt = 0; dt < DT; ++dt)
+#pragma unroll
+    for (int tt = 0; tt < 2; ++tt) {
+      Bf b;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const int row = min(n0 + 16 * tt + 8 * hi + e, ke - 1);
+        b.u[e] = *(const unsigned short*)(p.k + (size_t)row * k_pitch +
+                                          (size_t)kh * D + dt * 32 + lo32);
+      }
+      kfB[dt][tt] = b.v;
+    }
+
+  // q loop bounds for this k tile
+  int q_lo = qs, q_hi = qe;
+  if (atype == 1 || atype == 3) q_lo = max(q_lo, n0 - (ke - qe));
+  if (atype == 2 || atype == 3) q_hi = min(q_hi, (n0 + BWD_BN - 1) - (ks - qs) + 1);
+
+  f32x16 acc_dk[DT], acc_dv[DT];
+#pragma unroll
+  for (int dt = 0; dt < DT; ++dt) {
+    acc_dk[dt] = (f32x16)(0.f);
+    acc_dv[dt] = (f32x16)(0.f);
+  }
+
+  // per-wave LDS transpose tile [32 k][32 q] bf16, row padded by 2 elems
+  __shared__ __bf16 p_lds[32][34];
+
+  for (int m0 = q_lo; m0 < q_hi; m0 += BWD_BM) {
+    const int qrow = m0 + lo32;
+    const bool qvalid = qrow < q_hi;
+    const int qcl = min(qrow, qe - 1);
+    const bf16_t* qp = p.q + (size_t)qcl * q_pitch + (size_t)h * D;
+    const bf16_t* dop = p.dout + (size_t)qcl * q_pitch + (size_t)h * D;
+
+    // ---- S^T = K Q^T ; dP^T = V dO^T  (both [k=crow][q=lo32]) ----
+    f32x16 s = (f32x16)(0.f), dp = (f32x16)(0.f);
+#pragma unroll
+    for (int dd = 0; dd < DF; ++dd) {
+      bf16x8 qf = *(const bf16x8*)(qp + dd * 16 + hi * 8);
+      bf16x8 dof = *(const bf16x8*)(dop + dd * 16 + hi * 8);
+      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfA[dd], qf, s, 0, 0, 0);
+      dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfA[dd], dof, dp, 0, 0, 0);
+    }
+
+    const float lse_q = qvalid ? p.lse[(size_t)qrow * p.hq + h] : INFINITY;
+    const float dpsum_q = qvalid ? p.dpsum[(size_t)qrow * p.hq + h] : 0.f;
+    const bool row_live = qvalid && lse_q != INFINITY && lse_q != -INFINITY;
+
+    // ---- P^T and dS^T (regs, C layout) ----
+    float pv[16], dsv[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int kk = n0 + crow(r, hi);
+      bool ok = row_live && kk < ke;
+      if (atype == 1 || atype == 3) ok = ok && (kk - qrow <= ke - qe);
+      if (atype == 2 || atype == 3) ok = ok && (kk - qrow >= ks - qs);
+      float sv = s[r];
+      float dscale = p.scale;
+      float t;
+      if (HAS_SOFTCAP) {
+        const float th = tanhf(sv * cap_pre);
+        t = th * sl2;
+        dscale = p.scale * (1.f - th * th);
+      } else {
+        t = sv * sl2;
+      }
+      const float pij = ok ? exp2f(t - lse_q * log2e) : 0.f;
+      pv[r] = pij;
+      dsv[r] = pij * (dp[r] - dpsum_q) * dscale;
+    }
+
+    // ---- dV += P^T dO : transpose P via LDS, dO as B-frag [q][d] ----
+    // LDS ops of one wave complete in order: no explicit wait needed between
+    // the ds_writes below and the ds_reads of the same tile.
+#pragma unroll
+    for (int r = 0; r < 16; ++r) p_lds[crow(r, hi)][lo32] = (__bf16)pv[r];
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt) {
+#pragma unroll
+      for (int tt = 0; tt < 2; ++tt) {
+        Bf b;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int row = min(m0 + 16 * tt + 8 * hi + e, qe - 1);
+          b.u[e] = *(const unsigned short*)(p.dout + (size_t)row * q_pitch +
+                                            (size_t)h * D + dt * 32 + lo32);
+        }
+        // A-frag for k-rows: lane holds A[k=lo32][q-slice tt: 8hi+e]
+        bf16x8 pa = *(const bf16x8*)(&p_lds[lo32][0] + 16 * tt + 8 * hi);
+        acc_dv[dt] =
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, b.v, acc_dv[dt], 0, 0, 0);
+      }
+    }
+
+    // ---- dK += dS^T Q : transpose dS via LDS, Q as B-frag [q][d] ----
+#pragma unroll
+    for (int r = 0; r < 16; ++r) p_lds[crow(r, hi)][lo32] = (__bf16)dsv[r];
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt) {
+#pragma unroll
+      for (int tt = 0; tt < 2; ++tt) {
+        Bf b;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int row = min(m0 + 16 * tt + 8 * hi + e, qe - 1);
+          b.u[e] = *(const unsigned short*)(p.q + (size_t)row * q_pitch +
+                                            (size_t)h * D + dt * 32 + lo32);
+        }
+        bf16x8 dsa = *(const bf16x8*)(&p_lds[lo32][0] + 16 * tt + 8 * hi);
+        acc_dk[dt] =
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, b.v, acc_dk[dt], 0, 0, 0);
+      }
+    }
+
+    // ---- dQ += dS K (atomicAdd): dS^T regs -> A-frag in-register ----
+    f32x16 acc_dq[DT];
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt) acc_dq[dt] = (f32x16)(0.f);
+#pragma unroll
+    for (int tt = 0; tt < 2; ++tt) {
+      bf16x8 dsa = cframe_to_afrag(dsv, tt);
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt)
+        acc_dq[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, kfB[dt][tt],
+                                                             acc_dq[dt], 0, 0, 0);
+    }
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int qr = m0 + crow(r, hi);
+      if (qr >= q_hi) continue;
+      float* dst = p.dq + (size_t)qr * q_pitch + (size_t)h * D;
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        const float val = acc_dq[dt][r];
+        if (val != 0.f) unsafeAtomicAdd(dst + dt * 32 + lo32, val);
+      }
+    }
+  }
+
+  // ---- write dK/dV: atomicAdd (k_ranges may overlap; GQA heads collide) ----
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int kr = n0 + crow(r, hi);
+    if (kr >= ke) continue;
+    float* dkp = p.dk + (size_t)kr * k_pitch + (size_t)kh * D;
+    float* dvp = p.dv + (size_t)kr * k_pitch + (size_t)kh * D;
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt) {
+      if (acc_dk[dt][r] != 0.f) unsafeAtomicAdd(dkp + dt * 32 + lo32, acc_dk[dt][r]);
+      if (acc_dv[dt][r] != 0.f) unsafeAtomicAdd(dvp + dt * 32 + lo32, acc_dv[dt][r]);
+    }
+  }
+}
+
+// ---------------- launchers ----------------
+extern "C" int magi_ffa_bwd_preprocess(const magi_ffa_bwd_args* a) {
+  if (!a || !a->dout || !a->out || !a->dpsum) return -1;
+  BwdParams p{};
+  p.dout = (const bf16_t*)a->dout;
+  p.out = a->out;
+  p.dpsum = a->dpsum;
+  const long long n_rows = a->total_q * a->hq;
+  if (n_rows == 0) return 0;
+  dim3 grid((unsigned)((n_rows + 3) / 4)), block(256);
+  hipStream_t s = (hipStream_t)a->stream;
+  if (a->out_is_fp32)
+    hipLaunchKernelGGL((bwd_preprocess_kernel<true>), grid, block, 0, s, p,
+                       a->d, n_rows);
+  else
+    hipLaunchKernelGGL((bwd_preprocess_kernel<false>), grid, block, 0, s, p,
+                       a->d, n_rows);
+  return (int)hipGetLastError();
+}
+
+extern "C" int magi_ffa_bwd(const magi_ffa_bwd_args* a) {
+  if (!a || !a->dout || !a->q || !a->k || !a->v || !a->lse) return -1;
+  if (a->d != 64 && a->d != 128) return -2;
+  if (a->hq % a->hk != 0) return -3;
+  if (a->n_ranges <= 0) return 0;
+
+  BwdParams p{};
+  p.dout = (const bf16_t*)a->dout;
+  p.q = (const bf16_t*)a->q;
+  p.k = (const bf16_t*)a->k;
+  p.v = (const bf16_t*)a->v;
+  p.out = a->out;
+  p.lse = a->lse;
+  p.dq = a->dq;
+  p.dk = a->dk;
+  p.dv = a->dv;
+  p.dpsum = a->dpsum;
+  p.q_ranges = a->q_ranges;
+  p.k_ranges = a->k_ranges;
+  p.attn_type_map = a->attn_type_map;
+  p.hq = a->hq;
+  p.hk = a->hk;
+  p.gqa = a->hq / a->hk;
+  p.scale = a->softmax_scale;
+  p.softcap = a->softcap;
+  p.total_q = a->total_q;
+  p.total_k = a->total_k;
+
+  const int nblocks = (a->max_seqlen_k + BWD_BN - 1) / BWD_BN;
+  dim3 grid(nblocks, (unsigned)a->n_ranges, a->hq), block(64);
+  hipStream_t s = (hipStream_t)a->stream;
+  const bool sc = a->softcap > 0.f;
+  if (a->d == 64) {
+    if (sc)
+      hipLaunchKernelGGL((ffa_bwd_kernel<64, true>), grid, block, 0, s, p);
+    else
+      hipLaunchKernelGGL((ffa_bwd_kernel<64, false>), grid, block, 0, s, p);
+  } else {
+    if (sc)
+      hipLaunchKernelGGL((ffa_bwd_kernel<128, true>), grid, block, 0, s, p);
+    else
+      hipLaunchKernelGGL((ffa_bwd_kernel<128, false>), grid, block, 0, s, p);
+  }
+  return (int)hipGetLastError();
+}

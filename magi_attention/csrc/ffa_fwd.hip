@@ -1,0 +1,421 @@
+// ffa_fwd.hip — MI355X-native flex-flash-attention FORWARD kernel (gfx950).
+//
+// Re-designed from scratch for CDNA4 (NOT a port of the reference SM90 kernel):
+//   - wave64, MFMA v_mfma_f32_32x32x16_bf16 tiles
+//   - swapped QK^T (mfma(K, Q)) so each lane owns the scores of ONE q column
+//     -> softmax row-reduce is 15 in-register fmax + one shfl_xor(32)
+//   - P -> bf16 A-fragments rebuilt in-register via cvt_pk + permlane32_swap
+//     (no LDS round trip for P)
+//   - fp32 out accumulator merged in gmem under 2-slot range locks
+//     (replaces the reference's atomicCAS range-lock epilogue,
+//      epilogue_fwd.hpp:271-424), so overlapping q_ranges and cross-launch
+//     accumulation (the CP runtime's out_acc) both work.
+//
+// Numerics mirror the reference (softmax.h:150-331): base-2 exponentials with
+// softmax_scale_log2 = scale*log2(e); lse = (m + log2(l)) * ln2 natural-log;
+// softcap: score = tanh(s*scale/softcap), softmax scale becomes softcap
+// (mainloop_fwd_sm90_tma_gmma_ws.hpp:466-489); empty rows lse=-inf, out=0.
+//
+// Mask semantics (flex_flash_attn.py:1247-1341), in GLOBAL coords for slice
+// (q_range=[qs,qe), k_range=[ks,ke)):
+//   causal      : k - q <= ke - qe   (bottom-right aligned)
+//   inv_causal  : k - q >= ks - qs   (top-left aligned)
+//   bi_causal   : both
+
+#include <hip/hip_runtime.h>
+#include <math.h>
+
+#include "../../include/magi_ffa.h"
+
+#define FFA_BM 128   // q rows per workgroup (4 waves x 32)
+#define FFA_BN 32    // k rows per inner tile
+#define LOCK_GRAN 128
+
+using f32x16 = __attribute__((ext_vector_type(16))) float;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using bf16_t = __bf16;
+
+#define DEV_INLINE __device__ __forceinline__
+
+DEV_INLINE float warp_xor32(float v) { return __shfl_xor(v, 32, 64); }
+
+DEV_INLINE unsigned pack_bf16_pair(float lo, float hi) {
+  union { __bf16 b; unsigned short u; } a, b;
+  a.b = (__bf16)lo;
+  b.b = (__bf16)hi;
+  return ((unsigned)b.u << 16) | a.u;
+}
+
+// C/D fragment row for v_mfma_f32_32x32x16_bf16: reg r, lane-half hi
+DEV_INLINE int crow(int r, int hi) { return (r & 3) + 8 * (r >> 2) + 4 * hi; }
+
+struct FwdParams {
+  const bf16_t* q;
+  const bf16_t* k;
+  const bf16_t* v;
+  float* out_f32;     // used when OUT_BF16==false
+  bf16_t* out_bf16;   // used when OUT_BF16==true
+  float* lse;
+  const int* q_ranges;
+  const int* k_ranges;
+  const int* attn_type_map;
+  int* locks;
+  int hq, hk, gqa;    // gqa = hq / hk
+  int n_lock_slots;
+  float scale;        // softmax_scale
+  float softcap;
+  long long total_q, total_k;
+};
+
+template <int D, bool HAS_SOFTCAP, bool ATOMIC, bool OUT_BF16>
+__global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
+  constexpr int DF = D / 16;    // # of 16-wide d fragments
+  constexpr int DT = D / 32;    // # of 32-wide output d tiles
+  const int ri = blockIdx.y;
+  const int h = blockIdx.z;
+  const int qs = p.q_ranges[2 * ri], qe = p.q_ranges[2 * ri + 1];
+  const int m0 = qs + blockIdx.x * FFA_BM;
+  if (m0 >= qe) return;                       // uniform across block
+  const int ks = p.k_ranges[2 * ri], ke = p.k_ranges[2 * ri + 1];
+  const int atype = p.attn_type_map ? p.attn_type_map[ri] : 0;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int lo32 = lane & 31;
+  const int hi = lane >> 5;
+
+  const int q0 = m0 + wave * 32;
+  const int qrow = q0 + lo32;
+  const bool qvalid = (qrow < qe) && (q0 < qe);
+  const int qclamp = qvalid ? qrow : (qe - 1);
+
+  // softmax scales (reference mainloop_fwd...hpp:466-489)
+  const float sl2 = HAS_SOFTCAP ? p.softcap * 1.4426950408889634f
+                                : p.scale * 1.4426950408889634f;
+  const float cap_pre = HAS_SOFTCAP ? p.scale / p.softcap : 0.f;
+
+  // per-wave k loop bounds
+  int n_lo = ks, n_hi = ke;
+  if (ke > ks && q0 < qe) {
+    const int qhiw = min(q0 + 31, qe - 1);
+    if (atype == 1 || atype == 3) n_hi = min(n_hi, qhiw + (ke - qe) + 1);
+    if (atype == 2 || atype == 3) n_lo = max(n_lo, q0 + (ks - qs));
+  } else {
+    n_hi = n_lo;  // empty
+  }
+
+  // Q fragments in registers (8 x bf16x8 for D=128)
+  bf16x8 qf[DF];
+  {
+    const bf16_t* qp = p.q + (size_t)qclamp * p.hq * D + (size_t)h * D;
+#pragma unroll
+    for (int dd = 0; dd < DF; ++dd)
+      qf[dd] = *(const bf16x8*)(qp + dd * 16 + hi * 8);
+  }
+
+  const int kh = h / p.gqa;
+  const size_t k_pitch = (size_t)p.hk * D;
+  const bf16_t* kbase = p.k + (size_t)kh * D;
+  const bf16_t* vbase = p.v + (size_t)kh * D;
+
+  float m_run = -INFINITY;
+  float l_run = 0.f;
+  f32x16 acc_o[DT];
+#pragma unroll
+  for (int dt = 0; dt < DT; ++dt) acc_o[dt] = (f32x16)(0.f);
+
+  for (int n0 = n_lo; n0 < n_hi; n0 += FFA_BN) {
+    // ---- K fragments + QK^T (swapped: A=K rows, B=Q cols) ----
+    const int krow = n0 + lo32;
+    const int kcl = min(krow, ke - 1);
+    const bf16_t* kp = kbase + (size_t)kcl * k_pitch;
+    f32x16 s = (f32x16)(0.f);
+#pragma unroll
+    for (int dd = 0; dd < DF; ++dd) {
+      bf16x8 kf = *(const bf16x8*)(kp + dd * 16 + hi * 8);
+      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[dd], s, 0, 0, 0);
+    }
+
+    // ---- mask + scale into exp2 domain ----
+    float t[16];
+    float mx = -INFINITY;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int kk = n0 + crow(r, hi);
+      bool ok = qvalid && (kk >= n_lo) && (kk < n_hi);
+      if (atype == 1 || atype == 3) ok = ok && (kk - qrow <= ke - qe);
+      if (atype == 2 || atype == 3) ok = ok && (kk - qrow >= ks - qs);
+      float sv = s[r];
+      if (HAS_SOFTCAP) sv = tanhf(sv * cap_pre);
+      t[r] = ok ? sv * sl2 : -INFINITY;
+      mx = fmaxf(mx, t[r]);
+    }
+    mx = fmaxf(mx, warp_xor32(mx));
+
+    const float m_new = fmaxf(m_run, mx);
+    const float m_use = (m_new == -INFINITY) ? 0.f : m_new;
+    const float alpha = (m_run == -INFINITY) ? 0.f : exp2f(m_run - m_use);
+    m_run = m_new;
+
+    float pr[16];
+    float psum = 0.f;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      pr[r] = exp2f(t[r] - m_use);   // exp2(-inf)=0 for masked
+      psum += pr[r];
+    }
+    l_run = l_run * alpha + (psum + warp_xor32(psum));
+
+    // ---- rescale O by alpha (per q = crow layout) ----
+    if (alpha != 1.f) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int src = crow(r, hi);
+        const float aq = __uint_as_float(
+            __builtin_amdgcn_ds_bpermute(src << 2, __float_as_uint(alpha)));
+#pragma unroll
+        for (int dt = 0; dt < DT; ++dt) acc_o[dt][r] *= aq;
+      }
+    }
+
+    // ---- P -> bf16 A-fragments (cvt_pk + permlane32_swap) ----
+    bf16x8 pa[2];
+#pragma unroll
+    for (int tt = 0; tt < 2; ++tt) {
+      unsigned c0 = pack_bf16_pair(pr[8 * tt + 0], pr[8 * tt + 1]);
+      unsigned c1 = pack_bf16_pair(pr[8 * tt + 2], pr[8 * tt + 3]);
+      unsigned c2 = pack_bf16_pair(pr[8 * tt + 4], pr[8 * tt + 5]);
+      unsigned c3 = pack_bf16_pair(pr[8 * tt + 6], pr[8 * tt + 7]);
+      {
+        auto r2 = __builtin_amdgcn_permlane32_swap(c0, c2, false, false);
+        c0 = r2[0];
+        c2 = r2[1];
+      }
+      {
+        auto r2 = __builtin_amdgcn_permlane32_swap(c1, c3, false, false);
+        c1 = r2[0];
+        c3 = r2[1];
+      }
+      union {
+        unsigned u[4];
+        bf16x8 v;
+      } cvt;
+      cvt.u[0] = c0;
+      cvt.u[1] = c1;
+      cvt.u[2] = c2;
+      cvt.u[3] = c3;
+      pa[tt] = cvt.v;
+    }
+
+    // ---- PV: O[32q][32d] += P^T V per d-tile ----
+#pragma unroll
+    for (int tt = 0; tt < 2; ++tt) {
+      const int vr = n0 + 16 * tt + 8 * hi;
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        // B-fragment: lane reads V[vr + e][dt*32 + lo32], e = 0..7
+        union {
+          unsigned short u[8];
+          bf16x8 v;
+        } bv;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int row = min(vr + e, ke - 1);
+          bv.u[e] = *(const unsigned short*)(vbase + (size_t)row * k_pitch +
+                                             dt * 32 + lo32);
+        }
+        acc_o[dt] =
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[tt], bv.v, acc_o[dt], 0, 0, 0);
+      }
+    }
+  }
+
+  // ======================= epilogue =======================
+  const float lse_new =
+      (l_run > 0.f) ? (m_run + __log2f(l_run)) * 0.6931471805599453f : -INFINITY;
+  const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
+
+  const size_t out_row_pitch = (size_t)p.hq * D;
+
+  if (!ATOMIC) {
+    // direct store: q_ranges guaranteed disjoint; rows with l==0 keep initial
+    if (l_run > 0.f && qvalid && hi == 0)
+      p.lse[(size_t)qrow * p.hq + h] = lse_new;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int src = crow(r, hi);
+      const float ilq = __uint_as_float(
+          __builtin_amdgcn_ds_bpermute(src << 2, __float_as_uint(inv_l)));
+      const int qr = q0 + src;
+      const bool valid = (qr < qe) && ilq > 0.f;
+      if (!valid) continue;
+      const size_t base = (size_t)qr * out_row_pitch + (size_t)h * D;
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        const float val = acc_o[dt][r] * ilq;
+        if (OUT_BF16)
+          p.out_bf16[base + dt * 32 + lo32] = (bf16_t)val;
+        else
+          p.out_f32[base + dt * 32 + lo32] = val;
+      }
+    }
+    return;
+  }
+
+  // ---- lock-guarded read-merge-write (2-slot range locks) ----
+  const int row_last = min(m0 + FFA_BM, qe) - 1;
+  const int s0 = m0 / LOCK_GRAN, s1 = row_last / LOCK_GRAN;
+  int* lock0 = p.locks + (size_t)s0 * p.hq + h;
+  int* lock1 = p.locks + (size_t)s1 * p.hq + h;
+  if (threadIdx.x == 0) {
+    int expected = 0;
+    while (!__hip_atomic_compare_exchange_strong(
+        lock0, &expected, 1, __ATOMIC_ACQUIRE, __ATOMIC_RELAXED,
+        __HIP_MEMORY_SCOPE_AGENT))
+      expected = 0;
+    if (s1 != s0) {
+      expected = 0;
+      while (!__hip_atomic_compare_exchange_strong(
+          lock1, &expected, 1, __ATOMIC_ACQUIRE, __ATOMIC_RELAXED,
+          __HIP_MEMORY_SCOPE_AGENT))
+        expected = 0;
+    }
+  }
+  __syncthreads();
+
+  // merged lse per lane (q = lo32 layout)
+  float lse_prev = -INFINITY;
+  if (qvalid) lse_prev = p.lse[(size_t)qrow * p.hq + h];
+  float lse_m;
+  {
+    const float a = fmaxf(lse_prev, lse_new);
+    const float b = fminf(lse_prev, lse_new);
+    lse_m = (a == -INFINITY) ? -INFINITY : a + log1pf(expf(b - a));
+  }
+  const float w_prev = (lse_prev == -INFINITY) ? 0.f : expf(lse_prev - lse_m);
+  // fold 1/l into the new weight
+  const float w_new = (lse_new == -INFINITY) ? 0.f : expf(lse_new - lse_m) * inv_l;
+
+  if (qvalid && lse_m != -INFINITY && hi == 0)
+    p.lse[(size_t)qrow * p.hq + h] = lse_m;
+
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int src = crow(r, hi);
+    const float wp = __uint_as_float(
+        __builtin_amdgcn_ds_bpermute(src << 2, __float_as_uint(w_prev)));
+    const float wn = __uint_as_float(
+        __builtin_amdgcn_ds_bpermute(src << 2, __float_as_uint(w_new)));
+    const int qr = q0 + src;
+    if (qr >= qe || (wp == 0.f && wn == 0.f)) continue;
+    const size_t base = (size_t)qr * out_row_pitch + (size_t)h * D;
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt) {
+      float* ptr = p.out_f32 + base + dt * 32 + lo32;
+      // first writer (wp==0) must not read: `out` may be uninitialised
+      const float prev = (wp > 0.f) ? wp * (*ptr) : 0.f;
+      *ptr = prev + wn * acc_o[dt][r];
+    }
+  }
+
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    if (s1 != s0)
+      __hip_atomic_store(lock1, 0, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
+    __hip_atomic_store(lock0, 0, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
+  }
+}
+
+// ------------------------------------------------------------------
+// MFMA layout probe (GPU self-test; see tests/test_ffa_gpu.py)
+// ------------------------------------------------------------------
+__global__ void probe_mfma_kernel(const bf16_t* a, const bf16_t* b, float* d) {
+  const int lane = threadIdx.x & 63;
+  const int lo32 = lane & 31, hi = lane >> 5;
+  bf16x8 af = *(const bf16x8*)(a + lo32 * 16 + hi * 8);  // A[32][16] row-major
+  bf16x8 bf;
+  union {
+    unsigned short u[8];
+    bf16x8 v;
+  } bu;
+  for (int e = 0; e < 8; ++e)
+    bu.u[e] = *(const unsigned short*)(b + (hi * 8 + e) * 32 + lo32);  // B[16][32]
+  bf = bu.v;
+  f32x16 acc = (f32x16)(0.f);
+  acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, acc, 0, 0, 0);
+  for (int r = 0; r < 16; ++r) d[(size_t)crow(r, hi) * 32 + lo32] = acc[r];
+}
+
+extern "C" int magi_probe_mfma(const void* a, const void* b, void* d,
+                               void* stream) {
+  hipLaunchKernelGGL(probe_mfma_kernel, dim3(1), dim3(64), 0,
+                     (hipStream_t)stream, (const bf16_t*)a, (const bf16_t*)b,
+                     (float*)d);
+  return (int)hipGetLastError();
+}
+
+// ------------------------------------------------------------------
+// launcher
+// ------------------------------------------------------------------
+template <int D>
+static int launch_fwd_d(const magi_ffa_fwd_args* a, const FwdParams& p,
+                        dim3 grid, dim3 block, hipStream_t stream) {
+  const bool sc = a->softcap > 0.f;
+  const bool atomic = !a->disable_atomic_reduction;
+  const bool obf16 = !a->out_is_fp32;
+  if (atomic && obf16) return -10;  // atomic merge requires fp32 out
+#define LAUNCH(SC, AT, OB)                                              \
+  hipLaunchKernelGGL((ffa_fwd_kernel<D, SC, AT, OB>), grid, block, 0,   \
+                     stream, p)
+  if (atomic) {
+    if (sc) LAUNCH(true, true, false);
+    else LAUNCH(false, true, false);
+  } else if (obf16) {
+    if (sc) LAUNCH(true, false, true);
+    else LAUNCH(false, false, true);
+  } else {
+    if (sc) LAUNCH(true, false, false);
+    else LAUNCH(false, false, false);
+  }
+#undef LAUNCH
+  return (int)hipGetLastError();
+}
+
+extern "C" int magi_ffa_fwd(const magi_ffa_fwd_args* a) {
+  if (!a || !a->q || !a->k || !a->v || !a->out || !a->lse) return -1;
+  if (a->d != 64 && a->d != 128) return -2;
+  if (a->hq % a->hk != 0) return -3;
+  if (a->n_ranges <= 0) return 0;
+  if (!a->disable_atomic_reduction && !a->locks) return -4;
+
+  FwdParams p;
+  p.q = (const bf16_t*)a->q;
+  p.k = (const bf16_t*)a->k;
+  p.v = (const bf16_t*)a->v;
+  p.out_f32 = (float*)a->out;
+  p.out_bf16 = (bf16_t*)a->out;
+  p.lse = a->lse;
+  p.q_ranges = a->q_ranges;
+  p.k_ranges = a->k_ranges;
+  p.attn_type_map = a->attn_type_map;
+  p.locks = a->locks;
+  p.hq = a->hq;
+  p.hk = a->hk;
+  p.gqa = a->hq / a->hk;
+  p.n_lock_slots = (int)((a->total_q + LOCK_GRAN - 1) / LOCK_GRAN);
+  p.scale = a->softmax_scale;
+  p.softcap = a->softcap;
+  p.total_q = a->total_q;
+  p.total_k = a->total_k;
+
+  const int mblocks = (a->max_seqlen_q + FFA_BM - 1) / FFA_BM;
+  dim3 grid(mblocks, (unsigned)a->n_ranges, a->hq);
+  dim3 block(256);
+  hipStream_t stream = (hipStream_t)a->stream;
+  if (a->d == 64) return launch_fwd_d<64>(a, p, grid, block, stream);
+  return launch_fwd_d<128>(a, p, grid, block, stream);
+}
+
+extern "C" int magi_ffa_abi_version(void) { return 1; }

@@ -1,0 +1,278 @@
+# Reference surface: magi_attention/functional/flex_flash_attn.py
+# (_flex_flash_attn_forward:334, _flex_flash_attn_backward:594,
+#  FlexFlashAttnFunc:699, flex_flash_attn_func:1066).
+# MI355X-native rebuild: the JIT'd CUDA module is replaced by the C-ABI HIP
+# library (include/magi_ffa.h) bound via ctypes in magi_attention/_ffa_lib.py.
+from __future__ import annotations
+
+import math
+from typing import Optional, Tuple
+
+import torch
+
+from .. import _ffa_lib
+from .._ffa_lib import (
+    MagiFfaBwdArgs,
+    MagiFfaFwdArgs,
+    check,
+    current_stream_ptr,
+    ptr,
+)
+from ..common.forward_meta import AttnForwardMeta
+
+LOCK_GRAN = 128
+_lock_cache: dict[tuple[int, int], torch.Tensor] = {}
+
+
+def maybe_contiguous(x: Optional[torch.Tensor]) -> Optional[torch.Tensor]:
+    return x.contiguous() if x is not None and not x.is_contiguous() else x
+
+
+def _get_locks(total_q: int, hq: int, device: torch.device) -> torch.Tensor:
+    """Lock words for the fwd merge epilogue. Kernel always releases to 0, so
+    the buffer can be cached and reused without re-zeroing."""
+    slots = (total_q + LOCK_GRAN - 1) // LOCK_GRAN
+    key = (device.index or 0, 0)
+    buf = _lock_cache.get(key)
+    need = slots * hq
+    if buf is None or buf.numel() < need:
+        buf = torch.zeros(max(need, 1 << 16), dtype=torch.int32, device=device)
+        _lock_cache[key] = buf
+    return buf
+
+
+def _max_seqlen_of(ranges: torch.Tensor) -> int:
+    if ranges.numel() == 0:
+        return 0
+    return int((ranges[:, 1] - ranges[:, 0]).amax().item())
+
+
+def _flex_flash_attn_forward(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    sink: Optional[torch.Tensor],
+    sink_layout: str,
+    out: Optional[torch.Tensor],
+    lse: Optional[torch.Tensor],
+    q_ranges: torch.Tensor,
+    k_ranges: torch.Tensor,
+    attn_type_map: Optional[torch.Tensor],
+    softmax_scale: float,
+    softcap: float,
+    out_type: Optional[torch.dtype],
+    disable_fwd_atomic_reduction: bool,
+    deterministic: bool,
+    sm_margin: int,
+    max_seqlen_q: Optional[int] = None,
+    **_unused,
+) -> tuple[torch.Tensor, AttnForwardMeta]:
+    assert sink is None, "sink support lands in a later round"
+    assert q.dtype == torch.bfloat16, "bf16 compute only (fp8 path later)"
+    q, k, v, q_ranges, k_ranges = [
+        maybe_contiguous(x) for x in (q, k, v, q_ranges, k_ranges)
+    ]
+    attn_type_map = maybe_contiguous(attn_type_map)
+
+    tq, hq, d = q.shape
+    tk, hk, _ = k.shape
+
+    if out is None:
+        dtype = out_type or (q.dtype if disable_fwd_atomic_reduction else torch.float32)
+        out = torch.zeros(tq, hq, d, dtype=dtype, device=q.device)
+    if lse is None:
+        lse = torch.full((tq, hq), float("-inf"), dtype=torch.float32, device=q.device)
+
+    out_is_fp32 = out.dtype == torch.float32
+    if not disable_fwd_atomic_reduction:
+        assert out_is_fp32, "atomic (merge) forward requires an fp32 out accumulator"
+        locks = _get_locks(tq, hq, q.device)
+    else:
+        assert out.dtype in (torch.float32, torch.bfloat16)
+        locks = None
+
+    if max_seqlen_q is None:
+        max_seqlen_q = _max_seqlen_of(q_ranges)
+
+    args = MagiFfaFwdArgs(
+        q=ptr(q), k=ptr(k), v=ptr(v), out=ptr(out), lse=ptr(lse),
+        q_ranges=ptr(q_ranges), k_ranges=ptr(k_ranges),
+        attn_type_map=ptr(attn_type_map), locks=ptr(locks),
+        n_ranges=q_ranges.shape[0], total_q=tq, total_k=tk,
+        hq=hq, hk=hk, d=d, max_seqlen_q=max_seqlen_q,
+        softmax_scale=softmax_scale, softcap=softcap,
+        out_is_fp32=int(out_is_fp32),
+        disable_atomic_reduction=int(disable_fwd_atomic_reduction),
+        cu_margin=sm_margin, stream=current_stream_ptr(),
+    )
+    check(_ffa_lib.lib().magi_ffa_fwd(args), "magi_ffa_fwd")
+    return out, AttnForwardMeta(lse=lse, max_logits=None)
+
+
+def _flex_flash_attn_backward(
+    dout: torch.Tensor,
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    sink: Optional[torch.Tensor],
+    sink_layout: str,
+    out: torch.Tensor,
+    lse: torch.Tensor,
+    dq: Optional[torch.Tensor],
+    dk: Optional[torch.Tensor],
+    dv: Optional[torch.Tensor],
+    dsink: Optional[torch.Tensor],
+    q_ranges: torch.Tensor,
+    k_ranges: torch.Tensor,
+    attn_type_map: Optional[torch.Tensor],
+    softmax_scale: float,
+    softcap: float,
+    dq_type: Optional[torch.dtype],
+    dk_type: Optional[torch.dtype],
+    dv_type: Optional[torch.dtype],
+    disable_bwd_dkv_atomic_reduction: bool,
+    deterministic: bool,
+    sm_margin: int,
+    max_seqlen_k: Optional[int] = None,
+    **_unused,
+) -> tuple[torch.Tensor, torch.Tensor, torch.Tensor, Optional[torch.Tensor]]:
+    assert sink is None, "sink support lands in a later round"
+    dout, q, k, v, out, q_ranges, k_ranges = [
+        maybe_contiguous(x) for x in (dout, q, k, v, out, q_ranges, k_ranges)
+    ]
+    attn_type_map = maybe_contiguous(attn_type_map)
+    tq, hq, d = q.shape
+    tk, hk, _ = k.shape
+
+    # fp32 accumulators, atomically reduced by the kernel
+    dq = torch.zeros_like(q, dtype=torch.float32) if dq is None else dq
+    dk = torch.zeros_like(k, dtype=torch.float32) if dk is None else dk
+    dv = torch.zeros_like(v, dtype=torch.float32) if dv is None else dv
+    assert dq.dtype == dk.dtype == dv.dtype == torch.float32
+    dpsum = torch.empty(tq, hq, dtype=torch.float32, device=q.device)
+
+    if max_seqlen_k is None:
+        max_seqlen_k = _max_seqlen_of(k_ranges)
+
+    args = MagiFfaBwdArgs(
+        dout=ptr(dout), q=ptr(q), k=ptr(k), v=ptr(v), out=ptr(out),
+        lse=ptr(lse), dq=ptr(dq), dk=ptr(dk), dv=ptr(dv), dpsum=ptr(dpsum),
+        q_ranges=ptr(q_ranges), k_ranges=ptr(k_ranges),
+        attn_type_map=ptr(attn_type_map),
+        n_ranges=q_ranges.shape[0], total_q=tq, total_k=tk,
+        hq=hq, hk=hk, d=d, max_seqlen_k=max_seqlen_k,
+        out_is_fp32=int(out.dtype == torch.float32),
+        softmax_scale=softmax_scale, softcap=softcap,
+        cu_margin=sm_margin, stream=current_stream_ptr(),
+    )
+    lib = _ffa_lib.lib()
+    check(lib.magi_ffa_bwd_preprocess(args), "magi_ffa_bwd_preprocess")
+    check(lib.magi_ffa_bwd(args), "magi_ffa_bwd")
+    return dq, dk, dv, None
+
+
+class FlexFlashAttnFunc(torch.autograd.Function):
+    @staticmethod
+    def forward(
+        ctx,
+        q, k, v, sink, sink_layout,
+        q_ranges, k_ranges, attn_type_map,
+        softmax_scale=None, softcap=0.0, deterministic=False, sm_margin=0,
+        disable_fwd_atomic_reduction=False,
+        disable_bwd_dkv_atomic_reduction=False,
+        ref_block_size=None, max_seqlen_q=None,
+        auto_range_merge=False, swap_ab=False, pack_gqa=False, cat_gqa=False,
+        sparse_load=False, index_attn=False, swap_bwd_qk_loop=False,
+        return_max_logits=False, index_attn_indices_2d=None,
+        index_attn_max_topk=0,
+    ):
+        softmax_scale = (
+            q.shape[-1] ** (-0.5) if softmax_scale is None else softmax_scale
+        )
+        assert q_ranges is not None and k_ranges is not None
+        assert q_ranges.size(0) == k_ranges.size(0)
+        assert not return_max_logits, "max_logits support lands in a later round"
+
+        out, meta = _flex_flash_attn_forward(
+            q=q, k=k, v=v, sink=sink, sink_layout=sink_layout,
+            out=None, lse=None,
+            q_ranges=q_ranges, k_ranges=k_ranges, attn_type_map=attn_type_map,
+            softmax_scale=softmax_scale, softcap=softcap, out_type=None,
+            disable_fwd_atomic_reduction=disable_fwd_atomic_reduction,
+            deterministic=deterministic, sm_margin=sm_margin,
+            max_seqlen_q=max_seqlen_q,
+        )
+        lse = meta.lse
+        if out.dtype != q.dtype:
+            out = out.to(q.dtype)
+        ctx.save_for_backward(q, k, v, out, lse, q_ranges, k_ranges, attn_type_map)
+        ctx.softmax_scale = softmax_scale
+        ctx.softcap = softcap
+        ctx.deterministic = deterministic
+        ctx.sm_margin = sm_margin
+        ctx.sink_layout = sink_layout
+        return out, lse, None
+
+    @staticmethod
+    def backward(ctx, dout, *_):
+        q, k, v, out, lse, q_ranges, k_ranges, attn_type_map = ctx.saved_tensors
+        dq, dk, dv, _ = _flex_flash_attn_backward(
+            dout=dout, q=q, k=k, v=v, sink=None, sink_layout=ctx.sink_layout,
+            out=out, lse=lse, dq=None, dk=None, dv=None, dsink=None,
+            q_ranges=q_ranges, k_ranges=k_ranges, attn_type_map=attn_type_map,
+            softmax_scale=ctx.softmax_scale, softcap=ctx.softcap,
+            dq_type=None, dk_type=None, dv_type=None,
+            disable_bwd_dkv_atomic_reduction=False,
+            deterministic=ctx.deterministic, sm_margin=ctx.sm_margin,
+        )
+        dq = dq.to(q.dtype)
+        dk = dk.to(k.dtype)
+        dv = dv.to(v.dtype)
+        return (dq, dk, dv) + (None,) * 24
+
+
+def flex_flash_attn_func(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    q_ranges: Optional[torch.Tensor] = None,
+    k_ranges: Optional[torch.Tensor] = None,
+    attn_type_map: Optional[torch.Tensor] = None,
+    *,
+    index_attn_indices: Optional[torch.Tensor] = None,
+    q_block_size: int = 1,
+    k_block_size: int = 1,
+    sink: Optional[torch.Tensor] = None,
+    sink_layout: str = "sh",
+    softmax_scale: Optional[float] = None,
+    softcap: float = 0.0,
+    deterministic: bool = False,
+    sm_margin: int = 0,
+    disable_fwd_atomic_reduction: bool = False,
+    disable_bwd_dkv_atomic_reduction: bool = False,
+    ref_block_size: Optional[Tuple[int, int]] = None,
+    max_seqlen_q: Optional[int] = None,
+    auto_range_merge: bool = False,
+    swap_ab: bool = False,
+    pack_gqa: bool = False,
+    cat_gqa: bool = False,
+    sparse_load: bool = False,
+    index_attn: bool = False,
+    swap_bwd_qk_loop: bool = False,
+    return_max_logits: bool = False,
+) -> tuple[torch.Tensor, AttnForwardMeta]:
+    """Single-GPU flex-flash-attention (drop-in for the reference
+    flex_flash_attn_func, flex_flash_attn.py:1066; full mask semantics in the
+    reference docstring :1247-1341). Returns (out, AttnForwardMeta(lse=...))."""
+    assert index_attn_indices is None and not index_attn, (
+        "index_attn lands in a later round"
+    )
+    out, lse, max_logits = FlexFlashAttnFunc.apply(
+        q, k, v, sink, sink_layout, q_ranges, k_ranges, attn_type_map,
+        softmax_scale, softcap, deterministic, sm_margin,
+        disable_fwd_atomic_reduction, disable_bwd_dkv_atomic_reduction,
+        ref_block_size, max_seqlen_q, auto_range_merge, swap_ab, pack_gqa,
+        cat_gqa, sparse_load, index_attn, swap_bwd_qk_loop, return_max_logits,
+        None, 0,
+    )
+    return out, AttnForwardMeta(lse=lse, max_logits=max_logits)
