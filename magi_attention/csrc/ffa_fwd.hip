@@ -348,6 +348,26 @@ __global__ void probe_mfma_kernel(const bf16_t* a, const bf16_t* b, float* d) {
   for (int r = 0; r < 16; ++r) d[(size_t)crow(r, hi) * 32 + lo32] = acc[r];
 }
 
+// probe permlane32_swap + ds_bpermute semantics: in[64] u32 per lane.
+// out0/out1 = the two results of permlane32_swap(in, in2) where in2 = in+1000;
+// out2 = ds_bpermute(addr = (lane%32)<<2, in).
+__global__ void probe_lane_kernel(const unsigned* in, unsigned* out) {
+  const int lane = threadIdx.x & 63;
+  unsigned a = in[lane];
+  unsigned b = in[lane] + 1000u;
+  auto r2 = __builtin_amdgcn_permlane32_swap(a, b, false, false);
+  out[lane] = r2[0];
+  out[64 + lane] = r2[1];
+  out[128 + lane] =
+      __builtin_amdgcn_ds_bpermute((lane & 31) << 2, in[lane]);
+}
+
+extern "C" int magi_probe_lane(const void* in, void* out, void* stream) {
+  hipLaunchKernelGGL(probe_lane_kernel, dim3(1), dim3(64), 0,
+                     (hipStream_t)stream, (const unsigned*)in, (unsigned*)out);
+  return (int)hipGetLastError();
+}
+
 extern "C" int magi_probe_mfma(const void* a, const void* b, void* d,
                                void* stream) {
   hipLaunchKernelGGL(probe_mfma_kernel, dim3(1), dim3(64), 0,
