@@ -167,7 +167,9 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
     l_run = l_run * alpha + (psum + warp_xor32(psum));
 
     // ---- rescale O by alpha (per q = crow layout) ----
-    if (alpha != 1.f) {
+    // NOTE: wave-uniform condition — ds_bpermute reads other lanes' registers,
+    // so every lane must be active whenever any lane needs the rescale.
+    if (__any(alpha != 1.f)) {
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int src = crow(r, hi);
