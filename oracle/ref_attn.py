@@ -76,6 +76,7 @@ def ref_attn(
     softmax_scale: float | None = None,
     softcap: float = 0.0,
     high_precision: bool = True,
+    p_dtype: torch.dtype | None = None,
 ) -> tuple[torch.Tensor, torch.Tensor]:
     """Explicit-softmax attention; returns (out [tq,hq,d] in q.dtype,
     lse [tq,hq] fp32, natural log). GQA: hq must be a multiple of hk.
@@ -103,6 +104,9 @@ def ref_attn(
     lse = torch.logsumexp(s, dim=-1)  # [hq, tq]; -inf for empty rows
     p = torch.exp(s - lse.unsqueeze(-1))
     p = torch.nan_to_num(p, nan=0.0)  # empty rows: -inf - -inf = nan -> 0
+    if p_dtype is not None:
+        # emulate a kernel that quantises P before the PV matmul (bf16 MFMA)
+        p = p.to(p_dtype).to(p.dtype)
     out = torch.matmul(p, vf)  # [hq, tq, d]
     del neg_inf
     return (
@@ -113,7 +117,7 @@ def ref_attn(
 
 def ref_attn_with_grads(
     q, k, v, mask, dout, softmax_scale=None, softcap: float = 0.0,
-    high_precision: bool = True,
+    high_precision: bool = True, p_dtype=None,
 ):
     """Forward + backward through autograd. Returns (out, lse, dq, dk, dv)."""
     q_ = q.detach().clone().requires_grad_(True)
@@ -121,7 +125,7 @@ def ref_attn_with_grads(
     v_ = v.detach().clone().requires_grad_(True)
     out, lse = ref_attn(
         q_, k_, v_, mask, softmax_scale=softmax_scale, softcap=softcap,
-        high_precision=high_precision,
+        high_precision=high_precision, p_dtype=p_dtype,
     )
     out.backward(dout)
     return out.detach(), lse.detach(), q_.grad, k_.grad, v_.grad

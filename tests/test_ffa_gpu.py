@@ -93,7 +93,7 @@ def test_ffa_fwd_parity(case):
     qc, kc, vc = q.cpu(), k.cpu(), v.cpu()
     ref_o_hi, ref_lse = ref_attn(qc, kc, vc, mask, high_precision=True)
     ref_o_lo, _ = ref_attn(
-        qc.bfloat16(), kc.bfloat16(), vc.bfloat16(), mask, high_precision=False
+        qc, kc, vc, mask, high_precision=False, p_dtype=torch.bfloat16
     )
     assert_close_to_ref(out.cpu().float(), ref_o_hi.float(), ref_o_lo.float(),
                         f"{name}:out")
@@ -127,7 +127,7 @@ def test_ffa_fwd_bwd_parity(case):
     qc, kc, vc, doc = [t.detach().cpu() for t in (q, k, v, dout)]
     o_hi, _, dq_hi, dk_hi, dv_hi = ref_attn_with_grads(qc, kc, vc, mask, doc)
     o_lo, _, dq_lo, dk_lo, dv_lo = ref_attn_with_grads(
-        qc, kc, vc, mask, doc, high_precision=False
+        qc, kc, vc, mask, doc, high_precision=False, p_dtype=torch.bfloat16
     )
     assert_close_to_ref(q.grad.cpu().float(), dq_hi.float(), dq_lo.float(), f"{name}:dq")
     assert_close_to_ref(k.grad.cpu().float(), dk_hi.float(), dk_lo.float(), f"{name}:dk")
@@ -149,8 +149,8 @@ def test_fwd_softcap():
     mask = make_attn_mask(tq, tk, [[0, tq]], [[0, tk]], [1])
     o_hi, _ = ref_attn(q.cpu(), k.cpu(), v.cpu(), mask, softcap=20.0)
     o_lo, _ = ref_attn(
-        q.cpu().bfloat16(), k.cpu().bfloat16(), v.cpu().bfloat16(), mask,
-        softcap=20.0, high_precision=False,
+        q.cpu(), k.cpu(), v.cpu(), mask,
+        softcap=20.0, high_precision=False, p_dtype=torch.bfloat16,
     )
     assert_close_to_ref(out.cpu().float(), o_hi.float(), o_lo.float(), "softcap:out")
 
@@ -181,8 +181,8 @@ def test_cross_launch_accumulation():
     torch.cuda.synchronize()
     mask = make_attn_mask(tq, tk, [[0, tq]], [[0, tk]], [0])
     o_hi, lse_hi = ref_attn(q.cpu(), k.cpu(), v.cpu(), mask)
-    o_lo, _ = ref_attn(q.cpu().bfloat16(), k.cpu().bfloat16(), v.cpu().bfloat16(),
-                       mask, high_precision=False)
+    o_lo, _ = ref_attn(q.cpu(), k.cpu(), v.cpu(), mask, high_precision=False,
+                       p_dtype=torch.bfloat16)
     assert_close_to_ref(out.cpu(), o_hi.float(), o_lo.float(), "acc:out")
     torch.testing.assert_close(lse.cpu(), lse_hi, atol=5e-3, rtol=1e-3)
 
