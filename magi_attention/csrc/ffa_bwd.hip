@@ -205,7 +205,7 @@ __global__ __launch_bounds__(64) void ffa_bwd_kernel(BwdParams p) {
       bf16x8 qf = *(const bf16x8*)(qp + dd * 16 + hi * 8);
       bf16x8 dof = *(const bf16x8*)(dop + dd * 16 + hi * 8);
       s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfA[dd], qf, s, 0, 0, 0);
-      dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfA[dd], dof, dp, 0, 0, 0);
+      dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfA[dd], dof, dp, 0, 0, 0);
     }
 
     const float lse_q = qvalid ? p.lse[(size_t)qrow * p.hq + h] : INFINITY;
