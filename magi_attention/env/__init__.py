@@ -1,0 +1,98 @@
+# Env-flag system (reference magi_attention/env/{general,comm}.py — names kept
+# verbatim; flags are snapshotted into DistAttnRuntimeKey so flips produce new
+# plans, reference dist_attn_runtime_mgr.py:79-87).
+from __future__ import annotations
+
+import os
+from typing import Any
+
+
+def _get(name: str, default: str) -> str:
+    return os.environ.get(name, default)
+
+
+def _get_bool(name: str, default: bool = False) -> bool:
+    return _get(name, "1" if default else "0") in ("1", "true", "TRUE", "True")
+
+
+def _get_int(name: str, default: int) -> int:
+    return int(_get(name, str(default)))
+
+
+# ---- general (reference env/general.py) ----
+def is_sanity_check_enable() -> bool:
+    return _get_bool("MAGI_ATTENTION_SANITY_CHECK")
+
+
+def is_deterministic_mode_enable() -> bool:
+    return _get_bool("MAGI_ATTENTION_DETERMINISTIC_MODE")
+
+
+def is_cpp_backend_enable() -> bool:
+    return _get_bool("MAGI_ATTENTION_CPP_BACKEND")
+
+
+def is_auto_range_merge_enable() -> bool:
+    return _get_bool("MAGI_ATTENTION_AUTO_RANGE_MERGE")
+
+
+def kernel_backend() -> str:
+    return _get("MAGI_ATTENTION_KERNEL_BACKEND", "ffa")
+
+
+def min_chunks_per_rank() -> int:
+    return _get_int("MAGI_ATTENTION_MIN_CHUNKS_PER_RANK", 2)
+
+
+def dist_attn_runtime_dict_size() -> int:
+    return _get_int("MAGI_ATTENTION_DIST_ATTN_RUNTIME_DICT_SIZE", 100)
+
+
+def log_level() -> str:
+    return _get("MAGI_ATTENTION_LOG_LEVEL", "WARNING")
+
+
+# ---- comm (reference env/comm.py) ----
+def is_hierarchical_comm_enable() -> bool:
+    return _get_bool("MAGI_ATTENTION_HIERARCHICAL_COMM")
+
+
+def ffa_forward_sm_margin() -> int:
+    return _get_int("MAGI_ATTENTION_FFA_FORWARD_SM_MARGIN", 8)
+
+
+def ffa_backward_sm_margin() -> int:
+    return _get_int("MAGI_ATTENTION_FFA_BACKWARD_SM_MARGIN", 8)
+
+
+def is_qo_comm_enable() -> bool:
+    return _get_bool("MAGI_ATTENTION_QO_COMM")
+
+
+def is_forward_high_precision_reduce_enable() -> bool:
+    return _get_bool("MAGI_ATTENTION_FORWARD_HIGH_PRECISION_REDUCE")
+
+
+def is_backward_high_precision_reduce_enable() -> bool:
+    return _get_bool("MAGI_ATTENTION_BACKWARD_HIGH_PRECISION_REDUCE")
+
+
+def is_native_grpcoll_enable() -> bool:
+    return _get_bool("MAGI_ATTENTION_NATIVE_GRPCOLL")
+
+
+def snapshot() -> tuple[tuple[str, Any], ...]:
+    """Frozen snapshot of every flag that affects plan construction."""
+    return (
+        ("deterministic", is_deterministic_mode_enable()),
+        ("kernel_backend", kernel_backend()),
+        ("auto_range_merge", is_auto_range_merge_enable()),
+        ("hierarchical_comm", is_hierarchical_comm_enable()),
+        ("qo_comm", is_qo_comm_enable()),
+        ("fwd_hp_reduce", is_forward_high_precision_reduce_enable()),
+        ("bwd_hp_reduce", is_backward_high_precision_reduce_enable()),
+        ("fwd_sm_margin", ffa_forward_sm_margin()),
+        ("bwd_sm_margin", ffa_backward_sm_margin()),
+        ("native_grpcoll", is_native_grpcoll_enable()),
+        ("min_chunks_per_rank", min_chunks_per_rank()),
+    )

@@ -1,0 +1,115 @@
+# Planner layer (reference magi_attention/meta/_make_dispatch_meta.py:56 and
+# _make_attn_meta.py:40 — same roles, MI355X-first rebuild).
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import List
+
+from ..common.enum import AttnMaskType
+from ..common.ranges import AttnRanges
+from ..config import DistAttnConfig
+from .attn_solver import DistAttnSolver
+from .containers import CalcMeta, CommMeta
+from .geometry import MaskSlice, slice_from_raw
+from .solver.dispatch_solver import DispatchSolver
+
+
+@dataclass
+class DispatchMeta:
+    """Chunk -> rank assignment (reference meta/collection/dispatch_meta.py:24)."""
+
+    cp_size: int
+    cp_rank: int
+    chunk_size: int
+    total_seqlen: int          # padded
+    num_chunks: int
+    partitions: List[List[int]]
+    loads: List[float] = field(default_factory=list)
+
+    @property
+    def chunks_per_rank(self) -> int:
+        return self.num_chunks // self.cp_size
+
+    def host_ranges(self, rank: int) -> AttnRanges:
+        rr = AttnRanges()
+        for c in self.partitions[rank]:
+            rr.append(
+                AttnRanges.from_ranges(
+                    [(c * self.chunk_size, (c + 1) * self.chunk_size)]
+                )[0]
+            )
+        return rr.merge()
+
+
+def normalize_slices(
+    q_ranges: AttnRanges,
+    k_ranges: AttnRanges,
+    attn_mask_type,
+) -> List[MaskSlice]:
+    """Raw API triples -> aligned MaskSlices."""
+    if not isinstance(attn_mask_type, (list, tuple)):
+        attn_mask_type = [attn_mask_type] * len(q_ranges)
+    out: List[MaskSlice] = []
+    for qr, kr, t in zip(q_ranges, k_ranges, attn_mask_type):
+        ti = t.to_int_type() if isinstance(t, AttnMaskType) else (
+            AttnMaskType(t).to_int_type() if isinstance(t, str) else int(t)
+        )
+        out.extend(slice_from_raw(qr.start, qr.end, kr.start, kr.end, ti))
+    return out
+
+
+def make_dispatch_meta_from_qk_ranges(
+    slices: List[MaskSlice],
+    total_seqlen_padded: int,
+    cp_size: int,
+    cp_rank: int,
+    dist_attn_config: DistAttnConfig,
+) -> DispatchMeta:
+    """Chunk the (padded) token space and assign chunks to ranks balancing
+    per-chunk mask area (reference _make_dispatch_meta.py:56)."""
+    from .geometry import area_in_rows
+
+    chunk_size = dist_attn_config.dispatch_config.chunk_size
+    assert total_seqlen_padded % (chunk_size * cp_size) == 0, (
+        f"padded seqlen {total_seqlen_padded} not divisible by "
+        f"chunk_size*cp = {chunk_size}*{cp_size}"
+    )
+    num_chunks = total_seqlen_padded // chunk_size
+    workloads = []
+    for c in range(num_chunks):
+        a, b = c * chunk_size, (c + 1) * chunk_size
+        workloads.append(float(sum(area_in_rows(sl, a, b) for sl in slices)))
+    sol = DispatchSolver(dist_attn_config.dispatch_config.alg).solve(
+        workloads, cp_size
+    )
+    return DispatchMeta(
+        cp_size=cp_size,
+        cp_rank=cp_rank,
+        chunk_size=chunk_size,
+        total_seqlen=total_seqlen_padded,
+        num_chunks=num_chunks,
+        partitions=sol.partitions,
+        loads=sol.loads,
+    )
+
+
+def make_attn_meta_from_dispatch_meta(
+    slices: List[MaskSlice],
+    dispatch_meta: DispatchMeta,
+    dist_attn_config: DistAttnConfig,
+) -> tuple[DistAttnSolver, CalcMeta, CommMeta]:
+    """Reference _make_attn_meta.py:40: run the DistAttnSolver and emit this
+    rank's calc/comm tables."""
+    oc = dist_attn_config.overlap_config
+    degree = (oc.degree or 1) if oc.enable else 1
+    solver = DistAttnSolver(
+        slices=slices,
+        partitions=dispatch_meta.partitions,
+        chunk_size=dispatch_meta.chunk_size,
+        total_seqlen=dispatch_meta.total_seqlen,
+        cp_size=dispatch_meta.cp_size,
+        overlap_degree=degree,
+        min_stage_tokens=oc.min_chunk_size,
+    )
+    rank = dispatch_meta.cp_rank
+    return solver, solver.make_calc_meta(rank), solver.make_comm_meta(rank)

@@ -1,0 +1,104 @@
+# Planner output containers (reference meta/collection/{calc_meta.py:62 AttnArg,
+# :719 CalcMeta; comm_meta.py:572 CommMeta, :41 GroupCollectiveArg} — same roles,
+# rebuilt for the RCCL a2av transport).
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+
+@dataclass
+class AttnArg:
+    """Kernel arguments of one FFA call: (q_range, k_range, type) triples in
+    LOCAL coordinates (q: rank-local rows; k: host-local or stage-buffer rows).
+    Device tensors are materialised once per device and cached
+    (reference calc_meta.py:83-117 __post_init__)."""
+
+    q_ranges: List[Tuple[int, int]]
+    k_ranges: List[Tuple[int, int]]
+    attn_type_map: List[int]
+    max_seqlen_q: int = 0
+    total_area: int = 0
+    _cache: Dict[str, torch.Tensor] = field(default_factory=dict, repr=False)
+
+    def is_empty(self) -> bool:
+        return len(self.q_ranges) == 0
+
+    def to_device(self, device) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+        key = str(device)
+        if key not in self._cache:
+            self._cache[key] = (
+                torch.tensor(self.q_ranges, dtype=torch.int32, device=device).reshape(-1, 2),
+                torch.tensor(self.k_ranges, dtype=torch.int32, device=device).reshape(-1, 2),
+                torch.tensor(self.attn_type_map, dtype=torch.int32, device=device),
+            )
+        return self._cache[key]
+
+
+@dataclass
+class RowChunkMap:
+    """Row copy plan: (in_ranges over a source buffer, out_starts in a dest
+    buffer) — the unpack/pack tables driving range_gather/range_reduce."""
+
+    in_ranges: List[Tuple[int, int]]
+    out_starts: List[int]
+    total_rows: int
+    _cache: Dict[str, Tuple[torch.Tensor, torch.Tensor]] = field(
+        default_factory=dict, repr=False
+    )
+
+    def to_device(self, device) -> Tuple[torch.Tensor, torch.Tensor]:
+        key = str(device)
+        if key not in self._cache:
+            self._cache[key] = (
+                torch.tensor(self.in_ranges, dtype=torch.int32, device=device).reshape(-1, 2),
+                torch.tensor(self.out_starts, dtype=torch.int32, device=device),
+            )
+        return self._cache[key]
+
+
+@dataclass
+class GroupCastArg:
+    """One overlap stage's K/V multicast for ONE rank (a2av realisation;
+    reference comm_meta.py:41 GroupCollectiveArg + grpcoll/utils.py:593
+    calc_group_cast_a2a_args)."""
+
+    # send: my host-local k rows packed per dst rank (rows of the K tensor;
+    # V reuses the same tables at +total_local offset)
+    send_pack: RowChunkMap            # local kv rows -> send buffer
+    input_split_sizes: List[int]      # per dst rank (k rows)
+    # recv: rows arrive ordered by src rank; unpack into the globally-sorted
+    # stage buffer
+    recv_unpack: RowChunkMap          # recv buffer rows -> stage buffer
+    output_split_sizes: List[int]     # per src rank (k rows)
+    stage_tokens: int                 # stage buffer rows (k)
+
+
+@dataclass
+class GroupReduceArg:
+    """Reverse path: partial dK/dV rows of the stage buffer sent back to owner
+    ranks and sum-reduced into owner-local accumulators."""
+
+    send_pack: RowChunkMap            # stage-buffer rows -> send buffer (per dst owner)
+    input_split_sizes: List[int]
+    recv_reduce: RowChunkMap          # recv rows -> local kv row positions (sum)
+    output_split_sizes: List[int]
+    total_recv: int
+
+
+@dataclass
+class CommMeta:
+    stages_cast: List[GroupCastArg]
+    stages_reduce: List[GroupReduceArg]
+
+    @property
+    def overlap_degree(self) -> int:
+        return len(self.stages_cast)
+
+
+@dataclass
+class CalcMeta:
+    host_arg: AttnArg
+    stage_args: List[AttnArg]

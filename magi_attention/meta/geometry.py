@@ -71,13 +71,8 @@ class MaskSlice:
         w = sk - sq + 1
         if w >= 1:
             return sq * w
-        # degenerate band (sk < sq): some rows empty
-        for i in range(sq):
-            a = max(0, i + (sk - sq))
-            b = min(sk - 1, i)
-            if b >= a:
-                total += b - a + 1
-        return total
+        # degenerate band (sk < sq): allowed j in [i, i + sk - sq] is empty
+        return 0
 
 
 def normalize(
@@ -88,6 +83,8 @@ def normalize(
     exactly-aligned typed slices."""
     if qhi <= qlo or kb <= ka:
         return []
+    if lo is not None and up is not None and lo > up:
+        return []  # empty band (e.g. bi-causal with sk < sq shrunk to nothing)
     # clip empty rows
     if up is not None:
         qlo = max(qlo, ka - up)  # q+up >= ka

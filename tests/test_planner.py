@@ -1,0 +1,209 @@
+"""Planner end-to-end checks (CPU, single process): the dispatch + attn solver
+tables must reproduce the global mask exactly and the comm tables must be a
+consistent a2av exchange. The distributed runtime is only a transport for
+what is validated here."""
+import random
+
+import pytest
+import torch
+
+from magi_attention.config import DispatchConfig, DistAttnConfig, OverlapConfig
+from magi_attention.meta import (
+    make_attn_meta_from_dispatch_meta,
+    make_dispatch_meta_from_qk_ranges,
+    normalize_slices,
+)
+from magi_attention.meta.geometry import to_dense
+from magi_attention.common.ranges import AttnRanges
+from oracle import make_attn_mask, ref_attn
+from oracle.ref_attn import merge_out_lse
+
+
+def build_case(seed, total=1024, cp=4, chunk=128, degree=2):
+    rng = random.Random(seed)
+    # varlen docs with random types
+    n_docs = rng.randrange(2, 6)
+    cuts = sorted(rng.sample(range(64, total - 1), n_docs - 1))
+    bounds = [0] + cuts + [total]
+    q_ranges, k_ranges, types = [], [], []
+    for a, b in zip(bounds, bounds[1:]):
+        q_ranges.append([a, b])
+        k_ranges.append([a, b])
+        types.append(rng.choice([0, 1, 1, 2, 3]))
+    # plus a couple of cross-doc slices — kept PAIR-DISJOINT from existing
+    # slices (overlapping (q,k) pairs are invalid input: the kernel merges
+    # per-slice softmax contributions, so a pair present twice would be
+    # double-counted; reference semantics require disjoint pair sets)
+    occupied = make_attn_mask(total, total, q_ranges, k_ranges, types)
+    for _ in range(rng.randrange(0, 3)):
+        a = rng.randrange(0, total - 64)
+        b = rng.randrange(a + 32, min(a + 400, total) + 1)
+        c = rng.randrange(0, total - 64)
+        d = rng.randrange(c + 32, min(c + 400, total) + 1)
+        t = rng.choice([0, 1, 2, 3])
+        cand = make_attn_mask(total, total, [[a, b]], [[c, d]], [t])
+        if (cand & occupied).any():
+            continue
+        occupied |= cand
+        q_ranges.append([a, b])
+        k_ranges.append([c, d])
+        types.append(t)
+    cfg = DistAttnConfig(
+        dispatch_config=DispatchConfig(chunk_size=chunk),
+        overlap_config=OverlapConfig(degree=degree, min_chunk_size=64),
+    )
+    slices = normalize_slices(
+        AttnRanges.from_ranges(q_ranges), AttnRanges.from_ranges(k_ranges), types
+    )
+    return q_ranges, k_ranges, types, slices, cfg
+
+
+@pytest.mark.parametrize("seed", range(6))
+@pytest.mark.parametrize("cp", [1, 2, 4])
+def test_planner_covers_global_mask(seed, cp):
+    total, chunk = 1024, 128
+    q_ranges, k_ranges, types, slices, cfg = build_case(seed, total, cp, chunk)
+    dm = make_dispatch_meta_from_qk_ranges(slices, total, cp, 0, cfg)
+    ref_mask = make_attn_mask(total, total, q_ranges, k_ranges, types)
+
+    cover = torch.zeros(total, total, dtype=torch.int32)
+    for r in range(cp):
+        dm.cp_rank = r
+        solver, calc, comm = make_attn_meta_from_dispatch_meta(slices, dm, cfg)
+        plan = solver.plans[r]
+        hr = plan.host_ranges
+        # host slices are in global coords in the plan
+        for sl in plan.host_slices + plan.remote_slices:
+            cover += to_dense([sl], total, total).int()
+        # stage args must cover all remote slices exactly once:
+        st_tokens = sum(s.total_seqlen for s in plan.stages_need)
+        # zero-redundancy: remote need == union of remote slice k ranges minus host
+        need = AttnRanges()
+        for sl in plan.remote_slices:
+            need.append(AttnRanges.from_ranges([(sl.ks, sl.ke)])[0])
+        assert st_tokens == need.merge().total_seqlen
+        # comm table shape consistency
+        for s in range(comm.overlap_degree):
+            cast = comm.stages_cast[s]
+            assert sum(cast.output_split_sizes) == cast.recv_unpack.total_rows
+            assert sum(cast.input_split_sizes) == cast.send_pack.total_rows
+    # each (q,k) allowed pair computed exactly once across all ranks
+    assert torch.equal(cover.bool(), ref_mask)
+    assert int(cover.max()) <= 1
+
+
+@pytest.mark.parametrize("seed", range(4))
+@pytest.mark.parametrize("cp,degree", [(2, 1), (4, 2), (4, 3)])
+def test_planner_simulated_distributed_attention(seed, cp, degree):
+    """Simulate the full CP forward with the planner's own tables (gather +
+    a2av emulation + per-stage oracle attention + lse merge) and compare to
+    the global oracle."""
+    total, chunk = 1024, 128
+    hq, hk, d = 4, 2, 32
+    q_ranges, k_ranges, types, slices, cfg = build_case(seed, total, cp, chunk,
+                                                        degree)
+    g = torch.Generator().manual_seed(seed)
+    q = torch.randn(total, hq, d, generator=g, dtype=torch.float64)
+    k = torch.randn(total, hk, d, generator=g, dtype=torch.float64)
+    v = torch.randn(total, hk, d, generator=g, dtype=torch.float64)
+
+    ref_mask = make_attn_mask(total, total, q_ranges, k_ranges, types)
+    ref_o, ref_lse = ref_attn(q, k, v, ref_mask)
+
+    dm = make_dispatch_meta_from_qk_ranges(slices, total, cp, 0, cfg)
+
+    solvers = []
+    for r in range(cp):
+        dm.cp_rank = r
+        solvers.append(make_attn_meta_from_dispatch_meta(slices, dm, cfg))
+
+    # precompute per-rank local rows
+    def rows_of(rr):
+        idx = []
+        for p in rr:
+            idx.extend(range(p.start, p.end))
+        return torch.tensor(idx, dtype=torch.long)
+
+    host_rows = [rows_of(solvers[r][0].host_ranges_all[r]) for r in range(cp)]
+
+    # ---- emulate the a2av exchange per stage and check the stage buffers ----
+    deg = solvers[0][2].overlap_degree
+    for r in range(cp):
+        assert solvers[r][2].overlap_degree == deg
+
+    stage_kv = [[None] * deg for _ in range(cp)]
+    for s in range(deg):
+        # build send buffers
+        send_buf = []
+        for r in range(cp):
+            _, _, comm = solvers[r]
+            cast = comm.stages_cast[s]
+            kv_local = torch.cat(
+                [k[host_rows[r]], v[host_rows[r]]], dim=0
+            )  # [2L, hk, d]
+            buf = torch.zeros(cast.send_pack.total_rows, hk, d, dtype=k.dtype)
+            for (a, b), o in zip(cast.send_pack.in_ranges, cast.send_pack.out_starts):
+                buf[o:o + (b - a)] = kv_local[a:b]
+            send_buf.append(buf)
+        # exchange: dst r gets from each src o the o->r segment
+        for r in range(cp):
+            _, _, comm = solvers[r]
+            cast = comm.stages_cast[s]
+            recv = []
+            for o in range(cp):
+                ocast = solvers[o][2].stages_cast[s]
+                start = sum(ocast.input_split_sizes[:r])
+                recv.append(send_buf[o][start:start + ocast.input_split_sizes[r]])
+                assert ocast.input_split_sizes[r] == cast.output_split_sizes[o]
+            recv = torch.cat(recv) if recv else torch.zeros(0, hk, d)
+            S = cast.stage_tokens
+            st = torch.zeros(2 * S, hk, d, dtype=k.dtype)
+            for (a, b), o in zip(cast.recv_unpack.in_ranges,
+                                 cast.recv_unpack.out_starts):
+                st[o:o + (b - a)] = recv[a:b]
+            stage_kv[r][s] = st
+            # check against direct gather of the stage's global ranges
+            srows = rows_of(solvers[r][0].plans[r].stages_need[s])
+            if len(srows):
+                assert torch.equal(st[:S], k[srows])
+                assert torch.equal(st[S:], v[srows])
+
+    # ---- per-rank staged attention + merge, then compare ----
+    for r in range(cp):
+        solver, calc, comm = solvers[r]
+        rows = host_rows[r]
+        ql = q[rows]
+        kl, vl = k[rows], v[rows]
+        L = len(rows)
+        outs, lses = [], []
+
+        def run(arg, kk, vv):
+            if arg.is_empty():
+                return None
+            m = make_attn_mask(
+                L, kk.shape[0], arg.q_ranges, arg.k_ranges, arg.attn_type_map
+            )
+            return ref_attn(ql, kk, vv, m)
+
+        res = run(calc.host_arg, kl, vl)
+        if res:
+            outs.append(res[0])
+            lses.append(res[1])
+        for s in range(deg):
+            S = comm.stages_cast[s].stage_tokens
+            if S == 0:
+                continue
+            res = run(calc.stage_args[s], stage_kv[r][s][:S], stage_kv[r][s][S:])
+            if res:
+                outs.append(res[0])
+                lses.append(res[1])
+        if not outs:
+            continue
+        out_m, lse_m = merge_out_lse(outs, lses)
+        torch.testing.assert_close(
+            out_m.to(torch.float64), ref_o[rows], atol=1e-6, rtol=1e-6
+        )
+        fin = torch.isfinite(ref_lse[rows])
+        torch.testing.assert_close(
+            lse_m.float()[fin], ref_lse[rows][fin], atol=1e-5, rtol=1e-5
+        )
