@@ -1,0 +1,229 @@
+# Public drop-in API (reference magi_attention/api/magi_attn_interface.py —
+# signatures and semantics kept: magi_attn_varlen_key:160, magi_attn_flex_key:440,
+# dispatch:887, undispatch:924, calc_attn:1041, get_position_ids:1112,
+# get_most_recent_key:1136, clear_cache:1157, DistAttnRuntimeDictManager:64).
+from __future__ import annotations
+
+import warnings
+from typing import List, Optional, Tuple, Union
+
+import torch
+import torch.distributed as dist
+
+from ..common.enum import AttnMaskType
+from ..common.range import AttnRange
+from ..common.ranges import AttnRanges
+from ..config import DistAttnConfig
+from ..dist_attn_runtime_mgr import (
+    DistAttnRuntimeDict,
+    DistAttnRuntimeKey,
+    DistAttnRuntimeMgr,
+    init_dist_attn_runtime_key,
+    init_dist_attn_runtime_mgr,
+)
+from .functools import compute_pad_size, pad_at_dim, unpad_at_dim
+
+GeneralAttnMaskType = Union[str, AttnMaskType, List[Union[str, AttnMaskType]]]
+
+
+class DistAttnRuntimeDictManager:
+    """Per-cp-group LRU caches (reference magi_attn_interface.py:64-157)."""
+
+    def __init__(self, max_size_per_group: int = 100):
+        self.max_size = max_size_per_group
+        self._per_group: dict = {}
+
+    def _cache_for(self, key: DistAttnRuntimeKey) -> DistAttnRuntimeDict:
+        tag = key.cp_group_tag
+        if tag not in self._per_group:
+            self._per_group[tag] = DistAttnRuntimeDict(self.max_size)
+        return self._per_group[tag]
+
+    def get(self, key, default=None):
+        return self._cache_for(key).get(key, default)
+
+    def __contains__(self, key):
+        return key in self._cache_for(key)
+
+    def __setitem__(self, key, value):
+        self._cache_for(key)[key] = value
+
+    def __getitem__(self, key):
+        return self._cache_for(key)[key]
+
+    def get_most_recent_key(self, cp_group=None):
+        for cache in self._per_group.values():
+            k = cache.most_recent_key()
+            if k is not None:
+                return k
+        return None
+
+    def clear(self, cp_group=None):
+        self._per_group.clear()
+
+
+dist_attn_runtime_dict_mgr = DistAttnRuntimeDictManager()
+
+
+def _resolve_group(cp_group_or_mesh) -> dist.ProcessGroup:
+    if isinstance(cp_group_or_mesh, dist.ProcessGroup):
+        return cp_group_or_mesh
+    # 1D DeviceMesh
+    try:
+        return cp_group_or_mesh.get_group()
+    except Exception as e:  # pragma: no cover
+        raise ValueError(f"unsupported cp_group_or_mesh: {e}")
+
+
+def magi_attn_flex_key(
+    q_ranges: AttnRanges,
+    k_ranges: AttnRanges,
+    attn_mask_type: GeneralAttnMaskType,
+    total_seqlen_q: int,
+    total_seqlen_k: int,
+    num_heads_q: int,
+    num_heads_kv: int,
+    head_dim: int,
+    pad_size: int = 0,
+    cp_group_or_mesh=None,
+    dist_attn_config: DistAttnConfig = DistAttnConfig(),
+    is_same_source: bool = True,
+    is_q_permutable: bool = True,
+    is_k_permutable: bool = True,
+    chunk_size: Optional[int] = None,
+) -> DistAttnRuntimeKey:
+    """The most flexible key-creation interface (reference :440)."""
+    assert is_same_source, "cross-attn (is_same_source=False) lands in a later round"
+    assert total_seqlen_q == total_seqlen_k, "self-attn requires equal seqlens"
+    group = _resolve_group(cp_group_or_mesh)
+    cp_size = dist.get_world_size(group)
+    if chunk_size is not None:
+        warnings.warn(
+            "chunk_size arg is deprecated; pass DispatchConfig(chunk_size=...)",
+            DeprecationWarning,
+        )
+        from dataclasses import replace
+
+        dist_attn_config = replace(
+            dist_attn_config,
+            dispatch_config=type(dist_attn_config.dispatch_config)(
+                chunk_size=chunk_size,
+                alg=dist_attn_config.dispatch_config.alg,
+            ),
+        )
+    ck = dist_attn_config.dispatch_config.chunk_size
+    computed_pad = compute_pad_size(total_seqlen_q, cp_size, ck)
+    if pad_size not in (0, computed_pad):
+        warnings.warn(
+            "pad_size arg is deprecated and recomputed internally",
+            DeprecationWarning,
+        )
+    pad_size = computed_pad
+
+    key = init_dist_attn_runtime_key(
+        q_ranges, k_ranges, attn_mask_type, total_seqlen_q, total_seqlen_k,
+        pad_size, ck, num_heads_q, num_heads_kv, head_dim, group,
+        dist_attn_config,
+    )
+    if key not in dist_attn_runtime_dict_mgr:
+        dist_attn_runtime_dict_mgr[key] = init_dist_attn_runtime_mgr(
+            key, group, dist_attn_config
+        )
+    return key
+
+
+def magi_attn_varlen_key(
+    cu_seqlens_q: torch.Tensor,
+    cu_seqlens_k: torch.Tensor,
+    total_seqlen_q: int = None,
+    total_seqlen_k: int = None,
+    num_heads_q: int = 1,
+    num_heads_kv: int = 1,
+    head_dim: int = 128,
+    pad_size: int = 0,
+    cp_group_or_mesh=None,
+    causal: bool = False,
+    window_size: Tuple[int, int] = (-1, -1),
+    dist_attn_config: DistAttnConfig = DistAttnConfig(),
+) -> DistAttnRuntimeKey:
+    """Varlen (cu_seqlens) front-end (reference :160)."""
+    from .functools import infer_attn_mask_from_cu_seqlens
+
+    q_ranges, k_ranges, types, tq, tk = infer_attn_mask_from_cu_seqlens(
+        cu_seqlens_q, cu_seqlens_k, causal=causal, window_size=window_size
+    )
+    return magi_attn_flex_key(
+        q_ranges, k_ranges, types,
+        total_seqlen_q or tq, total_seqlen_k or tk,
+        num_heads_q, num_heads_kv, head_dim, pad_size, cp_group_or_mesh,
+        dist_attn_config,
+    )
+
+
+def dispatch(x: torch.Tensor, key: DistAttnRuntimeKey) -> torch.Tensor:
+    """Pad + scatter x to this rank's permuted local shard (reference :887)."""
+    mgr = dist_attn_runtime_dict_mgr[key]
+    x = pad_at_dim(x, 0, key.pad_size)
+    return mgr.dispatch_qo(x)
+
+
+def undispatch(x_local: torch.Tensor, key: DistAttnRuntimeKey) -> torch.Tensor:
+    """Gather local shards back to the global (unpadded) order (reference :924)."""
+    mgr = dist_attn_runtime_dict_mgr[key]
+    full = mgr.undispatch_qo(x_local)
+    return unpad_at_dim(full, 0, key.total_seqlen_q)
+
+
+def calc_attn(
+    q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, key: DistAttnRuntimeKey
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Distributed flex-flash-attention on dispatched shards (reference :1041).
+    Returns (out, lse)."""
+    mgr = dist_attn_runtime_dict_mgr[key]
+    return mgr.calc_attn(q, k, v)
+
+
+def magi_attn_flex_dispatch(x, *args, **kwargs):
+    """Key creation + dispatch in one call (reference :725)."""
+    key = magi_attn_flex_key(*args, **kwargs)
+    return dispatch(x, key), key
+
+
+def magi_attn_varlen_dispatch(x, *args, **kwargs):
+    key = magi_attn_varlen_key(*args, **kwargs)
+    return dispatch(x, key), key
+
+
+def get_position_ids(key: DistAttnRuntimeKey) -> torch.Tensor:
+    mgr = dist_attn_runtime_dict_mgr[key]
+    return mgr.get_position_ids()
+
+
+def get_most_recent_key(cp_group=None) -> Optional[DistAttnRuntimeKey]:
+    return dist_attn_runtime_dict_mgr.get_most_recent_key(cp_group)
+
+
+def clear_cache(cp_group=None) -> None:
+    dist_attn_runtime_dict_mgr.clear(cp_group)
+
+
+def roll(x: torch.Tensor, key: DistAttnRuntimeKey, shifts: int = 1) -> torch.Tensor:
+    """Cyclic shift along the GLOBAL sequence of a dispatched tensor
+    (reference :960; used by MTP). Round-1 implementation routes through
+    undispatch/dispatch; the P2P-roll fast path lands in a later round."""
+    mgr = dist_attn_runtime_dict_mgr[key]
+    full = mgr.undispatch_qo(x)
+    full = torch.roll(full, shifts=shifts, dims=0)
+    return mgr.dispatch_qo(full)
+
+
+def roll_simple(x: torch.Tensor, key: DistAttnRuntimeKey, shifts: int = 1):
+    return roll(x, key, shifts)
+
+
+def make_flex_key_for_new_mask_after_dispatch(*a, **k):
+    raise NotImplementedError("lands in a later round")
+
+
+def make_varlen_key_for_new_mask_after_dispatch(*a, **k):
+    raise NotImplementedError("lands in a later round")

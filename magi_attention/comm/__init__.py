@@ -1,0 +1,1 @@
+from .primitive import WorkWithPostProcessFn, group_cast, group_reduce  # noqa: F401

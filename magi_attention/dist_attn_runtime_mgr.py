@@ -1,0 +1,201 @@
+# Runtime manager (reference magi_attention/dist_attn_runtime_mgr.py:
+#  DistAttnRuntimeKey:62 frozen+hashed incl. env-flag snapshot :79-87,
+#  DistAttnRuntimeMgr:122, DistAttnRuntimeDict:410 LRU,
+#  init_dist_attn_runtime_key:484, init_dist_attn_runtime_mgr:545).
+from __future__ import annotations
+
+from collections import OrderedDict
+from dataclasses import dataclass, field
+from typing import Any, List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+
+from . import env
+from .common.enum import AttnMaskType
+from .common.ranges import AttnRanges
+from .config import DistAttnConfig
+from .functional.dispatch import dispatch_func, undispatch_func
+from .functional.dist_attn import DistAttnRuntime, dist_attn_func
+from .meta import (
+    DispatchMeta,
+    make_attn_meta_from_dispatch_meta,
+    make_dispatch_meta_from_qk_ranges,
+    normalize_slices,
+)
+
+
+@dataclass(frozen=True)
+class DistAttnRuntimeKey:
+    """Frozen, hashable identity of one (mask, config, flags, group) plan."""
+
+    q_ranges: Tuple[Tuple[int, int], ...]
+    k_ranges: Tuple[Tuple[int, int], ...]
+    attn_mask_type: Tuple[int, ...]
+    total_seqlen_q: int
+    total_seqlen_k: int
+    pad_size: int
+    chunk_size: int
+    num_heads_q: int
+    num_heads_kv: int
+    head_dim: int
+    cp_group_tag: Tuple[int, ...]
+    config_tag: str
+    env_flags: Tuple[Tuple[str, Any], ...]
+
+
+def _group_tag(group: dist.ProcessGroup) -> Tuple[int, ...]:
+    try:
+        return tuple(dist.get_process_group_ranks(group))
+    except Exception:
+        return (id(group),)
+
+
+class DistAttnRuntimeMgr:
+    """Holds the plan (dispatch meta + solver outputs) and executes
+    dispatch / undispatch / calc_attn (reference DistAttnRuntimeMgr:122)."""
+
+    def __init__(
+        self,
+        key: DistAttnRuntimeKey,
+        cp_group: dist.ProcessGroup,
+        dist_attn_config: DistAttnConfig,
+    ):
+        self.key = key
+        self.cp_group = cp_group
+        self.config = dist_attn_config
+        cp_size = dist.get_world_size(cp_group)
+        cp_rank = dist.get_rank(cp_group)
+
+        slices = normalize_slices(
+            AttnRanges.from_ranges([list(r) for r in key.q_ranges]),
+            AttnRanges.from_ranges([list(r) for r in key.k_ranges]),
+            list(key.attn_mask_type),
+        )
+        total_padded = key.total_seqlen_q + key.pad_size
+        self.dispatch_meta: DispatchMeta = make_dispatch_meta_from_qk_ranges(
+            slices, total_padded, cp_size, cp_rank, dist_attn_config
+        )
+        self.solver, calc_meta, comm_meta = make_attn_meta_from_dispatch_meta(
+            slices, self.dispatch_meta, dist_attn_config
+        )
+        self.runtime = DistAttnRuntime(
+            calc_meta=calc_meta,
+            comm_meta=comm_meta,
+            cp_group=cp_group,
+            total_local_q=total_padded // cp_size,
+        )
+
+    # ---- ops ----
+    def dispatch_qo(self, x: torch.Tensor) -> torch.Tensor:
+        return dispatch_func(
+            x, self.dispatch_meta.partitions, self.dispatch_meta.chunk_size,
+            self.cp_group,
+        )
+
+    dispatch_kv = dispatch_qo
+
+    def undispatch_qo(self, x_local: torch.Tensor) -> torch.Tensor:
+        return undispatch_func(
+            x_local, self.dispatch_meta.partitions, self.dispatch_meta.chunk_size,
+            self.cp_group,
+        )
+
+    undispatch_kv = undispatch_qo
+
+    def calc_attn(self, q, k, v) -> Tuple[torch.Tensor, torch.Tensor]:
+        return dist_attn_func(q, k, v, self.runtime)
+
+    def get_position_ids(self, device=None) -> torch.Tensor:
+        """Local row -> global (unpadded) position id
+        (reference api/magi_attn_interface.py:1112)."""
+        rank = self.dispatch_meta.cp_rank
+        idx = []
+        ck = self.dispatch_meta.chunk_size
+        for c in self.dispatch_meta.partitions[rank]:
+            idx.extend(range(c * ck, (c + 1) * ck))
+        return torch.tensor(idx, dtype=torch.long, device=device or "cpu")
+
+
+class DistAttnRuntimeDict:
+    """LRU cache key -> mgr (reference DistAttnRuntimeDict:410)."""
+
+    def __init__(self, max_size: Optional[int] = None):
+        self.max_size = max_size or env.dist_attn_runtime_dict_size()
+        self._d: "OrderedDict[DistAttnRuntimeKey, DistAttnRuntimeMgr]" = OrderedDict()
+
+    def get(self, key, default=None):
+        if key in self._d:
+            self._d.move_to_end(key)
+            return self._d[key]
+        return default
+
+    def __contains__(self, key):
+        return key in self._d
+
+    def __setitem__(self, key, value):
+        self._d[key] = value
+        self._d.move_to_end(key)
+        while len(self._d) > self.max_size:
+            self._d.popitem(last=False)
+
+    def __getitem__(self, key):
+        if key not in self._d:
+            raise ValueError(f"unknown DistAttnRuntimeKey {key}")
+        self._d.move_to_end(key)
+        return self._d[key]
+
+    def keys(self):
+        return list(self._d.keys())
+
+    def most_recent_key(self):
+        return next(reversed(self._d)) if self._d else None
+
+    def clear(self):
+        self._d.clear()
+
+
+def init_dist_attn_runtime_key(
+    q_ranges: AttnRanges,
+    k_ranges: AttnRanges,
+    attn_mask_type,
+    total_seqlen_q: int,
+    total_seqlen_k: int,
+    pad_size: int,
+    chunk_size: int,
+    num_heads_q: int,
+    num_heads_kv: int,
+    head_dim: int,
+    cp_group: dist.ProcessGroup,
+    dist_attn_config: DistAttnConfig,
+) -> DistAttnRuntimeKey:
+    if not isinstance(attn_mask_type, (list, tuple)):
+        attn_mask_type = [attn_mask_type] * len(q_ranges)
+    types = tuple(
+        t.to_int_type() if isinstance(t, AttnMaskType)
+        else AttnMaskType(t).to_int_type() if isinstance(t, str) else int(t)
+        for t in attn_mask_type
+    )
+    return DistAttnRuntimeKey(
+        q_ranges=tuple((r.start, r.end) for r in q_ranges),
+        k_ranges=tuple((r.start, r.end) for r in k_ranges),
+        attn_mask_type=types,
+        total_seqlen_q=total_seqlen_q,
+        total_seqlen_k=total_seqlen_k,
+        pad_size=pad_size,
+        chunk_size=chunk_size,
+        num_heads_q=num_heads_q,
+        num_heads_kv=num_heads_kv,
+        head_dim=head_dim,
+        cp_group_tag=_group_tag(cp_group),
+        config_tag=repr(dist_attn_config),
+        env_flags=env.snapshot(),
+    )
+
+
+def init_dist_attn_runtime_mgr(
+    key: DistAttnRuntimeKey,
+    cp_group: dist.ProcessGroup,
+    dist_attn_config: DistAttnConfig,
+) -> DistAttnRuntimeMgr:
+    return DistAttnRuntimeMgr(key, cp_group, dist_attn_config)

@@ -1,0 +1,221 @@
+# DistAttnRuntime / DistAttnFunc — the CP execution engine.
+# (Reference: functional/dist_attn.py:141 DistAttnRuntime, :2993 DistAttnFunc,
+#  :3608 dist_attn_func — multi-stage overlap of group-cast K/V fetch with
+#  partial FFA compute, out/lse accumulated in-kernel; backward re-fetches
+#  remote KV, computes partial dK/dV per stage and group-reduces them to owner
+#  ranks (:1463 _fetch_remote_kv, :1924/_2068 reduce paths).)
+#
+# MI355X-native: FFA HIP kernel accumulates (out fp32, lse) in place across
+# stage calls (lock-merge epilogue), so no separate out-correction pass is
+# needed; comm is RCCL a2av over xGMI (comm/primitive.py).
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+
+from .. import env
+from ..comm.primitive import WorkWithPostProcessFn, group_cast, group_reduce
+from ..meta.containers import AttnArg, CalcMeta, CommMeta
+
+# test-only hook: lets CPU (gloo) tests run the runtime with an oracle
+# attention implementation. NEVER set in production; the product path on GPU
+# is the HIP kernel and raises if the extension is missing.
+_test_attn_backend = None
+
+
+def register_test_attn_backend(backend) -> None:
+    global _test_attn_backend
+    _test_attn_backend = backend
+
+
+@dataclass
+class DistAttnRuntime:
+    calc_meta: CalcMeta
+    comm_meta: CommMeta
+    cp_group: dist.ProcessGroup
+    total_local_q: int
+    softmax_scale: Optional[float] = None
+
+    @property
+    def overlap_degree(self) -> int:
+        return self.comm_meta.overlap_degree
+
+    # ---------------- forward ----------------
+    def attn_fwd(
+        self, q: torch.Tensor, k: torch.Tensor, v: torch.Tensor
+    ) -> Tuple[torch.Tensor, torch.Tensor]:
+        tq, hq, d = q.shape
+        scale = self.softmax_scale or d ** (-0.5)
+        kv_local = torch.cat([k, v], dim=0)
+
+        # pre-issue ALL remote-stage group-casts (reference dist_attn.py:419-436)
+        works: List[WorkWithPostProcessFn] = [
+            group_cast(kv_local, self.comm_meta.stages_cast[s], self.cp_group)
+            for s in range(self.overlap_degree)
+        ]
+
+        out_acc = torch.zeros(tq, hq, d, dtype=torch.float32, device=q.device)
+        lse_acc = torch.full(
+            (tq, hq), float("-inf"), dtype=torch.float32, device=q.device
+        )
+        self._fwd_partial(q, k, v, self.calc_meta.host_arg, out_acc, lse_acc, scale)
+        for s in range(self.overlap_degree):
+            arg = self.calc_meta.stage_args[s]
+            stage_kv = works[s].wait_post_process()
+            S = self.comm_meta.stages_cast[s].stage_tokens
+            if S == 0 or arg.is_empty():
+                continue
+            self._fwd_partial(
+                q, stage_kv[:S], stage_kv[S:], arg, out_acc, lse_acc, scale
+            )
+        out = out_acc.to(q.dtype)
+        return out, lse_acc
+
+    def _fwd_partial(self, q, k, v, arg: AttnArg, out_acc, lse_acc, scale):
+        if arg.is_empty():
+            return
+        if _test_attn_backend is not None and not q.is_cuda:
+            _test_attn_backend.fwd_partial(q, k, v, arg, out_acc, lse_acc, scale)
+            return
+        from .flex_flash_attn import _flex_flash_attn_forward
+
+        qr, kr, tm = arg.to_device(q.device)
+        _flex_flash_attn_forward(
+            q=q, k=k, v=v, sink=None, sink_layout="sh",
+            out=out_acc, lse=lse_acc,
+            q_ranges=qr, k_ranges=kr, attn_type_map=tm,
+            softmax_scale=scale, softcap=0.0, out_type=torch.float32,
+            disable_fwd_atomic_reduction=False, deterministic=False,
+            sm_margin=env.ffa_forward_sm_margin(),
+            max_seqlen_q=arg.max_seqlen_q,
+        )
+
+    # ---------------- backward ----------------
+    def attn_bwd(
+        self,
+        dout: torch.Tensor,
+        q: torch.Tensor,
+        k: torch.Tensor,
+        v: torch.Tensor,
+        out: torch.Tensor,
+        lse: torch.Tensor,
+    ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+        tq, hq, d = q.shape
+        L = k.shape[0]
+        scale = self.softmax_scale or d ** (-0.5)
+        kv_local = torch.cat([k, v], dim=0)
+
+        # re-fetch remote KV (reference backward:3455 -> _fetch_remote_kv:1463)
+        works = [
+            group_cast(kv_local, self.comm_meta.stages_cast[s], self.cp_group)
+            for s in range(self.overlap_degree)
+        ]
+
+        dq_acc = torch.zeros(tq, hq, d, dtype=torch.float32, device=q.device)
+        dkv_acc = torch.zeros(2 * L, *k.shape[1:], dtype=torch.float32,
+                              device=q.device)
+        dpsum = self._bwd_dpsum(dout, out)
+        self._bwd_partial(
+            dout, q, k, v, out, lse, dpsum, self.calc_meta.host_arg,
+            dq_acc, dkv_acc[:L], dkv_acc[L:], scale,
+        )
+        rworks: List[WorkWithPostProcessFn] = []
+        for s in range(self.overlap_degree):
+            arg = self.calc_meta.stage_args[s]
+            stage_kv = works[s].wait_post_process()
+            S = self.comm_meta.stages_cast[s].stage_tokens
+            if S == 0:
+                continue
+            dkv_stage = torch.zeros(
+                2 * S, *k.shape[1:], dtype=torch.float32, device=q.device
+            )
+            if not arg.is_empty():
+                self._bwd_partial(
+                    dout, q, stage_kv[:S], stage_kv[S:], out, lse, dpsum, arg,
+                    dq_acc, dkv_stage[:S], dkv_stage[S:], scale,
+                )
+            rworks.append(
+                group_reduce(
+                    dkv_stage, dkv_acc, self.comm_meta.stages_reduce[s],
+                    self.cp_group,
+                )
+            )
+        for w in rworks:
+            w.wait_post_process()
+        dq = dq_acc.to(q.dtype)
+        dk = dkv_acc[:L].to(k.dtype)
+        dv = dkv_acc[L:].to(v.dtype)
+        return dq, dk, dv
+
+    def _bwd_dpsum(self, dout, out):
+        if _test_attn_backend is not None and not dout.is_cuda:
+            return (dout.float() * out.float()).sum(-1)
+        from .. import _ffa_lib
+        from .._ffa_lib import MagiFfaBwdArgs, check, current_stream_ptr, ptr
+
+        tq, hq, d = dout.shape
+        dpsum = torch.empty(tq, hq, dtype=torch.float32, device=dout.device)
+        args = MagiFfaBwdArgs(
+            dout=ptr(dout.contiguous()), out=ptr(out.contiguous()),
+            dpsum=ptr(dpsum),
+            total_q=tq, hq=hq, d=d,
+            out_is_fp32=int(out.dtype == torch.float32),
+            stream=current_stream_ptr(),
+        )
+        check(_ffa_lib.lib().magi_ffa_bwd_preprocess(args), "bwd_preprocess")
+        return dpsum
+
+    def _bwd_partial(self, dout, q, k, v, out, lse, dpsum, arg: AttnArg,
+                     dq, dk, dv, scale):
+        if arg.is_empty():
+            return
+        if _test_attn_backend is not None and not q.is_cuda:
+            _test_attn_backend.bwd_partial(
+                dout, q, k, v, out, lse, dpsum, arg, dq, dk, dv, scale
+            )
+            return
+        from .. import _ffa_lib
+        from .._ffa_lib import MagiFfaBwdArgs, check, current_stream_ptr, ptr
+
+        qr, kr, tm = arg.to_device(q.device)
+        tq, hq, d = q.shape
+        tk, hk, _ = k.shape
+        max_k = int(max(b - a for a, b in arg.k_ranges))
+        args = MagiFfaBwdArgs(
+            dout=ptr(dout), q=ptr(q), k=ptr(k.contiguous()),
+            v=ptr(v.contiguous()), out=ptr(out), lse=ptr(lse),
+            dq=ptr(dq), dk=ptr(dk), dv=ptr(dv), dpsum=ptr(dpsum),
+            q_ranges=ptr(qr), k_ranges=ptr(kr), attn_type_map=ptr(tm),
+            n_ranges=qr.shape[0], total_q=tq, total_k=tk,
+            hq=hq, hk=hk, d=d, max_seqlen_k=max_k,
+            out_is_fp32=int(out.dtype == torch.float32),
+            softmax_scale=scale, softcap=0.0,
+            cu_margin=env.ffa_backward_sm_margin(),
+            stream=current_stream_ptr(),
+        )
+        check(_ffa_lib.lib().magi_ffa_bwd(args), "magi_ffa_bwd")
+
+
+class DistAttnFunc(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, runtime: DistAttnRuntime):
+        out, lse = runtime.attn_fwd(q, k, v)
+        ctx.save_for_backward(q, k, v, out, lse)
+        ctx.runtime = runtime
+        return out, lse
+
+    @staticmethod
+    def backward(ctx, dout, _dlse):
+        q, k, v, out, lse = ctx.saved_tensors
+        dq, dk, dv = ctx.runtime.attn_bwd(dout.contiguous(), q, k, v, out, lse)
+        return dq, dk, dv, None
+
+
+def dist_attn_func(
+    q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, runtime: DistAttnRuntime
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Reference functional/dist_attn.py:3608."""
+    return DistAttnFunc.apply(q, k, v, runtime)
