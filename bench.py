@@ -1,0 +1,279 @@
+#!/usr/bin/env python3
+"""Driver-contract benchmark: the BASELINE.json headline metric
+("attention TFLOPS/sec/GPU fwd+bwd; cp=1->8 scaling at seqlen 64k") measured
+through the product path (magi_attention.api calc_attn) on synthetic data.
+
+  python bench.py --gpus N --steps K --warmup W
+N>1 is launched by the driver via torch.distributed.run (one rank per GPU,
+RCCL); this process then reads RANK/WORLD_SIZE from the env.
+
+Workload (BASELINE.json configs[2], the config the metric is quoted on):
+dense causal, seqlen 65536, hq=32, d=128, bf16, synthetic random-normal data.
+Strong scaling: the 64k sequence is context-parallel sharded over N GPUs.
+
+FLOPs convention (reference docs/source/blog/cp_benchmark.md:40-63):
+fwd = 4*MaskArea*hq*hd, bwd = 2.5*fwd. `value` is the WHOLE-JOB aggregate
+TFLOPS/s over all N GPUs; per-GPU = value / n_gpus.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+SEQLEN = 65536
+HQ = 32
+HKV = 32
+D = 128
+MASK_AREA = SEQLEN * (SEQLEN + 1) // 2
+FWD_FLOPS = 4 * MASK_AREA * HQ * D
+STEP_FLOPS = FWD_FLOPS * 3.5  # fwd + bwd (2.5x)
+MFMA_PEAK_BF16 = 2.5e15  # dense bf16 MFMA peak, MI355X (spec; 2:1-sparse excluded)
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(msg, file=sys.stderr, flush=True)
+
+
+def setup_dist(n_gpus: int):
+    import torch.distributed as dist
+
+    if n_gpus <= 1 and "WORLD_SIZE" not in os.environ:
+        return None, 0, 1
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(n_gpus)))
+    local = int(os.environ.get("LOCAL_RANK", str(rank)))
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29531")
+    torch.cuda.set_device(local)
+    dist.init_process_group("nccl", rank=rank, world_size=world)
+    return dist.group.WORLD, rank, world
+
+
+def make_key(group, world):
+    from magi_attention.api import magi_attn_flex_key
+    from magi_attention.common.ranges import AttnRanges
+    from magi_attention.config import (
+        DispatchConfig,
+        DistAttnConfig,
+        OverlapConfig,
+    )
+
+    cfg = DistAttnConfig(
+        dispatch_config=DispatchConfig(chunk_size=2048),
+        overlap_config=OverlapConfig(degree=2, min_chunk_size=1024),
+    )
+    if group is None:
+        import torch.distributed as dist
+
+        if not dist.is_initialized():
+            os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+            os.environ.setdefault("MASTER_PORT", "29531")
+            dist.init_process_group(
+                "nccl" if torch.cuda.is_available() else "gloo",
+                rank=0, world_size=1,
+            )
+        group = dist.group.WORLD
+    key = magi_attn_flex_key(
+        AttnRanges.from_ranges([[0, SEQLEN]]),
+        AttnRanges.from_ranges([[0, SEQLEN]]),
+        "causal",
+        SEQLEN, SEQLEN, HQ, HKV, D,
+        cp_group_or_mesh=group,
+        dist_attn_config=cfg,
+    )
+    return key, group
+
+
+def run_bench(args):
+    import torch.distributed as dist
+
+    group, rank, world = setup_dist(args.gpus)
+    device = torch.device("cuda", torch.cuda.current_device())
+    key, group = make_key(group, world)
+    from magi_attention.api import calc_attn, dispatch
+
+    g = torch.Generator(device="cpu").manual_seed(42 + rank)
+    q = (torch.randn(SEQLEN, HQ, D, generator=g) * 0.5).bfloat16().to(device)
+    k = (torch.randn(SEQLEN, HKV, D, generator=g) * 0.5).bfloat16().to(device)
+    v = (torch.randn(SEQLEN, HKV, D, generator=g) * 0.5).bfloat16().to(device)
+    ql = dispatch(q, key).requires_grad_(True)
+    kl = dispatch(k, key).requires_grad_(True)
+    vl = dispatch(v, key).requires_grad_(True)
+    dout = torch.randn_like(ql)
+    del q, k, v
+
+    def step():
+        out, lse = calc_attn(ql, kl, vl, key)
+        out.backward(dout)
+        ql.grad = kl.grad = vl.grad = None
+
+    for _ in range(args.warmup):
+        step()
+    torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    if world > 1:
+        t = torch.tensor([elapsed], device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+    ms_per_step = elapsed / args.steps * 1e3
+    value = STEP_FLOPS * args.steps / elapsed / 1e12  # whole-job TFLOPS/s
+    return rank, world, ms_per_step, value, device, (ql, kl, vl, dout, key)
+
+
+def measure_roofline(state, device):
+    """Per-launch duration of the dominant kernel (bwd mainloop) and the fwd
+    mainloop with HIP events on the launch stream; achieved = algorithmic
+    FLOPs per launch / duration."""
+    ql, kl, vl, dout, key = state
+    from magi_attention.dist_attn_runtime_mgr import DistAttnRuntimeMgr  # noqa
+    from magi_attention.api.magi_attn_interface import dist_attn_runtime_dict_mgr
+
+    mgr = dist_attn_runtime_dict_mgr[key]
+    rt = mgr.runtime
+    # cp=1: host_arg covers the whole local mask; one fwd launch + one bwd launch
+    out, lse = rt.attn_fwd(ql.detach(), kl.detach(), vl.detach())
+    torch.cuda.synchronize()
+
+    local_area = sum(
+        a.total_area
+        for a in [rt.calc_meta.host_arg] + list(rt.calc_meta.stage_args)
+    )
+    fwd_flops = 4 * local_area * HQ * D
+    ev0, ev1 = torch.cuda.Event(True), torch.cuda.Event(True)
+
+    scale = D ** -0.5
+    out_acc = torch.zeros_like(out, dtype=torch.float32)
+    lse_acc = torch.full_like(lse, float("-inf"))
+    reps = 3
+    ev0.record()
+    for _ in range(reps):
+        rt._fwd_partial(ql.detach(), kl.detach(), vl.detach(),
+                        rt.calc_meta.host_arg, out_acc, lse_acc, scale)
+    ev1.record()
+    torch.cuda.synchronize()
+    fwd_ms = ev0.elapsed_time(ev1) / reps
+    host_area = rt.calc_meta.host_arg.total_area
+    host_fwd_flops = 4 * host_area * HQ * D
+
+    # bwd mainloop (dominant)
+    dq = torch.zeros_like(ql, dtype=torch.float32)
+    dkv = torch.zeros(2 * kl.shape[0], HKV, D, dtype=torch.float32, device=device)
+    dpsum = rt._bwd_dpsum(dout, out)
+    ev0.record()
+    for _ in range(reps):
+        rt._bwd_partial(dout, ql.detach(), kl.detach(), vl.detach(), out, lse,
+                        dpsum, rt.calc_meta.host_arg, dq,
+                        dkv[: kl.shape[0]], dkv[kl.shape[0]:], scale)
+    ev1.record()
+    torch.cuda.synchronize()
+    bwd_ms = ev0.elapsed_time(ev1) / reps
+    bwd_flops = host_fwd_flops * 2.5
+
+    achieved = bwd_flops / (bwd_ms * 1e-3)
+    return {
+        "bound": "mfma",
+        "achieved": achieved,
+        "peak": MFMA_PEAK_BF16,
+        "unit": "FLOP/s",
+        "frac": achieved / MFMA_PEAK_BF16,
+        "traffic": None,
+        "detail": {
+            "kernel": "ffa_bwd_kernel (5-matmul mainloop)",
+            "bwd_ms_per_launch": bwd_ms,
+            "fwd_ms_per_launch": fwd_ms,
+            "fwd_achieved_flops_per_s": host_fwd_flops / (fwd_ms * 1e-3),
+        },
+    }
+
+
+def measure_cpu_baseline():
+    """Oracle (CPU restatement, kind="port") timed on the host cores over a
+    BOUNDED sample: one causal 4096-token slice of the same workload,
+    fwd+bwd, scaled to the TFLOPS metric."""
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    from oracle import make_attn_mask, ref_attn_with_grads
+
+    n = 4096
+    cores = os.cpu_count() or 1
+    torch.set_num_threads(cores)
+    g = torch.Generator().manual_seed(42)
+    q = torch.randn(n, HQ, D, generator=g) * 0.5
+    k = torch.randn(n, HKV, D, generator=g) * 0.5
+    v = torch.randn(n, HKV, D, generator=g) * 0.5
+    do = torch.randn(n, HQ, D, generator=g)
+    mask = make_attn_mask(n, n, [[0, n]], [[0, n]], [1])
+    # fp32 oracle pass for timing (the fp64 path is the parity oracle; fp32 is
+    # the fair CPU-throughput baseline)
+    t0 = time.perf_counter()
+    ref_attn_with_grads(q, k, v, mask, do, high_precision=False)
+    dt = time.perf_counter() - t0
+    area = n * (n + 1) // 2
+    flops = 4 * area * HQ * D * 3.5
+    return {
+        "value": flops / dt / 1e12,
+        "unit": "TFLOPS/s",
+        "cores": cores,
+        "kind": "port",
+        "sample": f"causal {n}-token slice of the 64k workload, fwd+bwd, "
+                  f"fp32 torch oracle, {dt:.1f}s",
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--skip-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    assert torch.cuda.is_available(), "bench.py needs an MI355X"
+    rank, world, ms_per_step, value, device, state = run_bench(args)
+
+    # all ranks participate (attn_fwd inside issues collectives); rank 0 reports
+    roofline = measure_roofline(state, device)
+    if rank == 0:
+        cpu_baseline = None if args.skip_cpu_baseline else measure_cpu_baseline()
+        line = {
+            "metric": "attention TFLOPS/sec fwd+bwd (whole-job; per-GPU = value/n_gpus); cp scaling at seqlen 64k",
+            "value": value,
+            "unit": "TFLOPS/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "workload": "dense_causal_64k_h32_d128",
+                "seqlen": SEQLEN,
+                "heads": HQ,
+                "kv_heads": HKV,
+                "head_dim": D,
+                "mask": "causal",
+                "global_batch": 1,
+                "parallelism": f"cp{world}",
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(line))
+
+
+if __name__ == "__main__":
+    main()
