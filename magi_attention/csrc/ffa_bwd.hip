@@ -1,28 +1,29 @@
-// ffa_bwd.hip — MI355X-native flex-flash-attention BACKWARD (gfx950).
+// ffa_bwd.hip — MI355X-native flex-flash-attention BACKWARD (gfx950), v1.
 //
-// Re-designed for CDNA4 (reference behaviour: flash_bwd_kernel_sm90.h:41 /
-// mainloop_bwd_sm90_tma_gmma_ws.hpp — 5-matmul recompute backward with fp32
-// atomic dQ (and dK/dV) accumulation; preprocess dPsum = rowsum(dO*O),
+// Behaviour: reference flash_bwd_kernel_sm90.h:41 /
+// mainloop_bwd_sm90_tma_gmma_ws.hpp (5-matmul recompute backward, fp32 atomic
+// dQ/dK/dV accumulation; preprocess dPsum = rowsum(dO*O),
 // flash_bwd_preprocess_kernel.h:42).
 //
-// Structure: ONE wave (64 threads) per (k-tile of 32 rows, range, q-head).
-// For its K/V tile the wave loops over the slice's q tiles of 32 rows:
-//   S^T  = K Q^T                        (swapped MFMA, like the fwd)
-//   P^T  = exp2(t - lse*log2e)          (recompute, base-2 like the fwd)
-//   dP^T = V dO^T
-//   dS^T = P^T (dP^T - dPsum) * dscale
-//   dV  += P^T dO     (P  transposed through a small LDS tile)
-//   dK  += dS^T Q     (dS transposed through the same LDS tile)
-//   dQ  += unsafeAtomicAdd(dS K)   (dS^T -> A-fragment in-register via permlane)
-// dK/dV accumulate in AGPRs across the q loop and are atomicAdd'd once.
+// Structure v1: ONE workgroup = 4 waves = 128 k rows (each wave one 32-row
+// k tile of the same range+head). The workgroup loops over the slice's q
+// tiles; per iteration Q[32][D] and dO[32][D] are staged ONCE into
+// XOR-swizzled LDS by all 256 threads (coalesced 16-B loads) and consumed by
+// all 4 waves — replacing v0's per-wave scalar global loads. Per wave:
+//   S^T  = K Q^T ; dP^T = V dO^T      (swapped MFMA, q-side operands from LDS)
+//   P^T, dS^T                          (base-2 recompute, like the fwd)
+//   dV  += P^T dO ; dK += dS^T Q       (P/dS transposed through per-wave LDS)
+//   dQ  += atomicAdd(dS K)             (dS^T -> A-frag in-register, permlane)
+// dK/dV accumulate in AGPRs across the q loop, atomic-added once at the end.
 
 #include <hip/hip_runtime.h>
 #include <math.h>
 
 #include "../../include/magi_ffa.h"
 
-#define BWD_BN 32  // k rows per wave
-#define BWD_BM 32  // q rows per inner tile
+#define BWD_BN 32    // k rows per wave
+#define BWD_BM 32    // q rows per tile
+#define BWD_WAVES 4  // k tiles per workgroup
 
 using f32x16 = __attribute__((ext_vector_type(16))) float;
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
@@ -39,9 +40,6 @@ DEV_INLINE unsigned pack_bf16_pair(float lo, float hi) {
   return ((unsigned)b.u << 16) | a.u;
 }
 
-// Rebuild an MFMA A-fragment (A[i=lo32][k=8*hi+e]) from values held in the
-// C/D layout (val[r] at row crow(r,hi), col lo32) — same cvt_pk +
-// permlane32_swap path as the forward kernel. tt selects k in [16*tt,16*tt+16).
 DEV_INLINE bf16x8 cframe_to_afrag(const float* val, int tt) {
   unsigned c0 = pack_bf16_pair(val[8 * tt + 0], val[8 * tt + 1]);
   unsigned c1 = pack_bf16_pair(val[8 * tt + 2], val[8 * tt + 3]);
@@ -73,7 +71,7 @@ struct BwdParams {
   const bf16_t* q;
   const bf16_t* k;
   const bf16_t* v;
-  const void* out;  // bf16 or f32 per OUT_F32
+  const void* out;
   const float* lse;
   float* dq;
   float* dk;
@@ -92,11 +90,10 @@ struct BwdParams {
 template <bool OUT_F32>
 __global__ __launch_bounds__(256) void bwd_preprocess_kernel(BwdParams p, int d,
                                                              long long n_rows) {
-  // one wave per (token, head) row
   const long long row = (long long)blockIdx.x * 4 + (threadIdx.x >> 6);
   if (row >= n_rows) return;
   const int lane = threadIdx.x & 63;
-  const int per = d / 64;  // elements per lane (1 for d=64, 2 for d=128)
+  const int per = d / 64;
   const bf16_t* do_p = p.dout + row * d;
   float acc = 0.f;
   for (int e = 0; e < per; ++e) {
@@ -114,23 +111,36 @@ __global__ __launch_bounds__(256) void bwd_preprocess_kernel(BwdParams p, int d,
   if (lane == 0) p.dpsum[row] = acc;
 }
 
+
 // ---------------- mainloop ----------------
 template <int D, bool HAS_SOFTCAP>
-__global__ __launch_bounds__(64) void ffa_bwd_kernel(BwdParams p) {
+__global__ __launch_bounds__(256, 1) void ffa_bwd_kernel(BwdParams p) {
   constexpr int DF = D / 16;
   constexpr int DT = D / 32;
+  constexpr int ROWB = D * 2;  // bytes per LDS tile row
+  // XOR swizzle within one LDS row: spreads a b128 lane group over the row's
+  // 16-B slots (T2/G4); mask keeps the XOR inside the row for D=64 too.
+  constexpr int SWZM = ROWB / 16 - 1;
+  auto swz = [](int row, int byte_off) {
+    return byte_off ^ ((row & SWZM) << 4);
+  };
   const int ri = blockIdx.y;
   const int h = blockIdx.z;
   const int ks = p.k_ranges[2 * ri], ke = p.k_ranges[2 * ri + 1];
-  const int n0 = ks + blockIdx.x * BWD_BN;
-  if (n0 >= ke) return;
+  const int nblk0 = ks + blockIdx.x * (BWD_BN * BWD_WAVES);
+  if (nblk0 >= ke) return;
   const int qs = p.q_ranges[2 * ri], qe = p.q_ranges[2 * ri + 1];
   if (qe <= qs) return;
   const int atype = p.attn_type_map ? p.attn_type_map[ri] : 0;
 
-  const int lane = threadIdx.x & 63;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
   const int lo32 = lane & 31;
   const int hi = lane >> 5;
+
+  const int n0 = nblk0 + wave * BWD_BN;  // this wave's k tile
+  const bool wave_live = n0 < ke;
 
   const float sl2 = HAS_SOFTCAP ? p.softcap * 1.4426950408889634f
                                 : p.scale * 1.4426950408889634f;
@@ -141,11 +151,14 @@ __global__ __launch_bounds__(64) void ffa_bwd_kernel(BwdParams p) {
   const size_t k_pitch = (size_t)p.hk * D;
   const size_t q_pitch = (size_t)p.hq * D;
 
-  const int krow = n0 + lo32;           // this lane's k row (A-layout row)
-  const bool kvalid = krow < ke;
-  const int kcl = kvalid ? krow : (ke - 1);
+  // LDS: Q tile + dO tile (XOR-swizzled rows) + per-wave transpose tiles
+  __shared__ __bf16 lds_q[BWD_BM * D];
+  __shared__ __bf16 lds_do[BWD_BM * D];
+  __shared__ __bf16 lds_p[BWD_WAVES][32][34];
 
-  // K/V fragments, A-layout (row lo32, d-slices by hi): 16B loads
+  // K/V fragments (A-layout) + K B-fragments, loaded once per block
+  const int krow = n0 + lo32;
+  const int kcl = min(krow, ke - 1);
   bf16x8 kfA[DF], vfA[DF];
   {
     const bf16_t* kp = p.k + (size_t)kcl * k_pitch + (size_t)kh * D;
@@ -156,7 +169,6 @@ __global__ __launch_bounds__(64) void ffa_bwd_kernel(BwdParams p) {
       vfA[dd] = *(const bf16x8*)(vp + dd * 16 + hi * 8);
     }
   }
-  // K as B-fragments for dQ += dS K: B[kk=8hi+e][j=d0+lo32] (column loads)
   union Bf {
     unsigned short u[8];
     bf16x8 v;
@@ -176,10 +188,16 @@ __global__ __launch_bounds__(64) void ffa_bwd_kernel(BwdParams p) {
       kfB[dt][tt] = b.v;
     }
 
-  // q loop bounds for this k tile
+  // block-level q loop bounds (union over the 4 waves' tiles)
   int q_lo = qs, q_hi = qe;
-  if (atype == 1 || atype == 3) q_lo = max(q_lo, n0 - (ke - qe));
-  if (atype == 2 || atype == 3) q_hi = min(q_hi, (n0 + BWD_BN - 1) - (ks - qs) + 1);
+  const int nlast = min(nblk0 + BWD_BN * BWD_WAVES, ke) - 1;
+  if (atype == 1 || atype == 3) q_lo = max(q_lo, nblk0 - (ke - qe));
+  if (atype == 2 || atype == 3) q_hi = min(q_hi, nlast - (ks - qs) + 1);
+  // wave-level bounds
+  int wq_lo = qs, wq_hi = qe;
+  if (atype == 1 || atype == 3) wq_lo = max(wq_lo, n0 - (ke - qe));
+  if (atype == 2 || atype == 3)
+    wq_hi = min(wq_hi, (n0 + BWD_BN - 1) - (ks - qs) + 1);
 
   f32x16 acc_dk[DT], acc_dv[DT];
 #pragma unroll
@@ -188,122 +206,137 @@ __global__ __launch_bounds__(64) void ffa_bwd_kernel(BwdParams p) {
     acc_dv[dt] = (f32x16)(0.f);
   }
 
-  // per-wave LDS transpose tile [32 k][32 q] bf16, row padded by 2 elems
-  __shared__ __bf16 p_lds[32][34];
-
   for (int m0 = q_lo; m0 < q_hi; m0 += BWD_BM) {
-    const int qrow = m0 + lo32;
-    const bool qvalid = qrow < q_hi;
-    const int qcl = min(qrow, qe - 1);
-    const bf16_t* qp = p.q + (size_t)qcl * q_pitch + (size_t)h * D;
-    const bf16_t* dop = p.dout + (size_t)qcl * q_pitch + (size_t)h * D;
-
-    // ---- S^T = K Q^T ; dP^T = V dO^T  (both [k=crow][q=lo32]) ----
-    f32x16 s = (f32x16)(0.f), dp = (f32x16)(0.f);
+    // ---- cooperative staging of Q/dO tile (swizzled 16-B chunks) ----
+    {
+      // 256 threads x 16B; chunks_per_row = D/8; rows_per_pass = 256/(D/8)
+      constexpr int CPR = D / 8;           // 16-B chunks per row
+      constexpr int RPP = 256 / CPR;       // rows staged per pass
+      const int row = tid / CPR;
+      const int col = tid % CPR;           // 16-B chunk index
 #pragma unroll
-    for (int dd = 0; dd < DF; ++dd) {
-      bf16x8 qf = *(const bf16x8*)(qp + dd * 16 + hi * 8);
-      bf16x8 dof = *(const bf16x8*)(dop + dd * 16 + hi * 8);
-      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfA[dd], qf, s, 0, 0, 0);
-      dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfA[dd], dof, dp, 0, 0, 0);
-    }
-
-    const float lse_q = qvalid ? p.lse[(size_t)qrow * p.hq + h] : INFINITY;
-    const float dpsum_q = qvalid ? p.dpsum[(size_t)qrow * p.hq + h] : 0.f;
-    const bool row_live = qvalid && lse_q != INFINITY && lse_q != -INFINITY;
-
-    // ---- P^T and dS^T (regs, C layout) ----
-    float pv[16], dsv[16];
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int kk = n0 + crow(r, hi);
-      bool ok = row_live && kk < ke;
-      if (atype == 1 || atype == 3) ok = ok && (kk - qrow <= ke - qe);
-      if (atype == 2 || atype == 3) ok = ok && (kk - qrow >= ks - qs);
-      float sv = s[r];
-      float dscale = p.scale;
-      float t;
-      if (HAS_SOFTCAP) {
-        const float th = tanhf(sv * cap_pre);
-        t = th * sl2;
-        dscale = p.scale * (1.f - th * th);
-      } else {
-        t = sv * sl2;
+      for (int pass = 0; pass < BWD_BM / RPP; ++pass) {
+        const int r = pass * RPP + row;
+        const int qrow = min(m0 + r, qe - 1);
+        const bf16_t* qp = p.q + (size_t)qrow * q_pitch + (size_t)h * D;
+        const bf16_t* dp = p.dout + (size_t)qrow * q_pitch + (size_t)h * D;
+        const int dst = swz(r, r * ROWB + col * 16);
+        *(bf16x8*)((char*)lds_q + dst) = *(const bf16x8*)(qp + col * 8);
+        *(bf16x8*)((char*)lds_do + dst) = *(const bf16x8*)(dp + col * 8);
       }
-      const float pij = ok ? exp2f(t - lse_q * log2e) : 0.f;
-      pv[r] = pij;
-      dsv[r] = pij * (dp[r] - dpsum_q) * dscale;
     }
+    __syncthreads();
 
-    // ---- dV += P^T dO : transpose P via LDS, dO as B-frag [q][d] ----
-    // LDS ops of one wave complete in order: no explicit wait needed between
-    // the ds_writes below and the ds_reads of the same tile.
+    if (wave_live && m0 + BWD_BM > wq_lo && m0 < wq_hi) {
+      const int qrow = m0 + lo32;
+      const bool qvalid = qrow < wq_hi && qrow < qe;
+
+      // ---- S^T = K Q^T ; dP^T = V dO^T (q-side B-frags from LDS rows) ----
+      f32x16 s = (f32x16)(0.f), dp = (f32x16)(0.f);
 #pragma unroll
-    for (int r = 0; r < 16; ++r) p_lds[crow(r, hi)][lo32] = (__bf16)pv[r];
+      for (int dd = 0; dd < DF; ++dd) {
+        const int off = swz(lo32, lo32 * ROWB + dd * 32 + hi * 16);
+        bf16x8 qf = *(const bf16x8*)((const char*)lds_q + off);
+        bf16x8 dof = *(const bf16x8*)((const char*)lds_do + off);
+        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfA[dd], qf, s, 0, 0, 0);
+        dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfA[dd], dof, dp, 0, 0, 0);
+      }
+
+      const float lse_q = qvalid ? p.lse[(size_t)qrow * p.hq + h] : INFINITY;
+      const float dpsum_q = qvalid ? p.dpsum[(size_t)qrow * p.hq + h] : 0.f;
+      const bool row_live = qvalid && lse_q != INFINITY && lse_q != -INFINITY;
+
+      float pv[16], dsv[16];
 #pragma unroll
-    for (int dt = 0; dt < DT; ++dt) {
-#pragma unroll
-      for (int tt = 0; tt < 2; ++tt) {
-        Bf b;
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const int row = min(m0 + 16 * tt + 8 * hi + e, qe - 1);
-          b.u[e] = *(const unsigned short*)(p.dout + (size_t)row * q_pitch +
-                                            (size_t)h * D + dt * 32 + lo32);
+      for (int r = 0; r < 16; ++r) {
+        const int kk = n0 + crow(r, hi);
+        bool ok = row_live && kk < ke;
+        if (atype == 1 || atype == 3) ok = ok && (kk - qrow <= ke - qe);
+        if (atype == 2 || atype == 3) ok = ok && (kk - qrow >= ks - qs);
+        float sv = s[r];
+        float dscale = p.scale;
+        float t;
+        if (HAS_SOFTCAP) {
+          const float th = tanhf(sv * cap_pre);
+          t = th * sl2;
+          dscale = p.scale * (1.f - th * th);
+        } else {
+          t = sv * sl2;
         }
-        // A-frag for k-rows: lane holds A[k=lo32][q-slice tt: 8hi+e]
-        bf16x8 pa = *(const bf16x8*)(&p_lds[lo32][0] + 16 * tt + 8 * hi);
-        acc_dv[dt] =
-            __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, b.v, acc_dv[dt], 0, 0, 0);
+        const float pij = ok ? exp2f(t - lse_q * log2e) : 0.f;
+        pv[r] = pij;
+        dsv[r] = pij * (dp[r] - dpsum_q) * dscale;
       }
-    }
 
-    // ---- dK += dS^T Q : transpose dS via LDS, Q as B-frag [q][d] ----
+      // ---- dV += P^T dO (P via per-wave LDS transpose; dO cols from LDS) ----
 #pragma unroll
-    for (int r = 0; r < 16; ++r) p_lds[crow(r, hi)][lo32] = (__bf16)dsv[r];
-#pragma unroll
-    for (int dt = 0; dt < DT; ++dt) {
-#pragma unroll
-      for (int tt = 0; tt < 2; ++tt) {
-        Bf b;
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const int row = min(m0 + 16 * tt + 8 * hi + e, qe - 1);
-          b.u[e] = *(const unsigned short*)(p.q + (size_t)row * q_pitch +
-                                            (size_t)h * D + dt * 32 + lo32);
-        }
-        bf16x8 dsa = *(const bf16x8*)(&p_lds[lo32][0] + 16 * tt + 8 * hi);
-        acc_dk[dt] =
-            __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, b.v, acc_dk[dt], 0, 0, 0);
-      }
-    }
-
-    // ---- dQ += dS K (atomicAdd): dS^T regs -> A-frag in-register ----
-    f32x16 acc_dq[DT];
-#pragma unroll
-    for (int dt = 0; dt < DT; ++dt) acc_dq[dt] = (f32x16)(0.f);
-#pragma unroll
-    for (int tt = 0; tt < 2; ++tt) {
-      bf16x8 dsa = cframe_to_afrag(dsv, tt);
-#pragma unroll
-      for (int dt = 0; dt < DT; ++dt)
-        acc_dq[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, kfB[dt][tt],
-                                                             acc_dq[dt], 0, 0, 0);
-    }
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int qr = m0 + crow(r, hi);
-      if (qr >= q_hi) continue;
-      float* dst = p.dq + (size_t)qr * q_pitch + (size_t)h * D;
+      for (int r = 0; r < 16; ++r) lds_p[wave][crow(r, hi)][lo32] = (__bf16)pv[r];
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
-        const float val = acc_dq[dt][r];
-        if (val != 0.f) unsafeAtomicAdd(dst + dt * 32 + lo32, val);
+#pragma unroll
+        for (int tt = 0; tt < 2; ++tt) {
+          Bf b;
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            const int r = 16 * tt + 8 * hi + e;
+            b.u[e] = *(const unsigned short*)(
+                (const char*)lds_do + swz(r, r * ROWB + (dt * 32 + lo32) * 2));
+          }
+          bf16x8 pa = *(const bf16x8*)(&lds_p[wave][lo32][0] + 16 * tt + 8 * hi);
+          acc_dv[dt] =
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, b.v, acc_dv[dt], 0, 0, 0);
+        }
+      }
+
+      // ---- dK += dS^T Q (dS via LDS transpose; Q cols from LDS) ----
+#pragma unroll
+      for (int r = 0; r < 16; ++r) lds_p[wave][crow(r, hi)][lo32] = (__bf16)dsv[r];
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+#pragma unroll
+        for (int tt = 0; tt < 2; ++tt) {
+          Bf b;
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            const int r = 16 * tt + 8 * hi + e;
+            b.u[e] = *(const unsigned short*)(
+                (const char*)lds_q + swz(r, r * ROWB + (dt * 32 + lo32) * 2));
+          }
+          bf16x8 dsa = *(const bf16x8*)(&lds_p[wave][lo32][0] + 16 * tt + 8 * hi);
+          acc_dk[dt] =
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, b.v, acc_dk[dt], 0, 0, 0);
+        }
+      }
+
+      // ---- dQ += dS K (atomicAdd; dS^T -> A-frag in-register) ----
+      f32x16 acc_dq[DT];
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) acc_dq[dt] = (f32x16)(0.f);
+#pragma unroll
+      for (int tt = 0; tt < 2; ++tt) {
+        bf16x8 dsa = cframe_to_afrag(dsv, tt);
+#pragma unroll
+        for (int dt = 0; dt < DT; ++dt)
+          acc_dq[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              dsa, kfB[dt][tt], acc_dq[dt], 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qr = m0 + crow(r, hi);
+        if (qr >= qe) continue;
+        float* dst = p.dq + (size_t)qr * q_pitch + (size_t)h * D;
+#pragma unroll
+        for (int dt = 0; dt < DT; ++dt) {
+          const float val = acc_dq[dt][r];
+          if (val != 0.f) unsafeAtomicAdd(dst + dt * 32 + lo32, val);
+        }
       }
     }
+    __syncthreads();
   }
 
-  // ---- write dK/dV: atomicAdd (k_ranges may overlap; GQA heads collide) ----
+  // ---- write dK/dV ----
+  if (!wave_live) return;
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     const int kr = n0 + crow(r, hi);
@@ -366,8 +399,9 @@ extern "C" int magi_ffa_bwd(const magi_ffa_bwd_args* a) {
   p.total_q = a->total_q;
   p.total_k = a->total_k;
 
-  const int nblocks = (a->max_seqlen_k + BWD_BN - 1) / BWD_BN;
-  dim3 grid(nblocks, (unsigned)a->n_ranges, a->hq), block(64);
+  const int span = BWD_BN * BWD_WAVES;
+  const int nblocks = (a->max_seqlen_k + span - 1) / span;
+  dim3 grid(nblocks, (unsigned)a->n_ranges, a->hq), block(64 * BWD_WAVES);
   hipStream_t s = (hipStream_t)a->stream;
   const bool sc = a->softcap > 0.f;
   if (a->d == 64) {
