@@ -18,6 +18,7 @@
 
 #include <hip/hip_runtime.h>
 #include <math.h>
+#include <stdlib.h>
 
 #include "../../include/magi_ffa.h"
 
@@ -84,6 +85,7 @@ struct BwdParams {
   float scale;
   float softcap;
   long long total_q, total_k;
+  int debug_ablate;  // perf ablation only: 1=skip dq stores, 2=skip dkv stores
 };
 
 // ---------------- preprocess: dpsum = rowsum(dO * O) ----------------
@@ -323,7 +325,7 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_kernel(BwdParams p) {
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qr = m0 + crow(r, hi);
-        if (qr >= qe) continue;
+        if (qr >= qe || (p.debug_ablate & 1)) continue;
         float* dst = p.dq + (size_t)qr * q_pitch + (size_t)h * D;
 #pragma unroll
         for (int dt = 0; dt < DT; ++dt) {
@@ -336,7 +338,7 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_kernel(BwdParams p) {
   }
 
   // ---- write dK/dV ----
-  if (!wave_live) return;
+  if (!wave_live || (p.debug_ablate & 2)) return;
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     const int kr = n0 + crow(r, hi);
@@ -398,6 +400,7 @@ extern "C" int magi_ffa_bwd(const magi_ffa_bwd_args* a) {
   p.softcap = a->softcap;
   p.total_q = a->total_q;
   p.total_k = a->total_k;
+  { const char* e = getenv("MAGI_BWD_ABLATE"); p.debug_ablate = e ? atoi(e) : 0; }
 
   const int span = BWD_BN * BWD_WAVES;
   const int nblocks = (a->max_seqlen_k + span - 1) / span;
