@@ -127,10 +127,13 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_kernel(BwdParams p) {
   auto swz = [](int row, int byte_off) {
     return byte_off ^ ((row & SWZM) << 4);
   };
-  const int ri = blockIdx.y;
-  const int h = blockIdx.z;
+  // blockIdx.x = HEAD: the dispatcher places block b on XCD b%8, so every
+  // block of one head shares one XCD/L2 — dq/dk/dv atomics stay XCD-local and
+  // K/V reads of a head hit a single L2 (T1 XCD-affinity via grid layout).
+  const int ri = blockIdx.z;
+  const int h = blockIdx.x;
   const int ks = p.k_ranges[2 * ri], ke = p.k_ranges[2 * ri + 1];
-  const int nblk0 = ks + blockIdx.x * (BWD_BN * BWD_WAVES);
+  const int nblk0 = ks + blockIdx.y * (BWD_BN * BWD_WAVES);
   if (nblk0 >= ke) return;
   const int qs = p.q_ranges[2 * ri], qe = p.q_ranges[2 * ri + 1];
   if (qe <= qs) return;
@@ -410,7 +413,8 @@ extern "C" int magi_ffa_bwd(const magi_ffa_bwd_args* a) {
 
   const int span = BWD_BN * BWD_WAVES;
   const int nblocks = (a->max_seqlen_k + span - 1) / span;
-  dim3 grid(nblocks, (unsigned)a->n_ranges, a->hq), block(64 * BWD_WAVES);
+  if (a->n_ranges > 65535) return -5;
+  dim3 grid(a->hq, nblocks, (unsigned)a->n_ranges), block(64 * BWD_WAVES);
   hipStream_t s = (hipStream_t)a->stream;
   const bool sc = a->softcap > 0.f;
   if (a->d == 64) {
