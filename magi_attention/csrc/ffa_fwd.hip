@@ -71,12 +71,14 @@ template <int D, bool HAS_SOFTCAP, bool ATOMIC, bool OUT_BF16>
 __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
   constexpr int DF = D / 16;    // # of 16-wide d fragments
   constexpr int DT = D / 32;    // # of 32-wide output d tiles
-  // blockIdx.x = HEAD: all blocks of one head share one XCD/L2 (K/V reuse,
-  // XCD-local lock/merge traffic) — see bwd kernel note.
-  const int ri = blockIdx.z;
-  const int h = blockIdx.x;
+  // grid = (m_blocks, ranges, heads): consecutive blocks are consecutive
+  // m-blocks, so all XCDs stream the same K prefix window concurrently
+  // (head-major XCD affinity measured SLOWER for fwd: K/V of one head
+  // overflows one XCD's 4 MB L2 at long seqlen).
+  const int ri = blockIdx.y;
+  const int h = blockIdx.z;
   const int qs = p.q_ranges[2 * ri], qe = p.q_ranges[2 * ri + 1];
-  const int m0 = qs + blockIdx.y * FFA_BM;
+  const int m0 = qs + blockIdx.x * FFA_BM;
   if (m0 >= qe) return;                       // uniform across block
   const int ks = p.k_ranges[2 * ri], ke = p.k_ranges[2 * ri + 1];
   const int atype = p.attn_type_map ? p.attn_type_map[ri] : 0;
@@ -436,7 +438,7 @@ extern "C" int magi_ffa_fwd(const magi_ffa_fwd_args* a) {
 
   const int mblocks = (a->max_seqlen_q + FFA_BM - 1) / FFA_BM;
   if (a->n_ranges > 65535) return -5;
-  dim3 grid(a->hq, mblocks, (unsigned)a->n_ranges);
+  dim3 grid(mblocks, (unsigned)a->n_ranges, a->hq);
   dim3 block(256);
   hipStream_t stream = (hipStream_t)a->stream;
   if (a->d == 64) return launch_fwd_d<64>(a, p, grid, block, stream);
