@@ -61,6 +61,7 @@ struct FwdParams {
   const int* attn_type_map;
   int* locks;
   int hq, hk, gqa;    // gqa = hq / hk
+  int head_major;     // 1: grid.x = head (XCD-affine; per-head KV fits L2)
   int n_lock_slots;
   float scale;        // softmax_scale
   float softcap;
@@ -71,14 +72,14 @@ template <int D, bool HAS_SOFTCAP, bool ATOMIC, bool OUT_BF16>
 __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
   constexpr int DF = D / 16;    // # of 16-wide d fragments
   constexpr int DT = D / 32;    // # of 32-wide output d tiles
-  // grid = (m_blocks, ranges, heads): consecutive blocks are consecutive
-  // m-blocks, so all XCDs stream the same K prefix window concurrently
-  // (head-major XCD affinity measured SLOWER for fwd: K/V of one head
-  // overflows one XCD's 4 MB L2 at long seqlen).
-  const int ri = blockIdx.y;
-  const int h = blockIdx.z;
+  // Adaptive grid: head-major (blockIdx.x = head -> one XCD per head: L2
+  // locality for K/V and merge traffic) when one head's K/V fits an XCD's
+  // 4 MB L2; m-block-major otherwise (all XCDs stream the same K window).
+  const int ri = p.head_major ? blockIdx.z : blockIdx.y;
+  const int h = p.head_major ? blockIdx.x : blockIdx.z;
+  const int mb = p.head_major ? blockIdx.y : blockIdx.x;
   const int qs = p.q_ranges[2 * ri], qe = p.q_ranges[2 * ri + 1];
-  const int m0 = qs + blockIdx.x * FFA_BM;
+  const int m0 = qs + mb * FFA_BM;
   if (m0 >= qe) return;                       // uniform across block
   const int ks = p.k_ranges[2 * ri], ke = p.k_ranges[2 * ri + 1];
   const int atype = p.attn_type_map ? p.attn_type_map[ri] : 0;
@@ -91,6 +92,7 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
   const int q0 = m0 + wave * 32;
   const int qrow = q0 + lo32;
   const bool qvalid = (qrow < qe) && (q0 < qe);
+  const bool qvalid_any = q0 < qe;
   const int qclamp = qvalid ? qrow : (qe - 1);
 
   // softmax scales (reference mainloop_fwd...hpp:466-489)
@@ -107,6 +109,22 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
   } else {
     n_hi = n_lo;  // empty
   }
+  // block-level union of the waves' bounds (staging is cooperative)
+  int b_lo = ks, b_hi = ke;
+  if (ke > ks && m0 < qe) {
+    const int qhib = min(m0 + FFA_BM - 1, qe - 1);
+    if (atype == 1 || atype == 3) b_hi = min(b_hi, qhib + (ke - qe) + 1);
+    if (atype == 2 || atype == 3) b_lo = max(b_lo, m0 + (ks - qs));
+  } else {
+    b_hi = b_lo;
+  }
+  constexpr int ROWB = D * 2;
+  constexpr int SWZM = ROWB / 16 - 1;
+  auto swz = [](int row, int byte_off) {
+    return byte_off ^ ((row & SWZM) << 4);
+  };
+  __shared__ __bf16 lds_k[FFA_BN * D];
+  __shared__ __bf16 lds_vt[D][40];
 
   // Q fragments in registers (8 x bf16x8 for D=128)
   bf16x8 qf[DF];
@@ -128,16 +146,39 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
 #pragma unroll
   for (int dt = 0; dt < DT; ++dt) acc_o[dt] = (f32x16)(0.f);
 
-  for (int n0 = n_lo; n0 < n_hi; n0 += FFA_BN) {
-    // ---- K fragments + QK^T (swapped: A=K rows, B=Q cols) ----
-    const int krow = n0 + lo32;
-    const int kcl = min(krow, ke - 1);
-    const bf16_t* kp = kbase + (size_t)kcl * k_pitch;
-    f32x16 s = (f32x16)(0.f);
+  for (int n0 = b_lo; n0 < b_hi; n0 += FFA_BN) {
+    // ---- cooperative staging: K rows (swizzled) + transposed V ----
+    {
+      constexpr int CPR = D / 8;       // 16-B chunks per row
+      constexpr int RPP = 256 / CPR;   // rows per pass
+      const int srow = threadIdx.x / CPR;
+      const int scol = threadIdx.x % CPR;
 #pragma unroll
-    for (int dd = 0; dd < DF; ++dd) {
-      bf16x8 kf = *(const bf16x8*)(kp + dd * 16 + hi * 8);
-      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[dd], s, 0, 0, 0);
+      for (int pass = 0; pass < FFA_BN / RPP; ++pass) {
+        const int r = pass * RPP + srow;
+        const int kr = min(n0 + r, ke - 1);
+        *(bf16x8*)((char*)lds_k + swz(r, r * ROWB + scol * 16)) =
+            *(const bf16x8*)(kbase + (size_t)kr * k_pitch + scol * 8);
+        const bf16x8 vv = *(const bf16x8*)(vbase + (size_t)kr * k_pitch + scol * 8);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) lds_vt[scol * 8 + e][r] = vv[e];
+      }
+    }
+    __syncthreads();
+    const bool live = (n0 + FFA_BN > n_lo) && (n0 < n_hi) && qvalid_any;
+
+    f32x16 s = (f32x16)(0.f);
+    if (live) {
+#pragma unroll
+      for (int dd = 0; dd < DF; ++dd) {
+        bf16x8 kf = *(const bf16x8*)(
+            (const char*)lds_k + swz(lo32, lo32 * ROWB + dd * 32 + hi * 16));
+        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[dd], s, 0, 0, 0);
+      }
+    }
+    if (!live) {
+      __syncthreads();
+      continue;
     }
 
     // ---- mask + scale into exp2 domain ----
@@ -213,27 +254,17 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
       pa[tt] = cvt.v;
     }
 
-    // ---- PV: O[32q][32d] += P^T V per d-tile ----
+    // ---- PV: O[32q][32d] += P^T V (B-frags from the transposed V tile) ----
 #pragma unroll
     for (int tt = 0; tt < 2; ++tt) {
-      const int vr = n0 + 16 * tt + 8 * hi;
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
-        // B-fragment: lane reads V[vr + e][dt*32 + lo32], e = 0..7
-        union {
-          unsigned short u[8];
-          bf16x8 v;
-        } bv;
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const int row = min(vr + e, ke - 1);
-          bv.u[e] = *(const unsigned short*)(vbase + (size_t)row * k_pitch +
-                                             dt * 32 + lo32);
-        }
+        bf16x8 bv = *(const bf16x8*)(&lds_vt[dt * 32 + lo32][16 * tt + 8 * hi]);
         acc_o[dt] =
-            __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[tt], bv.v, acc_o[dt], 0, 0, 0);
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[tt], bv, acc_o[dt], 0, 0, 0);
       }
     }
+    __syncthreads();
   }
 
   // ======================= epilogue =======================
@@ -438,7 +469,11 @@ extern "C" int magi_ffa_fwd(const magi_ffa_fwd_args* a) {
 
   const int mblocks = (a->max_seqlen_q + FFA_BM - 1) / FFA_BM;
   if (a->n_ranges > 65535) return -5;
-  dim3 grid(mblocks, (unsigned)a->n_ranges, a->hq);
+  // head-major XCD affinity pays when one head's K+V fits a 4 MB XCD L2
+  p.head_major = ((long long)a->total_k * a->d * 4 <= (4 << 20)) ? 1 : 0;
+  dim3 grid = p.head_major
+                  ? dim3(a->hq, mblocks, (unsigned)a->n_ranges)
+                  : dim3(mblocks, (unsigned)a->n_ranges, a->hq);
   dim3 block(256);
   hipStream_t stream = (hipStream_t)a->stream;
   if (a->d == 64) return launch_fwd_d<64>(a, p, grid, block, stream);
