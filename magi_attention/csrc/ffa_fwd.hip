@@ -123,8 +123,8 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
   auto swz = [](int row, int byte_off) {
     return byte_off ^ ((row & SWZM) << 4);
   };
-  __shared__ __bf16 lds_k[FFA_BN * D];
-  __shared__ __bf16 lds_vt[D][40];
+  __shared__ __bf16 lds_k[2][FFA_BN * D];
+  __shared__ __bf16 lds_vt[2][D][40];
 
   // Q fragments in registers (8 x bf16x8 for D=128)
   bf16x8 qf[DF];
@@ -146,25 +146,43 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
 #pragma unroll
   for (int dt = 0; dt < DT; ++dt) acc_o[dt] = (f32x16)(0.f);
 
-  for (int n0 = b_lo; n0 < b_hi; n0 += FFA_BN) {
-    // ---- cooperative staging: K rows (swizzled) + transposed V ----
-    {
-      constexpr int CPR = D / 8;       // 16-B chunks per row
-      constexpr int RPP = 256 / CPR;   // rows per pass
-      const int srow = threadIdx.x / CPR;
-      const int scol = threadIdx.x % CPR;
+  // ---- double-buffered staging pipeline (T14: issue loads for tile t+1
+  // into registers during tile t's compute; one barrier per iteration) ----
+  constexpr int CPR = D / 8;       // 16-B chunks per row
+  constexpr int RPP = 256 / CPR;   // rows per pass
+  constexpr int NPASS = FFA_BN / RPP;
+  const int srow = threadIdx.x / CPR;
+  const int scol = threadIdx.x % CPR;
+  bf16x8 kreg[NPASS], vreg[NPASS];
+
+  auto issue_loads = [&](int n0) {
 #pragma unroll
-      for (int pass = 0; pass < FFA_BN / RPP; ++pass) {
-        const int r = pass * RPP + srow;
-        const int kr = min(n0 + r, ke - 1);
-        *(bf16x8*)((char*)lds_k + swz(r, r * ROWB + scol * 16)) =
-            *(const bf16x8*)(kbase + (size_t)kr * k_pitch + scol * 8);
-        const bf16x8 vv = *(const bf16x8*)(vbase + (size_t)kr * k_pitch + scol * 8);
-#pragma unroll
-        for (int e = 0; e < 8; ++e) lds_vt[scol * 8 + e][r] = vv[e];
-      }
+    for (int pass = 0; pass < NPASS; ++pass) {
+      const int kr = min(n0 + pass * RPP + srow, ke - 1);
+      kreg[pass] = *(const bf16x8*)(kbase + (size_t)kr * k_pitch + scol * 8);
+      vreg[pass] = *(const bf16x8*)(vbase + (size_t)kr * k_pitch + scol * 8);
     }
-    __syncthreads();
+  };
+  auto write_stage = [&](int buf) {
+#pragma unroll
+    for (int pass = 0; pass < NPASS; ++pass) {
+      const int r = pass * RPP + srow;
+      *(bf16x8*)((char*)lds_k[buf] + swz(r, r * ROWB + scol * 16)) = kreg[pass];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) lds_vt[buf][scol * 8 + e][r] = vreg[pass][e];
+    }
+  };
+
+  int cur = 0;
+  if (b_lo < b_hi) {
+    issue_loads(b_lo);
+    write_stage(0);
+  }
+  __syncthreads();
+
+  for (int n0 = b_lo; n0 < b_hi; n0 += FFA_BN) {
+    const bool has_next = n0 + FFA_BN < b_hi;
+    if (has_next) issue_loads(n0 + FFA_BN);  // lands under this tile's compute
     const bool live = (n0 + FFA_BN > n_lo) && (n0 < n_hi) && qvalid_any;
 
     f32x16 s = (f32x16)(0.f);
@@ -172,12 +190,14 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
 #pragma unroll
       for (int dd = 0; dd < DF; ++dd) {
         bf16x8 kf = *(const bf16x8*)(
-            (const char*)lds_k + swz(lo32, lo32 * ROWB + dd * 32 + hi * 16));
+            (const char*)lds_k[cur] + swz(lo32, lo32 * ROWB + dd * 32 + hi * 16));
         s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[dd], s, 0, 0, 0);
       }
     }
     if (!live) {
+      if (has_next) write_stage(cur ^ 1);
       __syncthreads();
+      cur ^= 1;
       continue;
     }
 
@@ -259,12 +279,14 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
     for (int tt = 0; tt < 2; ++tt) {
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
-        bf16x8 bv = *(const bf16x8*)(&lds_vt[dt * 32 + lo32][16 * tt + 8 * hi]);
+        bf16x8 bv = *(const bf16x8*)(&lds_vt[cur][dt * 32 + lo32][16 * tt + 8 * hi]);
         acc_o[dt] =
             __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[tt], bv, acc_o[dt], 0, 0, 0);
       }
     }
+    if (has_next) write_stage(cur ^ 1);  // other buffer: overlaps this compute
     __syncthreads();
+    cur ^= 1;
   }
 
   // ======================= epilogue =======================
