@@ -430,8 +430,9 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dq_kernel(BwdParams p) {
         const int r = pass * RPP + row;
         const int src = swz(r, r * ROWB + col * 16);
         const bf16x8 kv8 = *(const bf16x8*)((const char*)lds_k + src);
+        const int bs = (col & 3) << 3;  // bank-spread swizzle (see fwd lds_vt)
 #pragma unroll
-        for (int e = 0; e < 8; ++e) lds_kt[col * 8 + e][r] = kv8[e];
+        for (int e = 0; e < 8; ++e) lds_kt[col * 8 + e][r ^ bs] = kv8[e];
       }
     }
     __syncthreads();
@@ -474,8 +475,9 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dq_kernel(BwdParams p) {
       bf16x8 dsa1 = cframe_to_afrag(dsv, 1);
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
-        bf16x8 b0 = *(const bf16x8*)(&lds_kt[dt * 32 + lo32][8 * hi]);
-        bf16x8 b1 = *(const bf16x8*)(&lds_kt[dt * 32 + lo32][16 + 8 * hi]);
+        const int kbs = ((lo32 >> 3) & 3) << 3;
+        bf16x8 b0 = *(const bf16x8*)(&lds_kt[dt * 32 + lo32][(8 * hi) ^ kbs]);
+        bf16x8 b1 = *(const bf16x8*)(&lds_kt[dt * 32 + lo32][(16 + 8 * hi) ^ kbs]);
         acc_dq[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa0, b0, acc_dq[dt], 0, 0, 0);
         acc_dq[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa1, b1, acc_dq[dt], 0, 0, 0);
       }

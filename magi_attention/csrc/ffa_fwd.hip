@@ -168,8 +168,13 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
     for (int pass = 0; pass < NPASS; ++pass) {
       const int r = pass * RPP + srow;
       *(bf16x8*)((char*)lds_k[buf] + swz(r, r * ROWB + scol * 16)) = kreg[pass];
+      // 8-block XOR swizzle on the k index: spreads the 16 column groups
+      // over distinct banks (unswizzled, every scol group hit the same bank:
+      // SQ_LDS_BANK_CONFLICT was 24% of fwd wave cycles)
+      const int bs = (scol & 3) << 3;
 #pragma unroll
-      for (int e = 0; e < 8; ++e) lds_vt[buf][scol * 8 + e][r] = vreg[pass][e];
+      for (int e = 0; e < 8; ++e)
+        lds_vt[buf][scol * 8 + e][r ^ bs] = vreg[pass][e];
     }
   };
 
@@ -279,7 +284,9 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
     for (int tt = 0; tt < 2; ++tt) {
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
-        bf16x8 bv = *(const bf16x8*)(&lds_vt[cur][dt * 32 + lo32][16 * tt + 8 * hi]);
+        bf16x8 bv = *(const bf16x8*)(
+            &lds_vt[cur][dt * 32 + lo32]
+                   [(16 * tt + 8 * hi) ^ (((lo32 >> 3) & 3) << 3)]);
         acc_o[dt] =
             __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[tt], bv, acc_o[dt], 0, 0, 0);
       }
