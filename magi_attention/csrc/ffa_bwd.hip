@@ -160,14 +160,23 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
   const size_t k_pitch = (size_t)p.hk * D;
   const size_t q_pitch = (size_t)p.hq * D;
 
-  // LDS: Q/dO tiles (XOR-swizzled rows) + their transposed copies (padded to
-  // 40-elem rows so b128 reads stay 16-B aligned and bank-spread) + per-wave
-  // P/dS transpose tiles + the block-combined dQ tile (4 waves' contributions
-  // summed in LDS, flushed by ONE global-atomic pass -> 4x fewer HBM atomics;
-  // ablation showed dq atomics were 45% of bwd time).
-  __shared__ __bf16 lds_q[BWD_BM * D];
-  __shared__ __bf16 lds_do[BWD_BM * D];
-  __shared__ __bf16 lds_p[BWD_WAVES][32][34];
+  // LDS: double-buffered Q/dO tiles (XOR-swizzled rows, filled by
+  // global_load_lds with the swizzle on the SOURCE address — rule 21) +
+  // per-wave P/dS transpose tiles. One barrier per iteration; the glds for
+  // tile t+1 flies under tile t's compute and is drained by the next
+  // barrier's implicit vmcnt(0).
+  // ONE shared object only: a second __shared__ array makes hipcc drain
+  // vmcnt(0) before every ds_read, destroying the glds pipeline (guide §5
+  // ".s-level traps" (a)).
+  __shared__ __attribute__((aligned(16))) char smem[
+      2 * 2 * BWD_BM * D * 2 + BWD_WAVES * 32 * 34 * 2];
+  auto lds_q = [&](int buf) -> __bf16* {
+    return (__bf16*)(smem + buf * BWD_BM * D * 2);
+  };
+  auto lds_do = [&](int buf) -> __bf16* {
+    return (__bf16*)(smem + (2 + buf) * BWD_BM * D * 2);
+  };
+  __bf16(*lds_p)[32][34] = (__bf16(*)[32][34])(smem + 4 * BWD_BM * D * 2);
 
   // K/V fragments (A-layout) + K B-fragments, loaded once per block
   const int krow = n0 + lo32;
@@ -205,26 +214,37 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
     acc_dv[dt] = (f32x16)(0.f);
   }
 
-  for (int m0 = q_lo; m0 < q_hi; m0 += BWD_BM) {
-    // ---- cooperative staging of Q/dO tile (swizzled 16-B chunks) ----
-    {
-      // 256 threads x 16B; chunks_per_row = D/8; rows_per_pass = 256/(D/8)
-      constexpr int CPR = D / 8;           // 16-B chunks per row
-      constexpr int RPP = 256 / CPR;       // rows staged per pass
-      const int row = tid / CPR;
-      const int col = tid % CPR;           // 16-B chunk index
+  // each wave's glds covers 4 rows (64 lanes x 16B = 1 KiB = 4 rows at D=128);
+  // wave w owns rows [8w, 8w+8) via 2 issues per tensor
+  constexpr int ROWS_PER_GLDS = 1024 / ROWB;
+  constexpr int GLDS_PER_WAVE = (BWD_BM / BWD_WAVES) / ROWS_PER_GLDS;
+  auto stage_glds = [&](int buf, int m0x) {
 #pragma unroll
-      for (int pass = 0; pass < BWD_BM / RPP; ++pass) {
-        const int r = pass * RPP + row;
-        const int qrow = min(m0 + r, qe - 1);
-        const bf16_t* qp = p.q + (size_t)qrow * q_pitch + (size_t)h * D;
-        const bf16_t* dp = p.dout + (size_t)qrow * q_pitch + (size_t)h * D;
-        const int dst = swz(r, r * ROWB + col * 16);
-        *(bf16x8*)((char*)lds_q + dst) = *(const bf16x8*)(qp + col * 8);
-        *(bf16x8*)((char*)lds_do + dst) = *(const bf16x8*)(dp + col * 8);
-      }
+    for (int gi = 0; gi < GLDS_PER_WAVE; ++gi) {
+      const int r0 = (BWD_BM / BWD_WAVES) * wave + ROWS_PER_GLDS * gi;
+      const int r = r0 + lane / (ROWB / 16);
+      const int c = lane % (ROWB / 16);
+      const int qrow = min(m0x + r, qe - 1);
+      const int csw = (c ^ (r & SWZM)) * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              p.q + (size_t)qrow * q_pitch + (size_t)h * D + csw),
+          (__attribute__((address_space(3))) unsigned int*)&lds_q(buf)[r0 * D],
+          16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              p.dout + (size_t)qrow * q_pitch + (size_t)h * D + csw),
+          (__attribute__((address_space(3))) unsigned int*)&lds_do(buf)[r0 * D],
+          16, 0, 0);
     }
-    __syncthreads();
+  };
+
+  int cur = 0;
+  if (q_lo < q_hi) stage_glds(0, q_lo);
+
+  for (int m0 = q_lo; m0 < q_hi; m0 += BWD_BM) {
+    __syncthreads();  // buf[cur] ready (barrier drains in-flight glds)
+    if (m0 + BWD_BM < q_hi) stage_glds(cur ^ 1, m0 + BWD_BM);
 
     if (wave_live && m0 + BWD_BM > wq_lo && m0 < wq_hi) {
       const int qrow = m0 + lo32;
@@ -235,8 +255,8 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
 #pragma unroll
       for (int dd = 0; dd < DF; ++dd) {
         const int off = swz(lo32, lo32 * ROWB + dd * 32 + hi * 16);
-        bf16x8 qf = *(const bf16x8*)((const char*)lds_q + off);
-        bf16x8 dof = *(const bf16x8*)((const char*)lds_do + off);
+        bf16x8 qf = *(const bf16x8*)((const char*)lds_q(cur) + off);
+        bf16x8 dof = *(const bf16x8*)((const char*)lds_do(cur) + off);
         s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfA[dd], qf, s, 0, 0, 0);
         dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfA[dd], dof, dp, 0, 0, 0);
       }
@@ -279,7 +299,7 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
           for (int e = 0; e < 8; ++e) {
             const int r = 16 * tt + 8 * hi + e;
             b.u[e] = *(const unsigned short*)(
-                (const char*)lds_do + swz(r, r * ROWB + (dt * 32 + lo32) * 2));
+                (const char*)lds_do(cur) + swz(r, r * ROWB + (dt * 32 + lo32) * 2));
           }
           bf16x8 pa = *(const bf16x8*)(&lds_p[wave][lo32][0] + 16 * tt + 8 * hi);
           acc_dv[dt] =
@@ -299,7 +319,7 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
           for (int e = 0; e < 8; ++e) {
             const int r = 16 * tt + 8 * hi + e;
             b.u[e] = *(const unsigned short*)(
-                (const char*)lds_q + swz(r, r * ROWB + (dt * 32 + lo32) * 2));
+                (const char*)lds_q(cur) + swz(r, r * ROWB + (dt * 32 + lo32) * 2));
           }
           bf16x8 dsa = *(const bf16x8*)(&lds_p[wave][lo32][0] + 16 * tt + 8 * hi);
           acc_dk[dt] =
@@ -308,7 +328,7 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
       }
 
     }
-    __syncthreads();
+    cur ^= 1;
   }
 
   // ---- write dK/dV ----
