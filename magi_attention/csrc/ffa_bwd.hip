@@ -394,9 +394,18 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dq_kernel(BwdParams p) {
   const size_t k_pitch = (size_t)p.hk * D;
   const size_t q_pitch = (size_t)p.hq * D;
 
-  __shared__ __bf16 lds_k[BWD_BN * D];
-  __shared__ __bf16 lds_v[BWD_BN * D];
-  __shared__ __bf16 lds_kt[D][40];
+  // single shared object (glds-pipeline trap, see dkv kernel)
+  __shared__ __attribute__((aligned(16))) char smem[
+      2 * 2 * BWD_BN * D * 2 + 2 * D * 40 * 2];
+  auto lds_k = [&](int buf) -> __bf16* {
+    return (__bf16*)(smem + buf * BWD_BN * D * 2);
+  };
+  auto lds_v = [&](int buf) -> __bf16* {
+    return (__bf16*)(smem + (2 + buf) * BWD_BN * D * 2);
+  };
+  auto lds_kt = [&](int buf) -> __bf16(*)[40] {
+    return (__bf16(*)[40])(smem + 4 * BWD_BN * D * 2 + buf * D * 40 * 2);
+  };
 
   // persistent per-wave operands: Q and dO fragments (B-layout rows)
   bf16x8 qf[DF], dof[DF];
@@ -427,32 +436,49 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dq_kernel(BwdParams p) {
 #pragma unroll
   for (int dt = 0; dt < DT; ++dt) acc_dq[dt] = (f32x16)(0.f);
 
+  constexpr int ROWS_PER_GLDS = 1024 / ROWB;
+  constexpr int GLDS_PER_WAVE = (BWD_BN / BWD_WAVES) / ROWS_PER_GLDS;
+  auto stage_glds = [&](int buf, int n0x) {
+#pragma unroll
+    for (int gi = 0; gi < GLDS_PER_WAVE; ++gi) {
+      const int r0 = (BWD_BN / BWD_WAVES) * wave + ROWS_PER_GLDS * gi;
+      const int r = r0 + lane / (ROWB / 16);
+      const int c = lane % (ROWB / 16);
+      const int kr = min(n0x + r, ke - 1);
+      const int csw = (c ^ (r & SWZM)) * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              p.k + (size_t)kr * k_pitch + (size_t)kh * D + csw),
+          (__attribute__((address_space(3))) unsigned int*)&lds_k(buf)[r0 * D],
+          16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              p.v + (size_t)kr * k_pitch + (size_t)kh * D + csw),
+          (__attribute__((address_space(3))) unsigned int*)&lds_v(buf)[r0 * D],
+          16, 0, 0);
+    }
+  };
+
+  int cur = 0;
+  if (k_lo < k_hi) stage_glds(0, k_lo);
+
   for (int n0 = k_lo; n0 < k_hi; n0 += BWD_BN) {
-    // ---- cooperative staging of K/V tile + transposed K ----
+    __syncthreads();  // glds for buf[cur] drained here
+    if (n0 + BWD_BN < k_hi) stage_glds(cur ^ 1, n0 + BWD_BN);
+    // transpose K into kt[cur] (kt B-frags for the dq MFMA)
     {
       constexpr int CPR = D / 8;
       constexpr int RPP = 256 / CPR;
       const int row = tid / CPR;
       const int col = tid % CPR;
-#pragma unroll
-      for (int pass = 0; pass < BWD_BN / RPP; ++pass) {
-        const int r = pass * RPP + row;
-        const int kr = min(n0 + r, ke - 1);
-        const bf16_t* kp = p.k + (size_t)kr * k_pitch + (size_t)kh * D;
-        const bf16_t* vp = p.v + (size_t)kr * k_pitch + (size_t)kh * D;
-        const int dst = swz(r, r * ROWB + col * 16);
-        *(bf16x8*)((char*)lds_k + dst) = *(const bf16x8*)(kp + col * 8);
-        *(bf16x8*)((char*)lds_v + dst) = *(const bf16x8*)(vp + col * 8);
-      }
-      __syncthreads();
+      const int bs = (col & 3) << 3;  // bank-spread swizzle
 #pragma unroll 1
       for (int pass = 0; pass < BWD_BN / RPP; ++pass) {
         const int r = pass * RPP + row;
-        const int src = swz(r, r * ROWB + col * 16);
-        const bf16x8 kv8 = *(const bf16x8*)((const char*)lds_k + src);
-        const int bs = (col & 3) << 3;  // bank-spread swizzle (see fwd lds_vt)
+        const bf16x8 kv8 = *(const bf16x8*)((const char*)lds_k(cur) +
+                                            swz(r, r * ROWB + col * 16));
 #pragma unroll
-        for (int e = 0; e < 8; ++e) lds_kt[col * 8 + e][r ^ bs] = kv8[e];
+        for (int e = 0; e < 8; ++e) lds_kt(cur)[col * 8 + e][r ^ bs] = kv8[e];
       }
     }
     __syncthreads();
@@ -463,8 +489,8 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dq_kernel(BwdParams p) {
 #pragma unroll
       for (int dd = 0; dd < DF; ++dd) {
         const int off = swz(lo32, lo32 * ROWB + dd * 32 + hi * 16);
-        bf16x8 kf = *(const bf16x8*)((const char*)lds_k + off);
-        bf16x8 vf = *(const bf16x8*)((const char*)lds_v + off);
+        bf16x8 kf = *(const bf16x8*)((const char*)lds_k(cur) + off);
+        bf16x8 vf = *(const bf16x8*)((const char*)lds_v(cur) + off);
         sA = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[dd], sA, 0, 0, 0);
         dpA = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dof[dd], dpA, 0, 0, 0);
       }
@@ -493,16 +519,21 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dq_kernel(BwdParams p) {
       // ---- dq += dS K (B-frags from the transposed K tile) ----
       bf16x8 dsa0 = cframe_to_afrag(dsv, 0);
       bf16x8 dsa1 = cframe_to_afrag(dsv, 1);
+      const int kbs = ((lo32 >> 3) & 3) << 3;
+      // two sweeps so consecutive MFMAs hit DIFFERENT accumulators
+      // (dependent-accumulator latency is 2x the issue interval)
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
-        const int kbs = ((lo32 >> 3) & 3) << 3;
-        bf16x8 b0 = *(const bf16x8*)(&lds_kt[dt * 32 + lo32][(8 * hi) ^ kbs]);
-        bf16x8 b1 = *(const bf16x8*)(&lds_kt[dt * 32 + lo32][(16 + 8 * hi) ^ kbs]);
+        bf16x8 b0 = *(const bf16x8*)(&lds_kt(cur)[dt * 32 + lo32][(8 * hi) ^ kbs]);
         acc_dq[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa0, b0, acc_dq[dt], 0, 0, 0);
+      }
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        bf16x8 b1 = *(const bf16x8*)(&lds_kt(cur)[dt * 32 + lo32][(16 + 8 * hi) ^ kbs]);
         acc_dq[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa1, b1, acc_dq[dt], 0, 0, 0);
       }
     }
-    __syncthreads();
+    cur ^= 1;
   }
 
   // ---- store dq once (atomicAdd: q_ranges of different slices may overlap) ----
