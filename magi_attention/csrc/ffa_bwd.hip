@@ -169,7 +169,7 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
   // vmcnt(0) before every ds_read, destroying the glds pipeline (guide §5
   // ".s-level traps" (a)).
   __shared__ __attribute__((aligned(16))) char smem[
-      2 * 2 * BWD_BM * D * 2 + BWD_WAVES * 32 * 34 * 2];
+      2 * 2 * BWD_BM * D * 2 + BWD_WAVES * 32 * 34 * 2 + 2 * D * 40 * 2];
   auto lds_q = [&](int buf) -> __bf16* {
     return (__bf16*)(smem + buf * BWD_BM * D * 2);
   };
@@ -177,6 +177,13 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
     return (__bf16*)(smem + (2 + buf) * BWD_BM * D * 2);
   };
   __bf16(*lds_p)[32][34] = (__bf16(*)[32][34])(smem + 4 * BWD_BM * D * 2);
+  // transposed copies of the CURRENT Q/dO tile (single-buffered; rebuilt per
+  // iteration): B-frags for dV/dK become b128 reads instead of 64 scalar
+  // reads per fragment
+  __bf16(*lds_qt)[40] =
+      (__bf16(*)[40])(smem + 4 * BWD_BM * D * 2 + BWD_WAVES * 32 * 34 * 2);
+  __bf16(*lds_dot)[40] = (__bf16(*)[40])(smem + 4 * BWD_BM * D * 2 +
+                                         BWD_WAVES * 32 * 34 * 2 + D * 40 * 2);
 
   // K/V fragments (A-layout) + K B-fragments, loaded once per block
   const int krow = n0 + lo32;
@@ -245,6 +252,26 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
   for (int m0 = q_lo; m0 < q_hi; m0 += BWD_BM) {
     __syncthreads();  // buf[cur] ready (barrier drains in-flight glds)
     if (m0 + BWD_BM < q_hi) stage_glds(cur ^ 1, m0 + BWD_BM);
+    {
+      constexpr int CPR = D / 8;
+      constexpr int RPP = 256 / CPR;
+      const int row = tid / CPR;
+      const int col = tid % CPR;
+      const int bs = (col & 3) << 3;
+#pragma unroll 1
+      for (int pass = 0; pass < BWD_BM / RPP; ++pass) {
+        const int r = pass * RPP + row;
+        const int src = swz(r, r * ROWB + col * 16);
+        const bf16x8 qv = *(const bf16x8*)((const char*)lds_q(cur) + src);
+        const bf16x8 dv8 = *(const bf16x8*)((const char*)lds_do(cur) + src);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          lds_qt[col * 8 + e][r ^ bs] = qv[e];
+          lds_dot[col * 8 + e][r ^ bs] = dv8[e];
+        }
+      }
+    }
+    __syncthreads();
 
     if (wave_live && m0 + BWD_BM > wq_lo && m0 < wq_hi) {
       const int qrow = m0 + lo32;
@@ -290,20 +317,16 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
       // ---- dV += P^T dO (P via per-wave LDS transpose; dO cols from LDS) ----
 #pragma unroll
       for (int r = 0; r < 16; ++r) lds_p[wave][crow(r, hi)][lo32] = (__bf16)pv[r];
+      const int kbs = ((lo32 >> 3) & 3) << 3;
 #pragma unroll
-      for (int dt = 0; dt < DT; ++dt) {
+      for (int tt = 0; tt < 2; ++tt) {  // tt outer: different accumulators back-to-back
+        bf16x8 pa = *(const bf16x8*)(&lds_p[wave][lo32][0] + 16 * tt + 8 * hi);
 #pragma unroll
-        for (int tt = 0; tt < 2; ++tt) {
-          Bf b;
-#pragma unroll
-          for (int e = 0; e < 8; ++e) {
-            const int r = 16 * tt + 8 * hi + e;
-            b.u[e] = *(const unsigned short*)(
-                (const char*)lds_do(cur) + swz(r, r * ROWB + (dt * 32 + lo32) * 2));
-          }
-          bf16x8 pa = *(const bf16x8*)(&lds_p[wave][lo32][0] + 16 * tt + 8 * hi);
+        for (int dt = 0; dt < DT; ++dt) {
+          bf16x8 b = *(const bf16x8*)(
+              &lds_dot[dt * 32 + lo32][(16 * tt + 8 * hi) ^ kbs]);
           acc_dv[dt] =
-              __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, b.v, acc_dv[dt], 0, 0, 0);
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, b, acc_dv[dt], 0, 0, 0);
         }
       }
 
@@ -311,19 +334,14 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
 #pragma unroll
       for (int r = 0; r < 16; ++r) lds_p[wave][crow(r, hi)][lo32] = (__bf16)dsv[r];
 #pragma unroll
-      for (int dt = 0; dt < DT; ++dt) {
+      for (int tt = 0; tt < 2; ++tt) {
+        bf16x8 dsa = *(const bf16x8*)(&lds_p[wave][lo32][0] + 16 * tt + 8 * hi);
 #pragma unroll
-        for (int tt = 0; tt < 2; ++tt) {
-          Bf b;
-#pragma unroll
-          for (int e = 0; e < 8; ++e) {
-            const int r = 16 * tt + 8 * hi + e;
-            b.u[e] = *(const unsigned short*)(
-                (const char*)lds_q(cur) + swz(r, r * ROWB + (dt * 32 + lo32) * 2));
-          }
-          bf16x8 dsa = *(const bf16x8*)(&lds_p[wave][lo32][0] + 16 * tt + 8 * hi);
+        for (int dt = 0; dt < DT; ++dt) {
+          bf16x8 b = *(const bf16x8*)(
+              &lds_qt[dt * 32 + lo32][(16 * tt + 8 * hi) ^ kbs]);
           acc_dk[dt] =
-              __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, b.v, acc_dk[dt], 0, 0, 0);
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, b, acc_dk[dt], 0, 0, 0);
         }
       }
 
