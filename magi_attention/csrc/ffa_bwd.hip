@@ -293,25 +293,41 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
       const bool row_live = qvalid && lse_q != INFINITY && lse_q != -INFINITY;
 
       float pv[16], dsv[16];
+      const bool interior =
+          (qrow < qe) && (m0 + 31 < wq_hi) && lse_q != INFINITY &&
+          lse_q != -INFINITY && (n0 + BWD_BN <= ke) &&
+          !((atype == 1 || atype == 3) && (n0 + BWD_BN - 1 > m0 + (ke - qe))) &&
+          !((atype == 2 || atype == 3) && (n0 < m0 + 31 + (ks - qs)));
+      const bool w_interior = __all(interior);
+      if (w_interior && !HAS_SOFTCAP) {
+        const float lsc = lse_q * log2e;
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int kk = n0 + crow(r, hi);
-        bool ok = row_live && kk < ke;
-        if (atype == 1 || atype == 3) ok = ok && (kk - qrow <= ke - qe);
-        if (atype == 2 || atype == 3) ok = ok && (kk - qrow >= ks - qs);
-        float sv = s[r];
-        float dscale = p.scale;
-        float t;
-        if (HAS_SOFTCAP) {
-          const float th = tanhf(sv * cap_pre);
-          t = th * sl2;
-          dscale = p.scale * (1.f - th * th);
-        } else {
-          t = sv * sl2;
+        for (int r = 0; r < 16; ++r) {
+          const float pij = exp2f(s[r] * sl2 - lsc);
+          pv[r] = pij;
+          dsv[r] = pij * (dp[r] - dpsum_q) * p.scale;
         }
-        const float pij = ok ? exp2f(t - lse_q * log2e) : 0.f;
-        pv[r] = pij;
-        dsv[r] = pij * (dp[r] - dpsum_q) * dscale;
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int kk = n0 + crow(r, hi);
+          bool ok = row_live && kk < ke;
+          if (atype == 1 || atype == 3) ok = ok && (kk - qrow <= ke - qe);
+          if (atype == 2 || atype == 3) ok = ok && (kk - qrow >= ks - qs);
+          float sv = s[r];
+          float dscale = p.scale;
+          float t;
+          if (HAS_SOFTCAP) {
+            const float th = tanhf(sv * cap_pre);
+            t = th * sl2;
+            dscale = p.scale * (1.f - th * th);
+          } else {
+            t = sv * sl2;
+          }
+          const float pij = ok ? exp2f(t - lse_q * log2e) : 0.f;
+          pv[r] = pij;
+          dsv[r] = pij * (dp[r] - dpsum_q) * dscale;
+        }
       }
 
       // ---- dV += P^T dO (P via per-wave LDS transpose; dO cols from LDS) ----
@@ -514,24 +530,37 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dq_kernel(BwdParams p) {
       }
 
       float dsv[16];
+      const bool interior =
+          (qrow < qe) && row_live && (n0 >= ks) && (n0 + BWD_BN <= ke) &&
+          !((atype == 1 || atype == 3) && (n0 + BWD_BN - 1 > m0 + (ke - qe))) &&
+          !((atype == 2 || atype == 3) && (n0 < m0 + 31 + (ks - qs)));
+      if (__all(interior) && !HAS_SOFTCAP) {
+        const float lsc = lse_q * log2e;
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int kk = n0 + crow(r, hi);
-        bool ok = row_live && kk < ke && kk >= ks;
-        if (atype == 1 || atype == 3) ok = ok && (kk - qrow <= ke - qe);
-        if (atype == 2 || atype == 3) ok = ok && (kk - qrow >= ks - qs);
-        float sv = sA[r];
-        float dscale = p.scale;
-        float t;
-        if (HAS_SOFTCAP) {
-          const float th = tanhf(sv * cap_pre);
-          t = th * sl2;
-          dscale = p.scale * (1.f - th * th);
-        } else {
-          t = sv * sl2;
+        for (int r = 0; r < 16; ++r) {
+          const float pij = exp2f(sA[r] * sl2 - lsc);
+          dsv[r] = pij * (dpA[r] - dpsum_q) * p.scale;
         }
-        const float pij = ok ? exp2f(t - lse_q * log2e) : 0.f;
-        dsv[r] = pij * (dpA[r] - dpsum_q) * dscale;
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int kk = n0 + crow(r, hi);
+          bool ok = row_live && kk < ke && kk >= ks;
+          if (atype == 1 || atype == 3) ok = ok && (kk - qrow <= ke - qe);
+          if (atype == 2 || atype == 3) ok = ok && (kk - qrow >= ks - qs);
+          float sv = sA[r];
+          float dscale = p.scale;
+          float t;
+          if (HAS_SOFTCAP) {
+            const float th = tanhf(sv * cap_pre);
+            t = th * sl2;
+            dscale = p.scale * (1.f - th * th);
+          } else {
+            t = sv * sl2;
+          }
+          const float pij = ok ? exp2f(t - lse_q * log2e) : 0.f;
+          dsv[r] = pij * (dpA[r] - dpsum_q) * dscale;
+        }
       }
 
       // ---- dq += dS K (B-frags from the transposed K tile) ----

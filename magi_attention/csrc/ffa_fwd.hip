@@ -207,18 +207,35 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
     }
 
     // ---- mask + scale into exp2 domain ----
+    // interior fast path: when the whole tile is provably unmasked for every
+    // lane (wave-uniform), skip the per-element compare/select chain — the
+    // kernels are instruction-issue-bound and masks only bind near edges.
+    const bool interior =
+        (q0 + 31 < qe) && (n0 >= n_lo) && (n0 + FFA_BN <= n_hi) &&
+        !((atype == 1 || atype == 3) && (n0 + FFA_BN - 1 > q0 + (ke - qe))) &&
+        !((atype == 2 || atype == 3) && (n0 < q0 + 31 + (ks - qs)));
     float t[16];
     float mx = -INFINITY;
+    if (interior) {
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int kk = n0 + crow(r, hi);
-      bool ok = qvalid && (kk >= n_lo) && (kk < n_hi);
-      if (atype == 1 || atype == 3) ok = ok && (kk - qrow <= ke - qe);
-      if (atype == 2 || atype == 3) ok = ok && (kk - qrow >= ks - qs);
-      float sv = s[r];
-      if (HAS_SOFTCAP) sv = tanhf(sv * cap_pre);
-      t[r] = ok ? sv * sl2 : -INFINITY;
-      mx = fmaxf(mx, t[r]);
+      for (int r = 0; r < 16; ++r) {
+        float sv = s[r];
+        if (HAS_SOFTCAP) sv = tanhf(sv * cap_pre);
+        t[r] = sv * sl2;
+        mx = fmaxf(mx, t[r]);
+      }
+    } else {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kk = n0 + crow(r, hi);
+        bool ok = qvalid && (kk >= n_lo) && (kk < n_hi);
+        if (atype == 1 || atype == 3) ok = ok && (kk - qrow <= ke - qe);
+        if (atype == 2 || atype == 3) ok = ok && (kk - qrow >= ks - qs);
+        float sv = s[r];
+        if (HAS_SOFTCAP) sv = tanhf(sv * cap_pre);
+        t[r] = ok ? sv * sl2 : -INFINITY;
+        mx = fmaxf(mx, t[r]);
+      }
     }
     mx = fmaxf(mx, warp_xor32(mx));
 
