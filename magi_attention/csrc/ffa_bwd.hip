@@ -169,21 +169,23 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
   // vmcnt(0) before every ds_read, destroying the glds pipeline (guide §5
   // ".s-level traps" (a)).
   __shared__ __attribute__((aligned(16))) char smem[
-      2 * 2 * BWD_BM * D * 2 + BWD_WAVES * 32 * 34 * 2 + 2 * D * 40 * 2];
+      2 * 2 * BWD_BM * D * 2 + 2 * BWD_BM * 4 + 2 * D * 40 * 2];
   auto lds_q = [&](int buf) -> __bf16* {
     return (__bf16*)(smem + buf * BWD_BM * D * 2);
   };
   auto lds_do = [&](int buf) -> __bf16* {
     return (__bf16*)(smem + (2 + buf) * BWD_BM * D * 2);
   };
-  __bf16(*lds_p)[32][34] = (__bf16(*)[32][34])(smem + 4 * BWD_BM * D * 2);
+  // lse / dpsum of the current q tile (staged once per iteration)
+  float* lds_lse = (float*)(smem + 4 * BWD_BM * D * 2);
+  float* lds_dps = (float*)(smem + 4 * BWD_BM * D * 2 + BWD_BM * 4);
   // transposed copies of the CURRENT Q/dO tile (single-buffered; rebuilt per
   // iteration): B-frags for dV/dK become b128 reads instead of 64 scalar
   // reads per fragment
   __bf16(*lds_qt)[40] =
-      (__bf16(*)[40])(smem + 4 * BWD_BM * D * 2 + BWD_WAVES * 32 * 34 * 2);
+      (__bf16(*)[40])(smem + 4 * BWD_BM * D * 2 + 2 * BWD_BM * 4);
   __bf16(*lds_dot)[40] = (__bf16(*)[40])(smem + 4 * BWD_BM * D * 2 +
-                                         BWD_WAVES * 32 * 34 * 2 + D * 40 * 2);
+                                         2 * BWD_BM * 4 + D * 40 * 2);
 
   // K/V fragments (A-layout) + K B-fragments, loaded once per block
   const int krow = n0 + lo32;
@@ -270,48 +272,58 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
           lds_dot[col * 8 + e][r ^ bs] = dv8[e];
         }
       }
+      if (tid < BWD_BM) {
+        const int qr = min(m0 + tid, qe - 1);
+        lds_lse[tid] = p.lse[(size_t)qr * p.hq + h];
+        lds_dps[tid] = p.dpsum[(size_t)qr * p.hq + h];
+      }
     }
     __syncthreads();
 
     if (wave_live && m0 + BWD_BM > wq_lo && m0 < wq_hi) {
-      const int qrow = m0 + lo32;
-      const bool qvalid = qrow < wq_hi && qrow < qe;
-
-      // ---- S^T = K Q^T ; dP^T = V dO^T (q-side B-frags from LDS rows) ----
+      // ---- S = Q K^T ; dP = dO V^T, UN-swapped: C layout [q=crow][k=lo32],
+      // so the dV/dK A-fragments come from the in-register permlane transform
+      // (cframe) instead of an LDS round-trip ----
       f32x16 s = (f32x16)(0.f), dp = (f32x16)(0.f);
 #pragma unroll
       for (int dd = 0; dd < DF; ++dd) {
         const int off = swz(lo32, lo32 * ROWB + dd * 32 + hi * 16);
         bf16x8 qf = *(const bf16x8*)((const char*)lds_q(cur) + off);
         bf16x8 dof = *(const bf16x8*)((const char*)lds_do(cur) + off);
-        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfA[dd], qf, s, 0, 0, 0);
-        dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfA[dd], dof, dp, 0, 0, 0);
+        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kfA[dd], s, 0, 0, 0);
+        dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vfA[dd], dp, 0, 0, 0);
       }
 
-      const float lse_q = qvalid ? p.lse[(size_t)qrow * p.hq + h] : INFINITY;
-      const float dpsum_q = qvalid ? p.dpsum[(size_t)qrow * p.hq + h] : 0.f;
-      const bool row_live = qvalid && lse_q != INFINITY && lse_q != -INFINITY;
-
+      const int kk = n0 + lo32;  // this lane's k column
       float pv[16], dsv[16];
       const bool interior =
-          (qrow < qe) && (m0 + 31 < wq_hi) && lse_q != INFINITY &&
-          lse_q != -INFINITY && (n0 + BWD_BN <= ke) &&
+          (m0 + BWD_BM <= wq_hi) && (m0 >= qs) && (n0 + BWD_BN <= ke) &&
           !((atype == 1 || atype == 3) && (n0 + BWD_BN - 1 > m0 + (ke - qe))) &&
-          !((atype == 2 || atype == 3) && (n0 < m0 + 31 + (ks - qs)));
-      const bool w_interior = __all(interior);
-      if (w_interior && !HAS_SOFTCAP) {
-        const float lsc = lse_q * log2e;
+          !((atype == 2 || atype == 3) && (n0 < m0 + BWD_BM - 1 + (ks - qs)));
+      bool all_live = interior;
+      if (interior) {
+        // rows' lse finite? checked via the staged tile by reg
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          const float pij = exp2f(s[r] * sl2 - lsc);
+          const float lq = lds_lse[crow(r, hi)];
+          all_live = all_live && (lq != INFINITY) && (lq != -INFINITY);
+        }
+      }
+      if (__all(all_live) && !HAS_SOFTCAP) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int rl = crow(r, hi);
+          const float pij = exp2f(s[r] * sl2 - lds_lse[rl] * log2e);
           pv[r] = pij;
-          dsv[r] = pij * (dp[r] - dpsum_q) * p.scale;
+          dsv[r] = pij * (dp[r] - lds_dps[rl]) * p.scale;
         }
       } else {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          const int kk = n0 + crow(r, hi);
-          bool ok = row_live && kk < ke;
+          const int qrow = m0 + crow(r, hi);
+          const float lq = (qrow < qe) ? lds_lse[crow(r, hi)] : INFINITY;
+          bool ok = (qrow < wq_hi) && (qrow >= qs) && lq != INFINITY &&
+                    lq != -INFINITY && kk < ke;
           if (atype == 1 || atype == 3) ok = ok && (kk - qrow <= ke - qe);
           if (atype == 2 || atype == 3) ok = ok && (kk - qrow >= ks - qs);
           float sv = s[r];
@@ -324,43 +336,49 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
           } else {
             t = sv * sl2;
           }
-          const float pij = ok ? exp2f(t - lse_q * log2e) : 0.f;
+          const float pij = ok ? exp2f(t - lq * log2e) : 0.f;
           pv[r] = pij;
-          dsv[r] = pij * (dp[r] - dpsum_q) * dscale;
+          dsv[r] = pij * (dp[r] - lds_dps[crow(r, hi)]) * dscale;
         }
       }
 
-      // ---- dV += P^T dO (P via per-wave LDS transpose; dO cols from LDS) ----
-#pragma unroll
-      for (int r = 0; r < 16; ++r) lds_p[wave][crow(r, hi)][lo32] = (__bf16)pv[r];
       const int kbs = ((lo32 >> 3) & 3) << 3;
-#pragma unroll
-      for (int tt = 0; tt < 2; ++tt) {  // tt outer: different accumulators back-to-back
-        bf16x8 pa = *(const bf16x8*)(&lds_p[wave][lo32][0] + 16 * tt + 8 * hi);
+      // ---- dV += P^T dO : A = cframe(pv) = [k=lo32][q-slice] ----
+      {
+        bf16x8 pa0 = cframe_to_afrag(pv, 0);
+        bf16x8 pa1 = cframe_to_afrag(pv, 1);
 #pragma unroll
         for (int dt = 0; dt < DT; ++dt) {
-          bf16x8 b = *(const bf16x8*)(
-              &lds_dot[dt * 32 + lo32][(16 * tt + 8 * hi) ^ kbs]);
+          bf16x8 b = *(const bf16x8*)(&lds_dot[dt * 32 + lo32][(8 * hi) ^ kbs]);
           acc_dv[dt] =
-              __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, b, acc_dv[dt], 0, 0, 0);
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa0, b, acc_dv[dt], 0, 0, 0);
         }
-      }
-
-      // ---- dK += dS^T Q (dS via LDS transpose; Q cols from LDS) ----
-#pragma unroll
-      for (int r = 0; r < 16; ++r) lds_p[wave][crow(r, hi)][lo32] = (__bf16)dsv[r];
-#pragma unroll
-      for (int tt = 0; tt < 2; ++tt) {
-        bf16x8 dsa = *(const bf16x8*)(&lds_p[wave][lo32][0] + 16 * tt + 8 * hi);
 #pragma unroll
         for (int dt = 0; dt < DT; ++dt) {
-          bf16x8 b = *(const bf16x8*)(
-              &lds_qt[dt * 32 + lo32][(16 * tt + 8 * hi) ^ kbs]);
-          acc_dk[dt] =
-              __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, b, acc_dk[dt], 0, 0, 0);
+          bf16x8 b =
+              *(const bf16x8*)(&lds_dot[dt * 32 + lo32][(16 + 8 * hi) ^ kbs]);
+          acc_dv[dt] =
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa1, b, acc_dv[dt], 0, 0, 0);
         }
       }
-
+      // ---- dK += dS^T Q ----
+      {
+        bf16x8 da0 = cframe_to_afrag(dsv, 0);
+        bf16x8 da1 = cframe_to_afrag(dsv, 1);
+#pragma unroll
+        for (int dt = 0; dt < DT; ++dt) {
+          bf16x8 b = *(const bf16x8*)(&lds_qt[dt * 32 + lo32][(8 * hi) ^ kbs]);
+          acc_dk[dt] =
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(da0, b, acc_dk[dt], 0, 0, 0);
+        }
+#pragma unroll
+        for (int dt = 0; dt < DT; ++dt) {
+          bf16x8 b =
+              *(const bf16x8*)(&lds_qt[dt * 32 + lo32][(16 + 8 * hi) ^ kbs]);
+          acc_dk[dt] =
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(da1, b, acc_dk[dt], 0, 0, 0);
+        }
+      }
     }
     cur ^= 1;
   }
