@@ -1,0 +1,59 @@
+"""CPU checks of the native boundary: the C-ABI library loads and exports
+every symbol include/magi_ffa.h declares; the magi_attn_ext surface module
+works (host-side parts)."""
+import ctypes
+import re
+from pathlib import Path
+
+import torch
+
+ROOT = Path(__file__).resolve().parent.parent
+
+
+def test_cabi_exports_all_declared_symbols():
+    header = (ROOT / "include" / "magi_ffa.h").read_text()
+    declared = re.findall(r"^int (magi_\w+)\(", header, re.M)
+    assert len(declared) >= 8, declared
+    lib = ctypes.CDLL(str(ROOT / "magi_attention" / "_libs" / "libmagi_ffa.so"))
+    for sym in declared:
+        assert getattr(lib, sym, None) is not None, f"missing symbol {sym}"
+
+
+def test_ext_module_surface_cpu():
+    from magi_attention import magi_attn_ext as ext
+
+    r = torch.tensor([[5, 9], [0, 3], [5, 9], [2, 4]], dtype=torch.int32)
+    order = ext.argsort_ranges(r)
+    assert order.tolist() == [1, 3, 0, 2]
+    q2, k2, t2 = ext.reorder_ranges_and_attn_type_maps(
+        r, r + 1, torch.tensor([0, 1, 2, 3], dtype=torch.int32), order
+    )
+    assert q2[:, 0].tolist() == [0, 2, 5, 5]
+    assert t2.tolist() == [1, 3, 0, 2]
+
+    sorted_r = q2
+    uniq, inv, cnt = ext.unique_consecutive_pairs(sorted_r)
+    assert int(cnt[0]) == uniq.shape[0]
+    assert torch.equal(uniq.long().index_select(0, inv.long()), sorted_r.long())
+
+    assert ext.is_valid_cu_seqlens([0, 3, 7], 7)
+    assert not ext.is_valid_cu_seqlens([1, 3], 3)
+
+    ext.start_event("x")
+    ext.stop_event("x")
+    assert ext.elapsed_ms("x") >= 0.0
+    ext.destroy_event("x")
+
+
+def test_product_fails_loudly_without_extension(monkeypatch):
+    """The compute path must never fall back silently when the HIP library is
+    absent (tier rule: no CPU fallback on the product path)."""
+    import magi_attention._ffa_lib as L
+
+    monkeypatch.setattr(L, "_lib", None)
+    monkeypatch.setattr(L, "_LIB_PATH", Path("/nonexistent/lib.so"))
+    try:
+        L.lib()
+        assert False, "expected RuntimeError"
+    except RuntimeError as e:
+        assert "no fallback" in str(e).lower() or "native" in str(e).lower()
