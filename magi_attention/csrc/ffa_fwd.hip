@@ -123,10 +123,8 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
   auto swz = [](int row, int byte_off) {
     return byte_off ^ ((row & SWZM) << 4);
   };
-  // stage 64 k rows per pipeline step (two 32-row compute sub-tiles share one
-  // barrier pair): k rows double-buffered, transposed V single-buffered
-  __shared__ __bf16 lds_k[2][2 * FFA_BN * D];
-  __shared__ __bf16 lds_vt[D][72];
+  __shared__ __bf16 lds_k[2][FFA_BN * D];
+  __shared__ __bf16 lds_vt[2][D][40];
 
   // Q fragments in registers (8 x bf16x8 for D=128)
   bf16x8 qf[DF];
@@ -148,12 +146,11 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
 #pragma unroll
   for (int dt = 0; dt < DT; ++dt) acc_o[dt] = (f32x16)(0.f);
 
-  // ---- double-buffered staging pipeline (T14): K rows for 64-row steps are
-  // issued into registers during the previous step's compute; the transposed
-  // V image is rebuilt from the V registers after the step's last PV read ----
-  constexpr int CPR = D / 8;        // 16-B chunks per row
-  constexpr int RPP = 256 / CPR;    // rows per pass
-  constexpr int NPASS = 2 * FFA_BN / RPP;
+  // ---- double-buffered staging pipeline (T14: issue loads for tile t+1
+  // into registers during tile t's compute; one barrier per iteration) ----
+  constexpr int CPR = D / 8;       // 16-B chunks per row
+  constexpr int RPP = 256 / CPR;   // rows per pass
+  constexpr int NPASS = FFA_BN / RPP;
   const int srow = threadIdx.x / CPR;
   const int scol = threadIdx.x % CPR;
   bf16x8 kreg[NPASS], vreg[NPASS];
@@ -167,14 +164,17 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
     }
   };
   auto write_stage = [&](int buf) {
-    const int bs = (scol & 3) << 3;
 #pragma unroll
     for (int pass = 0; pass < NPASS; ++pass) {
       const int r = pass * RPP + srow;
       *(bf16x8*)((char*)lds_k[buf] + swz(r, r * ROWB + scol * 16)) = kreg[pass];
+      // 8-block XOR swizzle on the k index: spreads the 16 column groups
+      // over distinct banks (unswizzled, every scol group hit the same bank:
+      // SQ_LDS_BANK_CONFLICT was 24% of fwd wave cycles)
+      const int bs = (scol & 3) << 3;
 #pragma unroll
       for (int e = 0; e < 8; ++e)
-        lds_vt[scol * 8 + e][r ^ bs] = vreg[pass][e];
+        lds_vt[buf][scol * 8 + e][r ^ bs] = vreg[pass][e];
     }
   };
 
@@ -185,128 +185,131 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
   }
   __syncthreads();
 
-  for (int nb = b_lo; nb < b_hi; nb += 2 * FFA_BN) {
-    const bool has_next = nb + 2 * FFA_BN < b_hi;
-    if (has_next) issue_loads(nb + 2 * FFA_BN);
+  for (int n0 = b_lo; n0 < b_hi; n0 += FFA_BN) {
+    const bool has_next = n0 + FFA_BN < b_hi;
+    if (has_next) issue_loads(n0 + FFA_BN);  // lands under this tile's compute
+    const bool live = (n0 + FFA_BN > n_lo) && (n0 < n_hi) && qvalid_any;
 
-#pragma unroll
-    for (int half = 0; half < 2; ++half) {
-      const int n0 = nb + half * FFA_BN;
-      if (n0 >= b_hi) break;
-      const int rbase = half * FFA_BN;
-      const bool live = (n0 + FFA_BN > n_lo) && (n0 < n_hi) && qvalid_any;
-      if (!live) continue;
-
-      f32x16 s = (f32x16)(0.f);
+    f32x16 s = (f32x16)(0.f);
+    if (live) {
 #pragma unroll
       for (int dd = 0; dd < DF; ++dd) {
         bf16x8 kf = *(const bf16x8*)(
-            (const char*)lds_k[cur] +
-            swz(rbase + lo32, (rbase + lo32) * ROWB + dd * 32 + hi * 16));
+            (const char*)lds_k[cur] + swz(lo32, lo32 * ROWB + dd * 32 + hi * 16));
         s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[dd], s, 0, 0, 0);
       }
+    }
+    if (!live) {
+      if (has_next) write_stage(cur ^ 1);
+      __syncthreads();
+      cur ^= 1;
+      continue;
+    }
 
-      // ---- mask + scale into exp2 domain ----
-      const bool interior =
-          (q0 + 31 < qe) && (n0 >= n_lo) && (n0 + FFA_BN <= n_hi) &&
-          !((atype == 1 || atype == 3) && (n0 + FFA_BN - 1 > q0 + (ke - qe))) &&
-          !((atype == 2 || atype == 3) && (n0 < q0 + 31 + (ks - qs)));
-      float t[16];
-      float mx = -INFINITY;
-      if (interior) {
-#pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          float sv = s[r];
-          if (HAS_SOFTCAP) sv = tanhf(sv * cap_pre);
-          t[r] = sv * sl2;
-          mx = fmaxf(mx, t[r]);
-        }
-      } else {
-#pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int kk = n0 + crow(r, hi);
-          bool ok = qvalid && (kk >= n_lo) && (kk < n_hi);
-          if (atype == 1 || atype == 3) ok = ok && (kk - qrow <= ke - qe);
-          if (atype == 2 || atype == 3) ok = ok && (kk - qrow >= ks - qs);
-          float sv = s[r];
-          if (HAS_SOFTCAP) sv = tanhf(sv * cap_pre);
-          t[r] = ok ? sv * sl2 : -INFINITY;
-          mx = fmaxf(mx, t[r]);
-        }
-      }
-      mx = fmaxf(mx, warp_xor32(mx));
-
-      const float m_new = fmaxf(m_run, mx);
-      const float m_use = (m_new == -INFINITY) ? 0.f : m_new;
-      const float alpha = (m_run == -INFINITY) ? 0.f : exp2f(m_run - m_use);
-      m_run = m_new;
-
-      float pr[16];
-      float psum = 0.f;
+    // ---- mask + scale into exp2 domain ----
+    // interior fast path: when the whole tile is provably unmasked for every
+    // lane (wave-uniform), skip the per-element compare/select chain — the
+    // kernels are instruction-issue-bound and masks only bind near edges.
+    const bool interior =
+        (q0 + 31 < qe) && (n0 >= n_lo) && (n0 + FFA_BN <= n_hi) &&
+        !((atype == 1 || atype == 3) && (n0 + FFA_BN - 1 > q0 + (ke - qe))) &&
+        !((atype == 2 || atype == 3) && (n0 < q0 + 31 + (ks - qs)));
+    float t[16];
+    float mx = -INFINITY;
+    if (interior) {
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        pr[r] = exp2f(t[r] - m_use);
-        psum += pr[r];
+        float sv = s[r];
+        if (HAS_SOFTCAP) sv = tanhf(sv * cap_pre);
+        t[r] = sv * sl2;
+        mx = fmaxf(mx, t[r]);
       }
-      l_run = l_run * alpha + (psum + warp_xor32(psum));
-
-      if (__any(alpha != 1.f)) {
+    } else {
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int src = crow(r, hi);
-          const float aq = __uint_as_float(
-              __builtin_amdgcn_ds_bpermute(src << 2, __float_as_uint(alpha)));
-#pragma unroll
-          for (int dt = 0; dt < DT; ++dt) acc_o[dt][r] *= aq;
-        }
+      for (int r = 0; r < 16; ++r) {
+        const int kk = n0 + crow(r, hi);
+        bool ok = qvalid && (kk >= n_lo) && (kk < n_hi);
+        if (atype == 1 || atype == 3) ok = ok && (kk - qrow <= ke - qe);
+        if (atype == 2 || atype == 3) ok = ok && (kk - qrow >= ks - qs);
+        float sv = s[r];
+        if (HAS_SOFTCAP) sv = tanhf(sv * cap_pre);
+        t[r] = ok ? sv * sl2 : -INFINITY;
+        mx = fmaxf(mx, t[r]);
       }
+    }
+    mx = fmaxf(mx, warp_xor32(mx));
 
-      // ---- P -> bf16 A-fragments (cvt_pk + permlane32_swap) ----
-      bf16x8 pa[2];
-#pragma unroll
-      for (int tt = 0; tt < 2; ++tt) {
-        unsigned c0 = pack_bf16_pair(pr[8 * tt + 0], pr[8 * tt + 1]);
-        unsigned c1 = pack_bf16_pair(pr[8 * tt + 2], pr[8 * tt + 3]);
-        unsigned c2 = pack_bf16_pair(pr[8 * tt + 4], pr[8 * tt + 5]);
-        unsigned c3 = pack_bf16_pair(pr[8 * tt + 6], pr[8 * tt + 7]);
-        {
-          auto r2 = __builtin_amdgcn_permlane32_swap(c0, c2, false, false);
-          c0 = r2[0];
-          c2 = r2[1];
-        }
-        {
-          auto r2 = __builtin_amdgcn_permlane32_swap(c1, c3, false, false);
-          c1 = r2[0];
-          c3 = r2[1];
-        }
-        union {
-          unsigned u[4];
-          bf16x8 v;
-        } cvt;
-        cvt.u[0] = c0;
-        cvt.u[1] = c1;
-        cvt.u[2] = c2;
-        cvt.u[3] = c3;
-        pa[tt] = cvt.v;
-      }
+    const float m_new = fmaxf(m_run, mx);
+    const float m_use = (m_new == -INFINITY) ? 0.f : m_new;
+    const float alpha = (m_run == -INFINITY) ? 0.f : exp2f(m_run - m_use);
+    m_run = m_new;
 
-      // ---- PV: B-frags from the transposed V tile ----
-      const int kbs = ((lo32 >> 3) & 3) << 3;
+    float pr[16];
+    float psum = 0.f;
 #pragma unroll
-      for (int tt = 0; tt < 2; ++tt) {
+    for (int r = 0; r < 16; ++r) {
+      pr[r] = exp2f(t[r] - m_use);   // exp2(-inf)=0 for masked
+      psum += pr[r];
+    }
+    l_run = l_run * alpha + (psum + warp_xor32(psum));
+
+    // ---- rescale O by alpha (per q = crow layout) ----
+    // NOTE: wave-uniform condition — ds_bpermute reads other lanes' registers,
+    // so every lane must be active whenever any lane needs the rescale.
+    if (__any(alpha != 1.f)) {
 #pragma unroll
-        for (int dt = 0; dt < DT; ++dt) {
-          bf16x8 bv = *(const bf16x8*)(
-              &lds_vt[dt * 32 + lo32][(rbase + 16 * tt + 8 * hi) ^ kbs]);
-          acc_o[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[tt], bv,
-                                                              acc_o[dt], 0, 0, 0);
-        }
+      for (int r = 0; r < 16; ++r) {
+        const int src = crow(r, hi);
+        const float aq = __uint_as_float(
+            __builtin_amdgcn_ds_bpermute(src << 2, __float_as_uint(alpha)));
+#pragma unroll
+        for (int dt = 0; dt < DT; ++dt) acc_o[dt][r] *= aq;
       }
     }
 
-    __syncthreads();  // all waves done reading lds_vt (and lds_k[cur])
-    if (has_next) write_stage(cur ^ 1);
-    __syncthreads();  // stage for the next step complete
+    // ---- P -> bf16 A-fragments (cvt_pk + permlane32_swap) ----
+    bf16x8 pa[2];
+#pragma unroll
+    for (int tt = 0; tt < 2; ++tt) {
+      unsigned c0 = pack_bf16_pair(pr[8 * tt + 0], pr[8 * tt + 1]);
+      unsigned c1 = pack_bf16_pair(pr[8 * tt + 2], pr[8 * tt + 3]);
+      unsigned c2 = pack_bf16_pair(pr[8 * tt + 4], pr[8 * tt + 5]);
+      unsigned c3 = pack_bf16_pair(pr[8 * tt + 6], pr[8 * tt + 7]);
+      {
+        auto r2 = __builtin_amdgcn_permlane32_swap(c0, c2, false, false);
+        c0 = r2[0];
+        c2 = r2[1];
+      }
+      {
+        auto r2 = __builtin_amdgcn_permlane32_swap(c1, c3, false, false);
+        c1 = r2[0];
+        c3 = r2[1];
+      }
+      union {
+        unsigned u[4];
+        bf16x8 v;
+      } cvt;
+      cvt.u[0] = c0;
+      cvt.u[1] = c1;
+      cvt.u[2] = c2;
+      cvt.u[3] = c3;
+      pa[tt] = cvt.v;
+    }
+
+    // ---- PV: O[32q][32d] += P^T V (B-frags from the transposed V tile) ----
+#pragma unroll
+    for (int tt = 0; tt < 2; ++tt) {
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        bf16x8 bv = *(const bf16x8*)(
+            &lds_vt[cur][dt * 32 + lo32]
+                   [(16 * tt + 8 * hi) ^ (((lo32 >> 3) & 3) << 3)]);
+        acc_o[dt] =
+            __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[tt], bv, acc_o[dt], 0, 0, 0);
+      }
+    }
+    if (has_next) write_stage(cur ^ 1);  // other buffer: overlaps this compute
+    __syncthreads();
     cur ^= 1;
   }
 
