@@ -89,7 +89,7 @@ def _worker(rank, ws, port, case_name, degree, q_data):
             case = CASES[case_name]
 
         total = case["total"]
-        hq, hk, d = 4, 2, 32
+        hq, hk, d = (8, 2, 64) if case_name == "mixed_types" else (4, 2, 32)
         g = torch.Generator().manual_seed(17)
         q = torch.randn(total, hq, d, generator=g, dtype=torch.float64)
         k = torch.randn(total, hk, d, generator=g, dtype=torch.float64)

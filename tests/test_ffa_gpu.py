@@ -215,3 +215,67 @@ def test_range_gather_reduce():
     ref2[40:45] = 2 * y[15:20].cpu()
     ref2[60:100] = 2 * y[20:60].cpu()
     torch.testing.assert_close(acc.cpu(), ref2)
+
+
+@requires_gpu
+def test_correct_out_lse_kernel():
+    """Fused merge of two partial (out,lse) sets vs the oracle merge
+    (reference functional/utils.py:371)."""
+    from magi_attention.ops import correct_out_lse
+    from oracle.ref_attn import merge_out_lse
+
+    g = torch.Generator().manual_seed(5)
+    t, h, d = 100, 4, 64
+    o1 = torch.randn(t, h, d, generator=g).float()
+    o2 = torch.randn(t, h, d, generator=g).float()
+    l1 = torch.randn(t, h, generator=g).float() * 3
+    l2 = torch.randn(t, h, generator=g).float() * 3
+    # some -inf rows on both sides
+    l1[:5] = float("-inf")
+    l2[3:8] = float("-inf")
+    o1[:5] = 0
+    o2[3:8] = 0
+    ref_o, ref_l = merge_out_lse([o1, o2], [l1, l2])
+
+    o1c, l1c = o1.cuda(), l1.cuda()
+    correct_out_lse(o1c, l1c, o2.cuda(), l2.cuda())
+    torch.cuda.synchronize()
+    torch.testing.assert_close(o1c.cpu().double(), ref_o, atol=1e-5, rtol=1e-5)
+    fin = torch.isfinite(ref_l)
+    torch.testing.assert_close(l1c.cpu()[fin].double(), ref_l[fin], atol=1e-5,
+                               rtol=1e-5)
+    assert torch.equal(torch.isfinite(l1c.cpu()), fin)
+
+
+@requires_gpu
+def test_range_reduce_lse_weighted():
+    """lse-weighted row reduce (reference _range_reduce.py:239): merging
+    partial (out,lse) rows into a destination must equal the oracle merge."""
+    from magi_attention.ops import range_reduce
+    from oracle.ref_attn import merge_out_lse
+
+    g = torch.Generator().manual_seed(6)
+    t, h, d = 64, 2, 32
+    dst_o = torch.randn(t, h, d, generator=g).float()
+    dst_l = torch.randn(t, h, generator=g).float()
+    src_o = torch.randn(40, h, d, generator=g).float()
+    src_l = torch.randn(40, h, generator=g).float()
+
+    in_ranges = torch.tensor([[0, 25], [25, 40]], dtype=torch.int32)
+    out_starts = torch.tensor([10, 45], dtype=torch.int32)
+
+    ref_o = dst_o.clone().double()
+    ref_l = dst_l.clone().double()
+    for (a, b), o in zip(in_ranges.tolist(), out_starts.tolist()):
+        mo, ml = merge_out_lse(
+            [ref_o[o:o + b - a], src_o[a:b]], [ref_l[o:o + b - a].float(), src_l[a:b]]
+        )
+        ref_o[o:o + b - a] = mo
+        ref_l[o:o + b - a] = ml
+
+    do, dl = dst_o.cuda(), dst_l.cuda()
+    range_reduce(src_o.cuda(), do, in_ranges.cuda(), out_starts.cuda(),
+                 op="lse", in_lse=src_l.cuda(), out_lse=dl)
+    torch.cuda.synchronize()
+    torch.testing.assert_close(do.cpu().double(), ref_o, atol=1e-5, rtol=1e-5)
+    torch.testing.assert_close(dl.cpu().double(), ref_l, atol=1e-5, rtol=1e-5)
