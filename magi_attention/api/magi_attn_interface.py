@@ -221,9 +221,50 @@ def roll_simple(x: torch.Tensor, key: DistAttnRuntimeKey, shifts: int = 1):
     return roll(x, key, shifts)
 
 
-def make_flex_key_for_new_mask_after_dispatch(*a, **k):
-    raise NotImplementedError("lands in a later round")
+def make_flex_key_for_new_mask_after_dispatch(
+    q_ranges: AttnRanges,
+    k_ranges: AttnRanges,
+    attn_mask_type: GeneralAttnMaskType,
+    total_seqlen_q: int,
+    total_seqlen_k: int,
+    key_for_dispatch: DistAttnRuntimeKey,
+    dist_attn_config: Optional[DistAttnConfig] = None,
+) -> DistAttnRuntimeKey:
+    """New mask, SAME dispatch solution (reference api:1315) — for hybrid-attn
+    models applying several masks within one training pass."""
+    base_mgr = dist_attn_runtime_dict_mgr[key_for_dispatch]
+    cfg = dist_attn_config or base_mgr.config
+    from dataclasses import replace
+
+    key = init_dist_attn_runtime_key(
+        q_ranges, k_ranges, attn_mask_type, total_seqlen_q, total_seqlen_k,
+        key_for_dispatch.pad_size, key_for_dispatch.chunk_size,
+        key_for_dispatch.num_heads_q, key_for_dispatch.num_heads_kv,
+        key_for_dispatch.head_dim, base_mgr.cp_group, cfg,
+    )
+    key = replace(key, dispatch_from=hash(key_for_dispatch))
+    if key not in dist_attn_runtime_dict_mgr:
+        dist_attn_runtime_dict_mgr[key] = DistAttnRuntimeMgr(
+            key, base_mgr.cp_group, cfg, reuse_dispatch_from=base_mgr
+        )
+    return key
 
 
-def make_varlen_key_for_new_mask_after_dispatch(*a, **k):
-    raise NotImplementedError("lands in a later round")
+def make_varlen_key_for_new_mask_after_dispatch(
+    cu_seqlens_q: torch.Tensor,
+    cu_seqlens_k: torch.Tensor,
+    key_for_dispatch: DistAttnRuntimeKey,
+    causal: bool = False,
+    window_size: Tuple[int, int] = (-1, -1),
+    dist_attn_config: Optional[DistAttnConfig] = None,
+) -> DistAttnRuntimeKey:
+    """Varlen front-end of make_flex_key_for_new_mask_after_dispatch
+    (reference api:1167)."""
+    from .functools import infer_attn_mask_from_cu_seqlens
+
+    q_ranges, k_ranges, types, tq, tk = infer_attn_mask_from_cu_seqlens(
+        cu_seqlens_q, cu_seqlens_k, causal=causal, window_size=window_size
+    )
+    return make_flex_key_for_new_mask_after_dispatch(
+        q_ranges, k_ranges, types, tq, tk, key_for_dispatch, dist_attn_config
+    )

@@ -42,6 +42,9 @@ class DistAttnRuntimeKey:
     cp_group_tag: Tuple[int, ...]
     config_tag: str
     env_flags: Tuple[Tuple[str, Any], ...]
+    # set when this key reuses another key's dispatch solution
+    # (make_*_key_for_new_mask_after_dispatch, reference api:1167,1315)
+    dispatch_from: Any = None
 
 
 def _group_tag(group: dist.ProcessGroup) -> Tuple[int, ...]:
@@ -60,6 +63,7 @@ class DistAttnRuntimeMgr:
         key: DistAttnRuntimeKey,
         cp_group: dist.ProcessGroup,
         dist_attn_config: DistAttnConfig,
+        reuse_dispatch_from: "DistAttnRuntimeMgr | None" = None,
     ):
         self.key = key
         self.cp_group = cp_group
@@ -73,9 +77,22 @@ class DistAttnRuntimeMgr:
             list(key.attn_mask_type),
         )
         total_padded = key.total_seqlen_q + key.pad_size
-        self.dispatch_meta: DispatchMeta = make_dispatch_meta_from_qk_ranges(
-            slices, total_padded, cp_size, cp_rank, dist_attn_config
-        )
+        if reuse_dispatch_from is not None:
+            # same dispatch solution, new mask (reference api:1167 semantics)
+            base = reuse_dispatch_from.dispatch_meta
+            assert base.total_seqlen == total_padded, (
+                "new mask must cover the same padded seqlen as the dispatch key"
+            )
+            self.dispatch_meta = DispatchMeta(
+                cp_size=base.cp_size, cp_rank=cp_rank,
+                chunk_size=base.chunk_size, total_seqlen=base.total_seqlen,
+                num_chunks=base.num_chunks, partitions=base.partitions,
+                loads=base.loads,
+            )
+        else:
+            self.dispatch_meta = make_dispatch_meta_from_qk_ranges(
+                slices, total_padded, cp_size, cp_rank, dist_attn_config
+            )
         self.solver, calc_meta, comm_meta = make_attn_meta_from_dispatch_meta(
             slices, self.dispatch_meta, dist_attn_config
         )

@@ -151,3 +151,56 @@ def test_dist_attn_cpu(case_name, ws, degree):
     port = _free_port()
     mp.spawn(_worker, args=(ws, port, case_name, degree, None), nprocs=ws,
              join=True)
+
+
+def _worker_new_mask(rank, ws, port, _a, _b, _c):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=ws)
+    try:
+        import magi_attention.functional.dist_attn as da
+        from magi_attention.api import (
+            calc_attn, dispatch, magi_attn_flex_key,
+            make_flex_key_for_new_mask_after_dispatch, undispatch,
+        )
+        from magi_attention.common.ranges import AttnRanges
+        from magi_attention.config import DispatchConfig, DistAttnConfig
+        from tests.dist_backend import OracleBackend
+
+        da.register_test_attn_backend(OracleBackend)
+        total, hq, hk, d = 512, 4, 2, 32
+        g = torch.Generator().manual_seed(23)
+        q = torch.randn(total, hq, d, generator=g, dtype=torch.float64)
+        k = torch.randn(total, hk, d, generator=g, dtype=torch.float64)
+        v = torch.randn(total, hk, d, generator=g, dtype=torch.float64)
+        cfg = DistAttnConfig(dispatch_config=DispatchConfig(chunk_size=64))
+        # dispatch under mask A (causal), compute under mask B (doc mask)
+        key_a = magi_attn_flex_key(
+            AttnRanges.from_ranges([[0, total]]),
+            AttnRanges.from_ranges([[0, total]]),
+            [1], total, total, hq, hk, d,
+            cp_group_or_mesh=dist.group.WORLD, dist_attn_config=cfg,
+        )
+        qB = [[0, 200], [200, 512]]
+        kB = [[0, 200], [200, 512]]
+        tB = [1, 0]
+        key_b = make_flex_key_for_new_mask_after_dispatch(
+            AttnRanges.from_ranges(qB), AttnRanges.from_ranges(kB), tB,
+            total, total, key_a,
+        )
+        ql = dispatch(q, key_a)
+        kl = dispatch(k, key_a)
+        vl = dispatch(v, key_a)
+        out_b, _ = calc_attn(ql, kl, vl, key_b)
+        full_b = undispatch(out_b, key_b)
+        mask_b = make_attn_mask(total, total, qB, kB, tB)
+        ref_b, _ = ref_attn(q, k, v, mask_b)
+        torch.testing.assert_close(full_b, ref_b, atol=1e-5, rtol=1e-4)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_new_mask_after_dispatch():
+    port = _free_port()
+    mp.spawn(_worker_new_mask, args=(2, port, None, None, None), nprocs=2,
+             join=True)
