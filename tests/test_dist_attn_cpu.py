@@ -36,6 +36,19 @@ CASES = {
         types=[0, 1, 3],
     ),
     "sliding_window": None,  # built in-proc from the sliding-window helper
+    # reference test_pipeline-style configs (scaled down):
+    "q_overlap": dict(  # continuous multi-masks with q overlap
+        total=640,
+        q_ranges=[[0, 256], [0, 256], [256, 640], [256, 640]],
+        k_ranges=[[0, 256], [256, 448], [256, 640], [0, 128]],
+        types=[1, 0, 1, 0],
+    ),
+    "full_from_pieces": dict(  # full mask assembled from small pieces
+        total=512,
+        q_ranges=[[0, 128], [128, 384], [384, 512]],
+        k_ranges=[[0, 512], [0, 512], [0, 512]],
+        types=[0, 0, 0],
+    ),
 }
 
 
