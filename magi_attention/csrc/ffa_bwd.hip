@@ -37,6 +37,29 @@ DEV_INLINE int crow(int r, int hi) { return (r & 3) + 8 * (r >> 2) + 4 * hi; }
 
 DEV_INLINE bool wave_alive(int m0, int qe) { return m0 < qe; }
 
+// ds_read_b64_tr_b16 pair -> one MFMA B-fragment (8 bf16).
+// Probed semantics (tests/gpu_probe_tr16.py): within each 16-lane group, the
+// addresses of lanes =0 (mod 4) give four ROW base addresses; lane l receives
+// element (row_j_base + (l&15)) for j=0..3. Two reads cover 8 rows.
+DEV_INLINE bf16x8 tr16_frag(int a0, int a1) {
+  unsigned long long v0, v1;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %3\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(v0), "=&v"(v1)
+      : "v"(a0), "v"(a1)
+      : "memory");
+  __builtin_amdgcn_sched_barrier(0);
+  union {
+    unsigned long long u[2];
+    bf16x8 v;
+  } r;
+  r.u[0] = v0;
+  r.u[1] = v1;
+  return r.v;
+}
+
 DEV_INLINE unsigned pack_bf16_pair(float lo, float hi) {
   union { __bf16 b; unsigned short u; } a, b;
   a.b = (__bf16)lo;
@@ -169,23 +192,21 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
   // vmcnt(0) before every ds_read, destroying the glds pipeline (guide §5
   // ".s-level traps" (a)).
   __shared__ __attribute__((aligned(16))) char smem[
-      2 * 2 * BWD_BM * D * 2 + 2 * BWD_BM * 4 + 2 * D * 40 * 2];
+      2 * 2 * BWD_BM * D * 2 + 2 * 2 * BWD_BM * 4];
   auto lds_q = [&](int buf) -> __bf16* {
     return (__bf16*)(smem + buf * BWD_BM * D * 2);
   };
   auto lds_do = [&](int buf) -> __bf16* {
     return (__bf16*)(smem + (2 + buf) * BWD_BM * D * 2);
   };
-  // lse / dpsum of the current q tile (staged once per iteration)
-  float* lds_lse = (float*)(smem + 4 * BWD_BM * D * 2);
-  float* lds_dps = (float*)(smem + 4 * BWD_BM * D * 2 + BWD_BM * 4);
-  // transposed copies of the CURRENT Q/dO tile (single-buffered; rebuilt per
-  // iteration): B-frags for dV/dK become b128 reads instead of 64 scalar
-  // reads per fragment
-  __bf16(*lds_qt)[40] =
-      (__bf16(*)[40])(smem + 4 * BWD_BM * D * 2 + 2 * BWD_BM * 4);
-  __bf16(*lds_dot)[40] = (__bf16(*)[40])(smem + 4 * BWD_BM * D * 2 +
-                                         2 * BWD_BM * 4 + D * 40 * 2);
+  // lse / dpsum of the q tile, double-buffered like the glds images (staged
+  // for tile t+1 during tile t)
+  auto lds_lse = [&](int buf) -> float* {
+    return (float*)(smem + 4 * BWD_BM * D * 2 + buf * 2 * BWD_BM * 4);
+  };
+  auto lds_dps = [&](int buf) -> float* {
+    return lds_lse(buf) + BWD_BM;
+  };
 
   // K/V fragments (A-layout) + K B-fragments, loaded once per block
   const int krow = n0 + lo32;
@@ -234,7 +255,7 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
       const int r = r0 + lane / (ROWB / 16);
       const int c = lane % (ROWB / 16);
       const int qrow = min(m0x + r, qe - 1);
-      const int csw = (c ^ (r & SWZM)) * 8;
+      const int csw = (c ^ ((r & 7) << 1)) * 8;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)(
               p.q + (size_t)qrow * q_pitch + (size_t)h * D + csw),
@@ -246,6 +267,12 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
           (__attribute__((address_space(3))) unsigned int*)&lds_do(buf)[r0 * D],
           16, 0, 0);
     }
+    // lse/dpsum for the same q tile
+    if (tid < BWD_BM) {
+      const int qr = min(m0x + tid, qe - 1);
+      lds_lse(buf)[tid] = p.lse[(size_t)qr * p.hq + h];
+      lds_dps(buf)[tid] = p.dpsum[(size_t)qr * p.hq + h];
+    }
   };
 
   int cur = 0;
@@ -254,31 +281,6 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
   for (int m0 = q_lo; m0 < q_hi; m0 += BWD_BM) {
     __syncthreads();  // buf[cur] ready (barrier drains in-flight glds)
     if (m0 + BWD_BM < q_hi) stage_glds(cur ^ 1, m0 + BWD_BM);
-    {
-      constexpr int CPR = D / 8;
-      constexpr int RPP = 256 / CPR;
-      const int row = tid / CPR;
-      const int col = tid % CPR;
-      const int bs = (col & 3) << 3;
-#pragma unroll 1
-      for (int pass = 0; pass < BWD_BM / RPP; ++pass) {
-        const int r = pass * RPP + row;
-        const int src = swz(r, r * ROWB + col * 16);
-        const bf16x8 qv = *(const bf16x8*)((const char*)lds_q(cur) + src);
-        const bf16x8 dv8 = *(const bf16x8*)((const char*)lds_do(cur) + src);
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          lds_qt[col * 8 + e][r ^ bs] = qv[e];
-          lds_dot[col * 8 + e][r ^ bs] = dv8[e];
-        }
-      }
-      if (tid < BWD_BM) {
-        const int qr = min(m0 + tid, qe - 1);
-        lds_lse[tid] = p.lse[(size_t)qr * p.hq + h];
-        lds_dps[tid] = p.dpsum[(size_t)qr * p.hq + h];
-      }
-    }
-    __syncthreads();
 
     if (wave_live && m0 + BWD_BM > wq_lo && m0 < wq_hi) {
       // ---- S = Q K^T ; dP = dO V^T, UN-swapped: C layout [q=crow][k=lo32],
@@ -302,10 +304,9 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
           !((atype == 2 || atype == 3) && (n0 < m0 + BWD_BM - 1 + (ks - qs)));
       bool all_live = interior;
       if (interior) {
-        // rows' lse finite? checked via the staged tile by reg
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          const float lq = lds_lse[crow(r, hi)];
+          const float lq = lds_lse(cur)[crow(r, hi)];
           all_live = all_live && (lq != INFINITY) && (lq != -INFINITY);
         }
       }
@@ -313,15 +314,15 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int rl = crow(r, hi);
-          const float pij = exp2f(s[r] * sl2 - lds_lse[rl] * log2e);
+          const float pij = exp2f(s[r] * sl2 - lds_lse(cur)[rl] * log2e);
           pv[r] = pij;
-          dsv[r] = pij * (dp[r] - lds_dps[rl]) * p.scale;
+          dsv[r] = pij * (dp[r] - lds_dps(cur)[rl]) * p.scale;
         }
       } else {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int qrow = m0 + crow(r, hi);
-          const float lq = (qrow < qe) ? lds_lse[crow(r, hi)] : INFINITY;
+          const float lq = (qrow < qe) ? lds_lse(cur)[crow(r, hi)] : INFINITY;
           bool ok = (qrow < wq_hi) && (qrow >= qs) && lq != INFINITY &&
                     lq != -INFINITY && kk < ke;
           if (atype == 1 || atype == 3) ok = ok && (kk - qrow <= ke - qe);
@@ -338,45 +339,58 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
           }
           const float pij = ok ? exp2f(t - lq * log2e) : 0.f;
           pv[r] = pij;
-          dsv[r] = pij * (dp[r] - lds_dps[crow(r, hi)]) * dscale;
+          dsv[r] = pij * (dp[r] - lds_dps(cur)[crow(r, hi)]) * dscale;
         }
       }
 
-      const int kbs = ((lo32 >> 3) & 3) << 3;
-      // ---- dV += P^T dO : A = cframe(pv) = [k=lo32][q-slice] ----
+      // ---- dV += P^T dO ; dK += dS^T Q ----
+      // B-fragments read straight off the ROW images with ds_read_b64_tr_b16:
+      // per 16-lane group, q-row bases rd*4+j supply the transpose; the 32-B
+      // runs stay contiguous under the 32-B-granular swizzle. Per (dt, rd):
+      //   addr = lds_base + row*ROWB + ((dcol*2) ^ ((row&7)<<5)) + (l&3)*8
+      // (the XOR distributes because the swizzle bits live inside dcol*2).
       {
         bf16x8 pa0 = cframe_to_afrag(pv, 0);
         bf16x8 pa1 = cframe_to_afrag(pv, 1);
-#pragma unroll
-        for (int dt = 0; dt < DT; ++dt) {
-          bf16x8 b = *(const bf16x8*)(&lds_dot[dt * 32 + lo32][(8 * hi) ^ kbs]);
-          acc_dv[dt] =
-              __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa0, b, acc_dv[dt], 0, 0, 0);
-        }
-#pragma unroll
-        for (int dt = 0; dt < DT; ++dt) {
-          bf16x8 b =
-              *(const bf16x8*)(&lds_dot[dt * 32 + lo32][(16 + 8 * hi) ^ kbs]);
-          acc_dv[dt] =
-              __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa1, b, acc_dv[dt], 0, 0, 0);
-        }
-      }
-      // ---- dK += dS^T Q ----
-      {
         bf16x8 da0 = cframe_to_afrag(dsv, 0);
         bf16x8 da1 = cframe_to_afrag(dsv, 1);
+        const int qhalf = (lane >> 4) & 1;       // which 16-d column half
+        const int jrow = (lane & 15) >> 2;       // canonical row for this lane
+        const int row0 = 8 * hi + jrow;          // rd = 0 rows
+        const int row1 = 8 * hi + 4 + jrow;      // rd = 1 rows
+        const int q_base = (int)(unsigned long long)(
+            (__attribute__((address_space(3))) char*)lds_q(cur));
+        const int do_base = (int)(unsigned long long)(
+            (__attribute__((address_space(3))) char*)lds_do(cur));
+        const int lane8 = (lane & 3) * 8;
+        const int sw0 = (row0 & 7) << 5;
+        const int sw1 = (row1 & 7) << 5;
+        const int rb0 = row0 * ROWB + lane8;
+        const int rb1 = row1 * ROWB + lane8;
+        // q rows 16..31 (second kk sub-tile)
+        const int sw2 = ((row0 + 16) & 7) << 5;
+        const int sw3 = ((row1 + 16) & 7) << 5;
+        const int rb2 = (row0 + 16) * ROWB + lane8;
+        const int rb3 = (row1 + 16) * ROWB + lane8;
 #pragma unroll
         for (int dt = 0; dt < DT; ++dt) {
-          bf16x8 b = *(const bf16x8*)(&lds_qt[dt * 32 + lo32][(8 * hi) ^ kbs]);
+          const int dcol = (dt * 32 + 16 * qhalf) * 2;
+          bf16x8 bdo = tr16_frag(do_base + rb0 + (dcol ^ sw0),
+                                 do_base + rb1 + (dcol ^ sw1));
+          acc_dv[dt] =
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa0, bdo, acc_dv[dt], 0, 0, 0);
+          bf16x8 bq = tr16_frag(q_base + rb0 + (dcol ^ sw0),
+                                q_base + rb1 + (dcol ^ sw1));
           acc_dk[dt] =
-              __builtin_amdgcn_mfma_f32_32x32x16_bf16(da0, b, acc_dk[dt], 0, 0, 0);
-        }
-#pragma unroll
-        for (int dt = 0; dt < DT; ++dt) {
-          bf16x8 b =
-              *(const bf16x8*)(&lds_qt[dt * 32 + lo32][(16 + 8 * hi) ^ kbs]);
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(da0, bq, acc_dk[dt], 0, 0, 0);
+          bf16x8 bdo1 = tr16_frag(do_base + rb2 + (dcol ^ sw2),
+                                  do_base + rb3 + (dcol ^ sw3));
+          acc_dv[dt] =
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa1, bdo1, acc_dv[dt], 0, 0, 0);
+          bf16x8 bq1 = tr16_frag(q_base + rb2 + (dcol ^ sw2),
+                                 q_base + rb3 + (dcol ^ sw3));
           acc_dk[dt] =
-              __builtin_amdgcn_mfma_f32_32x32x16_bf16(da1, b, acc_dk[dt], 0, 0, 0);
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(da1, bq1, acc_dk[dt], 0, 0, 0);
         }
       }
     }

@@ -527,3 +527,39 @@ extern "C" int magi_ffa_fwd(const magi_ffa_fwd_args* a) {
 }
 
 extern "C" int magi_ffa_abi_version(void) { return 1; }
+
+// ------------------------------------------------------------------
+// probe ds_read_b64_tr_b16 semantics (see tests/gpu_probe_tr16.py)
+// ------------------------------------------------------------------
+__global__ void probe_tr16_kernel(const unsigned short* in, unsigned short* out,
+                                  int mode) {
+  __shared__ __attribute__((aligned(16))) unsigned short lds[2048];
+  const int lane = threadIdx.x & 63;
+  for (int i = lane; i < 2048; i += 64) lds[i] = in[i];
+  __syncthreads();
+  int addr;  // byte address into LDS
+  if (mode == 0) addr = 0;                       // uniform base
+  else if (mode == 1) addr = lane * 8;           // lane-linear 8B
+  else addr = (lane & 15) * 8 + (lane >> 4) * 128;  // grouped guess
+  unsigned long long v;
+  const int a = (int)(unsigned long long)(
+                    (__attribute__((address_space(3))) char*)lds) +
+                addr;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %1\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=v"(v)
+      : "v"(a)
+      : "memory");
+  __builtin_amdgcn_sched_barrier(0);
+  for (int j = 0; j < 4; ++j)
+    out[lane * 4 + j] = (unsigned short)(v >> (16 * j));
+}
+
+extern "C" int magi_probe_tr16(const void* in, void* out, int mode,
+                               void* stream) {
+  hipLaunchKernelGGL(probe_tr16_kernel, dim3(1), dim3(64), 0,
+                     (hipStream_t)stream, (const unsigned short*)in,
+                     (unsigned short*)out, mode);
+  return (int)hipGetLastError();
+}
