@@ -255,7 +255,7 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
       const int r = r0 + lane / (ROWB / 16);
       const int c = lane % (ROWB / 16);
       const int qrow = min(m0x + r, qe - 1);
-      const int csw = (c ^ ((r & 7) << 1)) * 8;
+      const int csw = (c ^ ((r & SW32M) << 1)) * 8;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)(
               p.q + (size_t)qrow * q_pitch + (size_t)h * D + csw),
@@ -363,13 +363,13 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
         const int do_base = (int)(unsigned long long)(
             (__attribute__((address_space(3))) char*)lds_do(cur));
         const int lane8 = (lane & 3) * 8;
-        const int sw0 = (row0 & 7) << 5;
-        const int sw1 = (row1 & 7) << 5;
+        const int sw0 = (row0 & SW32M) << 5;
+        const int sw1 = (row1 & SW32M) << 5;
         const int rb0 = row0 * ROWB + lane8;
         const int rb1 = row1 * ROWB + lane8;
         // q rows 16..31 (second kk sub-tile)
-        const int sw2 = ((row0 + 16) & 7) << 5;
-        const int sw3 = ((row1 + 16) & 7) << 5;
+        const int sw2 = ((row0 + 16) & SW32M) << 5;
+        const int sw3 = ((row1 + 16) & SW32M) << 5;
         const int rb2 = (row0 + 16) * ROWB + lane8;
         const int rb3 = (row1 + 16) * ROWB + lane8;
 #pragma unroll
