@@ -148,9 +148,13 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
   constexpr int ROWB = D * 2;  // bytes per LDS tile row
   // XOR swizzle within one LDS row: spreads a b128 lane group over the row's
   // 16-B slots (T2/G4); mask keeps the XOR inside the row for D=64 too.
-  constexpr int SWZM = ROWB / 16 - 1;
+  // 32-B-granular XOR swizzle: spreads b128 lane groups over the row's 16-B
+  // slot pairs (<=2-way) while keeping every 32-B run physically contiguous,
+  // which the tr16 row reads require; mask scales with row size (D=64 rows
+  // are 128 B).
+  constexpr int SW32M = ROWB / 32 - 1;
   auto swz = [](int row, int byte_off) {
-    return byte_off ^ ((row & SWZM) << 4);
+    return byte_off ^ ((row & SW32M) << 5);
   };
   // blockIdx.x = HEAD: the dispatcher places block b on XCD b%8, so every
   // block of one head shares one XCD/L2 — dq/dk/dv atomics stay XCD-local and
