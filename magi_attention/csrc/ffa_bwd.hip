@@ -430,9 +430,9 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dq_kernel(BwdParams p) {
   constexpr int DF = D / 16;
   constexpr int DT = D / 32;
   constexpr int ROWB = D * 2;
-  constexpr int SWZM = ROWB / 16 - 1;
+  constexpr int SW32M = ROWB / 32 - 1;  // 32-B-granular swizzle (tr16 reads)
   auto swz = [](int row, int byte_off) {
-    return byte_off ^ ((row & SWZM) << 4);
+    return byte_off ^ ((row & SW32M) << 5);
   };
   const int ri = blockIdx.z;
   const int h = blockIdx.x;  // XCD-affine: one head -> one XCD
@@ -465,16 +465,12 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dq_kernel(BwdParams p) {
   const size_t q_pitch = (size_t)p.hq * D;
 
   // single shared object (glds-pipeline trap, see dkv kernel)
-  __shared__ __attribute__((aligned(16))) char smem[
-      2 * 2 * BWD_BN * D * 2 + 2 * D * 40 * 2];
+  __shared__ __attribute__((aligned(16))) char smem[2 * 2 * BWD_BN * D * 2];
   auto lds_k = [&](int buf) -> __bf16* {
     return (__bf16*)(smem + buf * BWD_BN * D * 2);
   };
   auto lds_v = [&](int buf) -> __bf16* {
     return (__bf16*)(smem + (2 + buf) * BWD_BN * D * 2);
-  };
-  auto lds_kt = [&](int buf) -> __bf16(*)[40] {
-    return (__bf16(*)[40])(smem + 4 * BWD_BN * D * 2 + buf * D * 40 * 2);
   };
 
   // persistent per-wave operands: Q and dO fragments (B-layout rows)
@@ -515,7 +511,7 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dq_kernel(BwdParams p) {
       const int r = r0 + lane / (ROWB / 16);
       const int c = lane % (ROWB / 16);
       const int kr = min(n0x + r, ke - 1);
-      const int csw = (c ^ (r & SWZM)) * 8;
+      const int csw = (c ^ ((r & SW32M) << 1)) * 8;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)(
               p.k + (size_t)kr * k_pitch + (size_t)kh * D + csw),
@@ -535,23 +531,6 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dq_kernel(BwdParams p) {
   for (int n0 = k_lo; n0 < k_hi; n0 += BWD_BN) {
     __syncthreads();  // glds for buf[cur] drained here
     if (n0 + BWD_BN < k_hi) stage_glds(cur ^ 1, n0 + BWD_BN);
-    // transpose K into kt[cur] (kt B-frags for the dq MFMA)
-    {
-      constexpr int CPR = D / 8;
-      constexpr int RPP = 256 / CPR;
-      const int row = tid / CPR;
-      const int col = tid % CPR;
-      const int bs = (col & 3) << 3;  // bank-spread swizzle
-#pragma unroll 1
-      for (int pass = 0; pass < BWD_BN / RPP; ++pass) {
-        const int r = pass * RPP + row;
-        const bf16x8 kv8 = *(const bf16x8*)((const char*)lds_k(cur) +
-                                            swz(r, r * ROWB + col * 16));
-#pragma unroll
-        for (int e = 0; e < 8; ++e) lds_kt(cur)[col * 8 + e][r ^ bs] = kv8[e];
-      }
-    }
-    __syncthreads();
 
     if (wave_alive(m0, qe) && n0 + BWD_BN > wk_lo && n0 < wk_hi) {
       // ---- S^T = K Q^T ; dP^T = V dO^T (K/V A-frags from LDS rows) ----
@@ -602,18 +581,36 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dq_kernel(BwdParams p) {
       // ---- dq += dS K (B-frags from the transposed K tile) ----
       bf16x8 dsa0 = cframe_to_afrag(dsv, 0);
       bf16x8 dsa1 = cframe_to_afrag(dsv, 1);
-      const int kbs = ((lo32 >> 3) & 3) << 3;
-      // two sweeps so consecutive MFMAs hit DIFFERENT accumulators
-      // (dependent-accumulator latency is 2x the issue interval)
+      // B-frags (B[kk=k][j=d]) via tr16 straight off the K row image
+      // (same recipe as the dkv kernel)
+      {
+        const int qhalf2 = (lane >> 4) & 1;
+        const int jrow = (lane & 15) >> 2;
+        const int row0 = 8 * hi + jrow;
+        const int row1 = 8 * hi + 4 + jrow;
+        const int k_base = (int)(unsigned long long)(
+            (__attribute__((address_space(3))) char*)lds_k(cur));
+        const int lane8 = (lane & 3) * 8;
+        const int sw0 = (row0 & SW32M) << 5;
+        const int sw1 = (row1 & SW32M) << 5;
+        const int rb0 = row0 * ROWB + lane8;
+        const int rb1 = row1 * ROWB + lane8;
+        const int sw2 = ((row0 + 16) & SW32M) << 5;
+        const int sw3 = ((row1 + 16) & SW32M) << 5;
+        const int rb2 = (row0 + 16) * ROWB + lane8;
+        const int rb3 = (row1 + 16) * ROWB + lane8;
 #pragma unroll
-      for (int dt = 0; dt < DT; ++dt) {
-        bf16x8 b0 = *(const bf16x8*)(&lds_kt(cur)[dt * 32 + lo32][(8 * hi) ^ kbs]);
-        acc_dq[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa0, b0, acc_dq[dt], 0, 0, 0);
-      }
-#pragma unroll
-      for (int dt = 0; dt < DT; ++dt) {
-        bf16x8 b1 = *(const bf16x8*)(&lds_kt(cur)[dt * 32 + lo32][(16 + 8 * hi) ^ kbs]);
-        acc_dq[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa1, b1, acc_dq[dt], 0, 0, 0);
+        for (int dt = 0; dt < DT; ++dt) {
+          const int dcol = (dt * 32 + 16 * qhalf2) * 2;
+          bf16x8 b0 = tr16_frag(k_base + rb0 + (dcol ^ sw0),
+                                k_base + rb1 + (dcol ^ sw1));
+          acc_dq[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa0, b0,
+                                                              acc_dq[dt], 0, 0, 0);
+          bf16x8 b1 = tr16_frag(k_base + rb2 + (dcol ^ sw2),
+                                k_base + rb3 + (dcol ^ sw3));
+          acc_dq[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa1, b1,
+                                                              acc_dq[dt], 0, 0, 0);
+        }
       }
     }
     cur ^= 1;
