@@ -49,6 +49,27 @@ DEV_INLINE unsigned pack_bf16_pair(float lo, float hi) {
 // C/D fragment row for v_mfma_f32_32x32x16_bf16: reg r, lane-half hi
 DEV_INLINE int crow(int r, int hi) { return (r & 3) + 8 * (r >> 2) + 4 * hi; }
 
+// ds_read_b64_tr_b16 pair -> one MFMA B-fragment (see ffa_bwd.hip for the
+// probed semantics).
+DEV_INLINE bf16x8 tr16_frag(int a0, int a1) {
+  unsigned long long v0, v1;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %3\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(v0), "=&v"(v1)
+      : "v"(a0), "v"(a1)
+      : "memory");
+  __builtin_amdgcn_sched_barrier(0);
+  union {
+    unsigned long long u[2];
+    bf16x8 v;
+  } r;
+  r.u[0] = v0;
+  r.u[1] = v1;
+  return r.v;
+}
+
 struct FwdParams {
   const bf16_t* q;
   const bf16_t* k;
@@ -119,12 +140,19 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
     b_hi = b_lo;
   }
   constexpr int ROWB = D * 2;
-  constexpr int SWZM = ROWB / 16 - 1;
+  // 32-B-granular swizzle: b128 lane groups spread (<=2-way) AND every 32-B
+  // run stays physically contiguous for the tr16 V reads (see bwd kernels)
+  constexpr int SW32M = ROWB / 32 - 1;
   auto swz = [](int row, int byte_off) {
-    return byte_off ^ ((row & SWZM) << 4);
+    return byte_off ^ ((row & SW32M) << 5);
   };
-  __shared__ __bf16 lds_k[2][FFA_BN * D];
-  __shared__ __bf16 lds_vt[2][D][40];
+  __shared__ __attribute__((aligned(16))) char fsmem[2 * 2 * FFA_BN * D * 2];
+  auto lds_k = [&](int buf) -> __bf16* {
+    return (__bf16*)(fsmem + buf * FFA_BN * D * 2);
+  };
+  auto lds_v = [&](int buf) -> __bf16* {
+    return (__bf16*)(fsmem + (2 + buf) * FFA_BN * D * 2);
+  };
 
   // Q fragments in registers (8 x bf16x8 for D=128)
   bf16x8 qf[DF];
@@ -167,14 +195,9 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
 #pragma unroll
     for (int pass = 0; pass < NPASS; ++pass) {
       const int r = pass * RPP + srow;
-      *(bf16x8*)((char*)lds_k[buf] + swz(r, r * ROWB + scol * 16)) = kreg[pass];
-      // 8-block XOR swizzle on the k index: spreads the 16 column groups
-      // over distinct banks (unswizzled, every scol group hit the same bank:
-      // SQ_LDS_BANK_CONFLICT was 24% of fwd wave cycles)
-      const int bs = (scol & 3) << 3;
-#pragma unroll
-      for (int e = 0; e < 8; ++e)
-        lds_vt[buf][scol * 8 + e][r ^ bs] = vreg[pass][e];
+      const int dst = swz(r, r * ROWB + scol * 16);
+      *(bf16x8*)((char*)lds_k(buf) + dst) = kreg[pass];
+      *(bf16x8*)((char*)lds_v(buf) + dst) = vreg[pass];
     }
   };
 
@@ -195,7 +218,7 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
 #pragma unroll
       for (int dd = 0; dd < DF; ++dd) {
         bf16x8 kf = *(const bf16x8*)(
-            (const char*)lds_k[cur] + swz(lo32, lo32 * ROWB + dd * 32 + hi * 16));
+            (const char*)lds_k(cur) + swz(lo32, lo32 * ROWB + dd * 32 + hi * 16));
         s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[dd], s, 0, 0, 0);
       }
     }
@@ -296,16 +319,28 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
       pa[tt] = cvt.v;
     }
 
-    // ---- PV: O[32q][32d] += P^T V (B-frags from the transposed V tile) ----
+    // ---- PV: B-frags via ds_read_b64_tr_b16 off the V row image ----
+    {
+      const int qhalf = (lane >> 4) & 1;
+      const int jrow = (lane & 15) >> 2;
+      const int v_base = (int)(unsigned long long)(
+          (__attribute__((address_space(3))) char*)lds_v(cur));
+      const int lane8 = (lane & 3) * 8;
 #pragma unroll
-    for (int tt = 0; tt < 2; ++tt) {
+      for (int tt = 0; tt < 2; ++tt) {
+        const int row0 = 16 * tt + 8 * hi + jrow;
+        const int row1 = row0 + 4;
+        const int sw0 = (row0 & SW32M) << 5;
+        const int sw1 = (row1 & SW32M) << 5;
+        const int rb0 = v_base + row0 * ROWB + lane8;
+        const int rb1 = v_base + row1 * ROWB + lane8;
 #pragma unroll
-      for (int dt = 0; dt < DT; ++dt) {
-        bf16x8 bv = *(const bf16x8*)(
-            &lds_vt[cur][dt * 32 + lo32]
-                   [(16 * tt + 8 * hi) ^ (((lo32 >> 3) & 3) << 3)]);
-        acc_o[dt] =
-            __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[tt], bv, acc_o[dt], 0, 0, 0);
+        for (int dt = 0; dt < DT; ++dt) {
+          const int dcol = (dt * 32 + 16 * qhalf) * 2;
+          bf16x8 bv = tr16_frag(rb0 + (dcol ^ sw0), rb1 + (dcol ^ sw1));
+          acc_o[dt] =
+              __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[tt], bv, acc_o[dt], 0, 0, 0);
+        }
       }
     }
     if (has_next) write_stage(cur ^ 1);  // other buffer: overlaps this compute
