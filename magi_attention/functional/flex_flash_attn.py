@@ -184,7 +184,7 @@ class FlexFlashAttnFunc(torch.autograd.Function):
         auto_range_merge=False, swap_ab=False, pack_gqa=False, cat_gqa=False,
         sparse_load=False, index_attn=False, swap_bwd_qk_loop=False,
         return_max_logits=False, index_attn_indices_2d=None,
-        index_attn_max_topk=0,
+        index_attn_max_topk=0, max_seqlen_k=None,
     ):
         softmax_scale = (
             q.shape[-1] ** (-0.5) if softmax_scale is None else softmax_scale
@@ -211,6 +211,8 @@ class FlexFlashAttnFunc(torch.autograd.Function):
         ctx.deterministic = deterministic
         ctx.sm_margin = sm_margin
         ctx.sink_layout = sink_layout
+        # avoid a per-backward device sync to size the bwd grid
+        ctx.max_seqlen_k = max_seqlen_k
         return out, lse, None
 
     @staticmethod
@@ -224,11 +226,12 @@ class FlexFlashAttnFunc(torch.autograd.Function):
             dq_type=None, dk_type=None, dv_type=None,
             disable_bwd_dkv_atomic_reduction=False,
             deterministic=ctx.deterministic, sm_margin=ctx.sm_margin,
+            max_seqlen_k=ctx.max_seqlen_k,
         )
         dq = dq.to(q.dtype)
         dk = dk.to(k.dtype)
         dv = dv.to(v.dtype)
-        return (dq, dk, dv) + (None,) * 24
+        return (dq, dk, dv) + (None,) * 25
 
 
 def flex_flash_attn_func(
@@ -260,6 +263,7 @@ def flex_flash_attn_func(
     index_attn: bool = False,
     swap_bwd_qk_loop: bool = False,
     return_max_logits: bool = False,
+    max_seqlen_k: Optional[int] = None,
 ) -> tuple[torch.Tensor, AttnForwardMeta]:
     """Single-GPU flex-flash-attention (drop-in for the reference
     flex_flash_attn_func, flex_flash_attn.py:1066; full mask semantics in the
@@ -273,6 +277,6 @@ def flex_flash_attn_func(
         disable_fwd_atomic_reduction, disable_bwd_dkv_atomic_reduction,
         ref_block_size, max_seqlen_q, auto_range_merge, swap_ab, pack_gqa,
         cat_gqa, sparse_load, index_attn, swap_bwd_qk_loop, return_max_logits,
-        None, 0,
+        None, 0, max_seqlen_k,
     )
     return out, AttnForwardMeta(lse=lse, max_logits=max_logits)

@@ -17,11 +17,12 @@ def bench(name, tq, tk, hq, hk, d, q_ranges, k_ranges, types, area, steps=10, wa
     tm = torch.tensor(types, dtype=torch.int32, device="cuda")
     do = torch.randn_like(q)
     ms = max(r[1]-r[0] for r in q_ranges)
+    msk = max(r[1]-r[0] for r in k_ranges)
     def fwd_only():
         with torch.no_grad():
-            flex_flash_attn_func(q, k, v, qr, kr, tm, max_seqlen_q=ms)
+            flex_flash_attn_func(q, k, v, qr, kr, tm, max_seqlen_q=ms, max_seqlen_k=msk)
     def fwd_bwd():
-        out, _ = flex_flash_attn_func(q, k, v, qr, kr, tm, max_seqlen_q=ms)
+        out, _ = flex_flash_attn_func(q, k, v, qr, kr, tm, max_seqlen_q=ms, max_seqlen_k=msk)
         out.backward(do)
         q.grad = k.grad = v.grad = None
     for fn, tag, mult in ((fwd_only, "fwd", 1.0),) + (((fwd_bwd, "fwd+bwd", 3.5),) if bwd else ()):
