@@ -85,9 +85,13 @@ typedef struct magi_ffa_bwd_args {
  * (reference flash_bwd_preprocess_kernel.h:42). */
 int magi_ffa_bwd_preprocess(const magi_ffa_bwd_args* args);
 
-/* 5-matmul backward mainloop (reference flash_bwd_kernel_sm90.h:41):
- * recompute P, dV += P^T dO, dP = dO V^T, dS = P*(dP-dpsum),
- * dK += dS^T Q, dQ += dS K (fp32 atomics). Requires dpsum filled. */
+/* 5-matmul backward (reference flash_bwd_kernel_sm90.h:41), split into two
+ * INDEPENDENT passes so the host may run them on separate streams
+ * concurrently: the dq pass (q-outer, dQ register-accumulated) and the dkv
+ * pass (k-outer, dK/dV register-accumulated). Both require dpsum filled.
+ * magi_ffa_bwd runs both sequentially on args->stream. */
+int magi_ffa_bwd_dq(const magi_ffa_bwd_args* args);
+int magi_ffa_bwd_dkv(const magi_ffa_bwd_args* args);
 int magi_ffa_bwd(const magi_ffa_bwd_args* args);
 
 /* ------------------------------------------------------------------ *

@@ -651,58 +651,87 @@ extern "C" int magi_ffa_bwd_preprocess(const magi_ffa_bwd_args* a) {
   return (int)hipGetLastError();
 }
 
-extern "C" int magi_ffa_bwd(const magi_ffa_bwd_args* a) {
+static int fill_bwd_params(const magi_ffa_bwd_args* a, BwdParams* p) {
   if (!a || !a->dout || !a->q || !a->k || !a->v || !a->lse) return -1;
   if (a->d != 64 && a->d != 128) return -2;
   if (a->hq % a->hk != 0) return -3;
-  if (a->n_ranges <= 0) return 0;
+  if (a->n_ranges <= 0) return 1;  // caller treats >0 as "nothing to do"
+  p->dout = (const bf16_t*)a->dout;
+  p->q = (const bf16_t*)a->q;
+  p->k = (const bf16_t*)a->k;
+  p->v = (const bf16_t*)a->v;
+  p->out = a->out;
+  p->lse = a->lse;
+  p->dq = a->dq;
+  p->dk = a->dk;
+  p->dv = a->dv;
+  p->dpsum = a->dpsum;
+  p->q_ranges = a->q_ranges;
+  p->k_ranges = a->k_ranges;
+  p->attn_type_map = a->attn_type_map;
+  p->hq = a->hq;
+  p->hk = a->hk;
+  p->gqa = a->hq / a->hk;
+  p->scale = a->softmax_scale;
+  p->softcap = a->softcap;
+  p->total_q = a->total_q;
+  p->total_k = a->total_k;
+  { const char* e = getenv("MAGI_BWD_ABLATE"); p->debug_ablate = e ? atoi(e) : 0; }
+  return 0;
+}
 
-  BwdParams p{};
-  p.dout = (const bf16_t*)a->dout;
-  p.q = (const bf16_t*)a->q;
-  p.k = (const bf16_t*)a->k;
-  p.v = (const bf16_t*)a->v;
-  p.out = a->out;
-  p.lse = a->lse;
-  p.dq = a->dq;
-  p.dk = a->dk;
-  p.dv = a->dv;
-  p.dpsum = a->dpsum;
-  p.q_ranges = a->q_ranges;
-  p.k_ranges = a->k_ranges;
-  p.attn_type_map = a->attn_type_map;
-  p.hq = a->hq;
-  p.hk = a->hk;
-  p.gqa = a->hq / a->hk;
-  p.scale = a->softmax_scale;
-  p.softcap = a->softcap;
-  p.total_q = a->total_q;
-  p.total_k = a->total_k;
-  { const char* e = getenv("MAGI_BWD_ABLATE"); p.debug_ablate = e ? atoi(e) : 0; }
+extern "C" int magi_ffa_bwd_dq(const magi_ffa_bwd_args* a) {
+  BwdParams p;
+  int rc = fill_bwd_params(a, &p);
+  if (rc) return rc > 0 ? 0 : rc;
 
-  const int span = BWD_BN * BWD_WAVES;
-  const int nblocks = (a->max_seqlen_k + span - 1) / span;
   const int qspan = BWD_BM * BWD_WAVES;
-  // max_seqlen_q is not in the bwd ABI; bound q blocks by total_q (tight
-  // enough: empty blocks exit on their first range check)
   const int qblocks = (int)((a->total_q + qspan - 1) / qspan);
   if (a->n_ranges > 65535) return -5;
-  dim3 grid_kv(a->hq, nblocks, (unsigned)a->n_ranges), block(64 * BWD_WAVES);
-  dim3 grid_q(a->hq, qblocks, (unsigned)a->n_ranges);
+  dim3 grid_q(a->hq, qblocks, (unsigned)a->n_ranges), block(64 * BWD_WAVES);
   hipStream_t s = (hipStream_t)a->stream;
   const bool sc = a->softcap > 0.f;
-#define LAUNCH_BWD(DD, SC)                                                   \
-  do {                                                                       \
-    hipLaunchKernelGGL((ffa_bwd_dq_kernel<DD, SC>), grid_q, block, 0, s, p); \
-    hipLaunchKernelGGL((ffa_bwd_dkv_kernel<DD, SC>), grid_kv, block, 0, s, p); \
-  } while (0)
   if (a->d == 64) {
-    if (sc) LAUNCH_BWD(64, true);
-    else LAUNCH_BWD(64, false);
+    if (sc)
+      hipLaunchKernelGGL((ffa_bwd_dq_kernel<64, true>), grid_q, block, 0, s, p);
+    else
+      hipLaunchKernelGGL((ffa_bwd_dq_kernel<64, false>), grid_q, block, 0, s, p);
   } else {
-    if (sc) LAUNCH_BWD(128, true);
-    else LAUNCH_BWD(128, false);
+    if (sc)
+      hipLaunchKernelGGL((ffa_bwd_dq_kernel<128, true>), grid_q, block, 0, s, p);
+    else
+      hipLaunchKernelGGL((ffa_bwd_dq_kernel<128, false>), grid_q, block, 0, s, p);
   }
-#undef LAUNCH_BWD
   return (int)hipGetLastError();
+}
+
+extern "C" int magi_ffa_bwd_dkv(const magi_ffa_bwd_args* a) {
+  BwdParams p;
+  int rc = fill_bwd_params(a, &p);
+  if (rc) return rc > 0 ? 0 : rc;
+  const int span = BWD_BN * BWD_WAVES;
+  const int nblocks = (a->max_seqlen_k + span - 1) / span;
+  if (a->n_ranges > 65535) return -5;
+  dim3 grid_kv(a->hq, nblocks, (unsigned)a->n_ranges), block(64 * BWD_WAVES);
+  hipStream_t s = (hipStream_t)a->stream;
+  const bool sc = a->softcap > 0.f;
+  if (a->d == 64) {
+    if (sc)
+      hipLaunchKernelGGL((ffa_bwd_dkv_kernel<64, true>), grid_kv, block, 0, s, p);
+    else
+      hipLaunchKernelGGL((ffa_bwd_dkv_kernel<64, false>), grid_kv, block, 0, s, p);
+  } else {
+    if (sc)
+      hipLaunchKernelGGL((ffa_bwd_dkv_kernel<128, true>), grid_kv, block, 0, s, p);
+    else
+      hipLaunchKernelGGL((ffa_bwd_dkv_kernel<128, false>), grid_kv, block, 0, s, p);
+  }
+  return (int)hipGetLastError();
+}
+
+extern "C" int magi_ffa_bwd(const magi_ffa_bwd_args* a) {
+  // sequential convenience form: dq pass then dkv pass on one stream
+  int rc = magi_ffa_bwd_dq(a);
+  if (rc) return rc;
+  return magi_ffa_bwd_dkv(a);
 }

@@ -196,7 +196,9 @@ class DistAttnRuntime:
             cu_margin=env.ffa_backward_sm_margin(),
             stream=current_stream_ptr(),
         )
-        check(_ffa_lib.lib().magi_ffa_bwd(args), "magi_ffa_bwd")
+        from .flex_flash_attn import run_bwd_passes
+
+        run_bwd_passes(args, q.device)
 
 
 class DistAttnFunc(torch.autograd.Function):

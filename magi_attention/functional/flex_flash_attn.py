@@ -5,10 +5,38 @@
 # library (include/magi_ffa.h) bound via ctypes in magi_attention/_ffa_lib.py.
 from __future__ import annotations
 
+import ctypes
 import math
 from typing import Optional, Tuple
 
 import torch
+
+_side_stream: dict = {}
+
+
+def _get_side_stream(device) -> torch.cuda.Stream:
+    key = device.index or 0
+    if key not in _side_stream:
+        _side_stream[key] = torch.cuda.Stream(device=device)
+    return _side_stream[key]
+
+
+def run_bwd_passes(args, device) -> None:
+    """Launch the independent dq / dkv backward passes on two streams so
+    their waves co-schedule across the chip (each alone under-fills it)."""
+    lib = _ffa_lib.lib()
+    main = torch.cuda.current_stream(device)
+    side = _get_side_stream(device)
+    ev = torch.cuda.Event()
+    ev.record(main)          # dpsum + inputs ready
+    side.wait_event(ev)
+    args.stream = ctypes.c_void_p(side.cuda_stream)
+    check(lib.magi_ffa_bwd_dq(args), "magi_ffa_bwd_dq")
+    args.stream = ctypes.c_void_p(main.cuda_stream)
+    check(lib.magi_ffa_bwd_dkv(args), "magi_ffa_bwd_dkv")
+    ev2 = torch.cuda.Event()
+    ev2.record(side)
+    main.wait_event(ev2)
 
 from .. import _ffa_lib
 from .._ffa_lib import (
@@ -167,7 +195,7 @@ def _flex_flash_attn_backward(
     )
     lib = _ffa_lib.lib()
     check(lib.magi_ffa_bwd_preprocess(args), "magi_ffa_bwd_preprocess")
-    check(lib.magi_ffa_bwd(args), "magi_ffa_bwd")
+    run_bwd_passes(args, q.device)
     return dq, dk, dv, None
 
 
