@@ -114,6 +114,7 @@ def _try_load() -> ctypes.CDLL | None:
         lib = ctypes.CDLL(str(_LIB_PATH))
         for name, argtypes in [
             ("magi_ffa_fwd", [ctypes.POINTER(MagiFfaFwdArgs)]),
+            ("magi_ffa_fwd_fp8", [ctypes.POINTER(MagiFfaFwdArgs)]),
             ("magi_ffa_bwd", [ctypes.POINTER(MagiFfaBwdArgs)]),
             ("magi_ffa_bwd_dq", [ctypes.POINTER(MagiFfaBwdArgs)]),
             ("magi_ffa_bwd_dkv", [ctypes.POINTER(MagiFfaBwdArgs)]),

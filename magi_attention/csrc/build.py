@@ -22,6 +22,7 @@ SOURCES = [
     CSRC / "ffa_fwd.hip",
     CSRC / "ffa_bwd.hip",
     CSRC / "range_ops.hip",
+    CSRC / "ffa_fwd_fp8.hip",
     CSRC / "ext_utils.hip",
 ]
 

@@ -96,7 +96,10 @@ def _flex_flash_attn_forward(
     **_unused,
 ) -> tuple[torch.Tensor, AttnForwardMeta]:
     assert sink is None, "sink support lands in a later round"
-    assert q.dtype == torch.bfloat16, "bf16 compute only (fp8 path later)"
+    is_fp8 = q.dtype == torch.float8_e4m3fn
+    assert q.dtype in (torch.bfloat16, torch.float8_e4m3fn), (
+        "bf16 or fp8-e4m3 inputs"
+    )
     q, k, v, q_ranges, k_ranges = [
         maybe_contiguous(x) for x in (q, k, v, q_ranges, k_ranges)
     ]
@@ -133,7 +136,12 @@ def _flex_flash_attn_forward(
         disable_atomic_reduction=int(disable_fwd_atomic_reduction),
         cu_margin=sm_margin, stream=current_stream_ptr(),
     )
-    check(_ffa_lib.lib().magi_ffa_fwd(args), "magi_ffa_fwd")
+    if is_fp8:
+        # MI355X extension: fp8 e4m3 MFMA path (reference has no fp8 compute)
+        assert not disable_fwd_atomic_reduction and out_is_fp32
+        check(_ffa_lib.lib().magi_ffa_fwd_fp8(args), "magi_ffa_fwd_fp8")
+    else:
+        check(_ffa_lib.lib().magi_ffa_fwd(args), "magi_ffa_fwd")
     return out, AttnForwardMeta(lse=lse, max_logits=None)
 
 
@@ -231,7 +239,9 @@ class FlexFlashAttnFunc(torch.autograd.Function):
             max_seqlen_q=max_seqlen_q,
         )
         lse = meta.lse
-        if out.dtype != q.dtype:
+        if q.dtype == torch.float8_e4m3fn:
+            out = out.to(torch.bfloat16)  # fp8 extension returns bf16 out
+        elif out.dtype != q.dtype:
             out = out.to(q.dtype)
         ctx.save_for_backward(q, k, v, out, lse, q_ranges, k_ranges, attn_type_map)
         ctx.softmax_scale = softmax_scale
@@ -246,6 +256,17 @@ class FlexFlashAttnFunc(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dout, *_):
         q, k, v, out, lse, q_ranges, k_ranges, attn_type_map = ctx.saved_tensors
+        in_dtype = q.dtype
+        if in_dtype == torch.float8_e4m3fn:
+            # fp8 backward runs on the bf16 kernels over upcast operands
+            # (standard mixed-precision practice; fp8 MFMA backward is a
+            # later-round item). Gradients are produced in bf16 precision and
+            # cast to the input dtype at return, as autograd requires.
+            q = q.to(torch.bfloat16)
+            k = k.to(torch.bfloat16)
+            v = v.to(torch.bfloat16)
+            out = out.to(torch.bfloat16) if out.dtype == torch.float8_e4m3fn else out
+            dout = dout.to(torch.bfloat16)
         dq, dk, dv, _ = _flex_flash_attn_backward(
             dout=dout, q=q, k=k, v=v, sink=None, sink_layout=ctx.sink_layout,
             out=out, lse=lse, dq=None, dk=None, dv=None, dsink=None,
@@ -256,9 +277,9 @@ class FlexFlashAttnFunc(torch.autograd.Function):
             deterministic=ctx.deterministic, sm_margin=ctx.sm_margin,
             max_seqlen_k=ctx.max_seqlen_k,
         )
-        dq = dq.to(q.dtype)
-        dk = dk.to(k.dtype)
-        dv = dv.to(v.dtype)
+        dq = dq.to(in_dtype)
+        dk = dk.to(in_dtype)
+        dv = dv.to(in_dtype)
         return (dq, dk, dv) + (None,) * 25
 
 
