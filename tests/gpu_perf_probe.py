@@ -42,3 +42,21 @@ rs = [[i*2048, (i+1)*2048] for i in range(8)]
 bench("varlen16k_h16_d128", 16384, 16384, 16, 16, 128, rs, rs, [1]*8, 8*area_causal(2048))
 # dense causal 64k h32 d128 (config 3 whole-job at cp=1)
 bench("dense64k_h32_d128", 65536, 65536, 32, 32, 128, [[0,65536]], [[0,65536]], [1], area_causal(65536), steps=3, warm=1)
+
+# fp8 extension (fwd only)
+def bench_fp8(name, n, hq, hk, d, steps=10):
+    q=(torch.randn(n,hq,d)*0.5).to(torch.float8_e4m3fn).cuda()
+    k=(torch.randn(n,hk,d)*0.5).to(torch.float8_e4m3fn).cuda()
+    v=(torch.randn(n,hk,d)*0.5).to(torch.float8_e4m3fn).cuda()
+    qr=torch.tensor([[0,n]],dtype=torch.int32,device="cuda"); kr=qr.clone()
+    tm=torch.tensor([1],dtype=torch.int32,device="cuda")
+    with torch.no_grad():
+        for _ in range(3): flex_flash_attn_func(q,k,v,qr,kr,tm,max_seqlen_q=n)
+        torch.cuda.synchronize(); t0=time.perf_counter()
+        for _ in range(steps): flex_flash_attn_func(q,k,v,qr,kr,tm,max_seqlen_q=n)
+        torch.cuda.synchronize()
+    dt=(time.perf_counter()-t0)/steps
+    fl=4*area_causal(n)*hq*d
+    print(f"{name} fp8-fwd: {dt*1e3:.2f} ms  {fl/dt/1e12:.1f} TFLOPS")
+
+bench_fp8("dense8k_h32_d128", 8192, 32, 32, 128)
