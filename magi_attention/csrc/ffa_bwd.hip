@@ -112,6 +112,7 @@ struct BwdParams {
   float softcap;
   long long total_q, total_k;
   int debug_ablate;  // perf ablation only: 1=skip dq stores, 2=skip dkv stores
+  int head_major;    // 1: blockIdx.x = head (XCD-affine); 0: work-major
 };
 
 // ---------------- preprocess: dpsum = rowsum(dO * O) ----------------
@@ -156,13 +157,14 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
   auto swz = [](int row, int byte_off) {
     return byte_off ^ ((row & SW32M) << 5);
   };
-  // blockIdx.x = HEAD: the dispatcher places block b on XCD b%8, so every
-  // block of one head shares one XCD/L2 — dq/dk/dv atomics stay XCD-local and
-  // K/V reads of a head hit a single L2 (T1 XCD-affinity via grid layout).
+  // Adaptive grid (see fwd kernel): head-major = XCD-affine atomics + K/V
+  // L2 locality per head; work-major when one head's operands overflow an
+  // XCD's L2.
   const int ri = blockIdx.z;
-  const int h = blockIdx.x;
+  const int h = p.head_major ? blockIdx.x : blockIdx.y;
+  const int wb = p.head_major ? blockIdx.y : blockIdx.x;
   const int ks = p.k_ranges[2 * ri], ke = p.k_ranges[2 * ri + 1];
-  const int nblk0 = ks + blockIdx.y * (BWD_BN * BWD_WAVES);
+  const int nblk0 = ks + wb * (BWD_BN * BWD_WAVES);
   if (nblk0 >= ke) return;
   const int qs = p.q_ranges[2 * ri], qe = p.q_ranges[2 * ri + 1];
   if (qe <= qs) return;
@@ -435,9 +437,10 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dq_kernel(BwdParams p) {
     return byte_off ^ ((row & SW32M) << 5);
   };
   const int ri = blockIdx.z;
-  const int h = blockIdx.x;  // XCD-affine: one head -> one XCD
+  const int h = p.head_major ? blockIdx.x : blockIdx.y;
+  const int wb = p.head_major ? blockIdx.y : blockIdx.x;
   const int qs = p.q_ranges[2 * ri], qe = p.q_ranges[2 * ri + 1];
-  const int mblk0 = qs + blockIdx.y * (BWD_BM * BWD_WAVES);
+  const int mblk0 = qs + wb * (BWD_BM * BWD_WAVES);
   if (mblk0 >= qe) return;
   const int ks = p.k_ranges[2 * ri], ke = p.k_ranges[2 * ri + 1];
   if (ke <= ks) return;
@@ -677,6 +680,9 @@ static int fill_bwd_params(const magi_ffa_bwd_args* a, BwdParams* p) {
   p->total_q = a->total_q;
   p->total_k = a->total_k;
   { const char* e = getenv("MAGI_BWD_ABLATE"); p->debug_ablate = e ? atoi(e) : 0; }
+  // head-major pays when one head's K+V (bf16) fits a 4 MB XCD L2
+  p->head_major = ((long long)a->total_k * a->d * 4 <= (4 << 20)) ? 1 : 0;
+  { const char* e = getenv("MAGI_BWD_HEAD_MAJOR"); if (e) p->head_major = atoi(e); }
   return 0;
 }
 
@@ -688,7 +694,9 @@ extern "C" int magi_ffa_bwd_dq(const magi_ffa_bwd_args* a) {
   const int qspan = BWD_BM * BWD_WAVES;
   const int qblocks = (int)((a->total_q + qspan - 1) / qspan);
   if (a->n_ranges > 65535) return -5;
-  dim3 grid_q(a->hq, qblocks, (unsigned)a->n_ranges), block(64 * BWD_WAVES);
+  dim3 grid_q = p.head_major ? dim3(a->hq, qblocks, (unsigned)a->n_ranges)
+                             : dim3(qblocks, a->hq, (unsigned)a->n_ranges);
+  dim3 block(64 * BWD_WAVES);
   hipStream_t s = (hipStream_t)a->stream;
   const bool sc = a->softcap > 0.f;
   if (a->d == 64) {
@@ -712,7 +720,9 @@ extern "C" int magi_ffa_bwd_dkv(const magi_ffa_bwd_args* a) {
   const int span = BWD_BN * BWD_WAVES;
   const int nblocks = (a->max_seqlen_k + span - 1) / span;
   if (a->n_ranges > 65535) return -5;
-  dim3 grid_kv(a->hq, nblocks, (unsigned)a->n_ranges), block(64 * BWD_WAVES);
+  dim3 grid_kv = p.head_major ? dim3(a->hq, nblocks, (unsigned)a->n_ranges)
+                              : dim3(nblocks, a->hq, (unsigned)a->n_ranges);
+  dim3 block(64 * BWD_WAVES);
   hipStream_t s = (hipStream_t)a->stream;
   const bool sc = a->softcap > 0.f;
   if (a->d == 64) {
