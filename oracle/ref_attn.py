@@ -99,7 +99,6 @@ def ref_attn(
     s = torch.matmul(qf, kf.transpose(-1, -2)) * scale  # [hq, tq, tk]
     if softcap > 0.0:
         s = softcap * torch.tanh(s / softcap)
-    neg_inf = torch.finfo(dt).min
     s = torch.where(mask.unsqueeze(0), s, torch.full_like(s, float("-inf")))
     lse = torch.logsumexp(s, dim=-1)  # [hq, tq]; -inf for empty rows
     p = torch.exp(s - lse.unsqueeze(-1))
@@ -108,7 +107,6 @@ def ref_attn(
         # emulate a kernel that quantises P before the PV matmul (bf16 MFMA)
         p = p.to(p_dtype).to(p.dtype)
     out = torch.matmul(p, vf)  # [hq, tq, d]
-    del neg_inf
     return (
         out.permute(1, 0, 2).to(q.dtype),
         lse.permute(1, 0).to(torch.float32),

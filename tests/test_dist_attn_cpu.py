@@ -49,6 +49,12 @@ CASES = {
         k_ranges=[[0, 512], [0, 512], [0, 512]],
         types=[0, 0, 0],
     ),
+    "unaligned_total": dict(  # exercises compute_pad_size > 0 (padded rows)
+        total=700,
+        q_ranges=[[0, 300], [300, 700]],
+        k_ranges=[[0, 300], [300, 700]],
+        types=[1, 1],
+    ),
 }
 
 
