@@ -227,10 +227,6 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
       vfA[dd] = *(const bf16x8*)(vp + dd * 16 + hi * 8);
     }
   }
-  union Bf {
-    unsigned short u[8];
-    bf16x8 v;
-  };
 
   // block-level q loop bounds (union over the 4 waves' tiles)
   int q_lo = qs, q_hi = qe;
