@@ -113,6 +113,9 @@ struct BwdParams {
   long long total_q, total_k;
   int debug_ablate;  // perf ablation only: 1=skip dq stores, 2=skip dkv stores
   int head_major;    // 1: blockIdx.x = head (XCD-affine); 0: work-major
+  int head_mult;     // deterministic GQA split: real head = off + idx*mult
+  int head_off;
+  int n_heads_launch;
 };
 
 // ---------------- preprocess: dpsum = rowsum(dO * O) ----------------
@@ -161,7 +164,8 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
   // L2 locality per head; work-major when one head's operands overflow an
   // XCD's L2.
   const int ri = blockIdx.z;
-  const int h = p.head_major ? blockIdx.x : blockIdx.y;
+  const int hidx = p.head_major ? blockIdx.x : blockIdx.y;
+  const int h = p.head_off + hidx * p.head_mult;
   const int wb = p.head_major ? blockIdx.y : blockIdx.x;
   const int ks = p.k_ranges[2 * ri], ke = p.k_ranges[2 * ri + 1];
   const int nblk0 = ks + wb * (BWD_BN * BWD_WAVES);
@@ -679,6 +683,16 @@ static int fill_bwd_params(const magi_ffa_bwd_args* a, BwdParams* p) {
   // head-major pays when one head's K+V (bf16) fits a 4 MB XCD L2
   p->head_major = ((long long)a->total_k * a->d * 4 <= (4 << 20)) ? 1 : 0;
   { const char* e = getenv("MAGI_BWD_HEAD_MAJOR"); if (e) p->head_major = atoi(e); }
+  // deterministic-mode GQA split: cu_margin's top bits carry (mult<<8)|off;
+  // 0 = all heads in one launch (default)
+  p->head_mult = 1;
+  p->head_off = 0;
+  p->n_heads_launch = a->hq;
+  if (a->cu_margin & 0xFFFF0000) {
+    p->head_mult = (a->cu_margin >> 24) & 0xFF;
+    p->head_off = (a->cu_margin >> 16) & 0xFF;
+    p->n_heads_launch = a->hq / p->head_mult;
+  }
   return 0;
 }
 
@@ -716,8 +730,9 @@ extern "C" int magi_ffa_bwd_dkv(const magi_ffa_bwd_args* a) {
   const int span = BWD_BN * BWD_WAVES;
   const int nblocks = (a->max_seqlen_k + span - 1) / span;
   if (a->n_ranges > 65535) return -5;
-  dim3 grid_kv = p.head_major ? dim3(a->hq, nblocks, (unsigned)a->n_ranges)
-                              : dim3(nblocks, a->hq, (unsigned)a->n_ranges);
+  dim3 grid_kv = p.head_major
+                     ? dim3(p.n_heads_launch, nblocks, (unsigned)a->n_ranges)
+                     : dim3(nblocks, p.n_heads_launch, (unsigned)a->n_ranges);
   dim3 block(64 * BWD_WAVES);
   hipStream_t s = (hipStream_t)a->stream;
   const bool sc = a->softcap > 0.f;

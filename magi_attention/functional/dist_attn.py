@@ -88,7 +88,8 @@ class DistAttnRuntime:
             out=out_acc, lse=lse_acc,
             q_ranges=qr, k_ranges=kr, attn_type_map=tm,
             softmax_scale=scale, softcap=0.0, out_type=torch.float32,
-            disable_fwd_atomic_reduction=False, deterministic=False,
+            disable_fwd_atomic_reduction=False,
+            deterministic=env.is_deterministic_mode_enable(),
             sm_margin=env.ffa_forward_sm_margin(),
             max_seqlen_q=arg.max_seqlen_q,
         )
@@ -196,9 +197,12 @@ class DistAttnRuntime:
             cu_margin=env.ffa_backward_sm_margin(),
             stream=current_stream_ptr(),
         )
-        from .flex_flash_attn import run_bwd_passes
+        from .flex_flash_attn import run_bwd_deterministic, run_bwd_passes
 
-        run_bwd_passes(args, q.device)
+        if env.is_deterministic_mode_enable():
+            run_bwd_deterministic(args, qr, kr, tm, hq, hk, q.device)
+        else:
+            run_bwd_passes(args, q.device)
 
 
 class DistAttnFunc(torch.autograd.Function):

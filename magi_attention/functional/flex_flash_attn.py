@@ -21,6 +21,36 @@ def _get_side_stream(device) -> torch.cuda.Stream:
     return _side_stream[key]
 
 
+def run_bwd_deterministic(args, q_ranges, k_ranges, attn_type_map,
+                          hq, hk, device) -> None:
+    """Deterministic backward: sequential launches over q-disjoint groups
+    (dq pass), k-disjoint groups x GQA head sub-launches (dkv pass) — every
+    accumulator element receives its additions in a fixed, stream-ordered
+    sequence."""
+    lib = _ffa_lib.lib()
+    gqa = hq // hk
+    q_groups = _color_ranges(q_ranges.cpu().tolist())
+    k_groups = _color_ranges(k_ranges.cpu().tolist())
+    base_margin = args.cu_margin & 0xFFFF
+
+    def launch(fn, groups, ranges_pairs, what, head_splits=1):
+        for g in groups:
+            idx = torch.tensor(g, dtype=torch.long, device=device)
+            args.q_ranges = ptr(_subset(ranges_pairs[0], idx))
+            args.k_ranges = ptr(_subset(ranges_pairs[1], idx))
+            args.attn_type_map = ptr(_subset(attn_type_map, idx))
+            args.n_ranges = len(g)
+            for j in range(head_splits):
+                if head_splits > 1:
+                    args.cu_margin = base_margin | (gqa << 24) | (j << 16)
+                check(fn(args), what)
+            args.cu_margin = base_margin
+
+    launch(lib.magi_ffa_bwd_dq, q_groups, (q_ranges, k_ranges), "bwd_dq[det]")
+    launch(lib.magi_ffa_bwd_dkv, k_groups, (q_ranges, k_ranges), "bwd_dkv[det]",
+           head_splits=gqa if gqa > 1 else 1)
+
+
 def run_bwd_passes(args, device) -> None:
     """Launch the independent dq / dkv backward passes on two streams so
     their waves co-schedule across the chip (each alone under-fills it)."""
@@ -73,6 +103,33 @@ def _max_seqlen_of(ranges: torch.Tensor) -> int:
     if ranges.numel() == 0:
         return 0
     return int((ranges[:, 1] - ranges[:, 0]).amax().item())
+
+
+def _color_ranges(ranges: list) -> list:
+    """Greedy interval partitioning: split slice indices into groups whose
+    ranges are pairwise DISJOINT. Deterministic mode runs one kernel launch
+    per group, sequentially on the stream, so reduction order is fixed by
+    construction — placement-independent, unlike device-side lock ordering
+    (the reference's deterministic.h approach assumes dispatch order, which
+    CDNA4 does not guarantee)."""
+    order = sorted(range(len(ranges)), key=lambda i: (ranges[i][0], ranges[i][1]))
+    groups: list = []   # list of (last_end, [indices])
+    for i in order:
+        s_, e_ = ranges[i]
+        placed = False
+        for g in groups:
+            if g[0] <= s_:
+                g[1].append(i)
+                g[0] = e_
+                placed = True
+                break
+        if not placed:
+            groups.append([e_, [i]])
+    return [sorted(g[1]) for g in groups]
+
+
+def _subset(t: Optional[torch.Tensor], idx: torch.Tensor) -> Optional[torch.Tensor]:
+    return None if t is None else t.index_select(0, idx)
 
 
 def _flex_flash_attn_forward(
@@ -136,12 +193,23 @@ def _flex_flash_attn_forward(
         disable_atomic_reduction=int(disable_fwd_atomic_reduction),
         cu_margin=sm_margin, stream=current_stream_ptr(),
     )
+    fwd_fn = (_ffa_lib.lib().magi_ffa_fwd_fp8 if is_fp8
+              else _ffa_lib.lib().magi_ffa_fwd)
     if is_fp8:
-        # MI355X extension: fp8 e4m3 MFMA path (reference has no fp8 compute)
         assert not disable_fwd_atomic_reduction and out_is_fp32
-        check(_ffa_lib.lib().magi_ffa_fwd_fp8(args), "magi_ffa_fwd_fp8")
+    if deterministic and q_ranges.shape[0] > 1:
+        # fixed merge order: one launch per q-disjoint slice group
+        groups = _color_ranges(q_ranges.cpu().tolist())
+        for g in groups:
+            idx = torch.tensor(g, dtype=torch.long, device=q.device)
+            ga = args
+            ga.q_ranges = ptr(_subset(q_ranges, idx))
+            ga.k_ranges = ptr(_subset(k_ranges, idx))
+            ga.attn_type_map = ptr(_subset(attn_type_map, idx))
+            ga.n_ranges = len(g)
+            check(fwd_fn(ga), "magi_ffa_fwd[det]")
     else:
-        check(_ffa_lib.lib().magi_ffa_fwd(args), "magi_ffa_fwd")
+        check(fwd_fn(args), "magi_ffa_fwd")
     return out, AttnForwardMeta(lse=lse, max_logits=None)
 
 
@@ -203,7 +271,11 @@ def _flex_flash_attn_backward(
     )
     lib = _ffa_lib.lib()
     check(lib.magi_ffa_bwd_preprocess(args), "magi_ffa_bwd_preprocess")
-    run_bwd_passes(args, q.device)
+    if deterministic:
+        run_bwd_deterministic(args, q_ranges, k_ranges, attn_type_map,
+                              hq, hk, q.device)
+    else:
+        run_bwd_passes(args, q.device)
     return dq, dk, dv, None
 
 
