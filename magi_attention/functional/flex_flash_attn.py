@@ -33,12 +33,18 @@ def run_bwd_deterministic(args, q_ranges, k_ranges, attn_type_map,
     k_groups = _color_ranges(k_ranges.cpu().tolist())
     base_margin = args.cu_margin & 0xFFFF
 
+    keep = []
+
     def launch(fn, groups, ranges_pairs, what, head_splits=1):
         for g in groups:
             idx = torch.tensor(g, dtype=torch.long, device=device)
-            args.q_ranges = ptr(_subset(ranges_pairs[0], idx))
-            args.k_ranges = ptr(_subset(ranges_pairs[1], idx))
-            args.attn_type_map = ptr(_subset(attn_type_map, idx))
+            sq = _subset(ranges_pairs[0], idx)
+            sk = _subset(ranges_pairs[1], idx)
+            st = _subset(attn_type_map, idx)
+            keep.append((sq, sk, st))  # outlive the async launches
+            args.q_ranges = ptr(sq)
+            args.k_ranges = ptr(sk)
+            args.attn_type_map = ptr(st)
             args.n_ranges = len(g)
             for j in range(head_splits):
                 if head_splits > 1:
@@ -198,16 +204,25 @@ def _flex_flash_attn_forward(
     if is_fp8:
         assert not disable_fwd_atomic_reduction and out_is_fp32
     if deterministic and q_ranges.shape[0] > 1:
-        # fixed merge order: one launch per q-disjoint slice group
+        # fixed merge order: one launch per q-disjoint slice group.
+        # NOTE: subset tensors must outlive the async kernel launches — hold
+        # references until the end of the loop (the ctypes kernel is invisible
+        # to torch's stream-aware allocator).
         groups = _color_ranges(q_ranges.cpu().tolist())
+        keep = []
         for g in groups:
             idx = torch.tensor(g, dtype=torch.long, device=q.device)
-            ga = args
-            ga.q_ranges = ptr(_subset(q_ranges, idx))
-            ga.k_ranges = ptr(_subset(k_ranges, idx))
-            ga.attn_type_map = ptr(_subset(attn_type_map, idx))
-            ga.n_ranges = len(g)
-            check(fwd_fn(ga), "magi_ffa_fwd[det]")
+            sq = _subset(q_ranges, idx)
+            sk = _subset(k_ranges, idx)
+            st = _subset(attn_type_map, idx)
+            keep.append((sq, sk, st))
+            args.q_ranges = ptr(sq)
+            args.k_ranges = ptr(sk)
+            args.attn_type_map = ptr(st)
+            args.n_ranges = len(g)
+            check(fwd_fn(args), "magi_ffa_fwd[det]")
+        torch.cuda.current_stream().synchronize() if False else None
+        del keep
     else:
         check(fwd_fn(args), "magi_ffa_fwd")
     return out, AttnForwardMeta(lse=lse, max_logits=None)
