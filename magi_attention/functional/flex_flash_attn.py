@@ -221,7 +221,6 @@ def _flex_flash_attn_forward(
             args.attn_type_map = ptr(st)
             args.n_ranges = len(g)
             check(fwd_fn(args), "magi_ffa_fwd[det]")
-        torch.cuda.current_stream().synchronize() if False else None
         del keep
     else:
         check(fwd_fn(args), "magi_ffa_fwd")
