@@ -156,6 +156,36 @@ def test_fwd_softcap():
 
 
 @requires_gpu
+def test_bwd_softcap():
+    """Softcap backward (tanh-derivative dscale path in both bwd passes)."""
+    from magi_attention.functional import flex_flash_attn_func
+    from oracle import ref_attn_with_grads
+
+    tq = tk = 192
+    q, k, v, dout, qr, kr, tm = make_flex_case(
+        tq, tk, 4, 2, 128, [[0, tq]], [[0, tk]], [1], seed=9
+    )
+    q.requires_grad_(True)
+    k.requires_grad_(True)
+    v.requires_grad_(True)
+    out, _ = flex_flash_attn_func(q, k, v, qr, kr, tm, softcap=15.0)
+    out.backward(dout)
+    torch.cuda.synchronize()
+    mask = make_attn_mask(tq, tk, [[0, tq]], [[0, tk]], [1])
+    qc, kc, vc, doc = [t.detach().cpu() for t in (q, k, v, dout)]
+    hi = ref_attn_with_grads(qc, kc, vc, mask, doc, softcap=15.0)
+    lo = ref_attn_with_grads(qc, kc, vc, mask, doc, softcap=15.0,
+                             high_precision=False, p_dtype=torch.bfloat16)
+    for g, ghi, glo, name in [
+        (q.grad, hi[2], lo[2], "dq"),
+        (k.grad, hi[3], lo[3], "dk"),
+        (v.grad, hi[4], lo[4], "dv"),
+    ]:
+        assert_close_to_ref(g.cpu().float(), ghi.float(), glo.float(),
+                            f"softcap:{name}")
+
+
+@requires_gpu
 def test_cross_launch_accumulation():
     """The CP runtime accumulates multiple kernel launches into one
     (out_acc, lse_acc) pair: splitting K across two calls must equal one call
