@@ -8,6 +8,9 @@ Run (cp=N, one node): torchrun --nproc-per-node N --master-addr 127.0.0.1 \
                           examples/torch_native_demo.py
 """
 import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 import torch.distributed as dist
