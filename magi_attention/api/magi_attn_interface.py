@@ -10,6 +10,7 @@ from typing import List, Optional, Tuple, Union
 import torch
 import torch.distributed as dist
 
+from .. import env
 from ..common.enum import AttnMaskType
 from ..common.range import AttnRange
 from ..common.ranges import AttnRanges
@@ -65,13 +66,28 @@ class DistAttnRuntimeDictManager:
 dist_attn_runtime_dict_mgr = DistAttnRuntimeDictManager()
 
 
-def _resolve_group(cp_group_or_mesh) -> dist.ProcessGroup:
+def _resolve_group(cp_group_or_mesh):
+    """Returns (flat cp_group, mesh_groups). mesh_groups is
+    (intra_group, inter_group, ws_intra, ws_inter) iff a 2D DeviceMesh was
+    passed AND MAGI_ATTENTION_HIERARCHICAL_COMM is on (reference
+    api:632 cp_mesh._flatten(), comm_meta.py:227-228 intra=dim1, inter=dim0).
+    """
     if isinstance(cp_group_or_mesh, dist.ProcessGroup):
-        return cp_group_or_mesh
-    # 1D DeviceMesh
+        return cp_group_or_mesh, None
     try:
-        return cp_group_or_mesh.get_group()
-    except Exception as e:  # pragma: no cover
+        mesh = cp_group_or_mesh
+        if getattr(mesh, "ndim", 1) == 2:
+            flat = mesh._flatten().get_group()
+            if env.is_hierarchical_comm_enable():
+                inter = mesh.get_group(0)
+                intra = mesh.get_group(1)
+                return flat, (
+                    intra, inter,
+                    dist.get_world_size(intra), dist.get_world_size(inter),
+                )
+            return flat, None
+        return mesh.get_group(), None
+    except AttributeError as e:  # pragma: no cover
         raise ValueError(f"unsupported cp_group_or_mesh: {e}")
 
 
@@ -95,7 +111,7 @@ def magi_attn_flex_key(
     """The most flexible key-creation interface (reference :440)."""
     assert is_same_source, "cross-attn (is_same_source=False) lands in a later round"
     assert total_seqlen_q == total_seqlen_k, "self-attn requires equal seqlens"
-    group = _resolve_group(cp_group_or_mesh)
+    group, mesh_groups = _resolve_group(cp_group_or_mesh)
     cp_size = dist.get_world_size(group)
     if chunk_size is not None:
         warnings.warn(
@@ -127,7 +143,7 @@ def magi_attn_flex_key(
     )
     if key not in dist_attn_runtime_dict_mgr:
         dist_attn_runtime_dict_mgr[key] = init_dist_attn_runtime_mgr(
-            key, group, dist_attn_config
+            key, group, dist_attn_config, mesh_groups=mesh_groups
         )
     return key
 
@@ -245,7 +261,8 @@ def make_flex_key_for_new_mask_after_dispatch(
     key = replace(key, dispatch_from=hash(key_for_dispatch))
     if key not in dist_attn_runtime_dict_mgr:
         dist_attn_runtime_dict_mgr[key] = DistAttnRuntimeMgr(
-            key, base_mgr.cp_group, cfg, reuse_dispatch_from=base_mgr
+            key, base_mgr.cp_group, cfg, reuse_dispatch_from=base_mgr,
+            mesh_groups=base_mgr.mesh_groups,
         )
     return key
 

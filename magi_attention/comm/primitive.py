@@ -13,7 +13,13 @@ from typing import Callable, List, Optional
 import torch
 import torch.distributed as dist
 
-from ..meta.containers import GroupCastArg, GroupReduceArg, RowChunkMap
+from ..meta.containers import (
+    GroupCastArg,
+    GroupReduceArg,
+    HierGroupCastArg,
+    HierGroupReduceArg,
+    RowChunkMap,
+)
 
 
 class WorkWithPostProcessFn:
@@ -41,11 +47,19 @@ def _rows_copy(src: torch.Tensor, dst: torch.Tensor, cmap: RowChunkMap,
     if not cmap.in_ranges:
         return dst
     if src.is_cuda:
+        from .. import env
         from ..common.range_op import range_gather, range_reduce
 
         in_r, out_s = cmap.to_device(src.device)
         if reduce == "copy":
             range_gather(src, in_r, out_s, cmap.total_rows, output=dst)
+        elif env.is_deterministic_mode_enable():
+            # ranges may overlap in the output; sequence them so the fp32 sum
+            # order is fixed by table order, not block scheduling
+            for i in range(len(cmap.in_ranges)):
+                a, b = cmap.in_ranges[i]
+                range_reduce(src, dst, in_r[i:i + 1], out_s[i:i + 1],
+                             op="sum", total_rows=b - a)
         else:
             range_reduce(src, dst, in_r, out_s, op="sum",
                          total_rows=cmap.total_rows)
@@ -133,6 +147,99 @@ def group_reduce(
         return dst
 
     return WorkWithPostProcessFn(work, post)
+
+
+def _a2av(recv: torch.Tensor, send: torch.Tensor, out_splits: List[int],
+          in_splits: List[int], group: dist.ProcessGroup, async_op=True):
+    """all_to_all_single over row dim; handles 0-row tensors and ws==1."""
+    if dist.get_world_size(group) == 1:
+        recv.copy_(send)
+        return None
+    flat = 1
+    for t in send.shape[1:]:
+        flat *= t
+    work = dist.all_to_all_single(
+        recv.view(recv.shape[0], flat), send.view(send.shape[0], flat),
+        output_split_sizes=out_splits, input_split_sizes=in_splits,
+        group=group, async_op=async_op,
+    )
+    return work if async_op else None
+
+
+def hier_group_cast(
+    kv_local: torch.Tensor,  # [2L, h, d]
+    arg: HierGroupCastArg,
+    intra_group: dist.ProcessGroup,
+    inter_group: dist.ProcessGroup,
+) -> WorkWithPostProcessFn:
+    """Hierarchical K/V multicast (2D mesh): pre-intra direct + one dedup copy
+    per remote node via the same-local-rank proxy + post-intra forward. Same
+    result tensor as group_cast on the flattened group (reference
+    _group_collective_hier.py:931 hier_group_cast_impl_with_a2av)."""
+    h_tail = kv_local.shape[1:]
+    send1 = kv_local.new_empty((arg.pre.send_pack.total_rows, *h_tail))
+    _rows_copy(kv_local, send1, arg.pre.send_pack)
+    recv1 = kv_local.new_empty((sum(arg.pre.output_split_sizes), *h_tail))
+    w1 = _a2av(recv1, send1, arg.pre.output_split_sizes,
+               arg.pre.input_split_sizes, intra_group)
+
+    send2 = kv_local.new_empty((arg.inter_send_pack.total_rows, *h_tail))
+    _rows_copy(kv_local, send2, arg.inter_send_pack)
+    recv2 = kv_local.new_empty((arg.inter_total_recv, *h_tail))
+    w2 = _a2av(recv2, send2, arg.inter_out_splits, arg.inter_in_splits,
+               inter_group)
+
+    def post() -> torch.Tensor:
+        if w2 is not None:
+            w2.wait()
+        send3 = kv_local.new_empty((arg.post_send_pack.total_rows, *h_tail))
+        _rows_copy(recv2, send3, arg.post_send_pack)
+        recv3 = kv_local.new_empty((sum(arg.post_out_splits), *h_tail))
+        w3 = _a2av(recv3, send3, arg.post_out_splits, arg.post_in_splits,
+                   intra_group)
+        stage = kv_local.new_zeros((2 * arg.stage_tokens, *h_tail))
+        if w1 is not None:
+            w1.wait()
+        _rows_copy(recv1, stage, arg.pre.recv_unpack)
+        if w3 is not None:
+            w3.wait()
+        _rows_copy(recv3, stage, arg.post_recv_unpack)
+        return stage
+
+    return WorkWithPostProcessFn(None, post)
+
+
+def hier_group_reduce(
+    partial: torch.Tensor,   # [2S, h, d] partial dK/dV of the stage buffer
+    dst: torch.Tensor,       # [2L, h, d] owner-local accumulator (fp32)
+    arg: HierGroupReduceArg,
+    intra_group: dist.ProcessGroup,
+    inter_group: dist.ProcessGroup,
+) -> WorkWithPostProcessFn:
+    """Mirror of hier_group_cast for partial dK/dV: the in-node proxy sums its
+    node's contributions before ONE inter-node copy per (node, owner) pair."""
+    h_tail = partial.shape[1:]
+    send1 = partial.new_empty((arg.pre_send_pack.total_rows, *h_tail))
+    _rows_copy(partial, send1, arg.pre_send_pack)
+    recv1 = partial.new_empty((arg.pre_total_recv, *h_tail))
+    w1 = _a2av(recv1, send1, arg.pre_out_splits, arg.pre_in_splits,
+               intra_group)
+
+    def post() -> torch.Tensor:
+        if w1 is not None:
+            w1.wait()
+        _rows_copy(recv1, dst, arg.pre_recv_direct, reduce="sum")
+        proxy = partial.new_zeros((arg.proxy_rows, *h_tail))
+        _rows_copy(recv1, proxy, arg.pre_recv_proxy, reduce="sum")
+        recv2 = partial.new_empty((arg.inter_total_recv, *h_tail))
+        w2 = _a2av(recv2, proxy, arg.inter_out_splits, arg.inter_in_splits,
+                   inter_group)
+        if w2 is not None:
+            w2.wait()
+        _rows_copy(recv2, dst, arg.inter_recv_reduce, reduce="sum")
+        return dst
+
+    return WorkWithPostProcessFn(None, post)
 
 
 def all_gather_v(

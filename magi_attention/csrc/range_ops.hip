@@ -52,14 +52,16 @@ __global__ __launch_bounds__(256) void range_rows_kernel(RangeParams p) {
       if (OP == 0) {
         dst[i] = src[i];
       } else {
-        // f32 sum-reduce: VecT = float4
+        // f32 sum-reduce: VecT = float4. Ranges may target OVERLAPPING output
+        // rows (several dst ranks returning partials for the same hosted
+        // rows), and ranges run in concurrent blocks -> must be atomic
+        // (-munsafe-fp-atomics => hardware global_atomic_add_f32).
         float4 a = *(const float4*)&src[i];
-        float4 b = *(const float4*)&dst[i];
-        b.x += a.x;
-        b.y += a.y;
-        b.z += a.z;
-        b.w += a.w;
-        *(float4*)&dst[i] = b;
+        float* d4 = (float*)&dst[i];
+        atomicAdd(d4 + 0, a.x);
+        atomicAdd(d4 + 1, a.y);
+        atomicAdd(d4 + 2, a.z);
+        atomicAdd(d4 + 3, a.w);
       }
     }
   }

@@ -64,10 +64,12 @@ class DistAttnRuntimeMgr:
         cp_group: dist.ProcessGroup,
         dist_attn_config: DistAttnConfig,
         reuse_dispatch_from: "DistAttnRuntimeMgr | None" = None,
+        mesh_groups: "Tuple[Any, Any, int, int] | None" = None,
     ):
         self.key = key
         self.cp_group = cp_group
         self.config = dist_attn_config
+        self.mesh_groups = mesh_groups
         cp_size = dist.get_world_size(cp_group)
         cp_rank = dist.get_rank(cp_group)
 
@@ -96,11 +98,21 @@ class DistAttnRuntimeMgr:
         self.solver, calc_meta, comm_meta = make_attn_meta_from_dispatch_meta(
             slices, self.dispatch_meta, dist_attn_config
         )
+        intra_group = inter_group = None
+        if mesh_groups is not None and env.is_hierarchical_comm_enable():
+            intra_group, inter_group, wi, wn = mesh_groups
+            casts_h, reduces_h = self.solver.make_hier_comm_meta(
+                cp_rank, wi, wn
+            )
+            comm_meta.stages_cast_hier = casts_h
+            comm_meta.stages_reduce_hier = reduces_h
         self.runtime = DistAttnRuntime(
             calc_meta=calc_meta,
             comm_meta=comm_meta,
             cp_group=cp_group,
             total_local_q=total_padded // cp_size,
+            intra_group=intra_group,
+            inter_group=inter_group,
         )
 
     # ---- ops ----
@@ -214,5 +226,7 @@ def init_dist_attn_runtime_mgr(
     key: DistAttnRuntimeKey,
     cp_group: dist.ProcessGroup,
     dist_attn_config: DistAttnConfig,
+    mesh_groups=None,
 ) -> DistAttnRuntimeMgr:
-    return DistAttnRuntimeMgr(key, cp_group, dist_attn_config)
+    return DistAttnRuntimeMgr(key, cp_group, dist_attn_config,
+                              mesh_groups=mesh_groups)

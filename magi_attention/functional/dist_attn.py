@@ -17,7 +17,13 @@ import torch
 import torch.distributed as dist
 
 from .. import env
-from ..comm.primitive import WorkWithPostProcessFn, group_cast, group_reduce
+from ..comm.primitive import (
+    WorkWithPostProcessFn,
+    group_cast,
+    group_reduce,
+    hier_group_cast,
+    hier_group_reduce,
+)
 from ..meta.containers import AttnArg, CalcMeta, CommMeta
 
 # test-only hook: lets CPU (gloo) tests run the runtime with an oracle
@@ -38,10 +44,36 @@ class DistAttnRuntime:
     cp_group: dist.ProcessGroup
     total_local_q: int
     softmax_scale: Optional[float] = None
+    # hierarchical 2D-mesh comm (set iff stages_cast_hier planned)
+    intra_group: Optional[dist.ProcessGroup] = None
+    inter_group: Optional[dist.ProcessGroup] = None
 
     @property
     def overlap_degree(self) -> int:
         return self.comm_meta.overlap_degree
+
+    @property
+    def use_hier(self) -> bool:
+        return (self.comm_meta.stages_cast_hier is not None
+                and self.intra_group is not None)
+
+    def _cast(self, kv_local: torch.Tensor, s: int) -> WorkWithPostProcessFn:
+        if self.use_hier:
+            return hier_group_cast(
+                kv_local, self.comm_meta.stages_cast_hier[s],
+                self.intra_group, self.inter_group,
+            )
+        return group_cast(kv_local, self.comm_meta.stages_cast[s],
+                          self.cp_group)
+
+    def _reduce(self, partial, dst, s: int) -> WorkWithPostProcessFn:
+        if self.use_hier:
+            return hier_group_reduce(
+                partial, dst, self.comm_meta.stages_reduce_hier[s],
+                self.intra_group, self.inter_group,
+            )
+        return group_reduce(partial, dst, self.comm_meta.stages_reduce[s],
+                            self.cp_group)
 
     # ---------------- forward ----------------
     def attn_fwd(
@@ -53,8 +85,7 @@ class DistAttnRuntime:
 
         # pre-issue ALL remote-stage group-casts (reference dist_attn.py:419-436)
         works: List[WorkWithPostProcessFn] = [
-            group_cast(kv_local, self.comm_meta.stages_cast[s], self.cp_group)
-            for s in range(self.overlap_degree)
+            self._cast(kv_local, s) for s in range(self.overlap_degree)
         ]
 
         out_acc = torch.zeros(tq, hq, d, dtype=torch.float32, device=q.device)
@@ -110,10 +141,7 @@ class DistAttnRuntime:
         kv_local = torch.cat([k, v], dim=0)
 
         # re-fetch remote KV (reference backward:3455 -> _fetch_remote_kv:1463)
-        works = [
-            group_cast(kv_local, self.comm_meta.stages_cast[s], self.cp_group)
-            for s in range(self.overlap_degree)
-        ]
+        works = [self._cast(kv_local, s) for s in range(self.overlap_degree)]
 
         dq_acc = torch.zeros(tq, hq, d, dtype=torch.float32, device=q.device)
         dkv_acc = torch.zeros(2 * L, *k.shape[1:], dtype=torch.float32,
@@ -138,12 +166,7 @@ class DistAttnRuntime:
                     dout, q, stage_kv[:S], stage_kv[S:], out, lse, dpsum, arg,
                     dq_acc, dkv_stage[:S], dkv_stage[S:], scale,
                 )
-            rworks.append(
-                group_reduce(
-                    dkv_stage, dkv_acc, self.comm_meta.stages_reduce[s],
-                    self.cp_group,
-                )
-            )
+            rworks.append(self._reduce(dkv_stage, dkv_acc, s))
         for w in rworks:
             w.wait_post_process()
         dq = dq_acc.to(q.dtype)

@@ -16,6 +16,8 @@ from .containers import (
     CommMeta,
     GroupCastArg,
     GroupReduceArg,
+    HierGroupCastArg,
+    HierGroupReduceArg,
     RowChunkMap,
 )
 from .geometry import MaskSlice, k_window, q_window
@@ -155,6 +157,20 @@ class DistAttnSolver:
             stage_args.append(to_local_args(stage_slices, st))
         return CalcMeta(host_arg=host_arg, stage_args=stage_args)
 
+    def _overlap_table(self, s: int) -> List[List[AttnRanges]]:
+        """ov[o][r] = the (merged) global k ranges owner o casts to rank r in
+        stage s. Every rank derives the full table deterministically."""
+        cp = self.cp_size
+        return [
+            [
+                self.plans[r].stages_need[s].find_overlap_ranges(
+                    self.host_ranges_all[o]
+                )
+                for r in range(cp)
+            ]
+            for o in range(cp)
+        ]
+
     # ---------------- comm meta ----------------
     def make_comm_meta(self, rank: int) -> CommMeta:
         cp = self.cp_size
@@ -164,16 +180,7 @@ class DistAttnSolver:
         L = my_hr.total_seqlen  # my local kv rows
 
         for s in range(self.overlap_degree):
-            # overlap table: ov[o][r] = what owner o sends to dst r this stage
-            ov = [
-                [
-                    self.plans[r].stages_need[s].find_overlap_ranges(
-                        self.host_ranges_all[o]
-                    )
-                    for r in range(cp)
-                ]
-                for o in range(cp)
-            ]
+            ov = self._overlap_table(s)
             # ---- cast: me as SENDER (owner) ----
             send_in: List[Tuple[int, int]] = []
             send_out: List[int] = []
@@ -278,3 +285,268 @@ class DistAttnSolver:
                 )
             )
         return CommMeta(stages_cast=casts, stages_reduce=reduces)
+
+    # ---------------- hierarchical comm meta ----------------
+    def make_hier_comm_meta(
+        self, rank: int, ws_intra: int, ws_inter: int
+    ) -> Tuple[List[HierGroupCastArg], List[HierGroupReduceArg]]:
+        """Hierarchical (2D-mesh) realisation of the same stages. Rank layout
+        is row-major over (inter, intra): node(r)=r//ws_intra, loc(r)=r%ws_intra
+        (reference comm_meta.py:227 intra=mesh dim 1, inter=mesh dim 0).
+
+        Cast: a row needed by several ranks of a remote node crosses the
+        inter-node wire ONCE, to the proxy with the owner's local rank, which
+        forwards it intra-node. Reduce mirrors this: partials for a remote
+        owner are first SUMMED on the in-node proxy, then one node-sum crosses
+        per contributing node. Unlike the reference
+        (_group_collective_hier.py:319 all_gather_object at call time), every
+        table is derived here at plan time from the global overlap table.
+        """
+        assert ws_intra * ws_inter == self.cp_size
+        wi, wn = ws_intra, ws_inter
+        n0, l0 = rank // wi, rank % wi
+
+        def g(n: int, l: int) -> int:
+            return n * wi + l
+
+        def union(ranges_list: List[AttnRanges]) -> AttnRanges:
+            u = AttnRanges()
+            for rr in ranges_list:
+                for p in rr:
+                    u.append(p.clone())
+            return u.merge()
+
+        my_hr = self.host_ranges_all[rank]
+        L = my_hr.total_seqlen
+        casts: List[HierGroupCastArg] = []
+        reduces: List[HierGroupReduceArg] = []
+
+        for s in range(self.overlap_degree):
+            ov = self._overlap_table(s)
+            my_need = self.plans[rank].stages_need[s]
+            S = my_need.total_seqlen
+
+            def loc_host(p):
+                lp = my_hr.make_range_local(p, is_self_merged=True)
+                return lp.start, lp.end
+
+            def loc_need(p):
+                lp = my_need.make_range_local(p)
+                return lp.start, lp.end
+
+            # ---------- cast phase 1: pre-intra (same-node direct) ----------
+            in_r, out_s, in_sp = [], [], []
+            cur = 0
+            for l in range(wi):
+                pieces = ov[rank][g(n0, l)]
+                tok = pieces.total_seqlen
+                vcur = cur + tok
+                for p in pieces:
+                    a, b = loc_host(p)
+                    in_r.append((a, b)); out_s.append(cur); cur += b - a
+                for p in pieces:
+                    a, b = loc_host(p)
+                    in_r.append((L + a, L + b)); out_s.append(vcur); vcur += b - a
+                cur = vcur
+                in_sp.append(2 * tok)
+            pre_send = RowChunkMap(in_r, out_s, cur)
+
+            in_r, out_s, out_sp = [], [], []
+            rcur = 0
+            for l in range(wi):
+                pieces = ov[g(n0, l)][rank]
+                tok = pieces.total_seqlen
+                for p in pieces:
+                    a, _ = loc_need(p)
+                    in_r.append((rcur, rcur + p.seqlen)); out_s.append(a)
+                    rcur += p.seqlen
+                for p in pieces:
+                    a, _ = loc_need(p)
+                    in_r.append((rcur, rcur + p.seqlen)); out_s.append(S + a)
+                    rcur += p.seqlen
+                out_sp.append(2 * tok)
+            pre = GroupCastArg(
+                send_pack=pre_send, input_split_sizes=in_sp,
+                recv_unpack=RowChunkMap(in_r, out_s, rcur),
+                output_split_sizes=out_sp, stage_tokens=S,
+            )
+
+            # ---------- cast phase 2: inter (one dedup copy per node) -------
+            # send: union over the destination node's ranks -> proxy (n, l0)
+            in_r, out_s, in_sp = [], [], []
+            cur = 0
+            for n in range(wn):
+                u = (union([ov[rank][g(n, l)] for l in range(wi)])
+                     if n != n0 else AttnRanges())
+                tok = u.total_seqlen
+                vcur = cur + tok
+                for p in u:
+                    a, b = loc_host(p)
+                    in_r.append((a, b)); out_s.append(cur); cur += b - a
+                for p in u:
+                    a, b = loc_host(p)
+                    in_r.append((L + a, L + b)); out_s.append(vcur); vcur += b - a
+                cur = vcur
+                in_sp.append(2 * tok)
+            inter_send = RowChunkMap(in_r, out_s, cur)
+            inter_in_splits = in_sp
+
+            # recv (me as proxy): from owner (n, l0): union over my node's dsts
+            proxy_u = [
+                (union([ov[g(n, l0)][g(n0, l)] for l in range(wi)])
+                 if n != n0 else AttnRanges())
+                for n in range(wn)
+            ]
+            inter_out_splits = [2 * u.total_seqlen for u in proxy_u]
+            inter_total_recv = sum(inter_out_splits)
+            off = [0] * wn
+            acc = 0
+            for n in range(wn):
+                off[n] = acc
+                acc += inter_out_splits[n]
+
+            # ---------- cast phase 3: post-intra (proxy -> final dst) -------
+            in_r, out_s, in_sp = [], [], []
+            cur = 0
+            for l in range(wi):
+                tok = sum(
+                    ov[g(n, l0)][g(n0, l)].total_seqlen
+                    for n in range(wn) if n != n0
+                )
+                vcur = cur + tok
+                for pass_v in (False, True):
+                    c = vcur if pass_v else cur
+                    for n in range(wn):
+                        if n == n0:
+                            continue
+                        u = proxy_u[n]
+                        U = u.total_seqlen
+                        base = off[n] + (U if pass_v else 0)
+                        for p in ov[g(n, l0)][g(n0, l)]:
+                            lp = u.make_range_local(p)
+                            in_r.append((base + lp.start, base + lp.end))
+                            out_s.append(c); c += p.seqlen
+                    if pass_v:
+                        vcur = c
+                    else:
+                        cur = c
+                cur = vcur
+                in_sp.append(2 * tok)
+            post_send = RowChunkMap(in_r, out_s, cur)
+            post_in_splits = in_sp
+
+            in_r, out_s, out_sp = [], [], []
+            rcur = 0
+            for l in range(wi):
+                tok = sum(
+                    ov[g(n, l)][rank].total_seqlen
+                    for n in range(wn) if n != n0
+                )
+                for pass_v in (False, True):
+                    for n in range(wn):
+                        if n == n0:
+                            continue
+                        for p in ov[g(n, l)][rank]:
+                            a, _ = loc_need(p)
+                            in_r.append((rcur, rcur + p.seqlen))
+                            out_s.append(S + a if pass_v else a)
+                            rcur += p.seqlen
+                out_sp.append(2 * tok)
+            casts.append(HierGroupCastArg(
+                pre=pre,
+                inter_send_pack=inter_send,
+                inter_in_splits=inter_in_splits,
+                inter_out_splits=inter_out_splits,
+                inter_total_recv=inter_total_recv,
+                post_send_pack=post_send,
+                post_in_splits=post_in_splits,
+                post_recv_unpack=RowChunkMap(in_r, out_s, rcur),
+                post_out_splits=out_sp,
+                stage_tokens=S,
+            ))
+
+            # ---------- reduce phase 1: pre-intra -------------------------
+            # send to local rank l: partials for owner (n, l) for EVERY node n
+            # (n == n0: l is the owner itself; else: l is the owner's proxy)
+            in_r, out_s, in_sp = [], [], []
+            cur = 0
+            for l in range(wi):
+                tok = sum(ov[g(n, l)][rank].total_seqlen for n in range(wn))
+                vcur = cur + tok
+                for pass_v in (False, True):
+                    c = vcur if pass_v else cur
+                    for n in range(wn):
+                        for p in ov[g(n, l)][rank]:
+                            a, b = loc_need(p)
+                            if pass_v:
+                                a, b = S + a, S + b
+                            in_r.append((a, b)); out_s.append(c); c += b - a
+                    if pass_v:
+                        vcur = c
+                    else:
+                        cur = c
+                cur = vcur
+                in_sp.append(2 * tok)
+            r_pre_send = RowChunkMap(in_r, out_s, cur)
+            r_pre_in_splits = in_sp
+
+            # recv from local rank lc: direct partials (owner me, n == n0) sum
+            # into my dkv; proxied partials (owner (n, l0), n != n0) sum into
+            # the proxy buffer laid out per node at off[n] (same proxy_u).
+            d_in, d_out = [], []
+            x_in, x_out = [], []
+            out_sp = []
+            rcur = 0
+            for lc in range(wi):
+                tok = sum(
+                    ov[g(n, l0)][g(n0, lc)].total_seqlen for n in range(wn)
+                )
+                for pass_v in (False, True):
+                    for n in range(wn):
+                        u = proxy_u[n]
+                        U = u.total_seqlen
+                        for p in ov[g(n, l0)][g(n0, lc)]:
+                            if n == n0:
+                                a, _b = loc_host(p)
+                                d_in.append((rcur, rcur + p.seqlen))
+                                d_out.append(L + a if pass_v else a)
+                            else:
+                                lp = u.make_range_local(p)
+                                base = off[n] + (U if pass_v else 0)
+                                x_in.append((rcur, rcur + p.seqlen))
+                                x_out.append(base + lp.start)
+                            rcur += p.seqlen
+                out_sp.append(2 * tok)
+            proxy_rows = inter_total_recv  # same layout as the cast proxy buf
+
+            # ---------- reduce phase 2: inter (node-sum -> owner) ----------
+            # send: the proxy buffer itself, segment n = [2*U_n] rows
+            r_inter_in_splits = list(inter_out_splits)
+            # recv: node-sums for MY hosted rows from each contributing node
+            in_r2, out_s2, out_sp2 = [], [], []
+            rcur2 = 0
+            for n in range(wn):
+                u = (union([ov[rank][g(n, l)] for l in range(wi)])
+                     if n != n0 else AttnRanges())
+                tok = u.total_seqlen
+                for pass_v in (False, True):
+                    for p in u:
+                        a, _b = loc_host(p)
+                        in_r2.append((rcur2, rcur2 + p.seqlen))
+                        out_s2.append(L + a if pass_v else a)
+                        rcur2 += p.seqlen
+                out_sp2.append(2 * tok)
+            reduces.append(HierGroupReduceArg(
+                pre_send_pack=r_pre_send,
+                pre_in_splits=r_pre_in_splits,
+                pre_recv_direct=RowChunkMap(d_in, d_out, rcur),
+                pre_recv_proxy=RowChunkMap(x_in, x_out, proxy_rows),
+                pre_out_splits=out_sp,
+                pre_total_recv=rcur,
+                proxy_rows=proxy_rows,
+                inter_in_splits=r_inter_in_splits,
+                inter_recv_reduce=RowChunkMap(in_r2, out_s2, rcur2),
+                inter_out_splits=out_sp2,
+                inter_total_recv=rcur2,
+            ))
+        return casts, reduces

@@ -89,9 +89,53 @@ class GroupReduceArg:
 
 
 @dataclass
+class HierGroupCastArg:
+    """Hierarchical (2D-mesh) K/V multicast for one stage: pre-intra a2av for
+    same-node destinations, inter a2av sending ONE deduplicated copy per remote
+    node to the same-local-rank proxy, post-intra a2av forwarding proxy rows to
+    final destinations (reference grpcoll/_group_collective_hier.py:49
+    HierGroupCastMetaSolver — but built fully at PLAN time from the solver's
+    global overlap table, no runtime all_gather_object)."""
+
+    pre: GroupCastArg                 # intra hop (recv lands in stage buffer)
+    inter_send_pack: RowChunkMap      # local kv rows -> inter send buffer
+    inter_in_splits: List[int]        # per peer node (2*tok rows)
+    inter_out_splits: List[int]
+    inter_total_recv: int
+    post_send_pack: RowChunkMap       # inter-recv rows -> post send buffer
+    post_in_splits: List[int]
+    post_recv_unpack: RowChunkMap     # post recv rows -> stage buffer
+    post_out_splits: List[int]
+    stage_tokens: int
+
+
+@dataclass
+class HierGroupReduceArg:
+    """Mirror path for partial dK/dV: pre-intra a2av routes partials to the
+    in-node proxy holding the owner's local rank, which SUM-reduces its node's
+    contributions (the traffic saving), then one inter a2av delivers each
+    node-sum to the owner."""
+
+    pre_send_pack: RowChunkMap        # stage partial rows -> pre send buffer
+    pre_in_splits: List[int]
+    pre_recv_direct: RowChunkMap      # pre recv -> local dkv rows (sum)
+    pre_recv_proxy: RowChunkMap       # pre recv -> proxy buffer rows (sum)
+    pre_out_splits: List[int]
+    pre_total_recv: int
+    proxy_rows: int                   # proxy buffer rows (== inter send buffer)
+    inter_in_splits: List[int]        # per peer node
+    inter_recv_reduce: RowChunkMap    # inter recv -> local dkv rows (sum)
+    inter_out_splits: List[int]
+    inter_total_recv: int
+
+
+@dataclass
 class CommMeta:
     stages_cast: List[GroupCastArg]
     stages_reduce: List[GroupReduceArg]
+    # hierarchical realisation of the same stages (set iff 2D mesh + env flag)
+    stages_cast_hier: Optional[List[HierGroupCastArg]] = None
+    stages_reduce_hier: Optional[List[HierGroupReduceArg]] = None
 
     @property
     def overlap_degree(self) -> int:
