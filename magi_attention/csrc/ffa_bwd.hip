@@ -33,6 +33,20 @@ using bf16_t = __bf16;
 
 #define DEV_INLINE __device__ __forceinline__
 
+// Software-pipelined LDS-DMA barrier: waits until at most VM vector-memory
+// ops are outstanding (= exactly the NEXT prefetch group, leaving the current
+// tile's group retired), makes LDS reads visible, then syncs the block.
+// vmcnt retires in issue order, so a constant distance works as long as every
+// iteration issues the same number of vector-memory ops — stage_glds is
+// therefore called unconditionally with clamped addresses past the end.
+// __syncthreads() would insert s_waitcnt vmcnt(0) and collapse the pipeline
+// to depth one (measured: 34-42%% SQ_WAIT in both backward kernels).
+template <int VM>
+DEV_INLINE void pipe_barrier() {
+  asm volatile("s_waitcnt vmcnt(%0)\n\ts_waitcnt lgkmcnt(0)\n\ts_barrier"
+               :: "n"(VM) : "memory");
+}
+
 DEV_INLINE int crow(int r, int hi) { return (r & 3) + 8 * (r >> 2) + 4 * hi; }
 
 DEV_INLINE bool wave_alive(int m0, int qe) { return m0 < qe; }
@@ -201,18 +215,20 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
   // ONE shared object only: a second __shared__ array makes hipcc drain
   // vmcnt(0) before every ds_read, destroying the glds pipeline (guide §5
   // ".s-level traps" (a)).
+  // 3 buffers: 2-deep prefetch (glds group t+2 issued while tile t computes)
   __shared__ __attribute__((aligned(16))) char smem[
-      2 * 2 * BWD_BM * D * 2 + 2 * 2 * BWD_BM * 4];
+      3 * 2 * BWD_BM * D * 2 + 3 * 2 * BWD_BM * 4];
   auto lds_q = [&](int buf) -> __bf16* {
-    return (__bf16*)(smem + buf * BWD_BM * D * 2);
+    return (__bf16*)(smem + (2 * buf) * BWD_BM * D * 2);
   };
   auto lds_do = [&](int buf) -> __bf16* {
-    return (__bf16*)(smem + (2 + buf) * BWD_BM * D * 2);
+    return (__bf16*)(smem + (2 * buf + 1) * BWD_BM * D * 2);
   };
-  // lse / dpsum of the q tile, double-buffered like the glds images (staged
-  // for tile t+1 during tile t)
+  // lse / dpsum of the q tile, staged by LDS-DMA in the SAME vmcnt group as
+  // the row images (one dword per lane: lanes 0-31 lse, 32-63 dpsum) so the
+  // per-iteration vector-memory count stays constant
   auto lds_lse = [&](int buf) -> float* {
-    return (float*)(smem + 4 * BWD_BM * D * 2 + buf * 2 * BWD_BM * 4);
+    return (float*)(smem + 6 * BWD_BM * D * 2 + buf * 2 * BWD_BM * 4);
   };
   auto lds_dps = [&](int buf) -> float* {
     return lds_lse(buf) + BWD_BM;
@@ -273,20 +289,29 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
           (__attribute__((address_space(3))) unsigned int*)&lds_do(buf)[r0 * D],
           16, 0, 0);
     }
-    // lse/dpsum for the same q tile
-    if (tid < BWD_BM) {
-      const int qr = min(m0x + tid, qe - 1);
-      lds_lse(buf)[tid] = p.lse[(size_t)qr * p.hq + h];
-      lds_dps(buf)[tid] = p.dpsum[(size_t)qr * p.hq + h];
+    // lse (lanes 0-31) / dpsum (lanes 32-63) of the same q tile via LDS-DMA
+    {
+      const int qr = min(m0x + lo32, qe - 1);
+      const float* src = (lane < 32) ? p.lse + (size_t)qr * p.hq + h
+                                     : p.dpsum + (size_t)qr * p.hq + h;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)src,
+          (__attribute__((address_space(3))) unsigned int*)lds_lse(buf),
+          4, 0, 0);
     }
   };
+  // vector-memory ops per stage_glds call (the pipe_barrier distance)
+  constexpr int PIPE_VM = 2 * GLDS_PER_WAVE + 1;
 
   int cur = 0;
-  if (q_lo < q_hi) stage_glds(0, q_lo);
+  if (q_lo < q_hi) {
+    stage_glds(0, q_lo);
+    stage_glds(1, q_lo + BWD_BM);  // clamped rows make overrun safe
+  }
 
   for (int m0 = q_lo; m0 < q_hi; m0 += BWD_BM) {
-    __syncthreads();  // buf[cur] ready (barrier drains in-flight glds)
-    if (m0 + BWD_BM < q_hi) stage_glds(cur ^ 1, m0 + BWD_BM);
+    pipe_barrier<PIPE_VM>();  // buf[cur] landed; buf[(cur+1)%3] in flight
+    stage_glds(cur == 0 ? 2 : cur - 1, m0 + 2 * BWD_BM);
 
     if (wave_live && m0 + BWD_BM > wq_lo && m0 < wq_hi) {
       // ---- S = Q K^T ; dP = dO V^T, UN-swapped: C layout [q=crow][k=lo32],
@@ -400,7 +425,7 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
         }
       }
     }
-    cur ^= 1;
+    cur = cur == 2 ? 0 : cur + 1;
   }
 
   // ---- write dK/dV ----
@@ -428,7 +453,7 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
 // cooperatively per iteration (row-major swizzled for the S^T/dP^T A-frags,
 // plus a transposed copy for the dQ B-frags).
 template <int D, bool HAS_SOFTCAP>
-__global__ __launch_bounds__(256, 1) void ffa_bwd_dq_kernel(BwdParams p) {
+__global__ __launch_bounds__(256, 2) void ffa_bwd_dq_kernel(BwdParams p) {
   constexpr int DF = D / 16;
   constexpr int DT = D / 32;
   constexpr int ROWB = D * 2;
@@ -467,13 +492,14 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dq_kernel(BwdParams p) {
   const size_t k_pitch = (size_t)p.hk * D;
   const size_t q_pitch = (size_t)p.hq * D;
 
-  // single shared object (glds-pipeline trap, see dkv kernel)
-  __shared__ __attribute__((aligned(16))) char smem[2 * 2 * BWD_BN * D * 2];
+  // single shared object (glds-pipeline trap, see dkv kernel); 3 buffers for
+  // the 2-deep constant-distance prefetch pipeline
+  __shared__ __attribute__((aligned(16))) char smem[3 * 2 * BWD_BN * D * 2];
   auto lds_k = [&](int buf) -> __bf16* {
-    return (__bf16*)(smem + buf * BWD_BN * D * 2);
+    return (__bf16*)(smem + (2 * buf) * BWD_BN * D * 2);
   };
   auto lds_v = [&](int buf) -> __bf16* {
-    return (__bf16*)(smem + (2 + buf) * BWD_BN * D * 2);
+    return (__bf16*)(smem + (2 * buf + 1) * BWD_BN * D * 2);
   };
 
   // persistent per-wave operands: Q and dO fragments (B-layout rows)
@@ -528,12 +554,16 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dq_kernel(BwdParams p) {
     }
   };
 
+  constexpr int PIPE_VM = 2 * GLDS_PER_WAVE;
   int cur = 0;
-  if (k_lo < k_hi) stage_glds(0, k_lo);
+  if (k_lo < k_hi) {
+    stage_glds(0, k_lo);
+    stage_glds(1, k_lo + BWD_BN);  // clamped rows make overrun safe
+  }
 
   for (int n0 = k_lo; n0 < k_hi; n0 += BWD_BN) {
-    __syncthreads();  // glds for buf[cur] drained here
-    if (n0 + BWD_BN < k_hi) stage_glds(cur ^ 1, n0 + BWD_BN);
+    pipe_barrier<PIPE_VM>();  // buf[cur] landed; buf[(cur+1)%3] in flight
+    stage_glds(cur == 0 ? 2 : cur - 1, n0 + 2 * BWD_BN);
 
     if (wave_alive(m0, qe) && n0 + BWD_BN > wk_lo && n0 < wk_hi) {
       // ---- S^T = K Q^T ; dP^T = V dO^T (K/V A-frags from LDS rows) ----
@@ -616,7 +646,7 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dq_kernel(BwdParams p) {
         }
       }
     }
-    cur ^= 1;
+    cur = cur == 2 ? 0 : cur + 1;
   }
 
   // ---- store dq once (atomicAdd: q_ranges of different slices may overlap) ----
