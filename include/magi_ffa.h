@@ -92,6 +92,11 @@ int magi_ffa_bwd_preprocess(const magi_ffa_bwd_args* args);
  * magi_ffa_bwd runs both sequentially on args->stream. */
 int magi_ffa_bwd_dq(const magi_ffa_bwd_args* args);
 int magi_ffa_bwd_dkv(const magi_ffa_bwd_args* args);
+/* dv/dk pass split of magi_ffa_bwd_dkv: each half re-derives S but fits
+ * 2 waves/SIMD on gfx950; launching both is result-identical to the fused
+ * kernel up to fp32 atomic-add ordering. */
+int magi_ffa_bwd_dv(const magi_ffa_bwd_args* args);
+int magi_ffa_bwd_dk(const magi_ffa_bwd_args* args);
 int magi_ffa_bwd(const magi_ffa_bwd_args* args);
 
 /* ------------------------------------------------------------------ *

@@ -118,6 +118,8 @@ def _try_load() -> ctypes.CDLL | None:
             ("magi_ffa_bwd", [ctypes.POINTER(MagiFfaBwdArgs)]),
             ("magi_ffa_bwd_dq", [ctypes.POINTER(MagiFfaBwdArgs)]),
             ("magi_ffa_bwd_dkv", [ctypes.POINTER(MagiFfaBwdArgs)]),
+            ("magi_ffa_bwd_dv", [ctypes.POINTER(MagiFfaBwdArgs)]),
+            ("magi_ffa_bwd_dk", [ctypes.POINTER(MagiFfaBwdArgs)]),
             ("magi_ffa_bwd_preprocess", [ctypes.POINTER(MagiFfaBwdArgs)]),
             ("magi_range_gather", [ctypes.POINTER(MagiRangeOpArgs)]),
             ("magi_range_reduce", [ctypes.POINTER(MagiRangeOpArgs)]),

@@ -159,8 +159,15 @@ __global__ __launch_bounds__(256) void bwd_preprocess_kernel(BwdParams p, int d,
 
 
 // ---------------- mainloop ----------------
-template <int D, bool HAS_SOFTCAP>
-__global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
+// MODE 0: fused dK+dV (1 wave/SIMD — 128 f32 accumulators).
+// MODE 1: dV only / MODE 2: dK only — each recomputes S but fits 2 waves/SIMD,
+// which beats the fused kernel's naked single-wave stalls (the dq kernel went
+// 4.1x faster at occupancy 2 with the same pipeline).
+template <int D, bool HAS_SOFTCAP, int MODE>
+__global__ __launch_bounds__(256, MODE == 0 ? 1 : 2)
+void ffa_bwd_dkv_kernel(BwdParams p) {
+  constexpr bool WANT_DV = MODE != 2;
+  constexpr bool WANT_DK = MODE != 1;
   constexpr int DF = D / 16;
   constexpr int DT = D / 32;
   constexpr int ROWB = D * 2;  // bytes per LDS tile row
@@ -237,14 +244,14 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
   // K/V fragments (A-layout) + K B-fragments, loaded once per block
   const int krow = n0 + lo32;
   const int kcl = min(krow, ke - 1);
-  bf16x8 kfA[DF], vfA[DF];
+  bf16x8 kfA[DF], vfA[WANT_DK ? DF : 1];
   {
     const bf16_t* kp = p.k + (size_t)kcl * k_pitch + (size_t)kh * D;
     const bf16_t* vp = p.v + (size_t)kcl * k_pitch + (size_t)kh * D;
 #pragma unroll
     for (int dd = 0; dd < DF; ++dd) {
       kfA[dd] = *(const bf16x8*)(kp + dd * 16 + hi * 8);
-      vfA[dd] = *(const bf16x8*)(vp + dd * 16 + hi * 8);
+      if constexpr (WANT_DK) vfA[dd] = *(const bf16x8*)(vp + dd * 16 + hi * 8);
     }
   }
 
@@ -259,11 +266,11 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
   if (atype == 2 || atype == 3)
     wq_hi = min(wq_hi, (n0 + BWD_BN - 1) - (ks - qs) + 1);
 
-  f32x16 acc_dk[DT], acc_dv[DT];
+  f32x16 acc_dk[WANT_DK ? DT : 1], acc_dv[WANT_DV ? DT : 1];
 #pragma unroll
   for (int dt = 0; dt < DT; ++dt) {
-    acc_dk[dt] = (f32x16)(0.f);
-    acc_dv[dt] = (f32x16)(0.f);
+    if constexpr (WANT_DK) acc_dk[dt] = (f32x16)(0.f);
+    if constexpr (WANT_DV) acc_dv[dt] = (f32x16)(0.f);
   }
 
   // each wave's glds covers 4 rows (64 lanes x 16B = 1 KiB = 4 rows at D=128);
@@ -322,13 +329,19 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
       for (int dd = 0; dd < DF; ++dd) {
         const int off = swz(lo32, lo32 * ROWB + dd * 32 + hi * 16);
         bf16x8 qf = *(const bf16x8*)((const char*)lds_q(cur) + off);
-        bf16x8 dof = *(const bf16x8*)((const char*)lds_do(cur) + off);
         s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kfA[dd], s, 0, 0, 0);
-        dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vfA[dd], dp, 0, 0, 0);
+        if constexpr (WANT_DK) {
+          bf16x8 dof = *(const bf16x8*)((const char*)lds_do(cur) + off);
+          dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vfA[dd], dp, 0, 0, 0);
+        }
       }
 
       const int kk = n0 + lo32;  // this lane's k column
-      float pv[16], dsv[16];
+      // S is dead once P is computed, dP once dS is — reuse their registers
+      // (all indices are compile-time constants after unrolling, so the
+      // pointer cast stays in VGPRs; checked: ScratchSize 0)
+      float* pv = (float*)&s;
+      float* dsv = (float*)&dp;
       const bool interior =
           (m0 + BWD_BM <= wq_hi) && (m0 >= qs) && (n0 + BWD_BN <= ke) &&
           !((atype == 1 || atype == 3) && (n0 + BWD_BN - 1 > m0 + (ke - qe))) &&
@@ -346,8 +359,9 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
         for (int r = 0; r < 16; ++r) {
           const int rl = crow(r, hi);
           const float pij = exp2f(s[r] * sl2 - lds_lse(cur)[rl] * log2e);
-          pv[r] = pij;
-          dsv[r] = pij * (dp[r] - lds_dps(cur)[rl]) * p.scale;
+          if constexpr (WANT_DV) pv[r] = pij;
+          if constexpr (WANT_DK)
+            dsv[r] = pij * (dp[r] - lds_dps(cur)[rl]) * p.scale;
         }
       } else {
 #pragma unroll
@@ -369,8 +383,9 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
             t = sv * sl2;
           }
           const float pij = ok ? exp2f(t - lq * log2e) : 0.f;
-          pv[r] = pij;
-          dsv[r] = pij * (dp[r] - lds_dps(cur)[crow(r, hi)]) * dscale;
+          if constexpr (WANT_DV) pv[r] = pij;
+          if constexpr (WANT_DK)
+            dsv[r] = pij * (dp[r] - lds_dps(cur)[crow(r, hi)]) * dscale;
         }
       }
 
@@ -381,10 +396,15 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
       //   addr = lds_base + row*ROWB + ((dcol*2) ^ ((row&7)<<5)) + (l&3)*8
       // (the XOR distributes because the swizzle bits live inside dcol*2).
       {
-        bf16x8 pa0 = cframe_to_afrag(pv, 0);
-        bf16x8 pa1 = cframe_to_afrag(pv, 1);
-        bf16x8 da0 = cframe_to_afrag(dsv, 0);
-        bf16x8 da1 = cframe_to_afrag(dsv, 1);
+        bf16x8 pa0, pa1, da0, da1;
+        if constexpr (WANT_DV) {
+          pa0 = cframe_to_afrag(pv, 0);
+          pa1 = cframe_to_afrag(pv, 1);
+        }
+        if constexpr (WANT_DK) {
+          da0 = cframe_to_afrag(dsv, 0);
+          da1 = cframe_to_afrag(dsv, 1);
+        }
         const int qhalf = (lane >> 4) & 1;       // which 16-d column half
         const int jrow = (lane & 15) >> 2;       // canonical row for this lane
         const int row0 = 8 * hi + jrow;          // rd = 0 rows
@@ -406,22 +426,30 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
 #pragma unroll
         for (int dt = 0; dt < DT; ++dt) {
           const int dcol = (dt * 32 + 16 * qhalf) * 2;
-          bf16x8 bdo = tr16_frag(do_base + rb0 + (dcol ^ sw0),
-                                 do_base + rb1 + (dcol ^ sw1));
-          acc_dv[dt] =
-              __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa0, bdo, acc_dv[dt], 0, 0, 0);
-          bf16x8 bq = tr16_frag(q_base + rb0 + (dcol ^ sw0),
-                                q_base + rb1 + (dcol ^ sw1));
-          acc_dk[dt] =
-              __builtin_amdgcn_mfma_f32_32x32x16_bf16(da0, bq, acc_dk[dt], 0, 0, 0);
-          bf16x8 bdo1 = tr16_frag(do_base + rb2 + (dcol ^ sw2),
-                                  do_base + rb3 + (dcol ^ sw3));
-          acc_dv[dt] =
-              __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa1, bdo1, acc_dv[dt], 0, 0, 0);
-          bf16x8 bq1 = tr16_frag(q_base + rb2 + (dcol ^ sw2),
-                                 q_base + rb3 + (dcol ^ sw3));
-          acc_dk[dt] =
-              __builtin_amdgcn_mfma_f32_32x32x16_bf16(da1, bq1, acc_dk[dt], 0, 0, 0);
+          if constexpr (WANT_DV) {
+            bf16x8 bdo = tr16_frag(do_base + rb0 + (dcol ^ sw0),
+                                   do_base + rb1 + (dcol ^ sw1));
+            acc_dv[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                pa0, bdo, acc_dv[dt], 0, 0, 0);
+          }
+          if constexpr (WANT_DK) {
+            bf16x8 bq = tr16_frag(q_base + rb0 + (dcol ^ sw0),
+                                  q_base + rb1 + (dcol ^ sw1));
+            acc_dk[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                da0, bq, acc_dk[dt], 0, 0, 0);
+          }
+          if constexpr (WANT_DV) {
+            bf16x8 bdo1 = tr16_frag(do_base + rb2 + (dcol ^ sw2),
+                                    do_base + rb3 + (dcol ^ sw3));
+            acc_dv[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                pa1, bdo1, acc_dv[dt], 0, 0, 0);
+          }
+          if constexpr (WANT_DK) {
+            bf16x8 bq1 = tr16_frag(q_base + rb2 + (dcol ^ sw2),
+                                   q_base + rb3 + (dcol ^ sw3));
+            acc_dk[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                da1, bq1, acc_dk[dt], 0, 0, 0);
+          }
         }
       }
     }
@@ -438,8 +466,14 @@ __global__ __launch_bounds__(256, 1) void ffa_bwd_dkv_kernel(BwdParams p) {
     float* dvp = p.dv + (size_t)kr * k_pitch + (size_t)kh * D;
 #pragma unroll
     for (int dt = 0; dt < DT; ++dt) {
-      if (acc_dk[dt][r] != 0.f) unsafeAtomicAdd(dkp + dt * 32 + lo32, acc_dk[dt][r]);
-      if (acc_dv[dt][r] != 0.f) unsafeAtomicAdd(dvp + dt * 32 + lo32, acc_dv[dt][r]);
+      if constexpr (WANT_DK) {
+        if (acc_dk[dt][r] != 0.f)
+          unsafeAtomicAdd(dkp + dt * 32 + lo32, acc_dk[dt][r]);
+      }
+      if constexpr (WANT_DV) {
+        if (acc_dv[dt][r] != 0.f)
+          unsafeAtomicAdd(dvp + dt * 32 + lo32, acc_dv[dt][r]);
+      }
     }
   }
 }
@@ -753,7 +787,8 @@ extern "C" int magi_ffa_bwd_dq(const magi_ffa_bwd_args* a) {
   return (int)hipGetLastError();
 }
 
-extern "C" int magi_ffa_bwd_dkv(const magi_ffa_bwd_args* a) {
+template <int MODE>
+static int launch_bwd_dkv(const magi_ffa_bwd_args* a) {
   BwdParams p;
   int rc = fill_bwd_params(a, &p);
   if (rc) return rc > 0 ? 0 : rc;
@@ -768,16 +803,32 @@ extern "C" int magi_ffa_bwd_dkv(const magi_ffa_bwd_args* a) {
   const bool sc = a->softcap > 0.f;
   if (a->d == 64) {
     if (sc)
-      hipLaunchKernelGGL((ffa_bwd_dkv_kernel<64, true>), grid_kv, block, 0, s, p);
+      hipLaunchKernelGGL((ffa_bwd_dkv_kernel<64, true, MODE>), grid_kv, block,
+                         0, s, p);
     else
-      hipLaunchKernelGGL((ffa_bwd_dkv_kernel<64, false>), grid_kv, block, 0, s, p);
+      hipLaunchKernelGGL((ffa_bwd_dkv_kernel<64, false, MODE>), grid_kv, block,
+                         0, s, p);
   } else {
     if (sc)
-      hipLaunchKernelGGL((ffa_bwd_dkv_kernel<128, true>), grid_kv, block, 0, s, p);
+      hipLaunchKernelGGL((ffa_bwd_dkv_kernel<128, true, MODE>), grid_kv, block,
+                         0, s, p);
     else
-      hipLaunchKernelGGL((ffa_bwd_dkv_kernel<128, false>), grid_kv, block, 0, s, p);
+      hipLaunchKernelGGL((ffa_bwd_dkv_kernel<128, false, MODE>), grid_kv,
+                         block, 0, s, p);
   }
   return (int)hipGetLastError();
+}
+
+extern "C" int magi_ffa_bwd_dkv(const magi_ffa_bwd_args* a) {
+  return launch_bwd_dkv<0>(a);
+}
+
+extern "C" int magi_ffa_bwd_dv(const magi_ffa_bwd_args* a) {
+  return launch_bwd_dkv<1>(a);
+}
+
+extern "C" int magi_ffa_bwd_dk(const magi_ffa_bwd_args* a) {
+  return launch_bwd_dkv<2>(a);
 }
 
 extern "C" int magi_ffa_bwd(const magi_ffa_bwd_args* a) {

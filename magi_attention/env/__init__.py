@@ -57,6 +57,12 @@ def is_hierarchical_comm_enable() -> bool:
     return _get_bool("MAGI_ATTENTION_HIERARCHICAL_COMM")
 
 
+def is_bwd_fused_dkv() -> bool:
+    """MAGI_BWD_FUSED_DKV=1 reverts to the single fused dK+dV kernel
+    (1 wave/SIMD) instead of the default dv/dk pass split (2 waves/SIMD)."""
+    return _get_bool("MAGI_BWD_FUSED_DKV")
+
+
 def ffa_forward_sm_margin() -> int:
     return _get_int("MAGI_ATTENTION_FFA_FORWARD_SM_MARGIN", 8)
 
