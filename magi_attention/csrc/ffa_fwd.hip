@@ -181,9 +181,15 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
   constexpr int NPASS = FFA_BN / RPP;
   const int srow = threadIdx.x / CPR;
   const int scol = threadIdx.x % CPR;
-  bf16x8 kreg[NPASS], vreg[NPASS];
+  // TWO register staging sets: loads for tile t+2 are issued while tile t
+  // computes and tile t+1's regs are written to LDS at loop TOP — the
+  // implicit vmcnt wait at write_stage then covers loads issued a FULL
+  // iteration earlier instead of the same iteration (the one-deep form
+  // measured 33% SQ_WAIT). Sets alternate by 2x loop unrolling so the
+  // register arrays stay compile-time addressed (rule 20).
+  bf16x8 kregA[NPASS], vregA[NPASS], kregB[NPASS], vregB[NPASS];
 
-  auto issue_loads = [&](int n0) {
+  auto issue_loads = [&](int n0, auto& kreg, auto& vreg) {
 #pragma unroll
     for (int pass = 0; pass < NPASS; ++pass) {
       const int kr = min(n0 + pass * RPP + srow, ke - 1);
@@ -191,7 +197,7 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
       vreg[pass] = *(const bf16x8*)(vbase + (size_t)kr * k_pitch + scol * 8);
     }
   };
-  auto write_stage = [&](int buf) {
+  auto write_stage = [&](int buf, auto& kreg, auto& vreg) {
 #pragma unroll
     for (int pass = 0; pass < NPASS; ++pass) {
       const int r = pass * RPP + srow;
@@ -203,14 +209,18 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
 
   int cur = 0;
   if (b_lo < b_hi) {
-    issue_loads(b_lo);
-    write_stage(0);
+    issue_loads(b_lo, kregA, vregA);
+    write_stage(0, kregA, vregA);
+    issue_loads(b_lo + FFA_BN, kregA, vregA);  // tile t1 (clamped if absent)
   }
   __syncthreads();
 
-  for (int n0 = b_lo; n0 < b_hi; n0 += FFA_BN) {
+  auto iter_body = [&](int n0, auto& kreg_w, auto& vreg_w, auto& kreg_l,
+                       auto& vreg_l) {
     const bool has_next = n0 + FFA_BN < b_hi;
-    if (has_next) issue_loads(n0 + FFA_BN);  // lands under this tile's compute
+    // stage tile t+1 (regs loaded last iteration) and start loads for t+2
+    if (has_next) write_stage(cur ^ 1, kreg_w, vreg_w);
+    issue_loads(n0 + 2 * FFA_BN, kreg_l, vreg_l);  // clamped overrun is safe
     const bool live = (n0 + FFA_BN > n_lo) && (n0 < n_hi) && qvalid_any;
 
     f32x16 s = (f32x16)(0.f);
@@ -223,10 +233,9 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
       }
     }
     if (!live) {
-      if (has_next) write_stage(cur ^ 1);
       __syncthreads();
       cur ^= 1;
-      continue;
+      return;
     }
 
     // ---- mask + scale into exp2 domain ----
@@ -343,9 +352,19 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
         }
       }
     }
-    if (has_next) write_stage(cur ^ 1);  // other buffer: overlaps this compute
     __syncthreads();
     cur ^= 1;
+  };
+
+  {
+    int n0 = b_lo;
+    while (n0 < b_hi) {
+      iter_body(n0, kregA, vregA, kregB, vregB);
+      n0 += FFA_BN;
+      if (n0 >= b_hi) break;
+      iter_body(n0, kregB, vregB, kregA, vregA);
+      n0 += FFA_BN;
+    }
   }
 
   // ======================= epilogue =======================
