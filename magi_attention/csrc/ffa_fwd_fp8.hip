@@ -142,9 +142,12 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_fp8_kernel(Fp8FwdParams p) {
   const int srow = threadIdx.x / CPR;
   const int scol = threadIdx.x % CPR;
   using u8x16 = __attribute__((ext_vector_type(16))) u8;
-  u8x16 kreg[NPASS], vreg[NPASS];
+  // two register staging sets (see ffa_fwd.hip): loads for tile t+2 fly
+  // under tile t's compute, write_stage at loop TOP waits on loads issued a
+  // full iteration earlier
+  u8x16 kregA[NPASS], vregA[NPASS], kregB[NPASS], vregB[NPASS];
 
-  auto issue_loads = [&](int n0) {
+  auto issue_loads = [&](int n0, auto& kreg, auto& vreg) {
 #pragma unroll
     for (int pass = 0; pass < NPASS; ++pass) {
       const int r = pass * RPP + srow;
@@ -153,7 +156,7 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_fp8_kernel(Fp8FwdParams p) {
       vreg[pass] = *(const u8x16*)(vbase + (size_t)kr * k_pitch + scol * 16);
     }
   };
-  auto write_stage = [&](int buf) {
+  auto write_stage = [&](int buf, auto& kreg, auto& vreg) {
 #pragma unroll
     for (int pass = 0; pass < NPASS; ++pass) {
       const int r = pass * RPP + srow;
@@ -173,20 +176,22 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_fp8_kernel(Fp8FwdParams p) {
 
   int cur = 0;
   if (b_lo < b_hi) {
-    issue_loads(b_lo);
-    write_stage(0);
+    issue_loads(b_lo, kregA, vregA);
+    write_stage(0, kregA, vregA);
+    issue_loads(b_lo + FP8_BN, kregA, vregA);
   }
   __syncthreads();
 
-  for (int n0 = b_lo; n0 < b_hi; n0 += FP8_BN) {
+  auto iter_body = [&](int n0, auto& kreg_w, auto& vreg_w, auto& kreg_l,
+                       auto& vreg_l) {
     const bool has_next = n0 + FP8_BN < b_hi;
-    if (has_next) issue_loads(n0 + FP8_BN);
+    if (has_next) write_stage(cur ^ 1, kreg_w, vreg_w);
+    issue_loads(n0 + 2 * FP8_BN, kreg_l, vreg_l);  // clamped overrun is safe
     const bool live = (n0 + FP8_BN > n_lo) && (n0 < n_hi) && qvalid_any;
     if (!live) {
-      if (has_next) write_stage(cur ^ 1);
       __syncthreads();
       cur ^= 1;
-      continue;
+      return;
     }
 
     // ---- S^T = K Q^T (swapped; A = K rows from LDS) ----
@@ -279,9 +284,19 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_fp8_kernel(Fp8FwdParams p) {
             __builtin_amdgcn_mfma_f32_32x32x16_fp8_fp8(pa[tt], bv, acc_o[dt], 0, 0, 0);
       }
     }
-    if (has_next) write_stage(cur ^ 1);
     __syncthreads();
     cur ^= 1;
+  };
+
+  {
+    int n0 = b_lo;
+    while (n0 < b_hi) {
+      iter_body(n0, kregA, vregA, kregB, vregB);
+      n0 += FP8_BN;
+      if (n0 >= b_hi) break;
+      iter_body(n0, kregB, vregB, kregA, vregA);
+      n0 += FP8_BN;
+    }
   }
 
   // ---- epilogue: fp8 sum is scaled by 2^MAX_OFFSET relative to exp2(t-m) ----
