@@ -125,6 +125,34 @@ typedef struct magi_range_op_args {
 int magi_range_gather(const magi_range_op_args* args);
 int magi_range_reduce(const magi_range_op_args* args);
 
+/* Attention-sink postprocess (reference flash_fwd_postprocess_kernel.h:39
+ * FlashAttnFwdPostprocess / flash_bwd_preprocess_kernel.h dsink path):
+ * value-less learnable logits folded into (out, lse) AFTER the (possibly
+ * multi-slice, atomic-merged) forward, so they enter the softmax denominator
+ * exactly once. Layouts: ssh=0 -> sink[s_sink, h]; ssh=1 -> sink[T, s_sink, h].
+ */
+typedef struct magi_sink_args {
+  void* out;            /* [T, h, d] bf16 (out_is_fp32=0) or f32, in/out */
+  float* lse;           /* [T, h], in/out (fwd) / in (bwd) */
+  const float* sink;    /* sink logits, f32 */
+  float* dsink;         /* same shape as sink, f32, zero-initialised (bwd) */
+  const float* dpsum;   /* [T, h] (bwd) */
+  int64_t total_rows;   /* T */
+  int32_t n_heads;
+  int32_t d;
+  int32_t s_sink;       /* <= 8 */
+  int32_t ssh;          /* 0 = "sh", 1 = "ssh" */
+  int32_t out_is_fp32;
+  void* stream;
+} magi_sink_args;
+
+/* fwd: lse_sink = log(sum_j exp(sink_j)); live rows: out *= exp(lse-lse'),
+ * lse' = log(exp(lse)+exp(lse_sink)); empty rows (lse=-inf): out=0, lse=lse_sink. */
+int magi_ffa_sink_postprocess(const magi_sink_args* args);
+/* bwd: dsink_j (+)= -exp(sink_j - lse_row) * dpsum_row (summed over rows for
+ * "sh", per row for "ssh"). */
+int magi_ffa_dsink(const magi_sink_args* args);
+
 /* Fused merge of two partial (out, lse) sets (reference functional/utils.py:371
  * correct_out_lse_kernel): out1/lse1 updated in place with out2/lse2 merged in. */
 typedef struct magi_correct_args {

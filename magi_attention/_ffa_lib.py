@@ -93,6 +93,23 @@ class MagiRangeOpArgs(ctypes.Structure):
     ]
 
 
+class MagiSinkArgs(ctypes.Structure):
+    _fields_ = [
+        ("out", ctypes.c_void_p),
+        ("lse", ctypes.c_void_p),
+        ("sink", ctypes.c_void_p),
+        ("dsink", ctypes.c_void_p),
+        ("dpsum", ctypes.c_void_p),
+        ("total_rows", ctypes.c_int64),
+        ("n_heads", ctypes.c_int32),
+        ("d", ctypes.c_int32),
+        ("s_sink", ctypes.c_int32),
+        ("ssh", ctypes.c_int32),
+        ("out_is_fp32", ctypes.c_int32),
+        ("stream", ctypes.c_void_p),
+    ]
+
+
 class MagiCorrectArgs(ctypes.Structure):
     _fields_ = [
         ("out1", ctypes.c_void_p),
@@ -124,6 +141,8 @@ def _try_load() -> ctypes.CDLL | None:
             ("magi_range_gather", [ctypes.POINTER(MagiRangeOpArgs)]),
             ("magi_range_reduce", [ctypes.POINTER(MagiRangeOpArgs)]),
             ("magi_correct_out_lse", [ctypes.POINTER(MagiCorrectArgs)]),
+            ("magi_ffa_sink_postprocess", [ctypes.POINTER(MagiSinkArgs)]),
+            ("magi_ffa_dsink", [ctypes.POINTER(MagiSinkArgs)]),
             (
                 "magi_probe_mfma",
                 [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p],

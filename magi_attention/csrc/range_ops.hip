@@ -181,3 +181,150 @@ extern "C" int magi_correct_out_lse(const magi_correct_args* a) {
                      (const float*)a->out2, a->lse2, n, a->d);
   return (int)hipGetLastError();
 }
+
+// ------------------------------------------------------------------
+// attention-sink postprocess + dsink (reference
+// flash_fwd_postprocess_kernel.h:39, flash_bwd_preprocess_kernel.h dsink
+// path): see include/magi_ffa.h for the math.
+// ------------------------------------------------------------------
+template <bool OUT_F32, bool SSH>
+__global__ __launch_bounds__(256) void sink_postprocess_kernel(
+    magi_sink_args a) {
+  const int lane = threadIdx.x & 63;
+  const int wavein = threadIdx.x >> 6;
+  const long long n_rowheads = a.total_rows * a.n_heads;
+  for (long long rh = blockIdx.x * 4 + wavein; rh < n_rowheads;
+       rh += (long long)gridDim.x * 4) {
+    const long long row = rh / a.n_heads;
+    const int h = rh % a.n_heads;
+    // lse_sink = logsumexp over the (<=8) sink logits of this (row, head)
+    float mx = -INFINITY;
+    float sv[8];
+    for (int j = 0; j < a.s_sink; ++j) {
+      sv[j] = SSH ? a.sink[(row * a.s_sink + j) * a.n_heads + h]
+                  : a.sink[(long long)j * a.n_heads + h];
+      mx = fmaxf(mx, sv[j]);
+    }
+    float l = 0.f;
+    for (int j = 0; j < a.s_sink; ++j) l += expf(sv[j] - mx);
+    const float lse_sink = (mx == -INFINITY) ? -INFINITY : mx + logf(l);
+
+    float* lse_p = a.lse + rh;
+    const float lse_old = *lse_p;
+    float w, lse_new;
+    if (lse_old == -INFINITY) {
+      w = 0.f;
+      lse_new = lse_sink;
+    } else if (lse_sink == -INFINITY) {
+      w = 1.f;
+      lse_new = lse_old;
+    } else {
+      const float m = fmaxf(lse_old, lse_sink);
+      lse_new = m + logf(expf(lse_old - m) + expf(lse_sink - m));
+      w = expf(lse_old - lse_new);
+    }
+    if (lane == 0) *lse_p = lse_new;
+    if (w != 1.f) {
+      if (OUT_F32) {
+        float* o = (float*)a.out + rh * a.d;
+        for (int i = lane; i < a.d; i += 64) o[i] *= w;
+      } else {
+        __bf16* o = (__bf16*)a.out + rh * a.d;
+        for (int i = lane; i < a.d; i += 64)
+          o[i] = (__bf16)((float)o[i] * w);
+      }
+    }
+  }
+}
+
+extern "C" int magi_ffa_sink_postprocess(const magi_sink_args* a) {
+  if (!a || !a->out || !a->lse || !a->sink) return -1;
+  if (a->s_sink <= 0 || a->s_sink > 8) return -2;
+  const long long n = a->total_rows * a->n_heads;
+  if (n == 0) return 0;
+  unsigned gx = (unsigned)min((long long)4096, (n + 3) / 4);
+  hipStream_t s = (hipStream_t)a->stream;
+  if (a->out_is_fp32) {
+    if (a->ssh)
+      hipLaunchKernelGGL((sink_postprocess_kernel<true, true>), dim3(gx),
+                         dim3(256), 0, s, *a);
+    else
+      hipLaunchKernelGGL((sink_postprocess_kernel<true, false>), dim3(gx),
+                         dim3(256), 0, s, *a);
+  } else {
+    if (a->ssh)
+      hipLaunchKernelGGL((sink_postprocess_kernel<false, true>), dim3(gx),
+                         dim3(256), 0, s, *a);
+    else
+      hipLaunchKernelGGL((sink_postprocess_kernel<false, false>), dim3(gx),
+                         dim3(256), 0, s, *a);
+  }
+  return (int)hipGetLastError();
+}
+
+// "sh": one block per (row-block, head); block-partial sums in LDS, one
+// atomicAdd per (block, j) — contention is blocks*h*s_sink, not rows.
+__global__ __launch_bounds__(256) void dsink_sh_kernel(magi_sink_args a) {
+  __shared__ float part[8][64];
+  const int h = blockIdx.y;
+  const int tid = threadIdx.x;
+  const long long row0 = (long long)blockIdx.x * 256;
+  float p[8];
+  for (int j = 0; j < a.s_sink; ++j) p[j] = 0.f;
+  const long long row = row0 + tid;
+  if (row < a.total_rows) {
+    const float lse = a.lse[row * a.n_heads + h];
+    const float dps = a.dpsum[row * a.n_heads + h];
+    if (lse != INFINITY && lse != -INFINITY) {
+      for (int j = 0; j < a.s_sink; ++j)
+        p[j] = -expf(a.sink[(long long)j * a.n_heads + h] - lse) * dps;
+    }
+  }
+  // wave reduce then cross-wave via LDS
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  for (int j = 0; j < a.s_sink; ++j) {
+    float x = p[j];
+    for (int off = 32; off; off >>= 1)
+      x += __shfl_down(x, off, 64);
+    if (lane == 0) part[j][w] = x;
+  }
+  __syncthreads();
+  if (tid < a.s_sink) {
+    float x = part[tid][0] + part[tid][1] + part[tid][2] + part[tid][3];
+    atomicAdd(a.dsink + (long long)tid * a.n_heads + h, x);
+  }
+}
+
+__global__ __launch_bounds__(256) void dsink_ssh_kernel(magi_sink_args a) {
+  const long long n = a.total_rows * a.n_heads;
+  for (long long rh = (long long)blockIdx.x * 256 + threadIdx.x; rh < n;
+       rh += (long long)gridDim.x * 256) {
+    const long long row = rh / a.n_heads;
+    const int h = rh % a.n_heads;
+    const float lse = a.lse[rh];
+    const float dps = a.dpsum[rh];
+    const bool live = lse != INFINITY && lse != -INFINITY;
+    for (int j = 0; j < a.s_sink; ++j) {
+      const long long idx = (row * a.s_sink + j) * a.n_heads + h;
+      a.dsink[idx] = live ? -expf(a.sink[idx] - lse) * dps : 0.f;
+    }
+  }
+}
+
+extern "C" int magi_ffa_dsink(const magi_sink_args* a) {
+  if (!a || !a->lse || !a->sink || !a->dsink || !a->dpsum) return -1;
+  if (a->s_sink <= 0 || a->s_sink > 8) return -2;
+  if (a->total_rows == 0) return 0;
+  hipStream_t s = (hipStream_t)a->stream;
+  if (a->ssh) {
+    const long long n = a->total_rows * a->n_heads;
+    unsigned gx = (unsigned)min((long long)4096, (n + 255) / 256);
+    hipLaunchKernelGGL(dsink_ssh_kernel, dim3(gx), dim3(256), 0, s, *a);
+  } else {
+    unsigned gx = (unsigned)((a->total_rows + 255) / 256);
+    hipLaunchKernelGGL(dsink_sh_kernel, dim3(gx, a->n_heads), dim3(256), 0, s,
+                       *a);
+  }
+  return (int)hipGetLastError();
+}

@@ -171,7 +171,6 @@ def _flex_flash_attn_forward(
     max_seqlen_q: Optional[int] = None,
     **_unused,
 ) -> tuple[torch.Tensor, AttnForwardMeta]:
-    assert sink is None, "sink support lands in a later round"
     is_fp8 = q.dtype == torch.float8_e4m3fn
     assert q.dtype in (torch.bfloat16, torch.float8_e4m3fn), (
         "bf16 or fp8-e4m3 inputs"
@@ -237,7 +236,40 @@ def _flex_flash_attn_forward(
         del keep
     else:
         check(fwd_fn(args), "magi_ffa_fwd")
+    if sink is not None:
+        _apply_sink_postprocess(out, lse, sink, sink_layout, tq, hq, d)
     return out, AttnForwardMeta(lse=lse, max_logits=None)
+
+
+def _check_sink(sink, sink_layout, tq, hq):
+    assert sink_layout in ("sh", "ssh"), f"unsupported sink_layout {sink_layout}"
+    sink = sink.contiguous().float()
+    if sink_layout == "sh":
+        assert sink.dim() == 2 and sink.shape[1] == hq, "sink must be [s_sink, hq]"
+        s_sink = sink.shape[0]
+    else:
+        assert sink.dim() == 3 and sink.shape[0] == tq and sink.shape[2] == hq, (
+            "sink must be [total_q, s_sink, hq]"
+        )
+        s_sink = sink.shape[1]
+    assert 1 <= s_sink <= 8, "seqlen_sink must be in [1, 8] (reference kMaxSeqlenSink)"
+    return sink, s_sink
+
+
+def _apply_sink_postprocess(out, lse, sink, sink_layout, tq, hq, d):
+    """Fold the sink logits into (out, lse) ONCE, after all slice merges
+    (reference flash_fwd_postprocess_kernel.h:39)."""
+    sink, s_sink = _check_sink(sink, sink_layout, tq, hq)
+    from .._ffa_lib import MagiSinkArgs
+
+    args = MagiSinkArgs(
+        out=ptr(out), lse=ptr(lse), sink=ptr(sink),
+        total_rows=tq, n_heads=hq, d=d, s_sink=s_sink,
+        ssh=int(sink_layout == "ssh"),
+        out_is_fp32=int(out.dtype == torch.float32),
+        stream=current_stream_ptr(),
+    )
+    check(_ffa_lib.lib().magi_ffa_sink_postprocess(args), "sink_postprocess")
 
 
 def _flex_flash_attn_backward(
@@ -267,7 +299,6 @@ def _flex_flash_attn_backward(
     max_seqlen_k: Optional[int] = None,
     **_unused,
 ) -> tuple[torch.Tensor, torch.Tensor, torch.Tensor, Optional[torch.Tensor]]:
-    assert sink is None, "sink support lands in a later round"
     dout, q, k, v, out, q_ranges, k_ranges = [
         maybe_contiguous(x) for x in (dout, q, k, v, out, q_ranges, k_ranges)
     ]
@@ -298,12 +329,26 @@ def _flex_flash_attn_backward(
     )
     lib = _ffa_lib.lib()
     check(lib.magi_ffa_bwd_preprocess(args), "magi_ffa_bwd_preprocess")
+    if sink is not None:
+        # dsink needs only (sink, lse, dpsum); the q/k/v gradients pick the
+        # sink up automatically through the corrected lse saved by forward
+        sink_f, s_sink = _check_sink(sink, sink_layout, tq, hq)
+        if dsink is None:
+            dsink = torch.zeros_like(sink_f)
+        from .._ffa_lib import MagiSinkArgs
+
+        sargs = MagiSinkArgs(
+            lse=ptr(lse), sink=ptr(sink_f), dsink=ptr(dsink), dpsum=ptr(dpsum),
+            total_rows=tq, n_heads=hq, d=d, s_sink=s_sink,
+            ssh=int(sink_layout == "ssh"), stream=current_stream_ptr(),
+        )
+        check(lib.magi_ffa_dsink(sargs), "magi_ffa_dsink")
     if deterministic:
         run_bwd_deterministic(args, q_ranges, k_ranges, attn_type_map,
                               hq, hk, q.device)
     else:
         run_bwd_passes(args, q.device)
-    return dq, dk, dv, None
+    return dq, dk, dv, dsink
 
 
 class FlexFlashAttnFunc(torch.autograd.Function):
@@ -342,7 +387,8 @@ class FlexFlashAttnFunc(torch.autograd.Function):
             out = out.to(torch.bfloat16)  # fp8 extension returns bf16 out
         elif out.dtype != q.dtype:
             out = out.to(q.dtype)
-        ctx.save_for_backward(q, k, v, out, lse, q_ranges, k_ranges, attn_type_map)
+        ctx.save_for_backward(q, k, v, out, lse, q_ranges, k_ranges,
+                              attn_type_map, sink)
         ctx.softmax_scale = softmax_scale
         ctx.softcap = softcap
         ctx.deterministic = deterministic
@@ -354,7 +400,8 @@ class FlexFlashAttnFunc(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dout, *_):
-        q, k, v, out, lse, q_ranges, k_ranges, attn_type_map = ctx.saved_tensors
+        (q, k, v, out, lse, q_ranges, k_ranges, attn_type_map,
+         sink) = ctx.saved_tensors
         in_dtype = q.dtype
         if in_dtype == torch.float8_e4m3fn:
             # fp8 backward runs on the bf16 kernels over upcast operands
@@ -366,8 +413,8 @@ class FlexFlashAttnFunc(torch.autograd.Function):
             v = v.to(torch.bfloat16)
             out = out.to(torch.bfloat16) if out.dtype == torch.float8_e4m3fn else out
             dout = dout.to(torch.bfloat16)
-        dq, dk, dv, _ = _flex_flash_attn_backward(
-            dout=dout, q=q, k=k, v=v, sink=None, sink_layout=ctx.sink_layout,
+        dq, dk, dv, dsink = _flex_flash_attn_backward(
+            dout=dout, q=q, k=k, v=v, sink=sink, sink_layout=ctx.sink_layout,
             out=out, lse=lse, dq=None, dk=None, dv=None, dsink=None,
             q_ranges=q_ranges, k_ranges=k_ranges, attn_type_map=attn_type_map,
             softmax_scale=ctx.softmax_scale, softcap=ctx.softcap,
@@ -379,7 +426,9 @@ class FlexFlashAttnFunc(torch.autograd.Function):
         dq = dq.to(in_dtype)
         dk = dk.to(in_dtype)
         dv = dv.to(in_dtype)
-        return (dq, dk, dv) + (None,) * 25
+        if sink is not None and dsink is not None:
+            dsink = dsink.to(sink.dtype)
+        return (dq, dk, dv, dsink) + (None,) * 24
 
 
 def flex_flash_attn_func(

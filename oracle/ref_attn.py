@@ -77,6 +77,8 @@ def ref_attn(
     softcap: float = 0.0,
     high_precision: bool = True,
     p_dtype: torch.dtype | None = None,
+    sink: torch.Tensor | None = None,
+    sink_layout: str = "sh",
 ) -> tuple[torch.Tensor, torch.Tensor]:
     """Explicit-softmax attention; returns (out [tq,hq,d] in q.dtype,
     lse [tq,hq] fp32, natural log). GQA: hq must be a multiple of hk.
@@ -100,12 +102,26 @@ def ref_attn(
     if softcap > 0.0:
         s = softcap * torch.tanh(s / softcap)
     s = torch.where(mask.unsqueeze(0), s, torch.full_like(s, float("-inf")))
+    if sink is not None:
+        # attention-sink logits participate in the softmax denominator as
+        # value-less extra columns — no softmax_scale, no softcap (reference
+        # flash_fwd_postprocess_kernel.h: lse_sink folded after the mainloop)
+        sf = sink.to(dt)
+        if sink_layout == "sh":      # [s_sink, hq]
+            extra = sf.permute(1, 0).unsqueeze(1).expand(hq, tq, sf.shape[0])
+        elif sink_layout == "ssh":   # [tq, s_sink, hq]
+            extra = sf.permute(2, 0, 1)
+        else:
+            raise ValueError(f"unsupported sink_layout {sink_layout}")
+        s = torch.cat([s, extra], dim=-1)
     lse = torch.logsumexp(s, dim=-1)  # [hq, tq]; -inf for empty rows
     p = torch.exp(s - lse.unsqueeze(-1))
     p = torch.nan_to_num(p, nan=0.0)  # empty rows: -inf - -inf = nan -> 0
     if p_dtype is not None:
         # emulate a kernel that quantises P before the PV matmul (bf16 MFMA)
         p = p.to(p_dtype).to(p.dtype)
+    if sink is not None:
+        p = p[..., : kf.shape[1]]    # sink columns carry no value rows
     out = torch.matmul(p, vf)  # [hq, tq, d]
     return (
         out.permute(1, 0, 2).to(q.dtype),
@@ -115,17 +131,22 @@ def ref_attn(
 
 def ref_attn_with_grads(
     q, k, v, mask, dout, softmax_scale=None, softcap: float = 0.0,
-    high_precision: bool = True, p_dtype=None,
+    high_precision: bool = True, p_dtype=None, sink=None, sink_layout="sh",
 ):
-    """Forward + backward through autograd. Returns (out, lse, dq, dk, dv)."""
+    """Forward + backward through autograd.
+    Returns (out, lse, dq, dk, dv[, dsink if sink given])."""
     q_ = q.detach().clone().requires_grad_(True)
     k_ = k.detach().clone().requires_grad_(True)
     v_ = v.detach().clone().requires_grad_(True)
+    s_ = sink.detach().clone().requires_grad_(True) if sink is not None else None
     out, lse = ref_attn(
         q_, k_, v_, mask, softmax_scale=softmax_scale, softcap=softcap,
         high_precision=high_precision, p_dtype=p_dtype,
+        sink=s_, sink_layout=sink_layout,
     )
     out.backward(dout)
+    if sink is not None:
+        return out.detach(), lse.detach(), q_.grad, k_.grad, v_.grad, s_.grad
     return out.detach(), lse.detach(), q_.grad, k_.grad, v_.grad
 
 
