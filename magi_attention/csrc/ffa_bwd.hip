@@ -222,23 +222,23 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   // ONE shared object only: a second __shared__ array makes hipcc drain
   // vmcnt(0) before every ds_read, destroying the glds pipeline (guide §5
   // ".s-level traps" (a)).
-  // 3 buffers: 2-deep prefetch (glds group t+2 issued while tile t computes)
+  // 2 buffers x 64-row Q/dO images: one barrier per 64 q rows — the doubled
+  // compute phase covers the prefetch latency a 32-row phase could not, with
+  // HALF the barrier parking (PMC: SQ_WAIT_ANY 35-46%% at 32-row iterations).
+  constexpr int QITER = 2 * BWD_BM;
   __shared__ __attribute__((aligned(16))) char smem[
-      3 * 2 * BWD_BM * D * 2 + 3 * 2 * BWD_BM * 4];
+      2 * 2 * QITER * D * 2 + 2 * 2 * QITER * 4];
   auto lds_q = [&](int buf) -> __bf16* {
-    return (__bf16*)(smem + (2 * buf) * BWD_BM * D * 2);
+    return (__bf16*)(smem + (2 * buf) * QITER * D * 2);
   };
   auto lds_do = [&](int buf) -> __bf16* {
-    return (__bf16*)(smem + (2 * buf + 1) * BWD_BM * D * 2);
+    return (__bf16*)(smem + (2 * buf + 1) * QITER * D * 2);
   };
-  // lse / dpsum of the q tile, staged by LDS-DMA in the SAME vmcnt group as
-  // the row images (one dword per lane: lanes 0-31 lse, 32-63 dpsum) so the
-  // per-iteration vector-memory count stays constant
+  // lse / dpsum, staged by LDS-DMA (one dword per lane: lanes 0-31 lse,
+  // 32-63 dpsum), one call per 32-row group: layout per buffer is
+  // [lse g0 (32) | dps g0 (32) | lse g1 (32) | dps g1 (32)]
   auto lds_lse = [&](int buf) -> float* {
-    return (float*)(smem + 6 * BWD_BM * D * 2 + buf * 2 * BWD_BM * 4);
-  };
-  auto lds_dps = [&](int buf) -> float* {
-    return lds_lse(buf) + BWD_BM;
+    return (float*)(smem + 4 * QITER * D * 2 + buf * 2 * QITER * 4);
   };
 
   // K/V fragments (A-layout) + K B-fragments, loaded once per block
@@ -276,11 +276,11 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   // each wave's glds covers 4 rows (64 lanes x 16B = 1 KiB = 4 rows at D=128);
   // wave w owns rows [8w, 8w+8) via 2 issues per tensor
   constexpr int ROWS_PER_GLDS = 1024 / ROWB;
-  constexpr int GLDS_PER_WAVE = (BWD_BM / BWD_WAVES) / ROWS_PER_GLDS;
+  constexpr int GLDS_PER_WAVE = (QITER / BWD_WAVES) / ROWS_PER_GLDS;
   auto stage_glds = [&](int buf, int m0x) {
 #pragma unroll
     for (int gi = 0; gi < GLDS_PER_WAVE; ++gi) {
-      const int r0 = (BWD_BM / BWD_WAVES) * wave + ROWS_PER_GLDS * gi;
+      const int r0 = (QITER / BWD_WAVES) * wave + ROWS_PER_GLDS * gi;
       const int r = r0 + lane / (ROWB / 16);
       const int c = lane % (ROWB / 16);
       const int qrow = min(m0x + r, qe - 1);
@@ -296,31 +296,36 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
           (__attribute__((address_space(3))) unsigned int*)&lds_do(buf)[r0 * D],
           16, 0, 0);
     }
-    // lse (lanes 0-31) / dpsum (lanes 32-63) of the same q tile via LDS-DMA
-    {
-      const int qr = min(m0x + lo32, qe - 1);
+    // lse (lanes 0-31) / dpsum (lanes 32-63), one LDS-DMA per 32-row group
+#pragma unroll
+    for (int sg = 0; sg < 2; ++sg) {
+      const int qr = min(m0x + sg * BWD_BM + lo32, qe - 1);
       const float* src = (lane < 32) ? p.lse + (size_t)qr * p.hq + h
                                      : p.dpsum + (size_t)qr * p.hq + h;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)src,
-          (__attribute__((address_space(3))) unsigned int*)lds_lse(buf),
+          (__attribute__((address_space(3))) unsigned int*)(
+              lds_lse(buf) + sg * 2 * BWD_BM),
           4, 0, 0);
     }
   };
-  // vector-memory ops per stage_glds call (the pipe_barrier distance)
-  constexpr int PIPE_VM = 2 * GLDS_PER_WAVE + 1;
 
   int cur = 0;
-  if (q_lo < q_hi) {
-    stage_glds(0, q_lo);
-    stage_glds(1, q_lo + BWD_BM);  // clamped rows make overrun safe
-  }
+  if (q_lo < q_hi) stage_glds(0, q_lo);
 
-  for (int m0 = q_lo; m0 < q_hi; m0 += BWD_BM) {
-    pipe_barrier<PIPE_VM>();  // buf[cur] landed; buf[(cur+1)%3] in flight
-    stage_glds(cur == 0 ? 2 : cur - 1, m0 + 2 * BWD_BM);
+  for (int m0 = q_lo; m0 < q_hi; m0 += QITER) {
+    pipe_barrier<0>();  // buf[cur] ready (prefetch had a full 64-row phase)
+    if (m0 + QITER < q_hi) stage_glds(cur ^ 1, m0 + QITER);
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+    const int ms = m0 + sub * BWD_BM;
+    if (ms >= q_hi) break;  // q_hi is block-uniform
+    const __bf16* lqb = lds_q(cur) + sub * BWD_BM * D;
+    const __bf16* ldb = lds_do(cur) + sub * BWD_BM * D;
+    const float* lse_t = lds_lse(cur) + sub * 2 * BWD_BM;
+    const float* dps_t = lse_t + BWD_BM;
 
-    if (wave_live && m0 + BWD_BM > wq_lo && m0 < wq_hi) {
+    if (wave_live && ms + BWD_BM > wq_lo && ms < wq_hi) {
       // ---- S = Q K^T ; dP = dO V^T, UN-swapped: C layout [q=crow][k=lo32],
       // so the dV/dK A-fragments come from the in-register permlane transform
       // (cframe) instead of an LDS round-trip ----
@@ -328,10 +333,10 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
 #pragma unroll
       for (int dd = 0; dd < DF; ++dd) {
         const int off = swz(lo32, lo32 * ROWB + dd * 32 + hi * 16);
-        bf16x8 qf = *(const bf16x8*)((const char*)lds_q(cur) + off);
+        bf16x8 qf = *(const bf16x8*)((const char*)lqb + off);
         s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kfA[dd], s, 0, 0, 0);
         if constexpr (WANT_DK) {
-          bf16x8 dof = *(const bf16x8*)((const char*)lds_do(cur) + off);
+          bf16x8 dof = *(const bf16x8*)((const char*)ldb + off);
           dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vfA[dd], dp, 0, 0, 0);
         }
       }
@@ -343,14 +348,14 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
       float* pv = (float*)&s;
       float* dsv = (float*)&dp;
       const bool interior =
-          (m0 + BWD_BM <= wq_hi) && (m0 >= qs) && (n0 + BWD_BN <= ke) &&
-          !((atype == 1 || atype == 3) && (n0 + BWD_BN - 1 > m0 + (ke - qe))) &&
-          !((atype == 2 || atype == 3) && (n0 < m0 + BWD_BM - 1 + (ks - qs)));
+          (ms + BWD_BM <= wq_hi) && (ms >= qs) && (n0 + BWD_BN <= ke) &&
+          !((atype == 1 || atype == 3) && (n0 + BWD_BN - 1 > ms + (ke - qe))) &&
+          !((atype == 2 || atype == 3) && (n0 < ms + BWD_BM - 1 + (ks - qs)));
       bool all_live = interior;
       if (interior) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          const float lq = lds_lse(cur)[crow(r, hi)];
+          const float lq = lse_t[crow(r, hi)];
           all_live = all_live && (lq != INFINITY) && (lq != -INFINITY);
         }
       }
@@ -358,16 +363,16 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int rl = crow(r, hi);
-          const float pij = exp2f(s[r] * sl2 - lds_lse(cur)[rl] * log2e);
+          const float pij = exp2f(s[r] * sl2 - lse_t[rl] * log2e);
           if constexpr (WANT_DV) pv[r] = pij;
           if constexpr (WANT_DK)
-            dsv[r] = pij * (dp[r] - lds_dps(cur)[rl]) * p.scale;
+            dsv[r] = pij * (dp[r] - dps_t[rl]) * p.scale;
         }
       } else {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          const int qrow = m0 + crow(r, hi);
-          const float lq = (qrow < qe) ? lds_lse(cur)[crow(r, hi)] : INFINITY;
+          const int qrow = ms + crow(r, hi);
+          const float lq = (qrow < qe) ? lse_t[crow(r, hi)] : INFINITY;
           bool ok = (qrow < wq_hi) && (qrow >= qs) && lq != INFINITY &&
                     lq != -INFINITY && kk < ke;
           if (atype == 1 || atype == 3) ok = ok && (kk - qrow <= ke - qe);
@@ -385,7 +390,7 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
           const float pij = ok ? exp2f(t - lq * log2e) : 0.f;
           if constexpr (WANT_DV) pv[r] = pij;
           if constexpr (WANT_DK)
-            dsv[r] = pij * (dp[r] - lds_dps(cur)[crow(r, hi)]) * dscale;
+            dsv[r] = pij * (dp[r] - dps_t[crow(r, hi)]) * dscale;
         }
       }
 
@@ -410,9 +415,9 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
         const int row0 = 8 * hi + jrow;          // rd = 0 rows
         const int row1 = 8 * hi + 4 + jrow;      // rd = 1 rows
         const int q_base = (int)(unsigned long long)(
-            (__attribute__((address_space(3))) char*)lds_q(cur));
+            (__attribute__((address_space(3))) const char*)lqb);
         const int do_base = (int)(unsigned long long)(
-            (__attribute__((address_space(3))) char*)lds_do(cur));
+            (__attribute__((address_space(3))) const char*)ldb);
         const int lane8 = (lane & 3) * 8;
         const int sw0 = (row0 & SW32M) << 5;
         const int sw1 = (row1 & SW32M) << 5;
@@ -453,7 +458,8 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
         }
       }
     }
-    cur = cur == 2 ? 0 : cur + 1;
+    }  // sub
+    cur ^= 1;
   }
 
   // ---- write dK/dV ----
@@ -526,14 +532,17 @@ __global__ __launch_bounds__(256, 2) void ffa_bwd_dq_kernel(BwdParams p) {
   const size_t k_pitch = (size_t)p.hk * D;
   const size_t q_pitch = (size_t)p.hq * D;
 
-  // single shared object (glds-pipeline trap, see dkv kernel); 3 buffers for
-  // the 2-deep constant-distance prefetch pipeline
-  __shared__ __attribute__((aligned(16))) char smem[3 * 2 * BWD_BN * D * 2];
+  // single shared object (glds-pipeline trap, see dkv kernel).
+  // 2 buffers x 64-row K/V images: one barrier per 64 k rows — the doubled
+  // compute phase covers the prefetch latency a 32-row phase could not, with
+  // HALF the barrier parking (PMC: SQ_WAIT_ANY 36% at 32-row iterations).
+  constexpr int KITER = 2 * BWD_BN;
+  __shared__ __attribute__((aligned(16))) char smem[2 * 2 * KITER * D * 2];
   auto lds_k = [&](int buf) -> __bf16* {
-    return (__bf16*)(smem + (2 * buf) * BWD_BN * D * 2);
+    return (__bf16*)(smem + (2 * buf) * KITER * D * 2);
   };
   auto lds_v = [&](int buf) -> __bf16* {
-    return (__bf16*)(smem + (2 * buf + 1) * BWD_BN * D * 2);
+    return (__bf16*)(smem + (2 * buf + 1) * KITER * D * 2);
   };
 
   // persistent per-wave operands: Q and dO fragments (B-layout rows)
@@ -566,11 +575,11 @@ __global__ __launch_bounds__(256, 2) void ffa_bwd_dq_kernel(BwdParams p) {
   for (int dt = 0; dt < DT; ++dt) acc_dq[dt] = (f32x16)(0.f);
 
   constexpr int ROWS_PER_GLDS = 1024 / ROWB;
-  constexpr int GLDS_PER_WAVE = (BWD_BN / BWD_WAVES) / ROWS_PER_GLDS;
+  constexpr int GLDS_PER_WAVE = (KITER / BWD_WAVES) / ROWS_PER_GLDS;
   auto stage_glds = [&](int buf, int n0x) {
 #pragma unroll
     for (int gi = 0; gi < GLDS_PER_WAVE; ++gi) {
-      const int r0 = (BWD_BN / BWD_WAVES) * wave + ROWS_PER_GLDS * gi;
+      const int r0 = (KITER / BWD_WAVES) * wave + ROWS_PER_GLDS * gi;
       const int r = r0 + lane / (ROWB / 16);
       const int c = lane % (ROWB / 16);
       const int kr = min(n0x + r, ke - 1);
@@ -588,34 +597,36 @@ __global__ __launch_bounds__(256, 2) void ffa_bwd_dq_kernel(BwdParams p) {
     }
   };
 
-  constexpr int PIPE_VM = 2 * GLDS_PER_WAVE;
   int cur = 0;
-  if (k_lo < k_hi) {
-    stage_glds(0, k_lo);
-    stage_glds(1, k_lo + BWD_BN);  // clamped rows make overrun safe
-  }
+  if (k_lo < k_hi) stage_glds(0, k_lo);
 
-  for (int n0 = k_lo; n0 < k_hi; n0 += BWD_BN) {
-    pipe_barrier<PIPE_VM>();  // buf[cur] landed; buf[(cur+1)%3] in flight
-    stage_glds(cur == 0 ? 2 : cur - 1, n0 + 2 * BWD_BN);
+  for (int n0 = k_lo; n0 < k_hi; n0 += KITER) {
+    pipe_barrier<0>();  // buf[cur] ready (prefetch had a full 64-row phase)
+    if (n0 + KITER < k_hi) stage_glds(cur ^ 1, n0 + KITER);
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+    const int ns = n0 + sub * BWD_BN;
+    if (ns >= k_hi) break;  // k_hi is block-uniform
+    const __bf16* lkb = lds_k(cur) + sub * BWD_BN * D;
+    const __bf16* lvb = lds_v(cur) + sub * BWD_BN * D;
 
-    if (wave_alive(m0, qe) && n0 + BWD_BN > wk_lo && n0 < wk_hi) {
+    if (wave_alive(m0, qe) && ns + BWD_BN > wk_lo && ns < wk_hi) {
       // ---- S^T = K Q^T ; dP^T = V dO^T (K/V A-frags from LDS rows) ----
       f32x16 sA = (f32x16)(0.f), dpA = (f32x16)(0.f);
 #pragma unroll
       for (int dd = 0; dd < DF; ++dd) {
         const int off = swz(lo32, lo32 * ROWB + dd * 32 + hi * 16);
-        bf16x8 kf = *(const bf16x8*)((const char*)lds_k(cur) + off);
-        bf16x8 vf = *(const bf16x8*)((const char*)lds_v(cur) + off);
+        bf16x8 kf = *(const bf16x8*)((const char*)lkb + off);
+        bf16x8 vf = *(const bf16x8*)((const char*)lvb + off);
         sA = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[dd], sA, 0, 0, 0);
         dpA = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dof[dd], dpA, 0, 0, 0);
       }
 
       float dsv[16];
       const bool interior =
-          (qrow < qe) && row_live && (n0 >= ks) && (n0 + BWD_BN <= ke) &&
-          !((atype == 1 || atype == 3) && (n0 + BWD_BN - 1 > m0 + (ke - qe))) &&
-          !((atype == 2 || atype == 3) && (n0 < m0 + 31 + (ks - qs)));
+          (qrow < qe) && row_live && (ns >= ks) && (ns + BWD_BN <= ke) &&
+          !((atype == 1 || atype == 3) && (ns + BWD_BN - 1 > m0 + (ke - qe))) &&
+          !((atype == 2 || atype == 3) && (ns < m0 + 31 + (ks - qs)));
       if (__all(interior) && !HAS_SOFTCAP) {
         const float lsc = lse_q * log2e;
 #pragma unroll
@@ -626,7 +637,7 @@ __global__ __launch_bounds__(256, 2) void ffa_bwd_dq_kernel(BwdParams p) {
       } else {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          const int kk = n0 + crow(r, hi);
+          const int kk = ns + crow(r, hi);
           bool ok = row_live && kk < ke && kk >= ks;
           if (atype == 1 || atype == 3) ok = ok && (kk - qrow <= ke - qe);
           if (atype == 2 || atype == 3) ok = ok && (kk - qrow >= ks - qs);
@@ -656,7 +667,7 @@ __global__ __launch_bounds__(256, 2) void ffa_bwd_dq_kernel(BwdParams p) {
         const int row0 = 8 * hi + jrow;
         const int row1 = 8 * hi + 4 + jrow;
         const int k_base = (int)(unsigned long long)(
-            (__attribute__((address_space(3))) char*)lds_k(cur));
+            (__attribute__((address_space(3))) const char*)lkb);
         const int lane8 = (lane & 3) * 8;
         const int sw0 = (row0 & SW32M) << 5;
         const int sw1 = (row1 & SW32M) << 5;
@@ -680,7 +691,8 @@ __global__ __launch_bounds__(256, 2) void ffa_bwd_dq_kernel(BwdParams p) {
         }
       }
     }
-    cur = cur == 2 ? 0 : cur + 1;
+    }  // sub
+    cur ^= 1;
   }
 
   // ---- store dq once (atomicAdd: q_ranges of different slices may overlap) ----

@@ -54,21 +54,20 @@ def run_bwd_deterministic(args, q_ranges, k_ranges, attn_type_map,
 
     launch(lib.magi_ffa_bwd_dq, q_groups, (q_ranges, k_ranges), "bwd_dq[det]")
     hs = gqa if gqa > 1 else 1
-    if env.is_bwd_fused_dkv():
-        launch(lib.magi_ffa_bwd_dkv, k_groups, (q_ranges, k_ranges),
-               "bwd_dkv[det]", head_splits=hs)
-    else:
+    if env.is_bwd_split_dkv():
         launch(lib.magi_ffa_bwd_dv, k_groups, (q_ranges, k_ranges),
                "bwd_dv[det]", head_splits=hs)
         launch(lib.magi_ffa_bwd_dk, k_groups, (q_ranges, k_ranges),
                "bwd_dk[det]", head_splits=hs)
+    else:
+        launch(lib.magi_ffa_bwd_dkv, k_groups, (q_ranges, k_ranges),
+               "bwd_dkv[det]", head_splits=hs)
 
 
 def run_bwd_passes(args, device) -> None:
-    """Launch the independent backward passes (dq / dv / dk by default — each
-    fits 2 waves/SIMD, unlike the fused dkv kernel's 1 — or dq / fused-dkv
-    with MAGI_BWD_FUSED_DKV=1) on two streams so their waves co-schedule
-    across the chip."""
+    """Launch the independent backward passes (dq / fused-dkv by default;
+    dq / dv / dk with MAGI_BWD_SPLIT_DKV=1) on two streams so their waves
+    co-schedule across the chip."""
     lib = _ffa_lib.lib()
     main = torch.cuda.current_stream(device)
     side = _get_side_stream(device)
@@ -78,11 +77,11 @@ def run_bwd_passes(args, device) -> None:
     args.stream = ctypes.c_void_p(side.cuda_stream)
     check(lib.magi_ffa_bwd_dq(args), "magi_ffa_bwd_dq")
     args.stream = ctypes.c_void_p(main.cuda_stream)
-    if env.is_bwd_fused_dkv():
-        check(lib.magi_ffa_bwd_dkv(args), "magi_ffa_bwd_dkv")
-    else:
+    if env.is_bwd_split_dkv():
         check(lib.magi_ffa_bwd_dv(args), "magi_ffa_bwd_dv")
         check(lib.magi_ffa_bwd_dk(args), "magi_ffa_bwd_dk")
+    else:
+        check(lib.magi_ffa_bwd_dkv(args), "magi_ffa_bwd_dkv")
     ev2 = torch.cuda.Event()
     ev2.record(side)
     main.wait_event(ev2)
