@@ -33,6 +33,9 @@ typedef struct magi_ffa_fwd_args {
   const int32_t* attn_type_map; /* [n] or NULL (= all full) */
   int32_t* locks;           /* zeroed int32 [ceil(total_q/128) * hq]; NULL iff
                                disable_atomic_reduction */
+  float* max_logits;        /* f32 [hq] init -inf, atomic-max of the scaled
+                               (softcapped) logits per head; NULL = off
+                               (reference return_max_logits) */
   int64_t n_ranges;
   int64_t total_q;
   int64_t total_k;

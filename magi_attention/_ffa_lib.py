@@ -29,6 +29,7 @@ class MagiFfaFwdArgs(ctypes.Structure):
         ("k_ranges", ctypes.c_void_p),
         ("attn_type_map", ctypes.c_void_p),
         ("locks", ctypes.c_void_p),
+        ("max_logits", ctypes.c_void_p),
         ("n_ranges", ctypes.c_int64),
         ("total_q", ctypes.c_int64),
         ("total_k", ctypes.c_int64),

@@ -81,6 +81,7 @@ struct FwdParams {
   const int* k_ranges;
   const int* attn_type_map;
   int* locks;
+  float* max_logits;
   int hq, hk, gqa;    // gqa = hq / hk
   int head_major;     // 1: grid.x = head (XCD-affine; per-head KV fits L2)
   int n_lock_slots;
@@ -368,6 +369,16 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
   }
 
   // ======================= epilogue =======================
+  if (p.max_logits) {
+    // per-head max of the scaled (softcapped) logits; m_run is the row max
+    // in the exp2 domain -> x ln2. Empty rows are -inf and drop out.
+    float ml = m_run * 0.6931471805599453f;
+#pragma unroll
+    for (int off = 32; off; off >>= 1)
+      ml = fmaxf(ml, __shfl_xor(ml, off, 64));
+    if (lane == 0 && ml != -INFINITY)
+      unsafeAtomicMax(p.max_logits + h, ml);
+  }
   const float lse_new =
       (l_run > 0.f) ? (m_run + __log2f(l_run)) * 0.6931471805599453f : -INFINITY;
   const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
@@ -558,6 +569,7 @@ extern "C" int magi_ffa_fwd(const magi_ffa_fwd_args* a) {
   p.k_ranges = a->k_ranges;
   p.attn_type_map = a->attn_type_map;
   p.locks = a->locks;
+  p.max_logits = a->max_logits;
   p.hq = a->hq;
   p.hk = a->hk;
   p.gqa = a->hq / a->hk;

@@ -55,6 +55,7 @@ struct Fp8FwdParams {
   const int* k_ranges;
   const int* attn_type_map;
   int* locks;
+  float* max_logits;
   int hq, hk, gqa;
   float scale;
   long long total_q, total_k;
@@ -299,6 +300,14 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_fp8_kernel(Fp8FwdParams p) {
     }
   }
 
+  if (p.max_logits) {
+    float ml = m_run * 0.6931471805599453f;  // see bf16 fwd epilogue
+#pragma unroll
+    for (int off = 32; off; off >>= 1)
+      ml = fmaxf(ml, __shfl_xor(ml, off, 64));
+    if (lane == 0 && ml != -INFINITY)
+      unsafeAtomicMax(p.max_logits + h, ml);
+  }
   // ---- epilogue: fp8 sum is scaled by 2^MAX_OFFSET relative to exp2(t-m) ----
   const float lse_new = (l_run > 0.f)
                             ? (m_run - MAX_OFFSET + __log2f(l_run)) *
@@ -389,6 +398,7 @@ extern "C" int magi_ffa_fwd_fp8(const magi_ffa_fwd_args* a) {
   p.k_ranges = a->k_ranges;
   p.attn_type_map = a->attn_type_map;
   p.locks = a->locks;
+  p.max_logits = a->max_logits;
   p.hq = a->hq;
   p.hk = a->hk;
   p.gqa = a->hq / a->hk;

@@ -168,6 +168,7 @@ def _flex_flash_attn_forward(
     deterministic: bool,
     sm_margin: int,
     max_seqlen_q: Optional[int] = None,
+    max_logits: Optional[torch.Tensor] = None,
     **_unused,
 ) -> tuple[torch.Tensor, AttnForwardMeta]:
     is_fp8 = q.dtype == torch.float8_e4m3fn
@@ -203,6 +204,7 @@ def _flex_flash_attn_forward(
         q=ptr(q), k=ptr(k), v=ptr(v), out=ptr(out), lse=ptr(lse),
         q_ranges=ptr(q_ranges), k_ranges=ptr(k_ranges),
         attn_type_map=ptr(attn_type_map), locks=ptr(locks),
+        max_logits=ptr(max_logits),
         n_ranges=q_ranges.shape[0], total_q=tq, total_k=tk,
         hq=hq, hk=hk, d=d, max_seqlen_q=max_seqlen_q,
         softmax_scale=softmax_scale, softcap=softcap,
@@ -237,7 +239,7 @@ def _flex_flash_attn_forward(
         check(fwd_fn(args), "magi_ffa_fwd")
     if sink is not None:
         _apply_sink_postprocess(out, lse, sink, sink_layout, tq, hq, d)
-    return out, AttnForwardMeta(lse=lse, max_logits=None)
+    return out, AttnForwardMeta(lse=lse, max_logits=max_logits)
 
 
 def _check_sink(sink, sink_layout, tq, hq):
@@ -370,7 +372,11 @@ class FlexFlashAttnFunc(torch.autograd.Function):
         )
         assert q_ranges is not None and k_ranges is not None
         assert q_ranges.size(0) == k_ranges.size(0)
-        assert not return_max_logits, "max_logits support lands in a later round"
+        max_logits = None
+        if return_max_logits:
+            assert q.shape[1] <= 128, "num_qheads must be <= 128 (reference cap)"
+            max_logits = torch.full((q.shape[1],), float("-inf"),
+                                    dtype=torch.float32, device=q.device)
 
         out, meta = _flex_flash_attn_forward(
             q=q, k=k, v=v, sink=sink, sink_layout=sink_layout,
@@ -379,7 +385,7 @@ class FlexFlashAttnFunc(torch.autograd.Function):
             softmax_scale=softmax_scale, softcap=softcap, out_type=None,
             disable_fwd_atomic_reduction=disable_fwd_atomic_reduction,
             deterministic=deterministic, sm_margin=sm_margin,
-            max_seqlen_q=max_seqlen_q,
+            max_seqlen_q=max_seqlen_q, max_logits=max_logits,
         )
         lse = meta.lse
         if q.dtype == torch.float8_e4m3fn:
@@ -395,7 +401,9 @@ class FlexFlashAttnFunc(torch.autograd.Function):
         ctx.sink_layout = sink_layout
         # avoid a per-backward device sync to size the bwd grid
         ctx.max_seqlen_k = max_seqlen_k
-        return out, lse, None
+        if max_logits is not None:
+            ctx.mark_non_differentiable(max_logits)
+        return out, lse, max_logits
 
     @staticmethod
     def backward(ctx, dout, *_):

@@ -363,3 +363,32 @@ def test_sink_fwd_bwd(layout):
         floor = 2e-2 if name == "dsink" and layout == "sh" else 3e-3
         assert_close_to_ref(gv.cpu().float(), ghi.float(), glo.float(),
                             f"sink:{name}", floor=floor)
+
+
+@requires_gpu
+def test_max_logits():
+    """return_max_logits (reference flex_flash_attn.py:397-408,
+    forward_meta.py:28): per-head max of the scaled logits over all allowed
+    (row, key) pairs."""
+    from magi_attention.functional import flex_flash_attn_func
+
+    tq = tk = 320
+    hq, hk, d = 4, 2, 128
+    q, k, v, dout, qr, kr, tm = make_flex_case(
+        tq, tk, hq, hk, d, [[0, 192], [192, 320]], [[0, 256], [64, 320]],
+        [1, 3], seed=5
+    )
+    out, meta = flex_flash_attn_func(q, k, v, qr, kr, tm,
+                                     return_max_logits=True)
+    torch.cuda.synchronize()
+    assert meta.max_logits is not None and meta.max_logits.shape == (hq,)
+
+    mask = make_attn_mask(tq, tk, [[0, 192], [192, 320]],
+                          [[0, 256], [64, 320]], [1, 3])
+    qc = q.detach().cpu().double().permute(1, 0, 2)
+    kc = k.detach().cpu().double().repeat_interleave(hq // hk, dim=1)
+    s = torch.matmul(qc, kc.permute(1, 0, 2).transpose(-1, -2)) * d ** -0.5
+    s = torch.where(mask.unsqueeze(0), s, torch.full_like(s, float("-inf")))
+    ref = s.amax(dim=(-1, -2)).float()
+    torch.testing.assert_close(meta.max_logits.cpu(), ref, atol=2e-2,
+                               rtol=2e-2)
