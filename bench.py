@@ -109,7 +109,7 @@ def run_bench(args):
     del q, k, v
 
     def step():
-        out, lse = calc_attn(ql, kl, vl, key)
+        out, meta = calc_attn(ql, kl, vl, key)
         out.backward(dout)
         ql.grad = kl.grad = vl.grad = None
 
@@ -143,7 +143,7 @@ def measure_roofline(state, device):
     mgr = dist_attn_runtime_dict_mgr[key]
     rt = mgr.runtime
     # cp=1: host_arg covers the whole local mask; one fwd launch + one bwd launch
-    out, lse = rt.attn_fwd(ql.detach(), kl.detach(), vl.detach())
+    out, lse, _ = rt.attn_fwd(ql.detach(), kl.detach(), vl.detach())
     torch.cuda.synchronize()
 
     local_area = sum(

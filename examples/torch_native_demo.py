@@ -56,7 +56,7 @@ def main():
     xl = dispatch(x, key)                       # local permuted shard
     q = (xl.reshape(-1, hq * hd) @ wq).reshape(-1, hq, hd)
     kv = xl.reshape(-1, hq, hd)[:, :hkv]
-    out, lse = calc_attn(q, kv.contiguous(), kv.contiguous(), key)
+    out, meta = calc_attn(q, kv.contiguous(), kv.contiguous(), key)
     loss = out.float().square().mean()
     loss.backward()
     full_out = undispatch(out, key)

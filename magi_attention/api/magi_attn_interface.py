@@ -109,7 +109,23 @@ def magi_attn_flex_key(
     chunk_size: Optional[int] = None,
 ) -> DistAttnRuntimeKey:
     """The most flexible key-creation interface (reference :440)."""
-    assert is_same_source, "cross-attn (is_same_source=False) lands in a later round"
+    # cross-attn settings: the reference raises NotImplementedError for every
+    # is_same_source=False case (_make_dispatch_meta.py:203-214) — mirrored
+    if not is_same_source:
+        if is_q_permutable and not is_k_permutable:
+            raise NotImplementedError(
+                "A cross-attn setting for encoder-decoder transformer like T5."
+            )
+        if not is_q_permutable and is_k_permutable:
+            raise NotImplementedError(
+                "A cross-attn setting for multi-modal transformer "
+                "with external encoders."
+            )
+        if is_q_permutable and is_k_permutable:
+            raise NotImplementedError(
+                "An unknown case as a pure cross-attn setting."
+            )
+        raise NotImplementedError("A trivial case with no need to dispatch.")
     assert total_seqlen_q == total_seqlen_k, "self-attn requires equal seqlens"
     group, mesh_groups = _resolve_group(cp_group_or_mesh)
     cp_size = dist.get_world_size(group)
@@ -191,12 +207,22 @@ def undispatch(x_local: torch.Tensor, key: DistAttnRuntimeKey) -> torch.Tensor:
 
 
 def calc_attn(
-    q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, key: DistAttnRuntimeKey
-) -> Tuple[torch.Tensor, torch.Tensor]:
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    key: DistAttnRuntimeKey,
+    sink: Optional[torch.Tensor] = None,
+    softmax_scale: Optional[float] = None,
+    softcap: float = 0.0,
+    return_max_logits: bool = False,
+):
     """Distributed flex-flash-attention on dispatched shards (reference :1041).
-    Returns (out, lse)."""
+    Returns (out, AttnForwardMeta(lse, max_logits)). `sink` is the replicated
+    [seqlen_sink, num_heads_q] sink tensor; `max_logits` is all-reduced MAX
+    across the cp group (forward_meta.py:28)."""
     mgr = dist_attn_runtime_dict_mgr[key]
-    return mgr.calc_attn(q, k, v)
+    return mgr.calc_attn(q, k, v, sink=sink, softmax_scale=softmax_scale,
+                         softcap=softcap, return_max_logits=return_max_logits)
 
 
 def magi_attn_flex_dispatch(x, *args, **kwargs):

@@ -132,8 +132,12 @@ class DistAttnRuntimeMgr:
 
     undispatch_kv = undispatch_qo
 
-    def calc_attn(self, q, k, v) -> Tuple[torch.Tensor, torch.Tensor]:
-        return dist_attn_func(q, k, v, self.runtime)
+    def calc_attn(self, q, k, v, sink=None, softmax_scale=None,
+                  softcap: float = 0.0, return_max_logits: bool = False):
+        return dist_attn_func(
+            q, k, v, self.runtime, sink=sink, softmax_scale=softmax_scale,
+            softcap=softcap, return_max_logits=return_max_logits,
+        )
 
     def get_position_ids(self, device=None) -> torch.Tensor:
         """Local row -> global (unpadded) position id

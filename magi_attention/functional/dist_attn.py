@@ -17,6 +17,7 @@ import torch
 import torch.distributed as dist
 
 from .. import env
+from ..common.forward_meta import AttnForwardMeta
 from ..comm.primitive import (
     WorkWithPostProcessFn,
     group_cast,
@@ -77,10 +78,14 @@ class DistAttnRuntime:
 
     # ---------------- forward ----------------
     def attn_fwd(
-        self, q: torch.Tensor, k: torch.Tensor, v: torch.Tensor
-    ) -> Tuple[torch.Tensor, torch.Tensor]:
+        self, q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+        sink: Optional[torch.Tensor] = None,
+        softmax_scale: Optional[float] = None,
+        softcap: float = 0.0,
+        return_max_logits: bool = False,
+    ) -> Tuple[torch.Tensor, torch.Tensor, Optional[torch.Tensor]]:
         tq, hq, d = q.shape
-        scale = self.softmax_scale or d ** (-0.5)
+        scale = softmax_scale or self.softmax_scale or d ** (-0.5)
         kv_local = torch.cat([k, v], dim=0)
 
         # pre-issue ALL remote-stage group-casts (reference dist_attn.py:419-436)
@@ -92,7 +97,12 @@ class DistAttnRuntime:
         lse_acc = torch.full(
             (tq, hq), float("-inf"), dtype=torch.float32, device=q.device
         )
-        self._fwd_partial(q, k, v, self.calc_meta.host_arg, out_acc, lse_acc, scale)
+        max_logits = None
+        if return_max_logits:
+            max_logits = torch.full((hq,), float("-inf"), dtype=torch.float32,
+                                    device=q.device)
+        self._fwd_partial(q, k, v, self.calc_meta.host_arg, out_acc, lse_acc,
+                          scale, softcap, max_logits)
         for s in range(self.overlap_degree):
             arg = self.calc_meta.stage_args[s]
             stage_kv = works[s].wait_post_process()
@@ -100,16 +110,44 @@ class DistAttnRuntime:
             if S == 0 or arg.is_empty():
                 continue
             self._fwd_partial(
-                q, stage_kv[:S], stage_kv[S:], arg, out_acc, lse_acc, scale
+                q, stage_kv[:S], stage_kv[S:], arg, out_acc, lse_acc, scale,
+                softcap, max_logits,
             )
+        if sink is not None:
+            # fold the replicated sink ONCE into this rank's (out, lse) — the
+            # q rows are rank-disjoint, so per-rank postprocess is exact
+            # (reference dist_attn.py: calc_lse_sink_compiled path)
+            self._sink_post(out_acc, lse_acc, sink)
+        if max_logits is not None:
+            dist.all_reduce(max_logits, op=dist.ReduceOp.MAX,
+                            group=self.cp_group)
         out = out_acc.to(q.dtype)
-        return out, lse_acc
+        return out, lse_acc, max_logits
 
-    def _fwd_partial(self, q, k, v, arg: AttnArg, out_acc, lse_acc, scale):
+    def _sink_post(self, out_acc, lse_acc, sink):
+        if _test_attn_backend is not None and not out_acc.is_cuda:
+            lse_sink = torch.logsumexp(sink.double(), dim=0)  # [hq]
+            lse_old = lse_acc.double()
+            lse_new = torch.logaddexp(lse_old, lse_sink.expand_as(lse_old))
+            w = torch.where(
+                lse_old == float("-inf"), torch.zeros_like(lse_old),
+                torch.exp(lse_old - lse_new),
+            )
+            out_acc.mul_(w.unsqueeze(-1).to(out_acc.dtype))
+            lse_acc.copy_(lse_new.to(lse_acc.dtype))
+            return
+        from .flex_flash_attn import _apply_sink_postprocess
+
+        tq, hq, d = out_acc.shape
+        _apply_sink_postprocess(out_acc, lse_acc, sink, "sh", tq, hq, d)
+
+    def _fwd_partial(self, q, k, v, arg: AttnArg, out_acc, lse_acc, scale,
+                     softcap=0.0, max_logits=None):
         if arg.is_empty():
             return
         if _test_attn_backend is not None and not q.is_cuda:
-            _test_attn_backend.fwd_partial(q, k, v, arg, out_acc, lse_acc, scale)
+            _test_attn_backend.fwd_partial(q, k, v, arg, out_acc, lse_acc,
+                                           scale, softcap, max_logits)
             return
         from .flex_flash_attn import _flex_flash_attn_forward
 
@@ -118,11 +156,12 @@ class DistAttnRuntime:
             q=q, k=k, v=v, sink=None, sink_layout="sh",
             out=out_acc, lse=lse_acc,
             q_ranges=qr, k_ranges=kr, attn_type_map=tm,
-            softmax_scale=scale, softcap=0.0, out_type=torch.float32,
+            softmax_scale=scale, softcap=softcap, out_type=torch.float32,
             disable_fwd_atomic_reduction=False,
             deterministic=env.is_deterministic_mode_enable(),
             sm_margin=env.ffa_forward_sm_margin(),
             max_seqlen_q=arg.max_seqlen_q,
+            max_logits=max_logits,
         )
 
     # ---------------- backward ----------------
@@ -134,10 +173,13 @@ class DistAttnRuntime:
         v: torch.Tensor,
         out: torch.Tensor,
         lse: torch.Tensor,
-    ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+        sink: Optional[torch.Tensor] = None,
+        softmax_scale: Optional[float] = None,
+        softcap: float = 0.0,
+    ):
         tq, hq, d = q.shape
         L = k.shape[0]
-        scale = self.softmax_scale or d ** (-0.5)
+        scale = softmax_scale or self.softmax_scale or d ** (-0.5)
         kv_local = torch.cat([k, v], dim=0)
 
         # re-fetch remote KV (reference backward:3455 -> _fetch_remote_kv:1463)
@@ -147,9 +189,12 @@ class DistAttnRuntime:
         dkv_acc = torch.zeros(2 * L, *k.shape[1:], dtype=torch.float32,
                               device=q.device)
         dpsum = self._bwd_dpsum(dout, out)
+        dsink = None
+        if sink is not None:
+            dsink = self._dsink(sink, lse, dpsum)
         self._bwd_partial(
             dout, q, k, v, out, lse, dpsum, self.calc_meta.host_arg,
-            dq_acc, dkv_acc[:L], dkv_acc[L:], scale,
+            dq_acc, dkv_acc[:L], dkv_acc[L:], scale, softcap,
         )
         rworks: List[WorkWithPostProcessFn] = []
         for s in range(self.overlap_degree):
@@ -164,15 +209,43 @@ class DistAttnRuntime:
             if not arg.is_empty():
                 self._bwd_partial(
                     dout, q, stage_kv[:S], stage_kv[S:], out, lse, dpsum, arg,
-                    dq_acc, dkv_stage[:S], dkv_stage[S:], scale,
+                    dq_acc, dkv_stage[:S], dkv_stage[S:], scale, softcap,
                 )
             rworks.append(self._reduce(dkv_stage, dkv_acc, s))
         for w in rworks:
             w.wait_post_process()
+        if dsink is not None:
+            # "sh" sink is replicated; its gradient sums over ALL q rows
+            dist.all_reduce(dsink, op=dist.ReduceOp.SUM, group=self.cp_group)
         dq = dq_acc.to(q.dtype)
         dk = dkv_acc[:L].to(k.dtype)
         dv = dkv_acc[L:].to(v.dtype)
-        return dq, dk, dv
+        return dq, dk, dv, dsink
+
+    def _dsink(self, sink, lse, dpsum):
+        if _test_attn_backend is not None and not lse.is_cuda:
+            live = torch.isfinite(lse.double())
+            # p[s, h, t] = exp(sink[s, h] - lse[t, h])
+            p = torch.exp(
+                sink.double().unsqueeze(-1) - lse.double().t().unsqueeze(0)
+            )
+            p = torch.where(live.t().unsqueeze(0), p, torch.zeros_like(p))
+            return (-(p * dpsum.double().t().unsqueeze(0)).sum(-1)).to(
+                torch.float32
+            )
+        from .. import _ffa_lib
+        from .._ffa_lib import MagiSinkArgs, check, current_stream_ptr, ptr
+
+        tq, hq = lse.shape
+        sink_f = sink.contiguous().float()
+        dsink = torch.zeros_like(sink_f)
+        sargs = MagiSinkArgs(
+            lse=ptr(lse), sink=ptr(sink_f), dsink=ptr(dsink), dpsum=ptr(dpsum),
+            total_rows=tq, n_heads=hq, d=0, s_sink=sink_f.shape[0], ssh=0,
+            stream=current_stream_ptr(),
+        )
+        check(_ffa_lib.lib().magi_ffa_dsink(sargs), "magi_ffa_dsink[dist]")
+        return dsink
 
     def _bwd_dpsum(self, dout, out):
         if _test_attn_backend is not None and not dout.is_cuda:
@@ -193,12 +266,12 @@ class DistAttnRuntime:
         return dpsum
 
     def _bwd_partial(self, dout, q, k, v, out, lse, dpsum, arg: AttnArg,
-                     dq, dk, dv, scale):
+                     dq, dk, dv, scale, softcap=0.0):
         if arg.is_empty():
             return
         if _test_attn_backend is not None and not q.is_cuda:
             _test_attn_backend.bwd_partial(
-                dout, q, k, v, out, lse, dpsum, arg, dq, dk, dv, scale
+                dout, q, k, v, out, lse, dpsum, arg, dq, dk, dv, scale, softcap
             )
             return
         from .. import _ffa_lib
@@ -216,7 +289,7 @@ class DistAttnRuntime:
             n_ranges=qr.shape[0], total_q=tq, total_k=tk,
             hq=hq, hk=hk, d=d, max_seqlen_k=max_k,
             out_is_fp32=int(out.dtype == torch.float32),
-            softmax_scale=scale, softcap=0.0,
+            softmax_scale=scale, softcap=softcap,
             cu_margin=env.ffa_backward_sm_margin(),
             stream=current_stream_ptr(),
         )
@@ -230,21 +303,42 @@ class DistAttnRuntime:
 
 class DistAttnFunc(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, runtime: DistAttnRuntime):
-        out, lse = runtime.attn_fwd(q, k, v)
-        ctx.save_for_backward(q, k, v, out, lse)
+    def forward(ctx, q, k, v, sink, runtime: DistAttnRuntime,
+                softmax_scale, softcap, return_max_logits):
+        out, lse, max_logits = runtime.attn_fwd(
+            q, k, v, sink=sink, softmax_scale=softmax_scale, softcap=softcap,
+            return_max_logits=return_max_logits,
+        )
+        ctx.save_for_backward(q, k, v, out, lse, sink)
         ctx.runtime = runtime
-        return out, lse
+        ctx.softmax_scale = softmax_scale
+        ctx.softcap = softcap
+        if max_logits is not None:
+            ctx.mark_non_differentiable(max_logits)
+        return out, lse, max_logits
 
     @staticmethod
-    def backward(ctx, dout, _dlse):
-        q, k, v, out, lse = ctx.saved_tensors
-        dq, dk, dv = ctx.runtime.attn_bwd(dout.contiguous(), q, k, v, out, lse)
-        return dq, dk, dv, None
+    def backward(ctx, dout, *_):
+        q, k, v, out, lse, sink = ctx.saved_tensors
+        dq, dk, dv, dsink = ctx.runtime.attn_bwd(
+            dout.contiguous(), q, k, v, out, lse, sink=sink,
+            softmax_scale=ctx.softmax_scale, softcap=ctx.softcap,
+        )
+        if sink is not None and dsink is not None:
+            dsink = dsink.to(sink.dtype)
+        return dq, dk, dv, dsink, None, None, None, None
 
 
 def dist_attn_func(
-    q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, runtime: DistAttnRuntime
-) -> Tuple[torch.Tensor, torch.Tensor]:
+    q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+    runtime: DistAttnRuntime,
+    sink: Optional[torch.Tensor] = None,
+    softmax_scale: Optional[float] = None,
+    softcap: float = 0.0,
+    return_max_logits: bool = False,
+):
     """Reference functional/dist_attn.py:3608."""
-    return DistAttnFunc.apply(q, k, v, runtime)
+    out, lse, max_logits = DistAttnFunc.apply(
+        q, k, v, sink, runtime, softmax_scale, softcap, return_max_logits
+    )
+    return out, AttnForwardMeta(lse=lse, max_logits=max_logits)

@@ -10,7 +10,8 @@ from oracle import make_attn_mask
 
 class OracleBackend:
     @staticmethod
-    def fwd_partial(q, k, v, arg, out_acc, lse_acc, scale):
+    def fwd_partial(q, k, v, arg, out_acc, lse_acc, scale, softcap=0.0,
+                    max_logits=None):
         L, hq, d = q.shape
         K = k.shape[0]
         hk = k.shape[1]
@@ -20,7 +21,12 @@ class OracleBackend:
         kf = k.double().repeat_interleave(g, 1).permute(1, 0, 2)
         vf = v.double().repeat_interleave(g, 1).permute(1, 0, 2)
         s = qf @ kf.transpose(-1, -2) * scale
+        if softcap > 0.0:
+            s = softcap * torch.tanh(s / softcap)
         s = torch.where(mask.unsqueeze(0), s, torch.full_like(s, float("-inf")))
+        if max_logits is not None:
+            ml = s.amax(dim=(-1, -2)).float()  # [hq]
+            torch.maximum(max_logits, ml, out=max_logits)
         lse_new = torch.logsumexp(s, -1)  # [hq, L]
         p = torch.nan_to_num(torch.exp(s - lse_new.unsqueeze(-1)), nan=0.0)
         o_new = (p @ vf).permute(1, 0, 2)  # [L, hq, d]
@@ -41,7 +47,8 @@ class OracleBackend:
         lse_acc.copy_(lse_m.to(lse_acc.dtype))
 
     @staticmethod
-    def bwd_partial(dout, q, k, v, out, lse, dpsum, arg, dq, dk, dv, scale):
+    def bwd_partial(dout, q, k, v, out, lse, dpsum, arg, dq, dk, dv, scale,
+                    softcap=0.0):
         L, hq, d = q.shape
         K = k.shape[0]
         hk = k.shape[1]
@@ -52,6 +59,12 @@ class OracleBackend:
         vf = v.double().repeat_interleave(g, 1).permute(1, 0, 2)
         dof = dout.double().permute(1, 0, 2)
         s = qf @ kf.transpose(-1, -2) * scale
+        if softcap > 0.0:
+            th = torch.tanh(s / softcap)
+            s = softcap * th
+            dscale = 1.0 - th * th  # d(capped)/d(scaled) per element
+        else:
+            dscale = 1.0
         lse_t = lse.double().permute(1, 0).unsqueeze(-1)  # [hq, L, 1]
         p = torch.where(
             mask.unsqueeze(0), torch.exp(s * 1.0 - lse_t), torch.zeros_like(s)
@@ -59,7 +72,7 @@ class OracleBackend:
         p = torch.nan_to_num(p, nan=0.0, posinf=0.0)
         dp = dof @ vf.transpose(-1, -2)  # [hq, L, K]
         dps = dpsum.double().permute(1, 0).unsqueeze(-1)
-        ds = p * (dp - dps) * scale
+        ds = p * (dp - dps) * dscale * scale
         dq_p = (ds @ kf).permute(1, 0, 2)
         dk_p = (ds.transpose(-1, -2) @ qf).permute(1, 0, 2)  # [K, hq, d]
         dv_p = (p.transpose(-1, -2) @ dof).permute(1, 0, 2)

@@ -130,7 +130,7 @@ def _worker(rank, ws, port, case_name, degree, q_data):
         ql = dispatch(q, key).requires_grad_(True)
         kl = dispatch(k, key).requires_grad_(True)
         vl = dispatch(v, key).requires_grad_(True)
-        out_l, lse_l = calc_attn(ql, kl, vl, key)
+        out_l, meta_l = calc_attn(ql, kl, vl, key)
         out_full = undispatch(out_l, key)
 
         # global oracle

@@ -77,7 +77,8 @@ def test_calc_attn_cp1_gpu(case):
     ql = dispatch(q, key).requires_grad_(True)
     kl = dispatch(k, key).requires_grad_(True)
     vl = dispatch(v, key).requires_grad_(True)
-    out_l, lse_l = calc_attn(ql, kl, vl, key)
+    out_l, meta_l = calc_attn(ql, kl, vl, key)
+    lse_l = meta_l.lse
     out = undispatch(out_l, key)
     out_l.backward(dispatch(dout, key))
     torch.cuda.synchronize()

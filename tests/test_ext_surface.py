@@ -57,3 +57,26 @@ def test_product_fails_loudly_without_extension(monkeypatch):
         assert False, "expected RuntimeError"
     except RuntimeError as e:
         assert "no fallback" in str(e).lower() or "native" in str(e).lower()
+
+
+def test_cross_attn_raises_like_reference():
+    """is_same_source=False: the reference raises NotImplementedError for
+    every cross-attn case (_make_dispatch_meta.py:203-214) — so do we."""
+    import pytest as _pytest
+
+    from magi_attention.api import magi_attn_flex_key
+    from magi_attention.common.ranges import AttnRanges
+
+    for qp, kp, msg in [
+        (True, False, "encoder-decoder"),
+        (False, True, "multi-modal"),
+        (True, True, "pure cross-attn"),
+        (False, False, "trivial"),
+    ]:
+        with _pytest.raises(NotImplementedError, match=msg):
+            magi_attn_flex_key(
+                AttnRanges.from_ranges([[0, 128]]),
+                AttnRanges.from_ranges([[0, 64]]),
+                [0], 128, 64, 4, 2, 32,
+                is_same_source=False, is_q_permutable=qp, is_k_permutable=kp,
+            )
