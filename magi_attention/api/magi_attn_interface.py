@@ -72,6 +72,8 @@ def _resolve_group(cp_group_or_mesh):
     passed AND MAGI_ATTENTION_HIERARCHICAL_COMM is on (reference
     api:632 cp_mesh._flatten(), comm_meta.py:227-228 intra=dim1, inter=dim0).
     """
+    if cp_group_or_mesh is None:
+        return dist.group.WORLD, None
     if isinstance(cp_group_or_mesh, dist.ProcessGroup):
         return cp_group_or_mesh, None
     try:

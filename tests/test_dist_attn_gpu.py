@@ -110,3 +110,71 @@ def test_calc_attn_cp1_gpu(case):
         vl.grad.cpu().float(), padded(dv_hi).float()[pos],
         padded(dv_lo).float()[pos], "cp1:dv",
     )
+
+
+@requires_gpu
+def test_calc_attn_cp1_sink_max_logits():
+    """Distributed runtime at cp=1 with sink + softcap + max_logits on the
+    HIP kernel path (sink postprocess + dsink kernels under the runtime)."""
+    from magi_attention.api import (
+        calc_attn,
+        dispatch,
+        get_position_ids,
+        magi_attn_flex_key,
+        undispatch,
+    )
+    from magi_attention.common.ranges import AttnRanges
+    from magi_attention.config import DispatchConfig, DistAttnConfig
+
+    pg = _init_pg()
+    total, hq, hk, d, s_sink = 1024, 4, 2, 128, 2
+    qr_l = [[0, 512], [512, 1024]]
+    kr_l = [[0, 512], [256, 1024]]
+    tt = [1, 0]
+    g = torch.Generator().manual_seed(91)
+    q = (torch.randn(total, hq, d, generator=g) * 0.5).bfloat16().cuda()
+    k = (torch.randn(total, hk, d, generator=g) * 0.5).bfloat16().cuda()
+    v = (torch.randn(total, hk, d, generator=g) * 0.5).bfloat16().cuda()
+    dout = (torch.randn(total, hq, d, generator=g) * 0.5).bfloat16().cuda()
+    sink = (torch.randn(s_sink, hq, generator=g) * 2).float().cuda()
+
+    cfg = DistAttnConfig(dispatch_config=DispatchConfig(chunk_size=256))
+    key = magi_attn_flex_key(
+        AttnRanges.from_ranges(qr_l), AttnRanges.from_ranges(kr_l), tt,
+        total, total, hq, hk, d, cp_group_or_mesh=pg, dist_attn_config=cfg,
+    )
+    ql = dispatch(q, key).requires_grad_(True)
+    kl = dispatch(k, key).requires_grad_(True)
+    vl = dispatch(v, key).requires_grad_(True)
+    sink_l = sink.clone().requires_grad_(True)
+    out_l, meta = calc_attn(ql, kl, vl, key, sink=sink_l,
+                            return_max_logits=True)
+    out_full = undispatch(out_l, key)
+    out_l.backward(dispatch(dout, key))
+    torch.cuda.synchronize()
+
+    mask = make_attn_mask(total, total, qr_l, kr_l, tt)
+    qc, kc, vc, doc = [t.cpu() for t in (q, k, v, dout)]
+    sc = sink.cpu()
+    hi = ref_attn_with_grads(qc, kc, vc, mask, doc, sink=sc)
+    lo = ref_attn_with_grads(qc, kc, vc, mask, doc, sink=sc,
+                             high_precision=False, p_dtype=torch.bfloat16)
+    assert_close_to_ref(out_full.cpu().float(), hi[0].float(), lo[0].float(),
+                        "cp1sink:out")
+    assert meta.max_logits is not None and meta.max_logits.shape == (hq,)
+
+    pos = get_position_ids(key).cpu()
+    pad = key.pad_size
+
+    def padded(t):
+        return torch.cat([t, torch.zeros(pad, *t.shape[1:], dtype=t.dtype)])
+
+    for got, i, name, fl in [
+        (ql.grad, 2, "dq", 3e-3), (kl.grad, 3, "dk", 3e-3),
+        (vl.grad, 4, "dv", 3e-3),
+    ]:
+        assert_close_to_ref(got.cpu().float(), padded(hi[i]).float()[pos],
+                            padded(lo[i]).float()[pos], f"cp1sink:{name}",
+                            floor=fl)
+    assert_close_to_ref(sink_l.grad.cpu(), hi[5].float(), lo[5].float(),
+                        "cp1sink:dsink", floor=2e-2)
