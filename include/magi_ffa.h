@@ -36,6 +36,9 @@ typedef struct magi_ffa_fwd_args {
   float* max_logits;        /* f32 [hq] init -inf, atomic-max of the scaled
                                (softcapped) logits per head; NULL = off
                                (reference return_max_logits) */
+  const int32_t* qk_starts; /* auto_range_merge: [n_ranges+1]; q_ranges are
+                               UNIQUE, k segments per unique q range; NULL =
+                               one k range per q range */
   int64_t n_ranges;
   int64_t total_q;
   int64_t total_k;
@@ -70,6 +73,9 @@ typedef struct magi_ffa_bwd_args {
   const int32_t* q_ranges;  /* [n, 2] */
   const int32_t* k_ranges;  /* [n, 2] */
   const int32_t* attn_type_map; /* [n] or NULL */
+  const int32_t* seg_starts;    /* auto_range_merge: [n_ranges+1] segment map
+                                   (dq: q->k segments; dkv: k->q segments);
+                                   NULL = one segment per range */
   int64_t n_ranges;
   int64_t total_q;
   int64_t total_k;

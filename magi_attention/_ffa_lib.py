@@ -30,6 +30,7 @@ class MagiFfaFwdArgs(ctypes.Structure):
         ("attn_type_map", ctypes.c_void_p),
         ("locks", ctypes.c_void_p),
         ("max_logits", ctypes.c_void_p),
+        ("qk_starts", ctypes.c_void_p),
         ("n_ranges", ctypes.c_int64),
         ("total_q", ctypes.c_int64),
         ("total_k", ctypes.c_int64),
@@ -61,6 +62,7 @@ class MagiFfaBwdArgs(ctypes.Structure):
         ("q_ranges", ctypes.c_void_p),
         ("k_ranges", ctypes.c_void_p),
         ("attn_type_map", ctypes.c_void_p),
+        ("seg_starts", ctypes.c_void_p),
         ("n_ranges", ctypes.c_int64),
         ("total_q", ctypes.c_int64),
         ("total_k", ctypes.c_int64),

@@ -130,6 +130,9 @@ struct BwdParams {
   int head_mult;     // deterministic GQA split: real head = off + idx*mult
   int head_off;
   int n_heads_launch;
+  const int* seg_starts;  // auto_range_merge: dq pass: ri = unique q range,
+                          // k segments; dkv pass: ri = unique k range, q
+                          // segments (reference bwd_kq_map). NULL = 1:1.
 };
 
 // ---------------- preprocess: dpsum = rowsum(dO * O) ----------------
@@ -191,9 +194,9 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   const int ks = p.k_ranges[2 * ri], ke = p.k_ranges[2 * ri + 1];
   const int nblk0 = ks + wb * (BWD_BN * BWD_WAVES);
   if (nblk0 >= ke) return;
-  const int qs = p.q_ranges[2 * ri], qe = p.q_ranges[2 * ri + 1];
-  if (qe <= qs) return;
-  const int atype = p.attn_type_map ? p.attn_type_map[ri] : 0;
+  const int seg0 = p.seg_starts ? p.seg_starts[ri] : ri;
+  const int seg1 = p.seg_starts ? p.seg_starts[ri + 1] : ri + 1;
+  int qs, qe, atype;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -255,16 +258,18 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
     }
   }
 
-  // block-level q loop bounds (union over the 4 waves' tiles)
-  int q_lo = qs, q_hi = qe;
+  // block + wave q loop bounds (recomputed per q segment)
   const int nlast = min(nblk0 + BWD_BN * BWD_WAVES, ke) - 1;
-  if (atype == 1 || atype == 3) q_lo = max(q_lo, nblk0 - (ke - qe));
-  if (atype == 2 || atype == 3) q_hi = min(q_hi, nlast - (ks - qs) + 1);
-  // wave-level bounds
-  int wq_lo = qs, wq_hi = qe;
-  if (atype == 1 || atype == 3) wq_lo = max(wq_lo, n0 - (ke - qe));
-  if (atype == 2 || atype == 3)
-    wq_hi = min(wq_hi, (n0 + BWD_BN - 1) - (ks - qs) + 1);
+  int q_lo = 0, q_hi = 0, wq_lo = 0, wq_hi = 0;
+  auto seg_bounds = [&]() {
+    q_lo = qs; q_hi = qe;
+    if (atype == 1 || atype == 3) q_lo = max(q_lo, nblk0 - (ke - qe));
+    if (atype == 2 || atype == 3) q_hi = min(q_hi, nlast - (ks - qs) + 1);
+    wq_lo = qs; wq_hi = qe;
+    if (atype == 1 || atype == 3) wq_lo = max(wq_lo, n0 - (ke - qe));
+    if (atype == 2 || atype == 3)
+      wq_hi = min(wq_hi, (n0 + BWD_BN - 1) - (ks - qs) + 1);
+  };
 
   f32x16 acc_dk[WANT_DK ? DT : 1], acc_dv[WANT_DV ? DT : 1];
 #pragma unroll
@@ -311,7 +316,15 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   };
 
   int cur = 0;
-  if (q_lo < q_hi) stage_glds(0, q_lo);
+  for (int seg = seg0; seg < seg1; ++seg) {
+  qs = p.q_ranges[2 * seg];
+  qe = p.q_ranges[2 * seg + 1];
+  atype = p.attn_type_map ? p.attn_type_map[seg] : 0;
+  if (qe <= qs) continue;  // uniform across block
+  seg_bounds();
+  if (q_lo >= q_hi) continue;
+  cur = 0;
+  stage_glds(0, q_lo);
 
   for (int m0 = q_lo; m0 < q_hi; m0 += QITER) {
     pipe_barrier<0>();  // buf[cur] ready (prefetch had a full 64-row phase)
@@ -461,6 +474,8 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
     }  // sub
     cur ^= 1;
   }
+  __syncthreads();  // this segment's LDS reads retired before restaging
+  }  // seg
 
   // ---- write dK/dV ----
   if (!wave_live || (p.debug_ablate & 2)) return;
@@ -507,9 +522,9 @@ __global__ __launch_bounds__(256, 2) void ffa_bwd_dq_kernel(BwdParams p) {
   const int qs = p.q_ranges[2 * ri], qe = p.q_ranges[2 * ri + 1];
   const int mblk0 = qs + wb * (BWD_BM * BWD_WAVES);
   if (mblk0 >= qe) return;
-  const int ks = p.k_ranges[2 * ri], ke = p.k_ranges[2 * ri + 1];
-  if (ke <= ks) return;
-  const int atype = p.attn_type_map ? p.attn_type_map[ri] : 0;
+  const int seg0 = p.seg_starts ? p.seg_starts[ri] : ri;
+  const int seg1 = p.seg_starts ? p.seg_starts[ri + 1] : ri + 1;
+  int ks, ke, atype;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -560,15 +575,18 @@ __global__ __launch_bounds__(256, 2) void ffa_bwd_dq_kernel(BwdParams p) {
   const float dpsum_q = qvalid ? p.dpsum[(size_t)qrow * p.hq + h] : 0.f;
   const bool row_live = qvalid && lse_q != INFINITY && lse_q != -INFINITY;
 
-  // block + wave k-loop bounds
-  int k_lo = ks, k_hi = ke;
+  // block + wave k-loop bounds (recomputed per k segment)
   const int mlast = min(mblk0 + BWD_BM * BWD_WAVES, qe) - 1;
-  if (atype == 1 || atype == 3) k_hi = min(k_hi, mlast + (ke - qe) + 1);
-  if (atype == 2 || atype == 3) k_lo = max(k_lo, mblk0 + (ks - qs));
-  int wk_lo = ks, wk_hi = ke;
-  if (atype == 1 || atype == 3)
-    wk_hi = min(wk_hi, min(m0 + BWD_BM, qe) - 1 + (ke - qe) + 1);
-  if (atype == 2 || atype == 3) wk_lo = max(wk_lo, m0 + (ks - qs));
+  int k_lo = 0, k_hi = 0, wk_lo = 0, wk_hi = 0;
+  auto seg_bounds = [&]() {
+    k_lo = ks; k_hi = ke;
+    if (atype == 1 || atype == 3) k_hi = min(k_hi, mlast + (ke - qe) + 1);
+    if (atype == 2 || atype == 3) k_lo = max(k_lo, mblk0 + (ks - qs));
+    wk_lo = ks; wk_hi = ke;
+    if (atype == 1 || atype == 3)
+      wk_hi = min(wk_hi, min(m0 + BWD_BM, qe) - 1 + (ke - qe) + 1);
+    if (atype == 2 || atype == 3) wk_lo = max(wk_lo, m0 + (ks - qs));
+  };
 
   f32x16 acc_dq[DT];
 #pragma unroll
@@ -598,7 +616,15 @@ __global__ __launch_bounds__(256, 2) void ffa_bwd_dq_kernel(BwdParams p) {
   };
 
   int cur = 0;
-  if (k_lo < k_hi) stage_glds(0, k_lo);
+  for (int seg = seg0; seg < seg1; ++seg) {
+  ks = p.k_ranges[2 * seg];
+  ke = p.k_ranges[2 * seg + 1];
+  atype = p.attn_type_map ? p.attn_type_map[seg] : 0;
+  if (ke <= ks) continue;  // uniform across block
+  seg_bounds();
+  if (k_lo >= k_hi) continue;
+  cur = 0;
+  stage_glds(0, k_lo);
 
   for (int n0 = k_lo; n0 < k_hi; n0 += KITER) {
     pipe_barrier<0>();  // buf[cur] ready (prefetch had a full 64-row phase)
@@ -694,6 +720,8 @@ __global__ __launch_bounds__(256, 2) void ffa_bwd_dq_kernel(BwdParams p) {
     }  // sub
     cur ^= 1;
   }
+  __syncthreads();  // this segment's LDS reads retired before restaging
+  }  // seg
 
   // ---- store dq once (atomicAdd: q_ranges of different slices may overlap) ----
   if (skip_dq) return;
@@ -755,6 +783,7 @@ static int fill_bwd_params(const magi_ffa_bwd_args* a, BwdParams* p) {
   p->softcap = a->softcap;
   p->total_q = a->total_q;
   p->total_k = a->total_k;
+  p->seg_starts = a->seg_starts;
   { const char* e = getenv("MAGI_BWD_ABLATE"); p->debug_ablate = e ? atoi(e) : 0; }
   // head-major pays when one head's K+V (bf16) fits a 4 MB XCD L2
   p->head_major = ((long long)a->total_k * a->d * 4 <= (4 << 20)) ? 1 : 0;

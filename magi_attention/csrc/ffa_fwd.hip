@@ -82,6 +82,9 @@ struct FwdParams {
   const int* attn_type_map;
   int* locks;
   float* max_logits;
+  const int* qk_starts;  // auto_range_merge: ri = unique q range; k segments
+                         // are k_ranges[qk_starts[ri]..qk_starts[ri+1]) with
+                         // per-segment attn types; NULL = one segment per ri
   int hq, hk, gqa;    // gqa = hq / hk
   int head_major;     // 1: grid.x = head (XCD-affine; per-head KV fits L2)
   int n_lock_slots;
@@ -103,8 +106,11 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
   const int qs = p.q_ranges[2 * ri], qe = p.q_ranges[2 * ri + 1];
   const int m0 = qs + mb * FFA_BM;
   if (m0 >= qe) return;                       // uniform across block
-  const int ks = p.k_ranges[2 * ri], ke = p.k_ranges[2 * ri + 1];
-  const int atype = p.attn_type_map ? p.attn_type_map[ri] : 0;
+  // auto_range_merge: iterate this unique q range's k segments in-kernel,
+  // online softmax carried across them (reference merge_range.cu semantics)
+  const int seg0 = p.qk_starts ? p.qk_starts[ri] : ri;
+  const int seg1 = p.qk_starts ? p.qk_starts[ri + 1] : ri + 1;
+  int ks, ke, atype;
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -122,24 +128,26 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
                                 : p.scale * 1.4426950408889634f;
   const float cap_pre = HAS_SOFTCAP ? p.scale / p.softcap : 0.f;
 
-  // per-wave k loop bounds
-  int n_lo = ks, n_hi = ke;
-  if (ke > ks && q0 < qe) {
-    const int qhiw = min(q0 + 31, qe - 1);
-    if (atype == 1 || atype == 3) n_hi = min(n_hi, qhiw + (ke - qe) + 1);
-    if (atype == 2 || atype == 3) n_lo = max(n_lo, q0 + (ks - qs));
-  } else {
-    n_hi = n_lo;  // empty
-  }
-  // block-level union of the waves' bounds (staging is cooperative)
-  int b_lo = ks, b_hi = ke;
-  if (ke > ks && m0 < qe) {
-    const int qhib = min(m0 + FFA_BM - 1, qe - 1);
-    if (atype == 1 || atype == 3) b_hi = min(b_hi, qhib + (ke - qe) + 1);
-    if (atype == 2 || atype == 3) b_lo = max(b_lo, m0 + (ks - qs));
-  } else {
-    b_hi = b_lo;
-  }
+  // per-wave / block k loop bounds (recomputed per k segment)
+  int n_lo = 0, n_hi = 0, b_lo = 0, b_hi = 0;
+  auto seg_bounds = [&]() {
+    n_lo = ks; n_hi = ke;
+    if (ke > ks && q0 < qe) {
+      const int qhiw = min(q0 + 31, qe - 1);
+      if (atype == 1 || atype == 3) n_hi = min(n_hi, qhiw + (ke - qe) + 1);
+      if (atype == 2 || atype == 3) n_lo = max(n_lo, q0 + (ks - qs));
+    } else {
+      n_hi = n_lo;  // empty
+    }
+    b_lo = ks; b_hi = ke;
+    if (ke > ks && m0 < qe) {
+      const int qhib = min(m0 + FFA_BM - 1, qe - 1);
+      if (atype == 1 || atype == 3) b_hi = min(b_hi, qhib + (ke - qe) + 1);
+      if (atype == 2 || atype == 3) b_lo = max(b_lo, m0 + (ks - qs));
+    } else {
+      b_hi = b_lo;
+    }
+  };
   constexpr int ROWB = D * 2;
   // 32-B-granular swizzle: b128 lane groups spread (<=2-way) AND every 32-B
   // run stays physically contiguous for the tr16 V reads (see bwd kernels)
@@ -209,12 +217,6 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
   };
 
   int cur = 0;
-  if (b_lo < b_hi) {
-    issue_loads(b_lo, kregA, vregA);
-    write_stage(0, kregA, vregA);
-    issue_loads(b_lo + FFA_BN, kregA, vregA);  // tile t1 (clamped if absent)
-  }
-  __syncthreads();
 
   auto iter_body = [&](int n0, auto& kreg_w, auto& vreg_w, auto& kreg_l,
                        auto& vreg_l) {
@@ -357,7 +359,20 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
     cur ^= 1;
   };
 
-  {
+  bool any_seg = false;
+  for (int seg = seg0; seg < seg1; ++seg) {
+    ks = p.k_ranges[2 * seg];
+    ke = p.k_ranges[2 * seg + 1];
+    atype = p.attn_type_map ? p.attn_type_map[seg] : 0;
+    if (ke <= ks) continue;  // uniform across block
+    seg_bounds();
+    if (b_lo >= b_hi) continue;
+    any_seg = true;
+    cur = 0;
+    issue_loads(b_lo, kregA, vregA);
+    write_stage(0, kregA, vregA);
+    issue_loads(b_lo + FFA_BN, kregA, vregA);  // tile t1 (clamped if absent)
+    __syncthreads();
     int n0 = b_lo;
     while (n0 < b_hi) {
       iter_body(n0, kregA, vregA, kregB, vregB);
@@ -366,7 +381,10 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
       iter_body(n0, kregB, vregB, kregA, vregA);
       n0 += FFA_BN;
     }
+    // LDS reads of this segment retired before the next segment's staging
+    __syncthreads();
   }
+  if (!any_seg) return;
 
   // ======================= epilogue =======================
   if (p.max_logits) {
@@ -570,6 +588,7 @@ extern "C" int magi_ffa_fwd(const magi_ffa_fwd_args* a) {
   p.attn_type_map = a->attn_type_map;
   p.locks = a->locks;
   p.max_logits = a->max_logits;
+  p.qk_starts = a->qk_starts;
   p.hq = a->hq;
   p.hk = a->hk;
   p.gqa = a->hq / a->hk;

@@ -399,6 +399,8 @@ extern "C" int magi_ffa_fwd_fp8(const magi_ffa_fwd_args* a) {
   p.attn_type_map = a->attn_type_map;
   p.locks = a->locks;
   p.max_logits = a->max_logits;
+  // fp8 fwd: auto_range_merge segments not supported yet
+  if (a->qk_starts) return -7;
   p.hq = a->hq;
   p.hk = a->hk;
   p.gqa = a->hq / a->hk;

@@ -64,19 +64,35 @@ def run_bwd_deterministic(args, q_ranges, k_ranges, attn_type_map,
                "bwd_dkv[det]", head_splits=hs)
 
 
-def run_bwd_passes(args, device) -> None:
+def run_bwd_passes(args, device, dq_tables=None, dkv_tables=None) -> None:
     """Launch the independent backward passes (dq / fused-dkv by default;
     dq / dv / dk with MAGI_BWD_SPLIT_DKV=1) on two streams so their waves
-    co-schedule across the chip."""
+    co-schedule across the chip. dq_tables / dkv_tables override the range
+    tables per pass for auto_range_merge: (ranges_outer, ranges_inner,
+    attn_type_map, seg_starts) — outer = the pass's OWN loop dim."""
     lib = _ffa_lib.lib()
     main = torch.cuda.current_stream(device)
     side = _get_side_stream(device)
     ev = torch.cuda.Event()
     ev.record(main)          # dpsum + inputs ready
     side.wait_event(ev)
+    keep = (dq_tables, dkv_tables)  # outlive the async launches
+
+    def set_tables(t, outer_is_q):
+        if t is None:
+            args.seg_starts = None
+            return
+        outer, inner, tm, starts = t
+        args.q_ranges = ptr(outer if outer_is_q else inner)
+        args.k_ranges = ptr(inner if outer_is_q else outer)
+        args.attn_type_map = ptr(tm)
+        args.seg_starts = ptr(starts)
+
     args.stream = ctypes.c_void_p(side.cuda_stream)
+    set_tables(dq_tables, True)
     check(lib.magi_ffa_bwd_dq(args), "magi_ffa_bwd_dq")
     args.stream = ctypes.c_void_p(main.cuda_stream)
+    set_tables(dkv_tables, False)
     if env.is_bwd_split_dkv():
         check(lib.magi_ffa_bwd_dv(args), "magi_ffa_bwd_dv")
         check(lib.magi_ffa_bwd_dk(args), "magi_ffa_bwd_dk")
@@ -85,6 +101,7 @@ def run_bwd_passes(args, device) -> None:
     ev2 = torch.cuda.Event()
     ev2.record(side)
     main.wait_event(ev2)
+    del keep
 
 from .. import _ffa_lib, env
 from .._ffa_lib import (
@@ -95,6 +112,38 @@ from .._ffa_lib import (
     ptr,
 )
 from ..common.forward_meta import AttnForwardMeta
+
+def merge_ranges(
+    outer_ranges: torch.Tensor,
+    inner_ranges: torch.Tensor,
+    attn_type_map: torch.Tensor,
+):
+    """Sort + deduplicate (outer, inner) attention block pairs (reference
+    flex_flash_attn.py:79 merge_ranges, built on the magi_attn_ext ops).
+    Returns (merged_outer [n,2] zero-padded, sorted_outer, sorted_inner,
+    sorted_attn_type_map, range_map inverse indices, unique_count [1])."""
+    from .. import magi_attn_ext
+
+    idx = magi_attn_ext.argsort_ranges(outer_ranges)
+    so, si, st = magi_attn_ext.reorder_ranges_and_attn_type_maps(
+        outer_ranges, inner_ranges, attn_type_map, idx
+    )
+    uniq, inverse, count = magi_attn_ext.unique_consecutive_pairs(so)
+    n = outer_ranges.shape[0]
+    merged = torch.zeros(n, 2, dtype=torch.int32, device=outer_ranges.device)
+    merged[: uniq.shape[0]] = uniq
+    return merged, so, si, st, inverse, count
+
+
+def _seg_starts(inverse: torch.Tensor, n: int) -> torch.Tensor:
+    """[n+1] cumulative segment starts from sorted inverse indices; entries
+    past the unique count repeat the total, giving EMPTY segments — so the
+    kernel grid can stay sized by n with no host sync on unique_count."""
+    counts = torch.bincount(inverse.long(), minlength=n)
+    starts = torch.zeros(n + 1, dtype=torch.int32, device=inverse.device)
+    starts[1:] = counts.cumsum(0).to(torch.int32)
+    return starts
+
 
 LOCK_GRAN = 128
 _lock_cache: dict[tuple[int, int], torch.Tensor] = {}
@@ -169,6 +218,7 @@ def _flex_flash_attn_forward(
     sm_margin: int,
     max_seqlen_q: Optional[int] = None,
     max_logits: Optional[torch.Tensor] = None,
+    qk_starts: Optional[torch.Tensor] = None,
     **_unused,
 ) -> tuple[torch.Tensor, AttnForwardMeta]:
     is_fp8 = q.dtype == torch.float8_e4m3fn
@@ -204,7 +254,7 @@ def _flex_flash_attn_forward(
         q=ptr(q), k=ptr(k), v=ptr(v), out=ptr(out), lse=ptr(lse),
         q_ranges=ptr(q_ranges), k_ranges=ptr(k_ranges),
         attn_type_map=ptr(attn_type_map), locks=ptr(locks),
-        max_logits=ptr(max_logits),
+        max_logits=ptr(max_logits), qk_starts=ptr(qk_starts),
         n_ranges=q_ranges.shape[0], total_q=tq, total_k=tk,
         hq=hq, hk=hk, d=d, max_seqlen_q=max_seqlen_q,
         softmax_scale=softmax_scale, softcap=softcap,
@@ -216,6 +266,11 @@ def _flex_flash_attn_forward(
               else _ffa_lib.lib().magi_ffa_fwd)
     if is_fp8:
         assert not disable_fwd_atomic_reduction and out_is_fp32
+    if qk_starts is not None:
+        assert not deterministic, (
+            "deterministic + auto_range_merge lands in a later round"
+        )
+        assert not is_fp8, "fp8 + auto_range_merge lands in a later round"
     if deterministic and q_ranges.shape[0] > 1:
         # fixed merge order: one launch per q-disjoint slice group.
         # NOTE: subset tensors must outlive the async kernel launches — hold
@@ -298,6 +353,7 @@ def _flex_flash_attn_backward(
     deterministic: bool,
     sm_margin: int,
     max_seqlen_k: Optional[int] = None,
+    auto_range_merge: bool = False,
     **_unused,
 ) -> tuple[torch.Tensor, torch.Tensor, torch.Tensor, Optional[torch.Tensor]]:
     dout, q, k, v, out, q_ranges, k_ranges = [
@@ -345,8 +401,27 @@ def _flex_flash_attn_backward(
         )
         check(lib.magi_ffa_dsink(sargs), "magi_ffa_dsink")
     if deterministic:
+        assert not auto_range_merge, (
+            "deterministic + auto_range_merge lands in a later round"
+        )
         run_bwd_deterministic(args, q_ranges, k_ranges, attn_type_map,
                               hq, hk, q.device)
+    elif auto_range_merge:
+        # dq pass: unique q ranges with k segments; dkv pass: unique k
+        # ranges with q segments (reference bwd_kq_map)
+        mq, _, sk_i, sq_t, inv_q, _ = merge_ranges(
+            q_ranges, k_ranges, attn_type_map
+        )
+        qk_starts = _seg_starts(inv_q, q_ranges.shape[0])
+        mk, _, sq_i, sk_t, inv_k, _ = merge_ranges(
+            k_ranges, q_ranges, attn_type_map
+        )
+        kq_starts = _seg_starts(inv_k, k_ranges.shape[0])
+        run_bwd_passes(
+            args, q.device,
+            dq_tables=(mq, sk_i, sq_t, qk_starts),
+            dkv_tables=(mk, sq_i, sk_t, kq_starts),
+        )
     else:
         run_bwd_passes(args, q.device)
     return dq, dk, dv, dsink
@@ -372,6 +447,17 @@ class FlexFlashAttnFunc(torch.autograd.Function):
         )
         assert q_ranges is not None and k_ranges is not None
         assert q_ranges.size(0) == k_ranges.size(0)
+        fwd_q_ranges, fwd_k_ranges, fwd_tm = q_ranges, k_ranges, attn_type_map
+        fwd_qk_starts = None
+        if auto_range_merge:
+            assert attn_type_map is not None, (
+                "auto_range_merge requires an explicit attn_type_map"
+            )
+            mq, _, sk_i, st, inv_q, _ = merge_ranges(
+                q_ranges, k_ranges, attn_type_map
+            )
+            fwd_q_ranges, fwd_k_ranges, fwd_tm = mq, sk_i, st
+            fwd_qk_starts = _seg_starts(inv_q, q_ranges.shape[0])
         max_logits = None
         if return_max_logits:
             assert q.shape[1] <= 128, "num_qheads must be <= 128 (reference cap)"
@@ -381,11 +467,13 @@ class FlexFlashAttnFunc(torch.autograd.Function):
         out, meta = _flex_flash_attn_forward(
             q=q, k=k, v=v, sink=sink, sink_layout=sink_layout,
             out=None, lse=None,
-            q_ranges=q_ranges, k_ranges=k_ranges, attn_type_map=attn_type_map,
+            q_ranges=fwd_q_ranges, k_ranges=fwd_k_ranges,
+            attn_type_map=fwd_tm,
             softmax_scale=softmax_scale, softcap=softcap, out_type=None,
             disable_fwd_atomic_reduction=disable_fwd_atomic_reduction,
             deterministic=deterministic, sm_margin=sm_margin,
             max_seqlen_q=max_seqlen_q, max_logits=max_logits,
+            qk_starts=fwd_qk_starts,
         )
         lse = meta.lse
         if q.dtype == torch.float8_e4m3fn:
@@ -399,6 +487,7 @@ class FlexFlashAttnFunc(torch.autograd.Function):
         ctx.deterministic = deterministic
         ctx.sm_margin = sm_margin
         ctx.sink_layout = sink_layout
+        ctx.auto_range_merge = auto_range_merge
         # avoid a per-backward device sync to size the bwd grid
         ctx.max_seqlen_k = max_seqlen_k
         if max_logits is not None:
@@ -429,6 +518,7 @@ class FlexFlashAttnFunc(torch.autograd.Function):
             disable_bwd_dkv_atomic_reduction=False,
             deterministic=ctx.deterministic, sm_margin=ctx.sm_margin,
             max_seqlen_k=ctx.max_seqlen_k,
+            auto_range_merge=ctx.auto_range_merge,
         )
         dq = dq.to(in_dtype)
         dk = dk.to(in_dtype)

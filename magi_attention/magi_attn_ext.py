@@ -61,9 +61,11 @@ class KernelBarrier:
 
 
 def argsort_ranges(ranges: torch.Tensor) -> torch.Tensor:
-    """Indices sorting [N,2] int32 ranges by start
-    (reference extensions/sort_and_reorder_ranges.cu)."""
-    return torch.argsort(ranges[:, 0].long(), stable=True).to(torch.int32)
+    """Indices sorting [N,2] int32 ranges lexicographically by (start, end)
+    (reference extensions/sort_and_reorder_ranges.cu) — identical pairs end
+    up adjacent, as unique_consecutive_pairs requires."""
+    key = ranges[:, 0].long() * (1 << 31) + ranges[:, 1].long()
+    return torch.argsort(key, stable=True).to(torch.int32)
 
 
 def reorder_ranges_and_attn_type_maps(

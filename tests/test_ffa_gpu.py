@@ -392,3 +392,54 @@ def test_max_logits():
     ref = s.amax(dim=(-1, -2)).float()
     torch.testing.assert_close(meta.max_logits.cpu(), ref, atol=2e-2,
                                rtol=2e-2)
+
+
+@requires_gpu
+def test_auto_range_merge():
+    """auto_range_merge (reference flex_flash_attn.py:79 merge_ranges +
+    mainloop qk_map): repeated q ranges iterate their k segments IN-KERNEL
+    with the online softmax carried across, instead of per-pair launches with
+    lock-merge. Must match both the oracle and the merge=False path."""
+    from magi_attention.functional import flex_flash_attn_func
+    from oracle import ref_attn_with_grads
+
+    tq = tk = 1024
+    hq, hk, d = 4, 2, 128
+    # block-sparse: q [0,256) attends 3 scattered k blocks; q [256,640) two;
+    # q [640,1024) causal tail; plus an OVERLAPPING extra q range
+    qr_l = [[0, 256], [0, 256], [0, 256], [256, 640], [256, 640],
+            [640, 1024], [128, 384]]
+    kr_l = [[0, 128], [384, 512], [640, 768], [0, 256], [512, 896],
+            [640, 1024], [896, 1024]]
+    tt = [0, 0, 1, 0, 0, 1, 0]
+    q, k, v, dout, qr, kr, tm = make_flex_case(
+        tq, tk, hq, hk, d, qr_l, kr_l, tt, seed=13
+    )
+
+    def run(arm):
+        qq = q.detach().clone().requires_grad_(True)
+        kk = k.detach().clone().requires_grad_(True)
+        vv = v.detach().clone().requires_grad_(True)
+        out, meta = flex_flash_attn_func(qq, kk, vv, qr, kr, tm,
+                                         auto_range_merge=arm)
+        out.backward(dout)
+        torch.cuda.synchronize()
+        return out.detach(), meta.lse, qq.grad, kk.grad, vv.grad
+
+    o_m, lse_m, dq_m, dk_m, dv_m = run(True)
+    o_u, lse_u, dq_u, dk_u, dv_u = run(False)
+
+    mask = make_attn_mask(tq, tk, qr_l, kr_l, tt)
+    qc, kc, vc, doc = [t.detach().cpu() for t in (q, k, v, dout)]
+    hi = ref_attn_with_grads(qc, kc, vc, mask, doc)
+    lo = ref_attn_with_grads(qc, kc, vc, mask, doc, high_precision=False,
+                             p_dtype=torch.bfloat16)
+    for got, ghi, glo, name in [
+        (o_m, hi[0], lo[0], "arm:out"), (lse_m.cpu(), hi[1], lo[1], "arm:lse"),
+        (dq_m, hi[2], lo[2], "arm:dq"), (dk_m, hi[3], lo[3], "arm:dk"),
+        (dv_m, hi[4], lo[4], "arm:dv"),
+    ]:
+        assert_close_to_ref(got.cpu().float(), ghi.float(), glo.float(), name)
+    # merged and unmerged paths agree within bf16 reduction noise
+    torch.testing.assert_close(o_m.float(), o_u.float(), atol=3e-2, rtol=3e-2)
+    torch.testing.assert_close(lse_m, lse_u, atol=1e-4, rtol=1e-4)
