@@ -565,6 +565,32 @@ def flex_flash_attn_func(
     assert index_attn_indices is None and not index_attn, (
         "index_attn lands in a later round"
     )
+    d = q.shape[-1]
+    if d not in (64, 128):
+        # arbitrary head dims run in the next bucket with zero feature
+        # padding — exact for attention (padding contributes 0 to scores;
+        # padded V columns are sliced off; gradients slice back through the
+        # autograd pad). Reference buckets head_size <=64/<=128/<=192
+        # (flash_api.cpp:322-334); the 192 bucket lands with a D=192 kernel.
+        assert d < 128, f"head_dim {d} > 128 lands in a later round"
+        assert q.dtype == torch.bfloat16, "padded head dims require bf16"
+        bucket = 64 if d <= 64 else 128
+        if softmax_scale is None:
+            softmax_scale = d ** (-0.5)  # scale from the REAL head dim
+        pad = bucket - d
+        qp = torch.nn.functional.pad(q, (0, pad))
+        kp = torch.nn.functional.pad(k, (0, pad))
+        vp = torch.nn.functional.pad(v, (0, pad))
+        out, meta = flex_flash_attn_func(
+            qp, kp, vp, q_ranges, k_ranges, attn_type_map,
+            sink=sink, sink_layout=sink_layout, softmax_scale=softmax_scale,
+            softcap=softcap, deterministic=deterministic, sm_margin=sm_margin,
+            disable_fwd_atomic_reduction=disable_fwd_atomic_reduction,
+            max_seqlen_q=max_seqlen_q, max_seqlen_k=max_seqlen_k,
+            auto_range_merge=auto_range_merge,
+            return_max_logits=return_max_logits,
+        )
+        return out[..., :d], meta
     out, lse, max_logits = FlexFlashAttnFunc.apply(
         q, k, v, sink, sink_layout, q_ranges, k_ranges, attn_type_map,
         softmax_scale, softcap, deterministic, sm_margin,

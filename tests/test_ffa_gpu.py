@@ -443,3 +443,35 @@ def test_auto_range_merge():
     # merged and unmerged paths agree within bf16 reduction noise
     torch.testing.assert_close(o_m.float(), o_u.float(), atol=3e-2, rtol=3e-2)
     torch.testing.assert_close(lse_m, lse_u, atol=1e-4, rtol=1e-4)
+
+
+@requires_gpu
+@pytest.mark.parametrize("hd", [48, 96])
+def test_odd_head_dims(hd):
+    """Head dims other than 64/128 run zero-padded in the next bucket
+    (reference flash_api.cpp:322-334 bucketing) — exact, incl. gradients."""
+    from magi_attention.functional import flex_flash_attn_func
+    from oracle import ref_attn_with_grads
+
+    tq = tk = 256
+    q, k, v, dout, qr, kr, tm = make_flex_case(
+        tq, tk, 4, 2, hd, [[0, tq]], [[0, tk]], [1], seed=3
+    )
+    q.requires_grad_(True)
+    k.requires_grad_(True)
+    v.requires_grad_(True)
+    out, meta = flex_flash_attn_func(q, k, v, qr, kr, tm)
+    assert out.shape[-1] == hd
+    out.backward(dout)
+    torch.cuda.synchronize()
+    mask = make_attn_mask(tq, tk, [[0, tq]], [[0, tk]], [1])
+    qc, kc, vc, doc = [t.detach().cpu() for t in (q, k, v, dout)]
+    hi = ref_attn_with_grads(qc, kc, vc, mask, doc)
+    lo = ref_attn_with_grads(qc, kc, vc, mask, doc, high_precision=False,
+                             p_dtype=torch.bfloat16)
+    for got, ghi, glo, name in [
+        (out.detach(), hi[0], lo[0], "out"), (q.grad, hi[2], lo[2], "dq"),
+        (k.grad, hi[3], lo[3], "dk"), (v.grad, hi[4], lo[4], "dv"),
+    ]:
+        assert_close_to_ref(got.cpu().float(), ghi.float(), glo.float(),
+                            f"hd{hd}:{name}")
