@@ -155,12 +155,16 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
   auto swz = [](int row, int byte_off) {
     return byte_off ^ ((row & SW32M) << 5);
   };
-  __shared__ __attribute__((aligned(16))) char fsmem[2 * 2 * FFA_BN * D * 2];
+  // 2 buffers x 64-row K/V images staged by LDS-DMA (same scheme as the
+  // backward kernels): one barrier per 64 k rows, the doubled compute phase
+  // covers the prefetch latency the 32-row register pipeline could not
+  constexpr int KITER = 2 * FFA_BN;
+  __shared__ __attribute__((aligned(16))) char fsmem[2 * 2 * KITER * D * 2];
   auto lds_k = [&](int buf) -> __bf16* {
-    return (__bf16*)(fsmem + buf * FFA_BN * D * 2);
+    return (__bf16*)(fsmem + (2 * buf) * KITER * D * 2);
   };
   auto lds_v = [&](int buf) -> __bf16* {
-    return (__bf16*)(fsmem + (2 + buf) * FFA_BN * D * 2);
+    return (__bf16*)(fsmem + (2 * buf + 1) * KITER * D * 2);
   };
 
   // Q fragments in registers (8 x bf16x8 for D=128)
@@ -183,62 +187,43 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
 #pragma unroll
   for (int dt = 0; dt < DT; ++dt) acc_o[dt] = (f32x16)(0.f);
 
-  // ---- double-buffered staging pipeline (T14: issue loads for tile t+1
-  // into registers during tile t's compute; one barrier per iteration) ----
-  constexpr int CPR = D / 8;       // 16-B chunks per row
-  constexpr int RPP = 256 / CPR;   // rows per pass
-  constexpr int NPASS = FFA_BN / RPP;
-  const int srow = threadIdx.x / CPR;
-  const int scol = threadIdx.x % CPR;
-  // TWO register staging sets: loads for tile t+2 are issued while tile t
-  // computes and tile t+1's regs are written to LDS at loop TOP — the
-  // implicit vmcnt wait at write_stage then covers loads issued a FULL
-  // iteration earlier instead of the same iteration (the one-deep form
-  // measured 33% SQ_WAIT). Sets alternate by 2x loop unrolling so the
-  // register arrays stay compile-time addressed (rule 20).
-  bf16x8 kregA[NPASS], vregA[NPASS], kregB[NPASS], vregB[NPASS];
-
-  auto issue_loads = [&](int n0, auto& kreg, auto& vreg) {
+  constexpr int ROWS_PER_GLDS = 1024 / ROWB;
+  constexpr int GLDS_PER_WAVE = (KITER / 4) / ROWS_PER_GLDS;
+  auto stage_glds = [&](int buf, int n0x) {
 #pragma unroll
-    for (int pass = 0; pass < NPASS; ++pass) {
-      const int kr = min(n0 + pass * RPP + srow, ke - 1);
-      kreg[pass] = *(const bf16x8*)(kbase + (size_t)kr * k_pitch + scol * 8);
-      vreg[pass] = *(const bf16x8*)(vbase + (size_t)kr * k_pitch + scol * 8);
-    }
-  };
-  auto write_stage = [&](int buf, auto& kreg, auto& vreg) {
-#pragma unroll
-    for (int pass = 0; pass < NPASS; ++pass) {
-      const int r = pass * RPP + srow;
-      const int dst = swz(r, r * ROWB + scol * 16);
-      *(bf16x8*)((char*)lds_k(buf) + dst) = kreg[pass];
-      *(bf16x8*)((char*)lds_v(buf) + dst) = vreg[pass];
+    for (int gi = 0; gi < GLDS_PER_WAVE; ++gi) {
+      const int r0 = (KITER / 4) * wave + ROWS_PER_GLDS * gi;
+      const int r = r0 + lane / (ROWB / 16);
+      const int c = lane % (ROWB / 16);
+      const int kr = min(n0x + r, ke - 1);
+      const int csw = (c ^ ((r & SW32M) << 1)) * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              kbase + (size_t)kr * k_pitch + csw),
+          (__attribute__((address_space(3))) unsigned int*)&lds_k(buf)[r0 * D],
+          16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              vbase + (size_t)kr * k_pitch + csw),
+          (__attribute__((address_space(3))) unsigned int*)&lds_v(buf)[r0 * D],
+          16, 0, 0);
     }
   };
 
   int cur = 0;
 
-  auto iter_body = [&](int n0, auto& kreg_w, auto& vreg_w, auto& kreg_l,
-                       auto& vreg_l) {
-    const bool has_next = n0 + FFA_BN < b_hi;
-    // stage tile t+1 (regs loaded last iteration) and start loads for t+2
-    if (has_next) write_stage(cur ^ 1, kreg_w, vreg_w);
-    issue_loads(n0 + 2 * FFA_BN, kreg_l, vreg_l);  // clamped overrun is safe
-    const bool live = (n0 + FFA_BN > n_lo) && (n0 < n_hi) && qvalid_any;
+  auto sub_body = [&](int ns, const __bf16* lkb, const __bf16* lvb) {
+    const bool live = (ns + FFA_BN > n_lo) && (ns < n_hi) && qvalid_any;
+    if (!live) return;
 
     f32x16 s = (f32x16)(0.f);
-    if (live) {
+    {
 #pragma unroll
       for (int dd = 0; dd < DF; ++dd) {
         bf16x8 kf = *(const bf16x8*)(
-            (const char*)lds_k(cur) + swz(lo32, lo32 * ROWB + dd * 32 + hi * 16));
+            (const char*)lkb + swz(lo32, lo32 * ROWB + dd * 32 + hi * 16));
         s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[dd], s, 0, 0, 0);
       }
-    }
-    if (!live) {
-      __syncthreads();
-      cur ^= 1;
-      return;
     }
 
     // ---- mask + scale into exp2 domain ----
@@ -246,9 +231,9 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
     // lane (wave-uniform), skip the per-element compare/select chain — the
     // kernels are instruction-issue-bound and masks only bind near edges.
     const bool interior =
-        (q0 + 31 < qe) && (n0 >= n_lo) && (n0 + FFA_BN <= n_hi) &&
-        !((atype == 1 || atype == 3) && (n0 + FFA_BN - 1 > q0 + (ke - qe))) &&
-        !((atype == 2 || atype == 3) && (n0 < q0 + 31 + (ks - qs)));
+        (q0 + 31 < qe) && (ns >= n_lo) && (ns + FFA_BN <= n_hi) &&
+        !((atype == 1 || atype == 3) && (ns + FFA_BN - 1 > q0 + (ke - qe))) &&
+        !((atype == 2 || atype == 3) && (ns < q0 + 31 + (ks - qs)));
     float t[16];
     float mx = -INFINITY;
     if (interior) {
@@ -262,7 +247,7 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
     } else {
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        const int kk = n0 + crow(r, hi);
+        const int kk = ns + crow(r, hi);
         bool ok = qvalid && (kk >= n_lo) && (kk < n_hi);
         if (atype == 1 || atype == 3) ok = ok && (kk - qrow <= ke - qe);
         if (atype == 2 || atype == 3) ok = ok && (kk - qrow >= ks - qs);
@@ -336,7 +321,7 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
       const int qhalf = (lane >> 4) & 1;
       const int jrow = (lane & 15) >> 2;
       const int v_base = (int)(unsigned long long)(
-          (__attribute__((address_space(3))) char*)lds_v(cur));
+          (__attribute__((address_space(3))) const char*)lvb);
       const int lane8 = (lane & 3) * 8;
 #pragma unroll
       for (int tt = 0; tt < 2; ++tt) {
@@ -355,8 +340,6 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
         }
       }
     }
-    __syncthreads();
-    cur ^= 1;
   };
 
   bool any_seg = false;
@@ -369,17 +352,18 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
     if (b_lo >= b_hi) continue;
     any_seg = true;
     cur = 0;
-    issue_loads(b_lo, kregA, vregA);
-    write_stage(0, kregA, vregA);
-    issue_loads(b_lo + FFA_BN, kregA, vregA);  // tile t1 (clamped if absent)
-    __syncthreads();
-    int n0 = b_lo;
-    while (n0 < b_hi) {
-      iter_body(n0, kregA, vregA, kregB, vregB);
-      n0 += FFA_BN;
-      if (n0 >= b_hi) break;
-      iter_body(n0, kregB, vregB, kregA, vregA);
-      n0 += FFA_BN;
+    stage_glds(0, b_lo);
+    for (int n0 = b_lo; n0 < b_hi; n0 += KITER) {
+      __syncthreads();  // buf[cur] glds drained (full 64-row phase of cover)
+      if (n0 + KITER < b_hi) stage_glds(cur ^ 1, n0 + KITER);
+#pragma unroll
+      for (int sub = 0; sub < 2; ++sub) {
+        const int ns = n0 + sub * FFA_BN;
+        if (ns >= b_hi) break;  // b_hi is block-uniform
+        sub_body(ns, lds_k(cur) + sub * FFA_BN * D,
+                 lds_v(cur) + sub * FFA_BN * D);
+      }
+      cur ^= 1;
     }
     // LDS reads of this segment retired before the next segment's staging
     __syncthreads();
