@@ -152,6 +152,10 @@ class DistAttnRuntime:
         from .flex_flash_attn import _flex_flash_attn_forward
 
         qr, kr, tm = arg.to_device(q.device)
+        qk_starts = None
+        if (env.is_auto_range_merge_enable()
+                and not env.is_deterministic_mode_enable()):
+            (qr, kr, tm, qk_starts), _ = arg.to_device_merged(q.device)
         _flex_flash_attn_forward(
             q=q, k=k, v=v, sink=None, sink_layout="sh",
             out=out_acc, lse=lse_acc,
@@ -161,7 +165,7 @@ class DistAttnRuntime:
             deterministic=env.is_deterministic_mode_enable(),
             sm_margin=env.ffa_forward_sm_margin(),
             max_seqlen_q=arg.max_seqlen_q,
-            max_logits=max_logits,
+            max_logits=max_logits, qk_starts=qk_starts,
         )
 
     # ---------------- backward ----------------
@@ -297,6 +301,9 @@ class DistAttnRuntime:
 
         if env.is_deterministic_mode_enable():
             run_bwd_deterministic(args, qr, kr, tm, hq, hk, q.device)
+        elif env.is_auto_range_merge_enable():
+            dq_t, dkv_t = arg.to_device_merged(q.device)
+            run_bwd_passes(args, q.device, dq_tables=dq_t, dkv_tables=dkv_t)
         else:
             run_bwd_passes(args, q.device)
 

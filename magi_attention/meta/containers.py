@@ -36,6 +36,25 @@ class AttnArg:
             )
         return self._cache[key]
 
+    def to_device_merged(self, device):
+        """Cached auto_range_merge tables for this arg: one (outer, inner,
+        types, seg_starts) tuple per pass direction (q-merged for fwd/dq,
+        k-merged for dkv) — built once per device, reused every step
+        (MAGI_ATTENTION_AUTO_RANGE_MERGE, reference env/general.py:206)."""
+        key = ("merged", str(device))
+        if key not in self._cache:
+            from ..functional.flex_flash_attn import _seg_starts, merge_ranges
+
+            qr, kr, tm = self.to_device(device)
+            n = qr.shape[0]
+            mq, _, sk, st, inv_q, _ = merge_ranges(qr, kr, tm)
+            mk, _, sq, st2, inv_k, _ = merge_ranges(kr, qr, tm)
+            self._cache[key] = (
+                (mq, sk, st, _seg_starts(inv_q, n)),
+                (mk, sq, st2, _seg_starts(inv_k, n)),
+            )
+        return self._cache[key]
+
 
 @dataclass
 class RowChunkMap:

@@ -178,3 +178,49 @@ def test_calc_attn_cp1_sink_max_logits():
                             floor=fl)
     assert_close_to_ref(sink_l.grad.cpu(), hi[5].float(), lo[5].float(),
                         "cp1sink:dsink", floor=2e-2)
+
+
+@requires_gpu
+def test_calc_attn_cp1_auto_range_merge():
+    """MAGI_ATTENTION_AUTO_RANGE_MERGE=1 under the runtime (cp=1): merged
+    segment tables per FFA call, results match the unmerged path."""
+    from magi_attention.api import calc_attn, dispatch, magi_attn_flex_key, undispatch
+    from magi_attention.common.ranges import AttnRanges
+    from magi_attention.config import DispatchConfig, DistAttnConfig
+
+    pg = _init_pg()
+    total, hq, hk, d = 1024, 4, 2, 128
+    # block-sparse rows: repeated q ranges with scattered k blocks
+    qr_l = [[0, 256], [0, 256], [256, 768], [256, 768], [768, 1024]]
+    kr_l = [[0, 128], [512, 768], [0, 512], [768, 1024], [768, 1024]]
+    tt = [0, 0, 0, 0, 1]
+    g = torch.Generator().manual_seed(17)
+    q = (torch.randn(total, hq, d, generator=g) * 0.5).bfloat16().cuda()
+    k = (torch.randn(total, hk, d, generator=g) * 0.5).bfloat16().cuda()
+    v = (torch.randn(total, hk, d, generator=g) * 0.5).bfloat16().cuda()
+    dout = (torch.randn(total, hq, d, generator=g) * 0.5).bfloat16().cuda()
+    cfg = DistAttnConfig(dispatch_config=DispatchConfig(chunk_size=256))
+
+    def run():
+        key = magi_attn_flex_key(
+            AttnRanges.from_ranges(qr_l), AttnRanges.from_ranges(kr_l), tt,
+            total, total, hq, hk, d, cp_group_or_mesh=pg,
+            dist_attn_config=cfg,
+        )
+        ql = dispatch(q, key).requires_grad_(True)
+        kl = dispatch(k, key).requires_grad_(True)
+        vl = dispatch(v, key).requires_grad_(True)
+        out_l, _ = calc_attn(ql, kl, vl, key)
+        full = undispatch(out_l, key)
+        out_l.backward(dispatch(dout, key))
+        torch.cuda.synchronize()
+        return full, ql.grad.clone(), kl.grad.clone(), vl.grad.clone()
+
+    base = run()
+    os.environ["MAGI_ATTENTION_AUTO_RANGE_MERGE"] = "1"
+    try:
+        merged = run()  # env snapshot differs -> fresh key/plan
+    finally:
+        del os.environ["MAGI_ATTENTION_AUTO_RANGE_MERGE"]
+    for a, b, name in zip(base, merged, ["out", "dq", "dk", "dv"]):
+        torch.testing.assert_close(a.float(), b.float(), atol=3e-2, rtol=3e-2)
