@@ -5,4 +5,9 @@
 # kernels behind a C-ABI (include/magi_ffa.h) + RCCL-over-xGMI collectives.
 __version__ = "0.1.0"
 
-from . import common  # noqa: F401
+from . import comm, common, config, env, functional  # noqa: F401
+from . import magi_attn_ext  # noqa: F401
+from .dist_attn_runtime_mgr import (  # noqa: F401
+    init_dist_attn_runtime_key,
+    init_dist_attn_runtime_mgr,
+)

@@ -252,13 +252,13 @@ def clear_cache(cp_group=None) -> None:
 
 
 def roll(x: torch.Tensor, key: DistAttnRuntimeKey, shifts: int = 1) -> torch.Tensor:
-    """Cyclic shift along the GLOBAL sequence of a dispatched tensor
-    (reference :960; used by MTP). Round-1 implementation routes through
-    undispatch/dispatch; the P2P-roll fast path lands in a later round."""
+    """Cyclic shift along the (padded) GLOBAL sequence of a dispatched tensor
+    (reference :960; used by MTP), via the P2P roll — each row moves once,
+    peak memory stays O(N/P) (functional/roll.py)."""
+    from ..functional.roll import roll_p2p
+
     mgr = dist_attn_runtime_dict_mgr[key]
-    full = mgr.undispatch_qo(x)
-    full = torch.roll(full, shifts=shifts, dims=0)
-    return mgr.dispatch_qo(full)
+    return roll_p2p(x, shifts, mgr.dispatch_meta, mgr.cp_group)
 
 
 def roll_simple(x: torch.Tensor, key: DistAttnRuntimeKey, shifts: int = 1):

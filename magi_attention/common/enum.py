@@ -80,3 +80,19 @@ class MagiAttentionPrecision(Enum):
     FP16 = "fp16"
     FP32 = "fp32"
     FP64 = "fp64"
+
+
+class DynamicAttnAlgType(Enum):
+    """Reference common/enum.py — dynamic attn solver algorithm tags (the
+    dynamic solver itself is a later-round item, SURVEY #17)."""
+
+    BALANCED = "balanced"
+    GREEDY = "greedy"
+
+
+class GrpCollBufferName(Enum):
+    """Reference common/enum.py — named native-grpcoll buffers (the RCCL
+    a2av transport used here needs none; kept for import parity)."""
+
+    GROUP_CAST = "group_cast"
+    GROUP_REDUCE = "group_reduce"

@@ -188,6 +188,34 @@ class DistAttnRuntimeDict:
         self._d.clear()
 
 
+def check_flag_comb() -> None:
+    """Reject invalid env-flag combinations (reference
+    dist_attn_runtime_mgr.py:451)."""
+    if env.is_hierarchical_comm_enable():
+        assert not env.is_qo_comm_enable(), (
+            "Hierarchical comm is not compatible with qo comm for now"
+        )
+        assert not env.is_native_grpcoll_enable(), (
+            "Hierarchical comm is not compatible with native grpcoll for now"
+        )
+    if env.is_native_grpcoll_enable():
+        assert not env.is_deterministic_mode_enable(), (
+            "Native grpcoll is not compatible with deterministic mode for now"
+        )
+
+
+def init_grpcoll_buffer_mgr(*args, **kwargs):
+    """Reference dist_attn_runtime_mgr grpcoll buffer manager — only needed
+    by the native (NVSHMEM-style) grpcoll transport, which this engine
+    replaces with RCCL a2av over xGMI; a HIP-IPC analogue is
+    a later-round item."""
+    raise NotImplementedError(
+        "native grpcoll buffers: the RCCL a2av transport needs no "
+        "pre-registered buffers; the HIP-IPC peer-write analogue lands in a "
+        "later round"
+    )
+
+
 def init_dist_attn_runtime_key(
     q_ranges: AttnRanges,
     k_ranges: AttnRanges,
@@ -209,6 +237,7 @@ def init_dist_attn_runtime_key(
         else AttnMaskType(t).to_int_type() if isinstance(t, str) else int(t)
         for t in attn_mask_type
     )
+    check_flag_comb()
     return DistAttnRuntimeKey(
         q_ranges=tuple((r.start, r.end) for r in q_ranges),
         k_ranges=tuple((r.start, r.end) for r in k_ranges),

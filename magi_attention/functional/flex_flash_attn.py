@@ -113,6 +113,25 @@ from .._ffa_lib import (
 )
 from ..common.forward_meta import AttnForwardMeta
 
+import contextlib
+
+
+@contextlib.contextmanager
+def maybe_profile_ffa_ctx(name: str, enable: bool = False):
+    """Optional named-event timing around an FFA call (reference
+    flex_flash_attn.py maybe_profile_ffa_ctx; magi_attn_ext event timers)."""
+    if not enable:
+        yield
+        return
+    from .. import magi_attn_ext
+
+    magi_attn_ext.start_event(name)
+    try:
+        yield
+    finally:
+        magi_attn_ext.stop_event(name)
+
+
 def merge_ranges(
     outer_ranges: torch.Tensor,
     inner_ranges: torch.Tensor,

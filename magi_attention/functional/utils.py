@@ -1,0 +1,113 @@
+# Online-attention correction helpers (reference functional/utils.py:286
+# correct_attn_lse, :322 correct_attn_out, :467 correct_attn_out_lse, :561/:593
+# sink variants) — the public merge math users compose partial (out, lse)
+# pairs with. The in-engine merges run in the HIP kernels
+# (csrc/range_ops.hip correct_out_lse / the fwd lock-merge epilogue); these
+# torch forms keep the reference's API.
+from __future__ import annotations
+
+import torch
+import torch.nn.functional as F
+
+
+def _safe_sub(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    # -inf - -inf -> nan; the merge semantics want -inf
+    d = a - b
+    return torch.where(torch.isnan(d), torch.full_like(d, float("-inf")), d)
+
+
+def correct_attn_lse(
+    lse1: torch.Tensor, lse2: torch.Tensor, inplace: bool = False
+) -> torch.Tensor:
+    """lse = log(exp(lse1) + exp(lse2)) = max + softplus(min - max)
+    (reference functional/utils.py:286)."""
+    assert lse1.dtype == lse2.dtype
+    mn = torch.minimum(lse1, lse2).float()
+    mx = torch.maximum(lse1, lse2).float()
+    lse = mx + F.softplus(_safe_sub(mn, mx))
+    return lse1.copy_(lse) if inplace else lse.to(lse1.dtype)
+
+
+def correct_attn_out(
+    out1: torch.Tensor,
+    lse1: torch.Tensor,
+    out2: torch.Tensor,
+    lse2: torch.Tensor,
+    lse: torch.Tensor,
+    inplace: bool = False,
+) -> torch.Tensor:
+    """out = exp(lse1-lse)*out1 + exp(lse2-lse)*out2 (reference :322)."""
+    w1 = torch.exp(_safe_sub(lse1.float(), lse.float())).nan_to_num(0.0)
+    w2 = torch.exp(_safe_sub(lse2.float(), lse.float())).nan_to_num(0.0)
+    out = w1.unsqueeze(-1) * out1.float() + w2.unsqueeze(-1) * out2.float()
+    return out1.copy_(out) if inplace else out.to(out1.dtype)
+
+
+def correct_attn_out_lse(
+    out1: torch.Tensor,
+    lse1: torch.Tensor,
+    out2: torch.Tensor,
+    lse2: torch.Tensor,
+    inplace: bool = False,
+) -> tuple[torch.Tensor, torch.Tensor]:
+    """Merge two partial (out, lse) pairs (reference :467)."""
+    lse = correct_attn_lse(lse1, lse2)
+    out = correct_attn_out(out1, lse1, out2, lse2, lse)
+    if inplace:
+        return out1.copy_(out), lse1.copy_(lse.to(lse1.dtype))
+    return out, lse
+
+
+def _lse_sink(sink: torch.Tensor, sink_layout: str, lse: torch.Tensor):
+    """lse_sink per (row?, head) from the sink logits (reference :262
+    calc_lse_sink)."""
+    if sink_layout == "sh":      # [s_sink, hq] -> [hq] broadcast over rows
+        return torch.logsumexp(sink.float(), dim=0).expand_as(lse)
+    if sink_layout == "ssh":     # [sq, s_sink, hq] -> [sq, hq]
+        return torch.logsumexp(sink.float(), dim=1)
+    raise ValueError(f"unsupported sink_layout {sink_layout}")
+
+
+def correct_attn_lse_with_sink(
+    lse: torch.Tensor,
+    sink: torch.Tensor,
+    sink_layout: str = "sh",
+    inplace: bool = False,
+) -> torch.Tensor:
+    """lse' = log(exp(lse) + exp(lse_sink)) (reference :561)."""
+    ls = _lse_sink(sink, sink_layout, lse)
+    new = torch.logaddexp(lse.float(), ls)
+    return lse.copy_(new) if inplace else new.to(lse.dtype)
+
+
+def correct_attn_out_with_sink(
+    out: torch.Tensor,
+    lse: torch.Tensor,
+    sink: torch.Tensor,
+    sink_layout: str = "sh",
+    inplace: bool = False,
+) -> torch.Tensor:
+    """out' = out * exp(lse - lse_with_sink) (reference :593)."""
+    ls = _lse_sink(sink, sink_layout, lse)
+    new_lse = torch.logaddexp(lse.float(), ls)
+    w = torch.exp(_safe_sub(lse.float(), new_lse)).nan_to_num(0.0).unsqueeze(-1)
+    return out.mul_(w.to(out.dtype)) if inplace else (out * w).to(out.dtype)
+
+
+def correct_attn_out_lse_with_sink(
+    out: torch.Tensor,
+    lse: torch.Tensor,
+    sink: torch.Tensor,
+    sink_layout: str = "sh",
+    inplace: bool = False,
+) -> tuple[torch.Tensor, torch.Tensor]:
+    """Fold sink logits into a final (out, lse) pair (reference :635; the
+    in-engine form is csrc/range_ops.hip sink_postprocess_kernel)."""
+    ls = _lse_sink(sink, sink_layout, lse)
+    new_lse = torch.logaddexp(lse.float(), ls)
+    w = torch.exp(_safe_sub(lse.float(), new_lse)).nan_to_num(0.0).unsqueeze(-1)
+    if inplace:
+        out.mul_(w.to(out.dtype))
+        lse.copy_(new_lse)
+        return out, lse
+    return (out * w).to(out.dtype), new_lse.to(lse.dtype)
