@@ -6,7 +6,7 @@
 __version__ = "0.1.0"
 
 from . import comm, common, config, env, functional  # noqa: F401
-from . import magi_attn_ext  # noqa: F401
+from . import magi_attn_comm, magi_attn_ext  # noqa: F401
 from .dist_attn_runtime_mgr import (  # noqa: F401
     init_dist_attn_runtime_key,
     init_dist_attn_runtime_mgr,

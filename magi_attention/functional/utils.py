@@ -111,3 +111,90 @@ def correct_attn_out_lse_with_sink(
         lse.copy_(new_lse)
         return out, lse
     return (out * w).to(out.dtype), new_lse.to(lse.dtype)
+
+
+def safe_subtract(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """a - b with (-inf) - (-inf) = -inf instead of nan (reference :25)."""
+    mask = (a == b) & (a == float("-inf"))
+    return (a - b).masked_fill(mask, float("-inf"))
+
+
+def safe_lse(x: torch.Tensor, dim: int = -1, keepdim: bool = False):
+    """logsumexp whose all--inf rows give -inf with zero grads
+    (reference :38)."""
+    all_neg_inf = (x == float("-inf")).all(dim=dim, keepdim=keepdim)
+
+    if x.requires_grad:
+        x.register_hook(lambda g: g.nan_to_num(0.0))
+    lse = torch.logsumexp(x, dim=dim, keepdim=keepdim)
+    return lse.masked_fill(all_neg_inf, float("-inf"))
+
+
+def safe_softmax(x: torch.Tensor, lse: torch.Tensor | None = None,
+                 dim: int = -1) -> torch.Tensor:
+    """softmax whose all--inf rows give all-zero rows with zero grads
+    (reference :66)."""
+    all_neg_inf = (x == float("-inf")).all(dim=dim, keepdim=True)
+    if x.requires_grad:
+        x.register_hook(lambda g: g.nan_to_num(0.0))
+    if lse is not None:
+        sm = torch.exp(safe_subtract(x, lse.unsqueeze(dim)))
+    else:
+        sm = F.softmax(x, dim=dim)
+    return sm.masked_fill(all_neg_inf, 0.0)
+
+
+def softmax_bwd(dout: torch.Tensor, out: torch.Tensor) -> torch.Tensor:
+    """Standard softmax backward (reference :107)."""
+    diag_out = torch.diag_embed(out)
+    outer_out = torch.einsum("...ij, ...ik -> ...ijk", out, out)
+    return torch.einsum("...ij, ...ijk -> ...ik", dout, diag_out - outer_out)
+
+
+def calc_lse_sink(sink: torch.Tensor, seqlen_q: int,
+                  sink_layout: str = "sh") -> torch.Tensor:
+    """[seqlen_q, hq] log-sum-exp of the sink logits (reference :238)."""
+    if sink_layout == "sh":
+        return safe_lse(sink, dim=0, keepdim=True).repeat(seqlen_q, 1)
+    if sink_layout == "ssh":
+        return safe_lse(sink, dim=1)
+    raise ValueError(f"unsupported sink_layout {sink_layout}")
+
+
+def calc_lse_rescale_weight(lse_to_rescale: torch.Tensor,
+                            rescaled_lse: torch.Tensor) -> torch.Tensor:
+    """exp(old_lse - new_lse) as [sq, hq, 1] (reference :~250)."""
+    return (
+        torch.exp(safe_subtract(lse_to_rescale.float(), rescaled_lse.float()))
+        .nan_to_num(0.0)
+        .unsqueeze(-1)
+    )
+
+
+def sink_bwd(
+    sink: torch.Tensor,
+    lse: torch.Tensor,
+    o: torch.Tensor,
+    do: torch.Tensor,
+    sink_layout: str = "sh",
+    dsink: torch.Tensor | None = None,
+) -> torch.Tensor:
+    """dsink_j = -exp(sink_j - lse_row) * rowsum(dO*O) (reference sink_bwd;
+    the in-engine form is csrc/range_ops.hip dsink_*_kernel)."""
+    dpsum = (do.float() * o.float()).sum(-1)          # [sq, hq]
+    live = torch.isfinite(lse.float())
+    if sink_layout == "sh":
+        # p[s, h, t] = exp(sink[s, h] - lse[t, h])
+        p = torch.exp(sink.float().unsqueeze(-1) - lse.float().t().unsqueeze(0))
+        p = torch.where(live.t().unsqueeze(0), p, torch.zeros_like(p))
+        g = -(p * dpsum.t().unsqueeze(0)).sum(-1)
+    elif sink_layout == "ssh":
+        p = torch.exp(sink.float() - lse.float().unsqueeze(1))
+        p = torch.where(live.unsqueeze(1), p, torch.zeros_like(p))
+        g = -(p * dpsum.unsqueeze(1))
+    else:
+        raise ValueError(f"unsupported sink_layout {sink_layout}")
+    g = g.to(sink.dtype)
+    if dsink is not None:
+        return dsink.copy_(g)
+    return g

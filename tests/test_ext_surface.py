@@ -80,3 +80,45 @@ def test_cross_attn_raises_like_reference():
                 [0], 128, 64, 4, 2, 32,
                 is_same_source=False, is_q_permutable=qp, is_k_permutable=kp,
             )
+
+
+def test_correct_attn_helpers_match_oracle_merge():
+    """functional.utils correct_attn_* vs the oracle's merge_out_lse."""
+    import torch
+
+    from magi_attention.functional import (
+        correct_attn_lse,
+        correct_attn_out_lse,
+        correct_attn_out_lse_with_sink,
+    )
+    from oracle import merge_out_lse
+
+    g = torch.Generator().manual_seed(3)
+    o1 = torch.randn(32, 4, 16, generator=g)
+    o2 = torch.randn(32, 4, 16, generator=g)
+    l1 = torch.randn(32, 4, generator=g)
+    l2 = torch.randn(32, 4, generator=g)
+    l1[5] = float("-inf")  # empty-row case
+    l2[7] = float("-inf")
+    l1[9] = l2[9] = float("-inf")
+    out, lse = correct_attn_out_lse(o1, l1, o2, l2)
+    ref_o, ref_l = merge_out_lse([o1, o2], [l1, l2])
+    torch.testing.assert_close(out.double(), ref_o, atol=1e-5, rtol=1e-5)
+    torch.testing.assert_close(
+        correct_attn_lse(l1, l2).double(), ref_l, atol=1e-5, rtol=1e-5
+    )
+    torch.testing.assert_close(lse.double(), ref_l, atol=1e-5, rtol=1e-5)
+
+    # sink fold == appending the sink columns in the oracle
+    from oracle import ref_attn
+
+    q = torch.randn(32, 4, 16, generator=g).double()
+    k = torch.randn(32, 4, 16, generator=g).double()
+    v = torch.randn(32, 4, 16, generator=g).double()
+    sink = torch.randn(2, 4, generator=g).float()
+    mask = torch.ones(32, 32, dtype=torch.bool)
+    o_ns, l_ns = ref_attn(q, k, v, mask)
+    o_s, l_s = ref_attn(q, k, v, mask, sink=sink)
+    o_f, l_f = correct_attn_out_lse_with_sink(o_ns.float(), l_ns, sink)
+    torch.testing.assert_close(o_f.double(), o_s, atol=1e-5, rtol=1e-5)
+    torch.testing.assert_close(l_f, l_s, atol=1e-5, rtol=1e-5)
