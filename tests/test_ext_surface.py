@@ -91,7 +91,7 @@ def test_correct_attn_helpers_match_oracle_merge():
         correct_attn_out_lse,
         correct_attn_out_lse_with_sink,
     )
-    from oracle import merge_out_lse
+    from oracle.ref_attn import merge_out_lse
 
     g = torch.Generator().manual_seed(3)
     o1 = torch.randn(32, 4, 16, generator=g)
