@@ -190,7 +190,7 @@ def measure_roofline(state, device):
         "frac": achieved / MFMA_PEAK_BF16,
         "traffic": None,
         "detail": {
-            "kernel": "ffa_bwd_dq_kernel + ffa_bwd_dkv_kernel (dq pass concurrent with fused dK+dV pass)",
+            "kernel": "ffa_bwd_dq_kernel + ffa_bwd_dkv_kernel<dv>/<dk> (dq pass concurrent with the dv/dk passes)",
             "bwd_ms_per_launch": bwd_ms,
             "fwd_ms_per_launch": fwd_ms,
             "fwd_achieved_flops_per_s": host_fwd_flops / (fwd_ms * 1e-3),
