@@ -166,8 +166,12 @@ __global__ __launch_bounds__(256) void bwd_preprocess_kernel(BwdParams p, int d,
 // MODE 1: dV only / MODE 2: dK only — each recomputes S but fits 2 waves/SIMD,
 // which beats the fused kernel's naked single-wave stalls (the dq kernel went
 // 4.1x faster at occupancy 2 with the same pipeline).
-template <int D, bool HAS_SOFTCAP, int MODE>
-__global__ __launch_bounds__(256, MODE == 0 ? 1 : 2)
+// WAVES: k-tiles per workgroup sharing ONE staged Q/dO image. The split
+// modes run 8 waves (a single 512-thread WG per CU = 2 waves/SIMD) so the
+// staging+barrier cost per MFMA halves; the fused mode's 437-VGPR waves
+// cannot co-reside 2/SIMD, so it stays at 4 waves.
+template <int D, bool HAS_SOFTCAP, int MODE, int WAVES>
+__global__ __launch_bounds__(64 * WAVES, 1)
 void ffa_bwd_dkv_kernel(BwdParams p) {
   constexpr bool WANT_DV = MODE != 2;
   constexpr bool WANT_DK = MODE != 1;
@@ -192,7 +196,7 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   const int h = p.head_off + hidx * p.head_mult;
   const int wb = p.head_major ? blockIdx.y : blockIdx.x;
   const int ks = p.k_ranges[2 * ri], ke = p.k_ranges[2 * ri + 1];
-  const int nblk0 = ks + wb * (BWD_BN * BWD_WAVES);
+  const int nblk0 = ks + wb * (BWD_BN * WAVES);
   if (nblk0 >= ke) return;
   const int seg0 = p.seg_starts ? p.seg_starts[ri] : ri;
   const int seg1 = p.seg_starts ? p.seg_starts[ri + 1] : ri + 1;
@@ -259,7 +263,7 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   }
 
   // block + wave q loop bounds (recomputed per q segment)
-  const int nlast = min(nblk0 + BWD_BN * BWD_WAVES, ke) - 1;
+  const int nlast = min(nblk0 + BWD_BN * WAVES, ke) - 1;
   int q_lo = 0, q_hi = 0, wq_lo = 0, wq_hi = 0;
   auto seg_bounds = [&]() {
     q_lo = qs; q_hi = qe;
@@ -281,11 +285,12 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   // each wave's glds covers 4 rows (64 lanes x 16B = 1 KiB = 4 rows at D=128);
   // wave w owns rows [8w, 8w+8) via 2 issues per tensor
   constexpr int ROWS_PER_GLDS = 1024 / ROWB;
-  constexpr int GLDS_PER_WAVE = (QITER / BWD_WAVES) / ROWS_PER_GLDS;
+  static_assert(QITER / WAVES >= ROWS_PER_GLDS, "stage rows per wave");
+  constexpr int GLDS_PER_WAVE = (QITER / WAVES) / ROWS_PER_GLDS;
   auto stage_glds = [&](int buf, int m0x) {
 #pragma unroll
     for (int gi = 0; gi < GLDS_PER_WAVE; ++gi) {
-      const int r0 = (QITER / BWD_WAVES) * wave + ROWS_PER_GLDS * gi;
+      const int r0 = (QITER / WAVES) * wave + ROWS_PER_GLDS * gi;
       const int r = r0 + lane / (ROWB / 16);
       const int c = lane % (ROWB / 16);
       const int qrow = min(m0x + r, qe - 1);
@@ -507,8 +512,9 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
 // ablation measured at ~45% of a fused backward. K/V tiles are staged
 // cooperatively per iteration (row-major swizzled for the S^T/dP^T A-frags,
 // plus a transposed copy for the dQ B-frags).
-template <int D, bool HAS_SOFTCAP>
-__global__ __launch_bounds__(256, 2) void ffa_bwd_dq_kernel(BwdParams p) {
+template <int D, bool HAS_SOFTCAP, int WAVES>
+__global__ __launch_bounds__(64 * WAVES, WAVES == 8 ? 1 : 2)
+void ffa_bwd_dq_kernel(BwdParams p) {
   constexpr int DF = D / 16;
   constexpr int DT = D / 32;
   constexpr int ROWB = D * 2;
@@ -520,7 +526,7 @@ __global__ __launch_bounds__(256, 2) void ffa_bwd_dq_kernel(BwdParams p) {
   const int h = p.head_major ? blockIdx.x : blockIdx.y;
   const int wb = p.head_major ? blockIdx.y : blockIdx.x;
   const int qs = p.q_ranges[2 * ri], qe = p.q_ranges[2 * ri + 1];
-  const int mblk0 = qs + wb * (BWD_BM * BWD_WAVES);
+  const int mblk0 = qs + wb * (BWD_BM * WAVES);
   if (mblk0 >= qe) return;
   const int seg0 = p.seg_starts ? p.seg_starts[ri] : ri;
   const int seg1 = p.seg_starts ? p.seg_starts[ri + 1] : ri + 1;
@@ -576,7 +582,7 @@ __global__ __launch_bounds__(256, 2) void ffa_bwd_dq_kernel(BwdParams p) {
   const bool row_live = qvalid && lse_q != INFINITY && lse_q != -INFINITY;
 
   // block + wave k-loop bounds (recomputed per k segment)
-  const int mlast = min(mblk0 + BWD_BM * BWD_WAVES, qe) - 1;
+  const int mlast = min(mblk0 + BWD_BM * WAVES, qe) - 1;
   int k_lo = 0, k_hi = 0, wk_lo = 0, wk_hi = 0;
   auto seg_bounds = [&]() {
     k_lo = ks; k_hi = ke;
@@ -593,11 +599,12 @@ __global__ __launch_bounds__(256, 2) void ffa_bwd_dq_kernel(BwdParams p) {
   for (int dt = 0; dt < DT; ++dt) acc_dq[dt] = (f32x16)(0.f);
 
   constexpr int ROWS_PER_GLDS = 1024 / ROWB;
-  constexpr int GLDS_PER_WAVE = (KITER / BWD_WAVES) / ROWS_PER_GLDS;
+  static_assert(KITER / WAVES >= ROWS_PER_GLDS, "stage rows per wave");
+  constexpr int GLDS_PER_WAVE = (KITER / WAVES) / ROWS_PER_GLDS;
   auto stage_glds = [&](int buf, int n0x) {
 #pragma unroll
     for (int gi = 0; gi < GLDS_PER_WAVE; ++gi) {
-      const int r0 = (KITER / BWD_WAVES) * wave + ROWS_PER_GLDS * gi;
+      const int r0 = (KITER / WAVES) * wave + ROWS_PER_GLDS * gi;
       const int r = r0 + lane / (ROWB / 16);
       const int c = lane % (ROWB / 16);
       const int kr = min(n0x + r, ke - 1);
@@ -806,25 +813,29 @@ extern "C" int magi_ffa_bwd_dq(const magi_ffa_bwd_args* a) {
   int rc = fill_bwd_params(a, &p);
   if (rc) return rc > 0 ? 0 : rc;
 
-  const int qspan = BWD_BM * BWD_WAVES;
+  // 8 waves share one staged K/V image (one 512-thread WG/CU, 2 waves/SIMD)
+  // for LONG ranges: halves staging+barrier cost per MFMA (measured
+  // 67->63 ms at 64k). Short ranges keep 4 waves — the 256-row block
+  // windows over-iterate masked edges on 2k varlen docs.
+  const int dqw = a->max_seqlen_k >= 8192 ? 8 : 4;
+  const int qspan = BWD_BM * dqw;
   const int qblocks = (int)((a->total_q + qspan - 1) / qspan);
   if (a->n_ranges > 65535) return -5;
   dim3 grid_q = p.head_major ? dim3(a->hq, qblocks, (unsigned)a->n_ranges)
                              : dim3(qblocks, a->hq, (unsigned)a->n_ranges);
-  dim3 block(64 * BWD_WAVES);
+  dim3 block(64 * dqw);
   hipStream_t s = (hipStream_t)a->stream;
   const bool sc = a->softcap > 0.f;
+#define LAUNCH_DQ(DD, SC, WW) \
+  hipLaunchKernelGGL((ffa_bwd_dq_kernel<DD, SC, WW>), grid_q, block, 0, s, p)
   if (a->d == 64) {
-    if (sc)
-      hipLaunchKernelGGL((ffa_bwd_dq_kernel<64, true>), grid_q, block, 0, s, p);
-    else
-      hipLaunchKernelGGL((ffa_bwd_dq_kernel<64, false>), grid_q, block, 0, s, p);
+    if (dqw == 8) { if (sc) LAUNCH_DQ(64, true, 8); else LAUNCH_DQ(64, false, 8); }
+    else          { if (sc) LAUNCH_DQ(64, true, 4); else LAUNCH_DQ(64, false, 4); }
   } else {
-    if (sc)
-      hipLaunchKernelGGL((ffa_bwd_dq_kernel<128, true>), grid_q, block, 0, s, p);
-    else
-      hipLaunchKernelGGL((ffa_bwd_dq_kernel<128, false>), grid_q, block, 0, s, p);
+    if (dqw == 8) { if (sc) LAUNCH_DQ(128, true, 8); else LAUNCH_DQ(128, false, 8); }
+    else          { if (sc) LAUNCH_DQ(128, true, 4); else LAUNCH_DQ(128, false, 4); }
   }
+#undef LAUNCH_DQ
   return (int)hipGetLastError();
 }
 
@@ -833,30 +844,32 @@ static int launch_bwd_dkv(const magi_ffa_bwd_args* a) {
   BwdParams p;
   int rc = fill_bwd_params(a, &p);
   if (rc) return rc > 0 ? 0 : rc;
-  const int span = BWD_BN * BWD_WAVES;
+  // dK pass (MODE 2) runs 8 waves per WG for LONG ranges (one staged Q/dO
+  // image shared by all; measured 95->83 ms at 64k, register spill gone).
+  // dV (MODE 1) stays at 4 — its shorter MFMA chain loses more to the wider
+  // block windows than it saves in staging. Fused (MODE 0) can't fit 8.
+  const bool big = a->max_seqlen_k >= 8192;
+  const int W = (MODE == 2 && big) ? 8 : 4;
+  const int span = BWD_BN * W;
   const int nblocks = (a->max_seqlen_k + span - 1) / span;
   if (a->n_ranges > 65535) return -5;
   dim3 grid_kv = p.head_major
                      ? dim3(p.n_heads_launch, nblocks, (unsigned)a->n_ranges)
                      : dim3(nblocks, p.n_heads_launch, (unsigned)a->n_ranges);
-  dim3 block(64 * BWD_WAVES);
+  dim3 block(64 * W);
   hipStream_t s = (hipStream_t)a->stream;
   const bool sc = a->softcap > 0.f;
+#define LAUNCH_DKV(DD, SC, WW) \
+  hipLaunchKernelGGL((ffa_bwd_dkv_kernel<DD, SC, MODE, WW>), grid_kv, block, \
+                     0, s, p)
   if (a->d == 64) {
-    if (sc)
-      hipLaunchKernelGGL((ffa_bwd_dkv_kernel<64, true, MODE>), grid_kv, block,
-                         0, s, p);
-    else
-      hipLaunchKernelGGL((ffa_bwd_dkv_kernel<64, false, MODE>), grid_kv, block,
-                         0, s, p);
+    if (W == 8) { if (sc) LAUNCH_DKV(64, true, 8); else LAUNCH_DKV(64, false, 8); }
+    else        { if (sc) LAUNCH_DKV(64, true, 4); else LAUNCH_DKV(64, false, 4); }
   } else {
-    if (sc)
-      hipLaunchKernelGGL((ffa_bwd_dkv_kernel<128, true, MODE>), grid_kv, block,
-                         0, s, p);
-    else
-      hipLaunchKernelGGL((ffa_bwd_dkv_kernel<128, false, MODE>), grid_kv,
-                         block, 0, s, p);
+    if (W == 8) { if (sc) LAUNCH_DKV(128, true, 8); else LAUNCH_DKV(128, false, 8); }
+    else        { if (sc) LAUNCH_DKV(128, true, 4); else LAUNCH_DKV(128, false, 4); }
   }
+#undef LAUNCH_DKV
   return (int)hipGetLastError();
 }
 
