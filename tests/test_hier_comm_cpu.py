@@ -241,3 +241,65 @@ def _worker_e2e(rank, ws, port, *_):
 def test_hier_end_to_end_mesh_api():
     port = _free_port()
     mp.spawn(_worker_e2e, args=(4, port), nprocs=4, join=True)
+
+
+def _worker_hier_det(rank, ws, port, *_):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["MAGI_ATTENTION_HIERARCHICAL_COMM"] = "1"
+    os.environ["MAGI_ATTENTION_DETERMINISTIC_MODE"] = "1"
+    dist.init_process_group("gloo", rank=rank, world_size=ws)
+    try:
+        from torch.distributed.device_mesh import init_device_mesh
+
+        import magi_attention.functional.dist_attn as da
+        from magi_attention.api import calc_attn, dispatch, magi_attn_flex_key
+        from magi_attention.common.ranges import AttnRanges
+        from magi_attention.config import (
+            DispatchConfig,
+            DistAttnConfig,
+            OverlapConfig,
+        )
+        from tests.dist_backend import OracleBackend
+
+        da.register_test_attn_backend(OracleBackend)
+        mesh = init_device_mesh("cpu", (2, 2),
+                                mesh_dim_names=("inter", "intra"))
+        total, hq, hk, d = MASK["total"], 4, 2, 32
+        g = torch.Generator().manual_seed(77)
+        q = torch.randn(total, hq, d, generator=g, dtype=torch.float64)
+        k = torch.randn(total, hk, d, generator=g, dtype=torch.float64)
+        v = torch.randn(total, hk, d, generator=g, dtype=torch.float64)
+        dout = torch.randn(total, hq, d, generator=g, dtype=torch.float64)
+        cfg = DistAttnConfig(
+            dispatch_config=DispatchConfig(chunk_size=64),
+            overlap_config=OverlapConfig(degree=2, min_chunk_size=32),
+        )
+
+        def run():
+            key = magi_attn_flex_key(
+                AttnRanges.from_ranges(MASK["q_ranges"]),
+                AttnRanges.from_ranges(MASK["k_ranges"]),
+                MASK["types"], total, total, hq, hk, d,
+                cp_group_or_mesh=mesh, dist_attn_config=cfg,
+            )
+            ql = dispatch(q, key).requires_grad_(True)
+            kl = dispatch(k, key).requires_grad_(True)
+            vl = dispatch(v, key).requires_grad_(True)
+            out_l, _ = calc_attn(ql, kl, vl, key)
+            (out_l * dispatch(dout, key)).sum().backward()
+            return (out_l.detach().clone(), ql.grad.clone(),
+                    kl.grad.clone(), vl.grad.clone())
+
+        a, b = run(), run()
+        for x, y, name in zip(a, b, ["out", "dq", "dk", "dv"]):
+            assert torch.equal(x, y), f"hier+det {name} not repeatable"
+    finally:
+        dist.destroy_process_group()
+
+
+def test_hier_plus_deterministic_repeatable():
+    """Hierarchical comm composed with deterministic mode: two identical runs
+    must be bitwise identical (plan + reduce order both fixed)."""
+    port = _free_port()
+    mp.spawn(_worker_hier_det, args=(4, port), nprocs=4, join=True)

@@ -207,3 +207,31 @@ def test_planner_simulated_distributed_attention(seed, cp, degree):
         torch.testing.assert_close(
             lse_m.float()[fin], ref_lse[rows][fin], atol=1e-5, rtol=1e-5
         )
+
+
+def test_merge_ranges_reference_example():
+    """merge_ranges on the reference's own docstring example
+    (flex_flash_attn.py:110-149)."""
+    import torch
+
+    from magi_attention.functional import merge_ranges
+
+    outer = torch.tensor([[20, 30], [10, 20], [10, 20], [20, 30]],
+                         dtype=torch.int32)
+    inner = torch.tensor([[100, 110], [120, 130], [140, 150], [160, 170]],
+                         dtype=torch.int32)
+    tm = torch.tensor([0, 1, 0, 0], dtype=torch.int32)
+    merged, so, si, st, inv, count = merge_ranges(outer, inner, tm)
+    assert count.item() == 2
+    assert merged[:2].tolist() == [[10, 20], [20, 30]]
+    assert merged[2:].tolist() == [[0, 0], [0, 0]]  # zero padding
+    assert so.tolist() == [[10, 20], [10, 20], [20, 30], [20, 30]]
+    # inner/type rows follow their outer rows (stable within equal outers)
+    assert si.tolist() == [[120, 130], [140, 150], [100, 110], [160, 170]]
+    assert st.tolist() == [1, 0, 0, 0]
+    assert inv.tolist() == [0, 0, 1, 1]
+
+    from magi_attention.functional.flex_flash_attn import _seg_starts
+
+    starts = _seg_starts(inv, 4)
+    assert starts.tolist() == [0, 2, 4, 4, 4]  # padded tail = empty segments
