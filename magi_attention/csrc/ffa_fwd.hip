@@ -93,8 +93,11 @@ struct FwdParams {
   long long total_q, total_k;
 };
 
-template <int D, bool HAS_SOFTCAP, bool ATOMIC, bool OUT_BF16>
-__global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
+// WAVES: q-tiles per workgroup sharing ONE staged K/V image (see the bwd
+// kernels): 8 waves (one 512-thread WG/CU = 2 waves/SIMD) halve the staging
+// and barrier cost per MFMA for long ranges; 4 for short ranges.
+template <int D, bool HAS_SOFTCAP, bool ATOMIC, bool OUT_BF16, int WAVES>
+__global__ __launch_bounds__(64 * WAVES, 8 / WAVES) void ffa_fwd_kernel(FwdParams p) {
   constexpr int DF = D / 16;    // # of 16-wide d fragments
   constexpr int DT = D / 32;    // # of 32-wide output d tiles
   // Adaptive grid: head-major (blockIdx.x = head -> one XCD per head: L2
@@ -104,7 +107,7 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
   const int h = p.head_major ? blockIdx.x : blockIdx.z;
   const int mb = p.head_major ? blockIdx.y : blockIdx.x;
   const int qs = p.q_ranges[2 * ri], qe = p.q_ranges[2 * ri + 1];
-  const int m0 = qs + mb * FFA_BM;
+  const int m0 = qs + mb * (32 * WAVES);
   if (m0 >= qe) return;                       // uniform across block
   // auto_range_merge: iterate this unique q range's k segments in-kernel,
   // online softmax carried across them (reference merge_range.cu semantics)
@@ -141,7 +144,7 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
     }
     b_lo = ks; b_hi = ke;
     if (ke > ks && m0 < qe) {
-      const int qhib = min(m0 + FFA_BM - 1, qe - 1);
+      const int qhib = min(m0 + 32 * WAVES - 1, qe - 1);
       if (atype == 1 || atype == 3) b_hi = min(b_hi, qhib + (ke - qe) + 1);
       if (atype == 2 || atype == 3) b_lo = max(b_lo, m0 + (ks - qs));
     } else {
@@ -188,11 +191,12 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
   for (int dt = 0; dt < DT; ++dt) acc_o[dt] = (f32x16)(0.f);
 
   constexpr int ROWS_PER_GLDS = 1024 / ROWB;
-  constexpr int GLDS_PER_WAVE = (KITER / 4) / ROWS_PER_GLDS;
+  static_assert(KITER / WAVES >= ROWS_PER_GLDS, "stage rows per wave");
+  constexpr int GLDS_PER_WAVE = (KITER / WAVES) / ROWS_PER_GLDS;
   auto stage_glds = [&](int buf, int n0x) {
 #pragma unroll
     for (int gi = 0; gi < GLDS_PER_WAVE; ++gi) {
-      const int r0 = (KITER / 4) * wave + ROWS_PER_GLDS * gi;
+      const int r0 = (KITER / WAVES) * wave + ROWS_PER_GLDS * gi;
       const int r = r0 + lane / (ROWB / 16);
       const int c = lane % (ROWB / 16);
       const int kr = min(n0x + r, ke - 1);
@@ -412,21 +416,17 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
     return;
   }
 
-  // ---- lock-guarded read-merge-write (2-slot range locks) ----
-  const int row_last = min(m0 + FFA_BM, qe) - 1;
+  // ---- lock-guarded read-merge-write (range locks; a 32*WAVES-row block
+  // spans up to 32*WAVES/LOCK_GRAN+1 slots, acquired in ascending order —
+  // the global order prevents deadlock between overlapping blocks) ----
+  const int row_last = min(m0 + 32 * WAVES, qe) - 1;
   const int s0 = m0 / LOCK_GRAN, s1 = row_last / LOCK_GRAN;
-  int* lock0 = p.locks + (size_t)s0 * p.hq + h;
-  int* lock1 = p.locks + (size_t)s1 * p.hq + h;
   if (threadIdx.x == 0) {
-    int expected = 0;
-    while (!__hip_atomic_compare_exchange_strong(
-        lock0, &expected, 1, __ATOMIC_ACQUIRE, __ATOMIC_RELAXED,
-        __HIP_MEMORY_SCOPE_AGENT))
-      expected = 0;
-    if (s1 != s0) {
-      expected = 0;
+    for (int sl = s0; sl <= s1; ++sl) {
+      int* lk = p.locks + (size_t)sl * p.hq + h;
+      int expected = 0;
       while (!__hip_atomic_compare_exchange_strong(
-          lock1, &expected, 1, __ATOMIC_ACQUIRE, __ATOMIC_RELAXED,
+          lk, &expected, 1, __ATOMIC_ACQUIRE, __ATOMIC_RELAXED,
           __HIP_MEMORY_SCOPE_AGENT))
         expected = 0;
     }
@@ -472,9 +472,9 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_kernel(FwdParams p) {
   __syncthreads();
   if (threadIdx.x == 0) {
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    if (s1 != s0)
-      __hip_atomic_store(lock1, 0, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
-    __hip_atomic_store(lock0, 0, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
+    for (int sl = s1; sl >= s0; --sl)
+      __hip_atomic_store(p.locks + (size_t)sl * p.hq + h, 0, __ATOMIC_RELEASE,
+                         __HIP_MEMORY_SCOPE_AGENT);
   }
 }
 
@@ -529,7 +529,7 @@ extern "C" int magi_probe_mfma(const void* a, const void* b, void* d,
 // ------------------------------------------------------------------
 // launcher
 // ------------------------------------------------------------------
-template <int D>
+template <int D, int W>
 static int launch_fwd_d(const magi_ffa_fwd_args* a, const FwdParams& p,
                         dim3 grid, dim3 block, hipStream_t stream) {
   const bool sc = a->softcap > 0.f;
@@ -537,8 +537,8 @@ static int launch_fwd_d(const magi_ffa_fwd_args* a, const FwdParams& p,
   const bool obf16 = !a->out_is_fp32;
   if (atomic && obf16) return -10;  // atomic merge requires fp32 out
 #define LAUNCH(SC, AT, OB)                                              \
-  hipLaunchKernelGGL((ffa_fwd_kernel<D, SC, AT, OB>), grid, block, 0,   \
-                     stream, p)
+  hipLaunchKernelGGL((ffa_fwd_kernel<D, SC, AT, OB, W>), grid, block,   \
+                     0, stream, p)
   if (atomic) {
     if (sc) LAUNCH(true, true, false);
     else LAUNCH(false, true, false);
@@ -582,17 +582,24 @@ extern "C" int magi_ffa_fwd(const magi_ffa_fwd_args* a) {
   p.total_q = a->total_q;
   p.total_k = a->total_k;
 
-  const int mblocks = (a->max_seqlen_q + FFA_BM - 1) / FFA_BM;
+  // 8 waves per WG for long ranges (staging shared over 2x the q rows),
+  // 4 for short ones (bounds masked-edge over-iteration on small docs)
+  const int fw = a->max_seqlen_q >= 8192 ? 8 : 4;
+  const int span = 32 * fw;
+  const int mblocks = (a->max_seqlen_q + span - 1) / span;
   if (a->n_ranges > 65535) return -5;
   // head-major XCD affinity pays when one head's K+V fits a 4 MB XCD L2
   p.head_major = ((long long)a->total_k * a->d * 4 <= (4 << 20)) ? 1 : 0;
   dim3 grid = p.head_major
                   ? dim3(a->hq, mblocks, (unsigned)a->n_ranges)
                   : dim3(mblocks, (unsigned)a->n_ranges, a->hq);
-  dim3 block(256);
+  dim3 block(64 * fw);
   hipStream_t stream = (hipStream_t)a->stream;
-  if (a->d == 64) return launch_fwd_d<64>(a, p, grid, block, stream);
-  return launch_fwd_d<128>(a, p, grid, block, stream);
+  if (a->d == 64)
+    return fw == 8 ? launch_fwd_d<64, 8>(a, p, grid, block, stream)
+                   : launch_fwd_d<64, 4>(a, p, grid, block, stream);
+  return fw == 8 ? launch_fwd_d<128, 8>(a, p, grid, block, stream)
+                 : launch_fwd_d<128, 4>(a, p, grid, block, stream);
 }
 
 extern "C" int magi_ffa_abi_version(void) { return 1; }
