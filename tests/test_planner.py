@@ -235,3 +235,91 @@ def test_merge_ranges_reference_example():
 
     starts = _seg_starts(inv, 4)
     assert starts.tolist() == [0, 2, 4, 4, 4]  # padded tail = empty segments
+
+
+def test_solver_coverage_fuzz():
+    """Seeded random masks: the union of every rank's host+stage slices must
+    equal the global mask exactly (coverage, no duplication), for random
+    range sets across cp sizes."""
+    import random
+
+    import torch
+
+    from magi_attention.common.ranges import AttnRanges
+    from magi_attention.config import (
+        DispatchConfig,
+        DistAttnConfig,
+        OverlapConfig,
+    )
+    from magi_attention.meta import (
+        make_attn_meta_from_dispatch_meta,
+        make_dispatch_meta_from_qk_ranges,
+        normalize_slices,
+    )
+    from oracle import make_attn_mask
+
+    rng = random.Random(1234)
+    for trial in range(12):
+        total = rng.choice([256, 384, 512])
+        # slices must be pairwise cell-disjoint (the reference's mask
+        # semantics: duplicated coverage would double-count in the softmax) —
+        # resample any slice that overlaps the accumulated area
+        n = rng.randint(1, 6)
+        qr, kr, tt = [], [], []
+        area = torch.zeros(total, total, dtype=torch.bool)
+        for _ in range(n):
+            for _attempt in range(20):
+                a = rng.randrange(0, total - 32)
+                b = rng.randrange(a + 16, min(a + 256, total) + 1)
+                c = rng.randrange(0, total - 32)
+                d_ = rng.randrange(c + 16, min(c + 256, total) + 1)
+                t = rng.choice([0, 1, 2, 3])
+                sub = make_attn_mask(total, total, [[a, b]], [[c, d_]], [t])
+                if not (area & sub).any():
+                    area |= sub
+                    qr.append([a, b])
+                    kr.append([c, d_])
+                    tt.append(t)
+                    break
+        if not qr:
+            continue
+        cp = rng.choice([2, 4])
+        cfg = DistAttnConfig(
+            dispatch_config=DispatchConfig(chunk_size=32),
+            overlap_config=OverlapConfig(
+                degree=rng.choice([1, 2]), min_chunk_size=16
+            ),
+        )
+        slices = normalize_slices(
+            AttnRanges.from_ranges(qr), AttnRanges.from_ranges(kr), tt
+        )
+        want = make_attn_mask(total, total, qr, kr, tt)
+        got = torch.zeros_like(want, dtype=torch.int32)
+        for rank in range(cp):
+            dm = make_dispatch_meta_from_qk_ranges(slices, total, cp, rank, cfg)
+            solver, calc, _ = make_attn_meta_from_dispatch_meta(slices, dm, cfg)
+            hr = solver.plans[rank].host_ranges
+            # host slices: q local -> global via host ranges; k local == global
+            # offsets within hosted/stage spaces
+            def add(arg, k_space):
+                rows = []
+                for piece in hr:
+                    rows.extend(range(piece.start, piece.end))
+                ksp = []
+                for piece in k_space:
+                    ksp.extend(range(piece.start, piece.end))
+                for (qs, qe), (ks, ke), t in zip(
+                    arg.q_ranges, arg.k_ranges, arg.attn_type_map
+                ):
+                    sub = make_attn_mask(
+                        qe - qs, ke - ks, [[0, qe - qs]], [[0, ke - ks]], [t]
+                    )
+                    for i in range(qe - qs):
+                        for j in range(ke - ks):
+                            if sub[i, j]:
+                                got[rows[qs + i], ksp[ks + j]] += 1
+            add(calc.host_arg, hr)
+            for s, arg in enumerate(calc.stage_args):
+                add(arg, solver.plans[rank].stages_need[s])
+        assert torch.equal(got.bool(), want), f"trial {trial}: coverage mismatch"
+        assert (got <= 1).all(), f"trial {trial}: duplicated area"
