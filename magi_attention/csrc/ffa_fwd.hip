@@ -582,9 +582,12 @@ extern "C" int magi_ffa_fwd(const magi_ffa_fwd_args* a) {
   p.total_q = a->total_q;
   p.total_k = a->total_k;
 
-  // 8 waves per WG for long ranges (staging shared over 2x the q rows),
-  // 4 for short ones (bounds masked-edge over-iteration on small docs)
-  const int fw = a->max_seqlen_q >= 8192 ? 8 : 4;
+  // Unlike the bwd dq/dk passes, the forward measured FASTER at 4 waves
+  // (58.1 vs 59.8 ms at 64k, same box A/B): its K/V staging is a smaller
+  // share of the loop than bwd's Q/dO staging, and the 256-row block widens
+  // the causal k-window per block. MAGI_FWD_WAVES=8 opts in for tuning.
+  int fw = 4;
+  { const char* e = getenv("MAGI_FWD_WAVES"); if (e && atoi(e)) fw = atoi(e); }
   const int span = 32 * fw;
   const int mblocks = (a->max_seqlen_q + span - 1) / span;
   if (a->n_ranges > 65535) return -5;
