@@ -129,11 +129,12 @@ def magi_attn_flex_key(
             )
         raise NotImplementedError("A trivial case with no need to dispatch.")
     assert total_seqlen_q == total_seqlen_k, "self-attn requires equal seqlens"
-    assert head_dim in (64, 128), (
-        "context-parallel runs support head_dim 64/128 in r1 (other dims run "
-        "zero-padded through the single-GPU functional API; the padded CP "
-        "path lands next round)"
-    )
+    if torch.cuda.is_available():  # kernel constraint; CPU test oracle is free
+        assert head_dim in (64, 128), (
+            "context-parallel runs support head_dim 64/128 in r1 (other dims "
+            "run zero-padded through the single-GPU functional API; the "
+            "padded CP path lands next round)"
+        )
     group, mesh_groups = _resolve_group(cp_group_or_mesh)
     cp_size = dist.get_world_size(group)
     if chunk_size is not None:
