@@ -59,6 +59,12 @@ def _band_to_triples(
     """Sliding-window region as aligned typed slices via the band-geometry
     engine (role of reference functools.py:180; bottom-right aligned window)."""
     sk = ke - ks
+    # Reference functools.py:218-224: when q_range is longer than k_range,
+    # only the LAST sk q rows participate (bottom-right alignment); earlier
+    # rows attend nothing and get no triple. The window-clamp shortcut below
+    # is only exact after this trim.
+    if qe - qs > sk:
+        qs = qe - sk
     left, right = window_size
     D = ke - qe  # bottom-right diagonal offset in global coords
     lo = None if (left == -1 or left >= sk - 1) else D - left

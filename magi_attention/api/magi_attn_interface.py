@@ -33,6 +33,12 @@ class DistAttnRuntimeDictManager:
     def __init__(self, max_size_per_group: int = 100):
         self.max_size = max_size_per_group
         self._per_group: dict = {}
+        self._tag_stamp: dict = {}   # tag -> monotonic last-touch stamp
+        self._stamp = 0
+
+    def _touch(self, tag):
+        self._stamp += 1
+        self._tag_stamp[tag] = self._stamp
 
     def _cache_for(self, key: DistAttnRuntimeKey) -> DistAttnRuntimeDict:
         tag = key.cp_group_tag
@@ -47,20 +53,41 @@ class DistAttnRuntimeDictManager:
         return key in self._cache_for(key)
 
     def __setitem__(self, key, value):
+        self._touch(key.cp_group_tag)
         self._cache_for(key)[key] = value
 
     def __getitem__(self, key):
+        self._touch(key.cp_group_tag)
         return self._cache_for(key)[key]
 
     def get_most_recent_key(self, cp_group=None):
-        for cache in self._per_group.values():
-            k = cache.most_recent_key()
-            if k is not None:
-                return k
+        """Most recent key of the REQUESTED cp_group; with cp_group=None the
+        globally most recently touched group's key (reference
+        magi_attn_interface.py:1136)."""
+        if cp_group is not None:
+            from ..dist_attn_runtime_mgr import _group_tag
+
+            cache = self._per_group.get(_group_tag(cp_group))
+            return cache.most_recent_key() if cache is not None else None
+        for tag in sorted(self._tag_stamp, key=self._tag_stamp.get,
+                          reverse=True):
+            cache = self._per_group.get(tag)
+            if cache is not None:
+                k = cache.most_recent_key()
+                if k is not None:
+                    return k
         return None
 
     def clear(self, cp_group=None):
-        self._per_group.clear()
+        if cp_group is None:
+            self._per_group.clear()
+            self._tag_stamp.clear()
+            return
+        from ..dist_attn_runtime_mgr import _group_tag
+
+        tag = _group_tag(cp_group)
+        self._per_group.pop(tag, None)
+        self._tag_stamp.pop(tag, None)
 
 
 dist_attn_runtime_dict_mgr = DistAttnRuntimeDictManager()

@@ -205,12 +205,13 @@ class DistAttnRuntime:
             arg = self.calc_meta.stage_args[s]
             stage_kv = works[s].wait_post_process()
             S = self.comm_meta.stages_cast[s].stage_tokens
-            if S == 0:
-                continue
+            # group_reduce is a COLLECTIVE: every rank must join every stage
+            # even with an empty local partial (this rank may still RECEIVE
+            # dK/dV contributions for KV rows it hosts).
             dkv_stage = torch.zeros(
                 2 * S, *k.shape[1:], dtype=torch.float32, device=q.device
             )
-            if not arg.is_empty():
+            if S > 0 and not arg.is_empty():
                 self._bwd_partial(
                     dout, q, stage_kv[:S], stage_kv[S:], out, lse, dpsum, arg,
                     dq_acc, dkv_stage[:S], dkv_stage[S:], scale, softcap,

@@ -223,3 +223,73 @@ def test_new_mask_after_dispatch():
     port = _free_port()
     mp.spawn(_worker_new_mask, args=(2, port, None, None, None), nprocs=2,
              join=True)
+
+
+def _worker_asym_reduce(rank, ws, port, _a, _b, _c):
+    """Regression: with SequentialDispatchAlg + causal, rank 0
+    hosts the earliest chunks so its REMOTE need is empty (stage_tokens == 0)
+    while the other ranks compute partial dK/dV for rank 0's KV rows. The
+    backward group_reduce is a collective: rank 0 must still join every stage
+    to RECEIVE those contributions."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=ws)
+    try:
+        import magi_attention.functional.dist_attn as da
+        from magi_attention.api import calc_attn, dispatch, magi_attn_flex_key
+        from magi_attention.common.ranges import AttnRanges
+        from magi_attention.config import (
+            DispatchConfig,
+            DistAttnConfig,
+            OverlapConfig,
+            SequentialDispatchAlg,
+        )
+        from tests.dist_backend import OracleBackend
+
+        da.register_test_attn_backend(OracleBackend)
+        total, hq, hk, d = 512, 4, 2, 32
+        g = torch.Generator().manual_seed(31)
+        q = torch.randn(total, hq, d, generator=g, dtype=torch.float64)
+        k = torch.randn(total, hk, d, generator=g, dtype=torch.float64)
+        v = torch.randn(total, hk, d, generator=g, dtype=torch.float64)
+        dout = torch.randn(total, hq, d, generator=g, dtype=torch.float64)
+        cfg = DistAttnConfig(
+            dispatch_config=DispatchConfig(
+                chunk_size=64, alg=SequentialDispatchAlg()
+            ),
+            overlap_config=OverlapConfig(degree=1, min_chunk_size=32),
+        )
+        key = magi_attn_flex_key(
+            AttnRanges.from_ranges([[0, total]]),
+            AttnRanges.from_ranges([[0, total]]),
+            [1], total, total, hq, hk, d,
+            cp_group_or_mesh=dist.group.WORLD, dist_attn_config=cfg,
+        )
+        ql = dispatch(q, key).requires_grad_(True)
+        kl = dispatch(k, key).requires_grad_(True)
+        vl = dispatch(v, key).requires_grad_(True)
+        out_l, _ = calc_attn(ql, kl, vl, key)
+        dout_l = dispatch(dout, key)
+        (out_l * dout_l).sum().backward()
+
+        mask = make_attn_mask(total, total, [[0, total]], [[0, total]], [1])
+        qg = q.clone().requires_grad_(True)
+        kg = k.clone().requires_grad_(True)
+        vg = v.clone().requires_grad_(True)
+        ro, _ = ref_attn(qg, kg, vg, mask)
+        (ro * dout).sum().backward()
+        from magi_attention.api import get_position_ids
+
+        pos = get_position_ids(key)
+        torch.testing.assert_close(ql.grad, qg.grad[pos], atol=1e-5, rtol=1e-4)
+        torch.testing.assert_close(kl.grad, kg.grad[pos], atol=1e-5, rtol=1e-4)
+        torch.testing.assert_close(vl.grad, vg.grad[pos], atol=1e-5, rtol=1e-4)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("ws", [2, 4])
+def test_asymmetric_stage_reduce(ws):
+    port = _free_port()
+    mp.spawn(_worker_asym_reduce, args=(ws, port, None, None, None), nprocs=ws,
+             join=True)
