@@ -55,31 +55,37 @@ DEV_INLINE bool wave_alive(int m0, int qe) { return m0 < qe; }
 // Probed semantics (tests/gpu_probe_tr16.py): within each 16-lane group, the
 // addresses of lanes =0 (mod 4) give four ROW base addresses; lane l receives
 // element (row_j_base + (l&15)) for j=0..3. Two reads cover 8 rows.
+// r2: the r1 inline-asm version forced `s_waitcnt lgkmcnt(0)` + a sched
+// fence per fragment — a full ~50-cycle LDS-latency park before every second
+// MFMA of the output matmuls. The clang builtin is the same instruction but
+// scheduler-visible: reads pipeline across fragments and waits batch.
 DEV_INLINE bf16x8 tr16_frag(int a0, int a1) {
-  unsigned long long v0, v1;
-  asm volatile(
-      "ds_read_b64_tr_b16 %0, %2\n\t"
-      "ds_read_b64_tr_b16 %1, %3\n\t"
-      "s_waitcnt lgkmcnt(0)"
-      : "=&v"(v0), "=&v"(v1)
-      : "v"(a0), "v"(a1)
-      : "memory");
-  __builtin_amdgcn_sched_barrier(0);
+  typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4_;
+  bf16x4_ v0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (__attribute__((address_space(3))) bf16x4_*)(unsigned)a0);
+  bf16x4_ v1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (__attribute__((address_space(3))) bf16x4_*)(unsigned)a1);
   union {
-    unsigned long long u[2];
+    bf16x4_ h[2];
     bf16x8 v;
   } r;
-  r.u[0] = v0;
-  r.u[1] = v1;
+  r.h[0] = v0;
+  r.h[1] = v1;
   return r.v;
 }
 
+// one packed convert (RNE, same as the scalar bf16 cast) — the C version
+// lowers to 2 cvt + shift + or, 4x the issue slots (guide: cvt_pk idiom)
 DEV_INLINE unsigned pack_bf16_pair(float lo, float hi) {
-  union { __bf16 b; unsigned short u; } a, b;
-  a.b = (__bf16)lo;
-  b.b = (__bf16)hi;
-  return ((unsigned)b.u << 16) | a.u;
+  unsigned r;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
 }
+
+// raw v_exp_f32: the libm exp2f expands to ~5 instructions of denormal-range
+// guards; specials (inf/NaN) behave identically, only sub-denormal precision
+// differs (P < 1e-38 ~ 0)
+DEV_INLINE float fast_exp2(float x) { return __builtin_amdgcn_exp2f(x); }
 
 DEV_INLINE bf16x8 cframe_to_afrag(const float* val, int tt) {
   unsigned c0 = pack_bf16_pair(val[8 * tt + 0], val[8 * tt + 1]);
@@ -162,19 +168,24 @@ __global__ __launch_bounds__(256) void bwd_preprocess_kernel(BwdParams p, int d,
 
 
 // ---------------- mainloop ----------------
-// MODE 0: fused dK+dV (1 wave/SIMD — 128 f32 accumulators).
-// MODE 1: dV only / MODE 2: dK only — each recomputes S but fits 2 waves/SIMD,
-// which beats the fused kernel's naked single-wave stalls (the dq kernel went
-// 4.1x faster at occupancy 2 with the same pipeline).
-// WAVES: k-tiles per workgroup sharing ONE staged Q/dO image. The split
-// modes run 8 waves (a single 512-thread WG per CU = 2 waves/SIMD) so the
-// staging+barrier cost per MFMA halves; the fused mode's 437-VGPR waves
-// cannot co-reside 2/SIMD, so it stays at 4 waves.
-template <int D, bool HAS_SOFTCAP, int MODE, int WAVES>
+// MODE 0: fused dK+dV — r2 redesign: V fragments live in a per-wave LDS tile
+// (staged once per block) instead of 32 persistent VGPRs, which brings the
+// fused kernel under the 256-reg budget of 2 waves/SIMD (r1's fused version
+// was 437 regs = 1 wave/SIMD, or 65 scratch spills RELOADED EVERY ITERATION
+// when forced to 256). Shares S, P, exp2 and the staged Q/dO image between
+// the dV and dK matmuls: 32 MFMAs per subtile vs the split modes' 40.
+// MODE 1: dV only / MODE 2: dK only — the r1 split, kept selectable.
+// WAVES: k-tiles per workgroup sharing ONE staged Q/dO image.
+// NBUF: LDS staging ring depth. 2 = r1 behaviour (barrier drains vmcnt(0),
+// prefetch has one compute phase to land). 3 = constant-distance vmcnt(G)
+// barrier: prefetch has TWO compute phases, and the barrier only waits for
+// the one stage it needs — attacks the 34-42% SQ_WAIT_ANY of the r1 PMC.
+template <int D, bool HAS_SOFTCAP, int MODE, int WAVES, int NBUF>
 __global__ __launch_bounds__(64 * WAVES, 1)
 void ffa_bwd_dkv_kernel(BwdParams p) {
   constexpr bool WANT_DV = MODE != 2;
   constexpr bool WANT_DK = MODE != 1;
+  constexpr bool V_IN_LDS = MODE == 0;
   constexpr int DF = D / 16;
   constexpr int DT = D / 32;
   constexpr int ROWB = D * 2;  // bytes per LDS tile row
@@ -229,12 +240,14 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   // ONE shared object only: a second __shared__ array makes hipcc drain
   // vmcnt(0) before every ds_read, destroying the glds pipeline (guide §5
   // ".s-level traps" (a)).
-  // 2 buffers x 64-row Q/dO images: one barrier per 64 q rows — the doubled
-  // compute phase covers the prefetch latency a 32-row phase could not, with
-  // HALF the barrier parking (PMC: SQ_WAIT_ANY 35-46%% at 32-row iterations).
+  // NBUF x 64-row Q/dO images: one barrier per 64 q rows — the doubled
+  // compute phase covers the prefetch latency a 32-row phase could not.
   constexpr int QITER = 2 * BWD_BM;
+  constexpr int VLDS = V_IN_LDS ? WAVES * BWD_BN * D * 2 : 0;
+  static_assert(NBUF * 2 * QITER * (D * 2 + 4) + VLDS <= 163840,
+                "LDS budget");
   __shared__ __attribute__((aligned(16))) char smem[
-      2 * 2 * QITER * D * 2 + 2 * 2 * QITER * 4];
+      NBUF * 2 * QITER * D * 2 + NBUF * 2 * QITER * 4 + (VLDS ? VLDS : 1)];
   auto lds_q = [&](int buf) -> __bf16* {
     return (__bf16*)(smem + (2 * buf) * QITER * D * 2);
   };
@@ -245,20 +258,47 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   // 32-63 dpsum), one call per 32-row group: layout per buffer is
   // [lse g0 (32) | dps g0 (32) | lse g1 (32) | dps g1 (32)]
   auto lds_lse = [&](int buf) -> float* {
-    return (float*)(smem + 4 * QITER * D * 2 + buf * 2 * QITER * 4);
+    return (float*)(smem + NBUF * 2 * QITER * D * 2 + buf * 2 * QITER * 4);
   };
+  // MODE 0: per-wave V tile (BWD_BN rows x D, same 32-B XOR swizzle as the
+  // Q/dO images so the dP B-fragment reads reuse the qf addressing)
+  __bf16* lds_vt = (__bf16*)(smem + NBUF * 2 * QITER * D * 2 +
+                             NBUF * 2 * QITER * 4) + wave * BWD_BN * D;
 
-  // K/V fragments (A-layout) + K B-fragments, loaded once per block
+  // K fragments (A-layout), loaded once per block; V fragments likewise in
+  // the split dK mode — the fused mode stages V into LDS instead (register
+  // budget).
   const int krow = n0 + lo32;
   const int kcl = min(krow, ke - 1);
-  bf16x8 kfA[DF], vfA[WANT_DK ? DF : 1];
-  {
-    const bf16_t* kp = p.k + (size_t)kcl * k_pitch + (size_t)kh * D;
+  // MODE 0 does NOT keep K fragments live across the q loop: the lane re-reads
+  // its own K row per subtile (8 b128 loads, L1/L2-resident) so the registers
+  // are subtile-local — keeping them loop-live was exactly what pushed the
+  // fused kernel to 437 regs / scratch spills.
+  const bf16_t* kp_row = p.k + (size_t)kcl * k_pitch + (size_t)kh * D + hi * 8;
+  bf16x8 kfA[V_IN_LDS ? 1 : DF], vfA[(WANT_DK && !V_IN_LDS) ? DF : 1];
+  if constexpr (!V_IN_LDS) {
     const bf16_t* vp = p.v + (size_t)kcl * k_pitch + (size_t)kh * D;
 #pragma unroll
     for (int dd = 0; dd < DF; ++dd) {
-      kfA[dd] = *(const bf16x8*)(kp + dd * 16 + hi * 8);
-      if constexpr (WANT_DK) vfA[dd] = *(const bf16x8*)(vp + dd * 16 + hi * 8);
+      kfA[dd] = *(const bf16x8*)(kp_row - hi * 8 + dd * 16 + hi * 8);
+      if constexpr (WANT_DK)
+        vfA[dd] = *(const bf16x8*)(vp + dd * 16 + hi * 8);
+    }
+  }
+  constexpr int ROWS_PER_GLDS_V = 1024 / ROWB;
+  if constexpr (V_IN_LDS) {
+#pragma unroll
+    for (int gi = 0; gi < BWD_BN / ROWS_PER_GLDS_V; ++gi) {
+      const int r0v = ROWS_PER_GLDS_V * gi;
+      const int r = r0v + lane / (ROWB / 16);
+      const int c = lane % (ROWB / 16);
+      const int kr = min(n0 + r, ke - 1);
+      const int csw = (c ^ ((r & SW32M) << 1)) * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              p.v + (size_t)kr * k_pitch + (size_t)kh * D + csw),
+          (__attribute__((address_space(3))) unsigned int*)&lds_vt[r0v * D],
+          16, 0, 0);
     }
   }
 
@@ -320,6 +360,9 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
     }
   };
 
+  // vm ops per stage call per wave (glds pairs + lse groups) — the NBUF=3
+  // constant-distance barrier counts on EVERY stage issuing exactly this many
+  constexpr int GOPS = 2 * GLDS_PER_WAVE + 2;
   int cur = 0;
   for (int seg = seg0; seg < seg1; ++seg) {
   qs = p.q_ranges[2 * seg];
@@ -330,10 +373,18 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   if (q_lo >= q_hi) continue;
   cur = 0;
   stage_glds(0, q_lo);
+  if constexpr (NBUF == 3) stage_glds(1, q_lo + QITER);  // rows clamp to qe-1
 
   for (int m0 = q_lo; m0 < q_hi; m0 += QITER) {
-    pipe_barrier<0>();  // buf[cur] ready (prefetch had a full 64-row phase)
-    if (m0 + QITER < q_hi) stage_glds(cur ^ 1, m0 + QITER);
+    // NBUF=3: wait ONLY for buf[cur] (leave the next stage in flight), then
+    // prefetch two stages ahead — always, with clamped rows, so the vmcnt
+    // distance stays constant. NBUF=2: full drain (prefetch had one phase).
+    pipe_barrier<NBUF == 3 ? GOPS : 0>();
+    if constexpr (NBUF == 3) {
+      stage_glds(cur == 0 ? 2 : cur - 1, m0 + 2 * QITER);
+    } else {
+      if (m0 + QITER < q_hi) stage_glds(cur ^ 1, m0 + QITER);
+    }
 #pragma unroll
     for (int sub = 0; sub < 2; ++sub) {
     const int ms = m0 + sub * BWD_BM;
@@ -348,14 +399,28 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
       // so the dV/dK A-fragments come from the in-register permlane transform
       // (cframe) instead of an LDS round-trip ----
       f32x16 s = (f32x16)(0.f), dp = (f32x16)(0.f);
+      bf16x8 kfs[V_IN_LDS ? DF : 1];
+      if constexpr (V_IN_LDS) {
+#pragma unroll
+        for (int dd = 0; dd < DF; ++dd)
+          kfs[dd] = *(const bf16x8*)(kp_row + dd * 16);
+      }
 #pragma unroll
       for (int dd = 0; dd < DF; ++dd) {
         const int off = swz(lo32, lo32 * ROWB + dd * 32 + hi * 16);
         bf16x8 qf = *(const bf16x8*)((const char*)lqb + off);
-        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kfA[dd], s, 0, 0, 0);
+        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            qf, V_IN_LDS ? kfs[dd] : kfA[dd], s, 0, 0, 0);
         if constexpr (WANT_DK) {
           bf16x8 dof = *(const bf16x8*)((const char*)ldb + off);
-          dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vfA[dd], dp, 0, 0, 0);
+          bf16x8 vf;
+          if constexpr (V_IN_LDS)
+            // lane holds V-tile row lo32 cols dd*16+hi*8 — identical
+            // addressing to qf, off the wave's swizzled V tile
+            vf = *(const bf16x8*)((const char*)lds_vt + off);
+          else
+            vf = vfA[dd];
+          dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vf, dp, 0, 0, 0);
         }
       }
 
@@ -381,7 +446,7 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int rl = crow(r, hi);
-          const float pij = exp2f(s[r] * sl2 - lse_t[rl] * log2e);
+          const float pij = fast_exp2(s[r] * sl2 - lse_t[rl] * log2e);
           if constexpr (WANT_DV) pv[r] = pij;
           if constexpr (WANT_DK)
             dsv[r] = pij * (dp[r] - dps_t[rl]) * p.scale;
@@ -405,7 +470,7 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
           } else {
             t = sv * sl2;
           }
-          const float pij = ok ? exp2f(t - lq * log2e) : 0.f;
+          const float pij = ok ? fast_exp2(t - lq * log2e) : 0.f;
           if constexpr (WANT_DV) pv[r] = pij;
           if constexpr (WANT_DK)
             dsv[r] = pij * (dp[r] - dps_t[crow(r, hi)]) * dscale;
@@ -419,15 +484,6 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
       //   addr = lds_base + row*ROWB + ((dcol*2) ^ ((row&7)<<5)) + (l&3)*8
       // (the XOR distributes because the swizzle bits live inside dcol*2).
       {
-        bf16x8 pa0, pa1, da0, da1;
-        if constexpr (WANT_DV) {
-          pa0 = cframe_to_afrag(pv, 0);
-          pa1 = cframe_to_afrag(pv, 1);
-        }
-        if constexpr (WANT_DK) {
-          da0 = cframe_to_afrag(dsv, 0);
-          da1 = cframe_to_afrag(dsv, 1);
-        }
         const int qhalf = (lane >> 4) & 1;       // which 16-d column half
         const int jrow = (lane & 15) >> 2;       // canonical row for this lane
         const int row0 = 8 * hi + jrow;          // rd = 0 rows
@@ -446,28 +502,59 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
         const int sw3 = ((row1 + 16) & SW32M) << 5;
         const int rb2 = (row0 + 16) * ROWB + lane8;
         const int rb3 = (row1 + 16) * ROWB + lane8;
+        // software-pipelined B-frag matmul: step j+1's fragment reads issue
+        // under step j's MFMA (see dq kernel note)
+        auto pipe_mm = [&](int tbase, bf16x8 a0, bf16x8 a1, f32x16* acc) {
+          bf16x8 bcur = tr16_frag(tbase + rb0 + ((16 * qhalf * 2) ^ sw0),
+                                  tbase + rb1 + ((16 * qhalf * 2) ^ sw1));
 #pragma unroll
-        for (int dt = 0; dt < DT; ++dt) {
-          const int dcol = (dt * 32 + 16 * qhalf) * 2;
-          if constexpr (WANT_DV) {
+          for (int j = 0; j < 2 * DT; ++j) {
+            bf16x8 bnext = bcur;
+            if (j + 1 < 2 * DT) {
+              const int dcoln = (((j + 1) >> 1) * 32 + 16 * qhalf) * 2;
+              if ((j + 1) & 1)
+                bnext = tr16_frag(tbase + rb2 + (dcoln ^ sw2),
+                                  tbase + rb3 + (dcoln ^ sw3));
+              else
+                bnext = tr16_frag(tbase + rb0 + (dcoln ^ sw0),
+                                  tbase + rb1 + (dcoln ^ sw1));
+            }
+            acc[j >> 1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                (j & 1) ? a1 : a0, bcur, acc[j >> 1], 0, 0, 0);
+            bcur = bnext;
+          }
+        };
+        if constexpr (MODE == 1) {
+          pipe_mm(do_base, cframe_to_afrag(pv, 0), cframe_to_afrag(pv, 1),
+                  acc_dv);
+        } else if constexpr (MODE == 2) {
+          pipe_mm(q_base, cframe_to_afrag(dsv, 0), cframe_to_afrag(dsv, 1),
+                  acc_dk);
+        } else {
+          // MODE 0 (fused dK+dV): INTERLEAVE the two independent MFMA
+          // streams — each fragment load's LDS latency is covered by the
+          // OTHER stream's in-flight MFMA, which a single serialized stream
+          // cannot do (the allocator coalesces any explicit double-buffer
+          // at this register pressure).
+          bf16x8 pa0 = cframe_to_afrag(pv, 0);
+          bf16x8 pa1 = cframe_to_afrag(pv, 1);
+          bf16x8 da0 = cframe_to_afrag(dsv, 0);
+          bf16x8 da1 = cframe_to_afrag(dsv, 1);
+#pragma unroll
+          for (int dt = 0; dt < DT; ++dt) {
+            const int dcol = (dt * 32 + 16 * qhalf) * 2;
             bf16x8 bdo = tr16_frag(do_base + rb0 + (dcol ^ sw0),
                                    do_base + rb1 + (dcol ^ sw1));
             acc_dv[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                 pa0, bdo, acc_dv[dt], 0, 0, 0);
-          }
-          if constexpr (WANT_DK) {
             bf16x8 bq = tr16_frag(q_base + rb0 + (dcol ^ sw0),
                                   q_base + rb1 + (dcol ^ sw1));
             acc_dk[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                 da0, bq, acc_dk[dt], 0, 0, 0);
-          }
-          if constexpr (WANT_DV) {
             bf16x8 bdo1 = tr16_frag(do_base + rb2 + (dcol ^ sw2),
                                     do_base + rb3 + (dcol ^ sw3));
             acc_dv[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                 pa1, bdo1, acc_dv[dt], 0, 0, 0);
-          }
-          if constexpr (WANT_DK) {
             bf16x8 bq1 = tr16_frag(q_base + rb2 + (dcol ^ sw2),
                                    q_base + rb3 + (dcol ^ sw3));
             acc_dk[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
@@ -477,7 +564,7 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
       }
     }
     }  // sub
-    cur ^= 1;
+    cur = (NBUF == 3) ? (cur == 2 ? 0 : cur + 1) : (cur ^ 1);
   }
   __syncthreads();  // this segment's LDS reads retired before restaging
   }  // seg
@@ -512,7 +599,7 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
 // ablation measured at ~45% of a fused backward. K/V tiles are staged
 // cooperatively per iteration (row-major swizzled for the S^T/dP^T A-frags,
 // plus a transposed copy for the dQ B-frags).
-template <int D, bool HAS_SOFTCAP, int WAVES>
+template <int D, bool HAS_SOFTCAP, int WAVES, int NBUF>
 __global__ __launch_bounds__(64 * WAVES, WAVES == 8 ? 1 : 2)
 void ffa_bwd_dq_kernel(BwdParams p) {
   constexpr int DF = D / 16;
@@ -558,7 +645,8 @@ void ffa_bwd_dq_kernel(BwdParams p) {
   // compute phase covers the prefetch latency a 32-row phase could not, with
   // HALF the barrier parking (PMC: SQ_WAIT_ANY 36% at 32-row iterations).
   constexpr int KITER = 2 * BWD_BN;
-  __shared__ __attribute__((aligned(16))) char smem[2 * 2 * KITER * D * 2];
+  static_assert(NBUF * 2 * KITER * D * 2 <= 163840, "LDS budget");
+  __shared__ __attribute__((aligned(16))) char smem[NBUF * 2 * KITER * D * 2];
   auto lds_k = [&](int buf) -> __bf16* {
     return (__bf16*)(smem + (2 * buf) * KITER * D * 2);
   };
@@ -622,6 +710,7 @@ void ffa_bwd_dq_kernel(BwdParams p) {
     }
   };
 
+  constexpr int GOPS = 2 * GLDS_PER_WAVE;  // vm ops per stage per wave
   int cur = 0;
   for (int seg = seg0; seg < seg1; ++seg) {
   ks = p.k_ranges[2 * seg];
@@ -632,10 +721,17 @@ void ffa_bwd_dq_kernel(BwdParams p) {
   if (k_lo >= k_hi) continue;
   cur = 0;
   stage_glds(0, k_lo);
+  if constexpr (NBUF == 3) stage_glds(1, k_lo + KITER);  // rows clamp to ke-1
 
   for (int n0 = k_lo; n0 < k_hi; n0 += KITER) {
-    pipe_barrier<0>();  // buf[cur] ready (prefetch had a full 64-row phase)
-    if (n0 + KITER < k_hi) stage_glds(cur ^ 1, n0 + KITER);
+    // NBUF=3: constant-distance barrier — wait only for buf[cur], keep the
+    // next stage in flight, prefetch two ahead (clamped). NBUF=2: full drain.
+    pipe_barrier<NBUF == 3 ? GOPS : 0>();
+    if constexpr (NBUF == 3) {
+      stage_glds(cur == 0 ? 2 : cur - 1, n0 + 2 * KITER);
+    } else {
+      if (n0 + KITER < k_hi) stage_glds(cur ^ 1, n0 + KITER);
+    }
 #pragma unroll
     for (int sub = 0; sub < 2; ++sub) {
     const int ns = n0 + sub * BWD_BN;
@@ -664,7 +760,7 @@ void ffa_bwd_dq_kernel(BwdParams p) {
         const float lsc = lse_q * log2e;
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          const float pij = exp2f(sA[r] * sl2 - lsc);
+          const float pij = fast_exp2(sA[r] * sl2 - lsc);
           dsv[r] = pij * (dpA[r] - dpsum_q) * p.scale;
         }
       } else {
@@ -684,7 +780,7 @@ void ffa_bwd_dq_kernel(BwdParams p) {
           } else {
             t = sv * sl2;
           }
-          const float pij = ok ? exp2f(t - lse_q * log2e) : 0.f;
+          const float pij = ok ? fast_exp2(t - lse_q * log2e) : 0.f;
           dsv[r] = pij * (dpA[r] - dpsum_q) * dscale;
         }
       }
@@ -710,22 +806,31 @@ void ffa_bwd_dq_kernel(BwdParams p) {
         const int sw3 = ((row1 + 16) & SW32M) << 5;
         const int rb2 = (row0 + 16) * ROWB + lane8;
         const int rb3 = (row1 + 16) * ROWB + lane8;
+        // software-pipelined: step j+1's fragment reads issue under step j's
+        // MFMA so the ~50-cycle LDS latency hides under the matrix op (the
+        // serialized read->wait->mfma form parks the wave per fragment)
+        bf16x8 bcur = tr16_frag(k_base + rb0 + ((16 * qhalf2 * 2) ^ sw0),
+                                k_base + rb1 + ((16 * qhalf2 * 2) ^ sw1));
 #pragma unroll
-        for (int dt = 0; dt < DT; ++dt) {
-          const int dcol = (dt * 32 + 16 * qhalf2) * 2;
-          bf16x8 b0 = tr16_frag(k_base + rb0 + (dcol ^ sw0),
-                                k_base + rb1 + (dcol ^ sw1));
-          acc_dq[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa0, b0,
-                                                              acc_dq[dt], 0, 0, 0);
-          bf16x8 b1 = tr16_frag(k_base + rb2 + (dcol ^ sw2),
-                                k_base + rb3 + (dcol ^ sw3));
-          acc_dq[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa1, b1,
-                                                              acc_dq[dt], 0, 0, 0);
+        for (int j = 0; j < 2 * DT; ++j) {
+          bf16x8 bnext = bcur;
+          if (j + 1 < 2 * DT) {
+            const int dcoln = (((j + 1) >> 1) * 32 + 16 * qhalf2) * 2;
+            if ((j + 1) & 1)
+              bnext = tr16_frag(k_base + rb2 + (dcoln ^ sw2),
+                                k_base + rb3 + (dcoln ^ sw3));
+            else
+              bnext = tr16_frag(k_base + rb0 + (dcoln ^ sw0),
+                                k_base + rb1 + (dcoln ^ sw1));
+          }
+          acc_dq[j >> 1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              (j & 1) ? dsa1 : dsa0, bcur, acc_dq[j >> 1], 0, 0, 0);
+          bcur = bnext;
         }
       }
     }
     }  // sub
-    cur ^= 1;
+    cur = (NBUF == 3) ? (cur == 2 ? 0 : cur + 1) : (cur ^ 1);
   }
   __syncthreads();  // this segment's LDS reads retired before restaging
   }  // seg
@@ -817,7 +922,11 @@ extern "C" int magi_ffa_bwd_dq(const magi_ffa_bwd_args* a) {
   // for LONG ranges: halves staging+barrier cost per MFMA (measured
   // 67->63 ms at 64k). Short ranges keep 4 waves — the 256-row block
   // windows over-iterate masked edges on 2k varlen docs.
+  // W8 runs the 3-slot LDS ring (constant-distance vmcnt barrier); W4 cannot
+  // (3 buffers exceed the 80 KB/WG budget of 2 blocks/CU) and keeps 2.
   const int dqw = a->max_seqlen_k >= 8192 ? 8 : 4;
+  int nbuf = 3;
+  { const char* e = getenv("MAGI_BWD_NBUF"); if (e) nbuf = atoi(e); }
   const int qspan = BWD_BM * dqw;
   const int qblocks = (int)((a->total_q + qspan - 1) / qspan);
   if (a->n_ranges > 65535) return -5;
@@ -826,15 +935,19 @@ extern "C" int magi_ffa_bwd_dq(const magi_ffa_bwd_args* a) {
   dim3 block(64 * dqw);
   hipStream_t s = (hipStream_t)a->stream;
   const bool sc = a->softcap > 0.f;
-#define LAUNCH_DQ(DD, SC, WW) \
-  hipLaunchKernelGGL((ffa_bwd_dq_kernel<DD, SC, WW>), grid_q, block, 0, s, p)
-  if (a->d == 64) {
-    if (dqw == 8) { if (sc) LAUNCH_DQ(64, true, 8); else LAUNCH_DQ(64, false, 8); }
-    else          { if (sc) LAUNCH_DQ(64, true, 4); else LAUNCH_DQ(64, false, 4); }
-  } else {
-    if (dqw == 8) { if (sc) LAUNCH_DQ(128, true, 8); else LAUNCH_DQ(128, false, 8); }
-    else          { if (sc) LAUNCH_DQ(128, true, 4); else LAUNCH_DQ(128, false, 4); }
-  }
+#define LAUNCH_DQ(DD, SC, WW, NB) \
+  hipLaunchKernelGGL((ffa_bwd_dq_kernel<DD, SC, WW, NB>), grid_q, block, 0, s, p)
+#define PICK_DQ(DD) \
+  do { \
+    if (dqw == 8) { \
+      if (nbuf == 3) { if (sc) LAUNCH_DQ(DD, true, 8, 3); else LAUNCH_DQ(DD, false, 8, 3); } \
+      else           { if (sc) LAUNCH_DQ(DD, true, 8, 2); else LAUNCH_DQ(DD, false, 8, 2); } \
+    } else { \
+      if (sc) LAUNCH_DQ(DD, true, 4, 2); else LAUNCH_DQ(DD, false, 4, 2); \
+    } \
+  } while (0)
+  if (a->d == 64) PICK_DQ(64); else PICK_DQ(128);
+#undef PICK_DQ
 #undef LAUNCH_DQ
   return (int)hipGetLastError();
 }
@@ -844,12 +957,16 @@ static int launch_bwd_dkv(const magi_ffa_bwd_args* a) {
   BwdParams p;
   int rc = fill_bwd_params(a, &p);
   if (rc) return rc > 0 ? 0 : rc;
-  // dK pass (MODE 2) runs 8 waves per WG for LONG ranges (one staged Q/dO
-  // image shared by all; measured 95->83 ms at 64k, register spill gone).
-  // dV (MODE 1) stays at 4 — its shorter MFMA chain loses more to the wider
-  // block windows than it saves in staging. Fused (MODE 0) can't fit 8.
+  // dK (MODE 2) and fused (MODE 0) run 8 waves per WG for LONG ranges (one
+  // staged Q/dO image shared by all). dV (MODE 1) stays at 4 — its shorter
+  // MFMA chain loses more to the wider block windows than it saves in
+  // staging. The fused mode keeps NBUF=2 (its V tiles + a 3-ring exceed the
+  // 160 KB LDS); the split W8 modes default to the 3-slot ring.
   const bool big = a->max_seqlen_k >= 8192;
-  const int W = (MODE == 2 && big) ? 8 : 4;
+  const int W = ((MODE == 2 || MODE == 0) && big) ? 8 : 4;
+  int nbuf = (MODE == 0 || W == 4) ? 2 : 3;
+  { const char* e = getenv("MAGI_BWD_NBUF");
+    if (e && nbuf == 3) nbuf = atoi(e); }
   const int span = BWD_BN * W;
   const int nblocks = (a->max_seqlen_k + span - 1) / span;
   if (a->n_ranges > 65535) return -5;
@@ -859,16 +976,28 @@ static int launch_bwd_dkv(const magi_ffa_bwd_args* a) {
   dim3 block(64 * W);
   hipStream_t s = (hipStream_t)a->stream;
   const bool sc = a->softcap > 0.f;
-#define LAUNCH_DKV(DD, SC, WW) \
-  hipLaunchKernelGGL((ffa_bwd_dkv_kernel<DD, SC, MODE, WW>), grid_kv, block, \
-                     0, s, p)
-  if (a->d == 64) {
-    if (W == 8) { if (sc) LAUNCH_DKV(64, true, 8); else LAUNCH_DKV(64, false, 8); }
-    else        { if (sc) LAUNCH_DKV(64, true, 4); else LAUNCH_DKV(64, false, 4); }
-  } else {
-    if (W == 8) { if (sc) LAUNCH_DKV(128, true, 8); else LAUNCH_DKV(128, false, 8); }
-    else        { if (sc) LAUNCH_DKV(128, true, 4); else LAUNCH_DKV(128, false, 4); }
-  }
+#define LAUNCH_DKV(DD, SC, WW, NB) \
+  hipLaunchKernelGGL((ffa_bwd_dkv_kernel<DD, SC, MODE, WW, NB>), grid_kv, \
+                     block, 0, s, p)
+#define PICK_DKV(DD) \
+  do { \
+    if (W == 8) { \
+      bool done = false; \
+      if constexpr (MODE != 0) { /* MODE0 + 3-ring exceeds 160 KB LDS */ \
+        if (nbuf == 3) { \
+          if (sc) LAUNCH_DKV(DD, true, 8, 3); else LAUNCH_DKV(DD, false, 8, 3); \
+          done = true; \
+        } \
+      } \
+      if (!done) { \
+        if (sc) LAUNCH_DKV(DD, true, 8, 2); else LAUNCH_DKV(DD, false, 8, 2); \
+      } \
+    } else { \
+      if (sc) LAUNCH_DKV(DD, true, 4, 2); else LAUNCH_DKV(DD, false, 4, 2); \
+    } \
+  } while (0)
+  if (a->d == 64) PICK_DKV(64); else PICK_DKV(128);
+#undef PICK_DKV
 #undef LAUNCH_DKV
   return (int)hipGetLastError();
 }

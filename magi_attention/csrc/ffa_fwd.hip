@@ -39,34 +39,40 @@ using bf16_t = __bf16;
 
 DEV_INLINE float warp_xor32(float v) { return __shfl_xor(v, 32, 64); }
 
+// one packed convert (RNE, same as the scalar bf16 cast) — the C version
+// lowers to 2 cvt + shift + or, 4x the issue slots (guide: cvt_pk idiom)
 DEV_INLINE unsigned pack_bf16_pair(float lo, float hi) {
-  union { __bf16 b; unsigned short u; } a, b;
-  a.b = (__bf16)lo;
-  b.b = (__bf16)hi;
-  return ((unsigned)b.u << 16) | a.u;
+  unsigned r;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
 }
+
+// raw v_exp_f32: the libm exp2f expands to ~5 instructions of denormal-range
+// guards; specials (inf/NaN) behave identically, only sub-denormal precision
+// differs (P < 1e-38 ~ 0)
+DEV_INLINE float fast_exp2(float x) { return __builtin_amdgcn_exp2f(x); }
 
 // C/D fragment row for v_mfma_f32_32x32x16_bf16: reg r, lane-half hi
 DEV_INLINE int crow(int r, int hi) { return (r & 3) + 8 * (r >> 2) + 4 * hi; }
 
 // ds_read_b64_tr_b16 pair -> one MFMA B-fragment (see ffa_bwd.hip for the
 // probed semantics).
+// r2: the r1 inline-asm version forced `s_waitcnt lgkmcnt(0)` + a sched
+// fence per fragment — a full ~50-cycle LDS-latency park before every second
+// MFMA of the output matmuls. The clang builtin is the same instruction but
+// scheduler-visible: reads pipeline across fragments and waits batch.
 DEV_INLINE bf16x8 tr16_frag(int a0, int a1) {
-  unsigned long long v0, v1;
-  asm volatile(
-      "ds_read_b64_tr_b16 %0, %2\n\t"
-      "ds_read_b64_tr_b16 %1, %3\n\t"
-      "s_waitcnt lgkmcnt(0)"
-      : "=&v"(v0), "=&v"(v1)
-      : "v"(a0), "v"(a1)
-      : "memory");
-  __builtin_amdgcn_sched_barrier(0);
+  typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4_;
+  bf16x4_ v0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (__attribute__((address_space(3))) bf16x4_*)(unsigned)a0);
+  bf16x4_ v1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (__attribute__((address_space(3))) bf16x4_*)(unsigned)a1);
   union {
-    unsigned long long u[2];
+    bf16x4_ h[2];
     bf16x8 v;
   } r;
-  r.u[0] = v0;
-  r.u[1] = v1;
+  r.h[0] = v0;
+  r.h[1] = v1;
   return r.v;
 }
 
@@ -265,14 +271,14 @@ __global__ __launch_bounds__(64 * WAVES, 8 / WAVES) void ffa_fwd_kernel(FwdParam
 
     const float m_new = fmaxf(m_run, mx);
     const float m_use = (m_new == -INFINITY) ? 0.f : m_new;
-    const float alpha = (m_run == -INFINITY) ? 0.f : exp2f(m_run - m_use);
+    const float alpha = (m_run == -INFINITY) ? 0.f : fast_exp2(m_run - m_use);
     m_run = m_new;
 
     float pr[16];
     float psum = 0.f;
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
-      pr[r] = exp2f(t[r] - m_use);   // exp2(-inf)=0 for masked
+      pr[r] = fast_exp2(t[r] - m_use);   // exp2(-inf)=0 for masked
       psum += pr[r];
     }
     l_run = l_run * alpha + (psum + warp_xor32(psum));

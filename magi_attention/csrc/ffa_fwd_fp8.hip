@@ -32,6 +32,8 @@ using u8 = unsigned char;
 
 #define DEV_INLINE __device__ __forceinline__
 
+DEV_INLINE float fast_exp2(float x) { return __builtin_amdgcn_exp2f(x); }
+
 DEV_INLINE float warp_xor32(float v) { return __shfl_xor(v, 32, 64); }
 DEV_INLINE int crow(int r, int hi) { return (r & 3) + 8 * (r >> 2) + 4 * hi; }
 
@@ -231,7 +233,7 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_fp8_kernel(Fp8FwdParams p) {
 
     const float m_new = fmaxf(m_run, mx);
     const float m_use = (m_new == -INFINITY) ? 0.f : m_new;
-    const float alpha = (m_run == -INFINITY) ? 0.f : exp2f(m_run - m_use);
+    const float alpha = (m_run == -INFINITY) ? 0.f : fast_exp2(m_run - m_use);
     m_run = m_new;
 
     float pr[16];
@@ -239,7 +241,7 @@ __global__ __launch_bounds__(256, 2) void ffa_fwd_fp8_kernel(Fp8FwdParams p) {
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       // fp8 convention: exp2 output in [0, 256] (max_offset = 8)
-      pr[r] = exp2f(t[r] - m_use + MAX_OFFSET);
+      pr[r] = fast_exp2(t[r] - m_use + MAX_OFFSET);
       psum += pr[r];
     }
     l_run = l_run * alpha + (psum + warp_xor32(psum));
