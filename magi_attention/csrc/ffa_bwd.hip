@@ -216,6 +216,11 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
+  // static priority for the second-dispatched half of an 8-wave WG: the
+  // younger waves are the VALU-arbitration losers on every segment
+  // (MI355X guide, two-waves-per-SIMD item 4)
+  if (WAVES == 8 && wave >= 4) asm volatile("s_setprio 1");
+
   const int lo32 = lane & 31;
   const int hi = lane >> 5;
 
@@ -399,29 +404,41 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
       // so the dV/dK A-fragments come from the in-register permlane transform
       // (cframe) instead of an LDS round-trip ----
       f32x16 s = (f32x16)(0.f), dp = (f32x16)(0.f);
-      bf16x8 kfs[V_IN_LDS ? DF : 1];
       if constexpr (V_IN_LDS) {
+        // fused mode: the lane's K fragments re-load per subtile (keeping
+        // them loop-persistent overflows the 256-reg/2-wave budget). Issue
+        // the 8 loads FIRST, run the dP chain (LDS-only operands) while they
+        // fly, then the S chain consumes them with the latency paid off.
+        bf16x8 kfs[DF];
 #pragma unroll
         for (int dd = 0; dd < DF; ++dd)
           kfs[dd] = *(const bf16x8*)(kp_row + dd * 16);
-      }
+#pragma unroll
+        for (int dd = 0; dd < DF; ++dd) {
+          const int off = swz(lo32, lo32 * ROWB + dd * 32 + hi * 16);
+          bf16x8 dof = *(const bf16x8*)((const char*)ldb + off);
+          // lane holds V-tile row lo32 cols dd*16+hi*8 — identical
+          // addressing to qf, off the wave's swizzled V tile
+          bf16x8 vf = *(const bf16x8*)((const char*)lds_vt + off);
+          dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vf, dp, 0, 0, 0);
+        }
+#pragma unroll
+        for (int dd = 0; dd < DF; ++dd) {
+          const int off = swz(lo32, lo32 * ROWB + dd * 32 + hi * 16);
+          bf16x8 qf = *(const bf16x8*)((const char*)lqb + off);
+          s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kfs[dd], s, 0, 0, 0);
+        }
+      } else {
 #pragma unroll
       for (int dd = 0; dd < DF; ++dd) {
         const int off = swz(lo32, lo32 * ROWB + dd * 32 + hi * 16);
         bf16x8 qf = *(const bf16x8*)((const char*)lqb + off);
-        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            qf, V_IN_LDS ? kfs[dd] : kfA[dd], s, 0, 0, 0);
+        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kfA[dd], s, 0, 0, 0);
         if constexpr (WANT_DK) {
           bf16x8 dof = *(const bf16x8*)((const char*)ldb + off);
-          bf16x8 vf;
-          if constexpr (V_IN_LDS)
-            // lane holds V-tile row lo32 cols dd*16+hi*8 — identical
-            // addressing to qf, off the wave's swizzled V tile
-            vf = *(const bf16x8*)((const char*)lds_vt + off);
-          else
-            vf = vfA[dd];
-          dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vf, dp, 0, 0, 0);
+          dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vfA[dd], dp, 0, 0, 0);
         }
+      }
       }
 
       const int kk = n0 + lo32;  // this lane's k column
@@ -622,6 +639,11 @@ void ffa_bwd_dq_kernel(BwdParams p) {
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
+  // static priority for the second-dispatched half of an 8-wave WG: the
+  // younger waves are the VALU-arbitration losers on every segment
+  // (MI355X guide, two-waves-per-SIMD item 4)
+  if (WAVES == 8 && wave >= 4) asm volatile("s_setprio 1");
+
   const int lo32 = lane & 31;
   const int hi = lane >> 5;
 
