@@ -182,20 +182,194 @@ def measure_roofline(state, device):
     bwd_flops = host_fwd_flops * 2.5
 
     achieved = bwd_flops / (bwd_ms * 1e-3)
+    # traffic: per-launch FETCH+WRITE bytes of the dominant kernel (fused
+    # dkv), from the committed rocprofv3 PMC passes on this same workload
+    # (profiles/r2_pmc_traffic.json; FETCH corrected 2x per the gfx950
+    # half-count of wide coalesced reads; L3 hits included by the counter)
+    traffic = None
+    traffic_note = None
+    try:
+        import json as _json
+        with open(os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                               "profiles", "r2_pmc_traffic.json")) as f:
+            t = _json.load(f)["per_launch_bytes"]["ffa_bwd_dkv_fused"]
+        traffic = t["fetch_corrected"] + t["write_raw"]
+        traffic_note = "rocprofv3 FETCH_SIZE(x2)+WRITE_SIZE per launch, same workload (profiles/r2_pmc_traffic.json); L2-miss bytes incl. L3 hits"
+    except Exception:
+        pass
     return {
         "bound": "mfma",
         "achieved": achieved,
         "peak": MFMA_PEAK_BF16,
         "unit": "FLOP/s",
         "frac": achieved / MFMA_PEAK_BF16,
-        "traffic": None,
+        "traffic": traffic,
         "detail": {
-            "kernel": "ffa_bwd_dq_kernel + ffa_bwd_dkv_kernel<dv>/<dk> (dq pass concurrent with the dv/dk passes)",
+            "kernel": "ffa_bwd_dkv_kernel<fused> (dominant; dq pass co-scheduled on a side stream)",
             "bwd_ms_per_launch": bwd_ms,
             "fwd_ms_per_launch": fwd_ms,
             "fwd_achieved_flops_per_s": host_fwd_flops / (fwd_ms * 1e-3),
+            "traffic_note": traffic_note,
         },
     }
+
+
+def measure_extra_configs(device):
+    """BASELINE configs 2/4/5 + the north-star 8k dense fwd point, measured
+    at cp=1 through the product path. Emitted inside the contract line under
+    "extra_configs" so the driver's parse carries them."""
+    import torch.distributed as dist
+    from magi_attention.functional import flex_flash_attn_func
+
+    out = {}
+
+    def time_fn(fn, steps=5, warm=2):
+        for _ in range(warm):
+            fn()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(steps):
+            fn()
+        torch.cuda.synchronize()
+        return (time.perf_counter() - t0) / steps
+
+    # ---- 8k dense fwd (north-star roofline point) ----
+    n = 8192
+    g = torch.Generator().manual_seed(42)
+    q = (torch.randn(n, HQ, D, generator=g) * 0.5).bfloat16().to(device)
+    k = (torch.randn(n, HKV, D, generator=g) * 0.5).bfloat16().to(device)
+    v = (torch.randn(n, HKV, D, generator=g) * 0.5).bfloat16().to(device)
+    qr = torch.tensor([[0, n]], dtype=torch.int32, device=device)
+    tm = torch.tensor([1], dtype=torch.int32, device=device)
+
+    def fwd8k():
+        with torch.no_grad():
+            flex_flash_attn_func(q, k, v, qr, qr.clone(), tm, max_seqlen_q=n)
+
+    dt = time_fn(fwd8k, steps=10, warm=3)
+    fl = 4 * (n * (n + 1) // 2) * HQ * D
+    out["dense_8k_fwd"] = {
+        "tflops": fl / dt / 1e12, "ms": dt * 1e3,
+        "frac_of_mfma_peak": fl / dt / MFMA_PEAK_BF16,
+        "config": "seqlen 8192 h32 d128 bf16 causal, fwd only",
+    }
+    del q, k, v
+
+    # ---- config 2: varlen packed 16k (8 causal docs of 2k), hq16 d128 ----
+    n, hq, hkv = 16384, 16, 16
+    q = (torch.randn(n, hq, D, generator=g) * 0.5).bfloat16().to(device)
+    k = (torch.randn(n, hkv, D, generator=g) * 0.5).bfloat16().to(device)
+    v = (torch.randn(n, hkv, D, generator=g) * 0.5).bfloat16().to(device)
+    rs = [[i * 2048, (i + 1) * 2048] for i in range(8)]
+    qr = torch.tensor(rs, dtype=torch.int32, device=device)
+    tm = torch.tensor([1] * 8, dtype=torch.int32, device=device)
+    qg = q.clone().requires_grad_(True)
+    kg = k.clone().requires_grad_(True)
+    vg = v.clone().requires_grad_(True)
+    do = torch.randn_like(q)
+
+    def vfwdbwd():
+        o, _ = flex_flash_attn_func(qg, kg, vg, qr, qr.clone(), tm,
+                                    max_seqlen_q=2048, max_seqlen_k=2048)
+        o.backward(do)
+        qg.grad = kg.grad = vg.grad = None
+
+    dt = time_fn(vfwdbwd, steps=5, warm=2)
+    area = 8 * (2048 * 2049 // 2)
+    fl = 4 * area * hq * D * 3.5
+    out["varlen_16k_fwdbwd"] = {
+        "tflops": fl / dt / 1e12, "ms": dt * 1e3,
+        "config": "BASELINE config 2: varlen 8x2k causal docs, 16k tokens, "
+                  "h16 d128 bf16, fwd+bwd, cp1",
+    }
+    del q, k, v, qg, kg, vg, do
+
+    # ---- config 4: sliding-window + doc mask, mixed seqlens to 32k ----
+    from magi_attention.api import (
+        calc_attn, dispatch, infer_attn_mask_from_sliding_window,
+        magi_attn_flex_key,
+    )
+    from magi_attention.common.range import AttnRange
+    from magi_attention.common.ranges import AttnRanges
+    from magi_attention.config import (
+        DispatchConfig, DistAttnConfig, OverlapConfig,
+    )
+
+    hq4, hkv4 = 32, 32
+    doc_lens = [32768, 16384, 8192, 4096, 2048, 2048]  # fixed-seed mixed docs
+    total = sum(doc_lens)
+    q_ranges = AttnRanges()
+    k_ranges = AttnRanges()
+    types = []
+    pos = 0
+    for L in doc_lens:
+        qrs, krs, tts = infer_attn_mask_from_sliding_window(
+            AttnRange(pos, pos + L), AttnRange(pos, pos + L), (1024, 0)
+        )
+        q_ranges.extend(qrs)
+        k_ranges.extend(krs)
+        types.extend(tts)
+        pos += L
+    cfg = DistAttnConfig(
+        dispatch_config=DispatchConfig(chunk_size=2048),
+        overlap_config=OverlapConfig(degree=2, min_chunk_size=1024),
+    )
+    key4 = magi_attn_flex_key(
+        q_ranges, k_ranges, types, total, total, hq4, hkv4, D,
+        cp_group_or_mesh=dist.group.WORLD, dist_attn_config=cfg,
+    )
+    g4 = torch.Generator().manual_seed(42)
+    q = (torch.randn(total, hq4, D, generator=g4) * 0.5).bfloat16().to(device)
+    k = (torch.randn(total, hkv4, D, generator=g4) * 0.5).bfloat16().to(device)
+    v = (torch.randn(total, hkv4, D, generator=g4) * 0.5).bfloat16().to(device)
+    ql = dispatch(q, key4).requires_grad_(True)
+    kl = dispatch(k, key4).requires_grad_(True)
+    vl = dispatch(v, key4).requires_grad_(True)
+    do4 = torch.randn_like(ql)
+
+    from magi_attention.api.magi_attn_interface import dist_attn_runtime_dict_mgr
+    rt4 = dist_attn_runtime_dict_mgr[key4].runtime
+    area4 = sum(a.total_area for a in
+                [rt4.calc_meta.host_arg] + list(rt4.calc_meta.stage_args))
+
+    def swstep():
+        o, _ = calc_attn(ql, kl, vl, key4)
+        o.backward(do4)
+        ql.grad = kl.grad = vl.grad = None
+
+    dt = time_fn(swstep, steps=5, warm=2)
+    fl = 4 * area4 * hq4 * D * 3.5
+    out["sw_doc_32k_fwdbwd"] = {
+        "tflops": fl / dt / 1e12, "ms": dt * 1e3,
+        "config": "BASELINE config 4: sliding-window(1024)+doc mask, docs "
+                  f"{doc_lens}, h32 d128 bf16, fwd+bwd, planner-solved, cp1 "
+                  "(cp4 form is the driver's multi-GPU run)",
+    }
+    del q, k, v, ql, kl, vl, do4
+
+    # ---- config 5: fp8 e4m3 128k GQA 64/8 ----
+    n5, hq5, hkv5 = 131072, 64, 8
+    q5 = (torch.randn(n5, hq5, D, generator=g) * 0.5).to(torch.float8_e4m3fn).to(device)
+    k5 = (torch.randn(n5, hkv5, D, generator=g) * 0.5).to(torch.float8_e4m3fn).to(device)
+    v5 = (torch.randn(n5, hkv5, D, generator=g) * 0.5).to(torch.float8_e4m3fn).to(device)
+    qr5 = torch.tensor([[0, n5]], dtype=torch.int32, device=device)
+    tm5 = torch.tensor([1], dtype=torch.int32, device=device)
+
+    def fp8fwd():
+        with torch.no_grad():
+            flex_flash_attn_func(q5, k5, v5, qr5, qr5.clone(), tm5,
+                                 max_seqlen_q=n5)
+
+    dt = time_fn(fp8fwd, steps=3, warm=1)
+    fl = 4 * (n5 * (n5 + 1) // 2) * hq5 * D
+    out["fp8_128k_gqa_fwd"] = {
+        "tflops": fl / dt / 1e12, "ms": dt * 1e3,
+        "config": "BASELINE config 5: fp8 e4m3 Q/K/V, seqlen 131072, GQA "
+                  "64/8, d128, causal, fwd (bwd runs the bf16 kernels over "
+                  "upcast operands), cp1",
+    }
+    del q5, k5, v5
+    return out
 
 
 def measure_cpu_baseline():
@@ -237,6 +411,7 @@ def main():
     ap.add_argument("--steps", type=int, default=5)
     ap.add_argument("--warmup", type=int, default=2)
     ap.add_argument("--skip-cpu-baseline", action="store_true")
+    ap.add_argument("--skip-extra-configs", action="store_true")
     args = ap.parse_args()
 
     assert torch.cuda.is_available(), "bench.py needs an MI355X"
@@ -244,6 +419,12 @@ def main():
 
     # all ranks participate (attn_fwd inside issues collectives); rank 0 reports
     roofline = measure_roofline(state, device)
+    extra = None
+    if world == 1 and not args.skip_extra_configs:
+        try:
+            extra = measure_extra_configs(device)
+        except Exception as e:  # never fail the contract line
+            extra = {"error": repr(e)}
     if rank == 0:
         cpu_baseline = None if args.skip_cpu_baseline else measure_cpu_baseline()
         line = {
@@ -271,6 +452,7 @@ def main():
             },
             "roofline": roofline,
             "cpu_baseline": cpu_baseline,
+            "extra_configs": extra,
         }
         print(json.dumps(line))
 

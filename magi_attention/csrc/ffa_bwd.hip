@@ -180,9 +180,13 @@ __global__ __launch_bounds__(256) void bwd_preprocess_kernel(BwdParams p, int d,
 // prefetch has one compute phase to land). 3 = constant-distance vmcnt(G)
 // barrier: prefetch has TWO compute phases, and the barrier only waits for
 // the one stage it needs — attacks the 34-42% SQ_WAIT_ANY of the r1 PMC.
-template <int D, bool HAS_SOFTCAP, int MODE, int WAVES, int NBUF>
+template <int D, bool HAS_SOFTCAP, int MODE, int WAVES, int NBUF, int NT = 0>
 __global__ __launch_bounds__(64 * WAVES, 1)
 void ffa_bwd_dkv_kernel(BwdParams p) {
+  // NT: stage with the non-temporal policy (aux=2) — staged rows are read
+  // once per WG; keeping them out of L2 protects the resident K strips and
+  // V tiles from streaming eviction (PMC: 174 GB/launch L2-miss traffic)
+  constexpr int STAGE_AUX = NT ? 2 : 0;
   constexpr bool WANT_DV = MODE != 2;
   constexpr bool WANT_DK = MODE != 1;
   constexpr bool V_IN_LDS = MODE == 0;
@@ -303,7 +307,7 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
           (const __attribute__((address_space(1))) unsigned int*)(
               p.v + (size_t)kr * k_pitch + (size_t)kh * D + csw),
           (__attribute__((address_space(3))) unsigned int*)&lds_vt[r0v * D],
-          16, 0, 0);
+          16, 0, STAGE_AUX);
     }
   }
 
@@ -344,12 +348,12 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
           (const __attribute__((address_space(1))) unsigned int*)(
               p.q + (size_t)qrow * q_pitch + (size_t)h * D + csw),
           (__attribute__((address_space(3))) unsigned int*)&lds_q(buf)[r0 * D],
-          16, 0, 0);
+          16, 0, STAGE_AUX);
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)(
               p.dout + (size_t)qrow * q_pitch + (size_t)h * D + csw),
           (__attribute__((address_space(3))) unsigned int*)&lds_do(buf)[r0 * D],
-          16, 0, 0);
+          16, 0, STAGE_AUX);
     }
     // lse (lanes 0-31) / dpsum (lanes 32-63), one LDS-DMA per 32-row group
 #pragma unroll
@@ -361,7 +365,7 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
           (const __attribute__((address_space(1))) unsigned int*)src,
           (__attribute__((address_space(3))) unsigned int*)(
               lds_lse(buf) + sg * 2 * BWD_BM),
-          4, 0, 0);
+          4, 0, STAGE_AUX);
     }
   };
 
@@ -407,8 +411,9 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
       if constexpr (V_IN_LDS) {
         // fused mode: the lane's K fragments re-load per subtile (keeping
         // them loop-persistent overflows the 256-reg/2-wave budget). Issue
-        // the 8 loads FIRST, run the dP chain (LDS-only operands) while they
-        // fly, then the S chain consumes them with the latency paid off.
+        // the 8 loads FIRST, then run the INTERLEAVED dP/S chains — dP's
+        // LDS-only operands cover the K loads' latency, and the two
+        // independent MFMA streams cover each other's LDS reads.
         bf16x8 kfs[DF];
 #pragma unroll
         for (int dd = 0; dd < DF; ++dd)
@@ -421,10 +426,6 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
           // addressing to qf, off the wave's swizzled V tile
           bf16x8 vf = *(const bf16x8*)((const char*)lds_vt + off);
           dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vf, dp, 0, 0, 0);
-        }
-#pragma unroll
-        for (int dd = 0; dd < DF; ++dd) {
-          const int off = swz(lo32, lo32 * ROWB + dd * 32 + hi * 16);
           bf16x8 qf = *(const bf16x8*)((const char*)lqb + off);
           s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kfs[dd], s, 0, 0, 0);
         }
@@ -616,9 +617,10 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
 // ablation measured at ~45% of a fused backward. K/V tiles are staged
 // cooperatively per iteration (row-major swizzled for the S^T/dP^T A-frags,
 // plus a transposed copy for the dQ B-frags).
-template <int D, bool HAS_SOFTCAP, int WAVES, int NBUF>
+template <int D, bool HAS_SOFTCAP, int WAVES, int NBUF, int NT = 0>
 __global__ __launch_bounds__(64 * WAVES, WAVES == 8 ? 1 : 2)
 void ffa_bwd_dq_kernel(BwdParams p) {
+  constexpr int STAGE_AUX = NT ? 2 : 0;  // see dkv kernel note
   constexpr int DF = D / 16;
   constexpr int DT = D / 32;
   constexpr int ROWB = D * 2;
@@ -723,12 +725,12 @@ void ffa_bwd_dq_kernel(BwdParams p) {
           (const __attribute__((address_space(1))) unsigned int*)(
               p.k + (size_t)kr * k_pitch + (size_t)kh * D + csw),
           (__attribute__((address_space(3))) unsigned int*)&lds_k(buf)[r0 * D],
-          16, 0, 0);
+          16, 0, STAGE_AUX);
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)(
               p.v + (size_t)kr * k_pitch + (size_t)kh * D + csw),
           (__attribute__((address_space(3))) unsigned int*)&lds_v(buf)[r0 * D],
-          16, 0, 0);
+          16, 0, STAGE_AUX);
     }
   };
 
@@ -872,6 +874,280 @@ void ffa_bwd_dq_kernel(BwdParams p) {
   }
 }
 
+
+// ---------------- dQ pass, 64-row q tiles (r2) ----------------
+// Each wave owns a 64-row q tile as two interleaved 32-row halves at ONE
+// wave/SIMD (512-reg budget). Why this beats the 32-row/2-wave shape:
+// every K/V fragment read (b128 A-frags and tr16 B-frags) is SHARED by the
+// two halves' MFMAs — memory ops per MFMA halve — and the two halves are
+// independent MFMA streams, so each read's LDS latency hides under the
+// other half's matrix op without a partner wave.
+template <int D, bool HAS_SOFTCAP, int NBUF, int NT = 0>
+__global__ __launch_bounds__(256, 1)
+void ffa_bwd_dq64_kernel(BwdParams p) {
+  constexpr int STAGE_AUX = NT ? 2 : 0;
+  constexpr int DF = D / 16;
+  constexpr int DT = D / 32;
+  constexpr int ROWB = D * 2;
+  constexpr int SW32M = ROWB / 32 - 1;
+  constexpr int WAVES = 4;
+  constexpr int TQ = 64;  // q rows per wave
+  auto swz = [](int row, int byte_off) {
+    return byte_off ^ ((row & SW32M) << 5);
+  };
+  const int ri = blockIdx.z;
+  const int h = p.head_major ? blockIdx.x : blockIdx.y;
+  const int wb = p.head_major ? blockIdx.y : blockIdx.x;
+  const int qs = p.q_ranges[2 * ri], qe = p.q_ranges[2 * ri + 1];
+  const int mblk0 = qs + wb * (TQ * WAVES);
+  if (mblk0 >= qe) return;
+  const int seg0 = p.seg_starts ? p.seg_starts[ri] : ri;
+  const int seg1 = p.seg_starts ? p.seg_starts[ri + 1] : ri + 1;
+  int ks, ke, atype;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int lo32 = lane & 31;
+  const int hi = lane >> 5;
+
+  const int m0 = mblk0 + wave * TQ;  // this wave's 64-row q tile
+  const bool skip_dq = (p.debug_ablate & 1) != 0;
+
+  const float sl2 = HAS_SOFTCAP ? p.softcap * 1.4426950408889634f
+                                : p.scale * 1.4426950408889634f;
+  const float cap_pre = HAS_SOFTCAP ? p.scale / p.softcap : 0.f;
+  const float log2e = 1.4426950408889634f;
+
+  const int kh = h / p.gqa;
+  const size_t k_pitch = (size_t)p.hk * D;
+  const size_t q_pitch = (size_t)p.hq * D;
+
+  constexpr int KITER = 2 * BWD_BN;
+  static_assert(NBUF * 2 * KITER * D * 2 <= 163840, "LDS budget");
+  __shared__ __attribute__((aligned(16))) char smem[NBUF * 2 * KITER * D * 2];
+  auto lds_k = [&](int buf) -> __bf16* {
+    return (__bf16*)(smem + (2 * buf) * KITER * D * 2);
+  };
+  auto lds_v = [&](int buf) -> __bf16* {
+    return (__bf16*)(smem + (2 * buf + 1) * KITER * D * 2);
+  };
+
+  // persistent per-wave operands, both halves
+  bf16x8 qf[2][DF], dof[2][DF];
+  float lse_q[2], dpsum_q[2];
+  bool row_live[2], qvalid[2];
+  int qrow_h[2];
+#pragma unroll
+  for (int hh = 0; hh < 2; ++hh) {
+    const int qrow = m0 + 32 * hh + lo32;
+    qrow_h[hh] = qrow;
+    qvalid[hh] = qrow < qe;
+    const int qcl = qvalid[hh] ? qrow : (qe - 1);
+    const bf16_t* qp = p.q + (size_t)qcl * q_pitch + (size_t)h * D;
+    const bf16_t* dp = p.dout + (size_t)qcl * q_pitch + (size_t)h * D;
+#pragma unroll
+    for (int dd = 0; dd < DF; ++dd) {
+      qf[hh][dd] = *(const bf16x8*)(qp + dd * 16 + hi * 8);
+      dof[hh][dd] = *(const bf16x8*)(dp + dd * 16 + hi * 8);
+    }
+    lse_q[hh] = qvalid[hh] ? p.lse[(size_t)qrow * p.hq + h] : INFINITY;
+    dpsum_q[hh] = qvalid[hh] ? p.dpsum[(size_t)qrow * p.hq + h] : 0.f;
+    row_live[hh] = qvalid[hh] && lse_q[hh] != INFINITY &&
+                   lse_q[hh] != -INFINITY;
+  }
+
+  const int mlast = min(mblk0 + TQ * WAVES, qe) - 1;
+  int k_lo = 0, k_hi = 0, wk_lo = 0, wk_hi = 0;
+  auto seg_bounds = [&]() {
+    k_lo = ks; k_hi = ke;
+    if (atype == 1 || atype == 3) k_hi = min(k_hi, mlast + (ke - qe) + 1);
+    if (atype == 2 || atype == 3) k_lo = max(k_lo, mblk0 + (ks - qs));
+    wk_lo = ks; wk_hi = ke;
+    if (atype == 1 || atype == 3)
+      wk_hi = min(wk_hi, min(m0 + TQ, qe) - 1 + (ke - qe) + 1);
+    if (atype == 2 || atype == 3) wk_lo = max(wk_lo, m0 + (ks - qs));
+  };
+
+  f32x16 acc_dq[2][DT];
+#pragma unroll
+  for (int hh = 0; hh < 2; ++hh)
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt) acc_dq[hh][dt] = (f32x16)(0.f);
+
+  constexpr int ROWS_PER_GLDS = 1024 / ROWB;
+  constexpr int GLDS_PER_WAVE = (KITER / WAVES) / ROWS_PER_GLDS;
+  auto stage_glds = [&](int buf, int n0x) {
+#pragma unroll
+    for (int gi = 0; gi < GLDS_PER_WAVE; ++gi) {
+      const int r0 = (KITER / WAVES) * wave + ROWS_PER_GLDS * gi;
+      const int r = r0 + lane / (ROWB / 16);
+      const int c = lane % (ROWB / 16);
+      const int kr = min(n0x + r, ke - 1);
+      const int csw = (c ^ ((r & SW32M) << 1)) * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              p.k + (size_t)kr * k_pitch + (size_t)kh * D + csw),
+          (__attribute__((address_space(3))) unsigned int*)&lds_k(buf)[r0 * D],
+          16, 0, STAGE_AUX);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              p.v + (size_t)kr * k_pitch + (size_t)kh * D + csw),
+          (__attribute__((address_space(3))) unsigned int*)&lds_v(buf)[r0 * D],
+          16, 0, STAGE_AUX);
+    }
+  };
+
+  constexpr int GOPS = 2 * GLDS_PER_WAVE;
+  int cur = 0;
+  for (int seg = seg0; seg < seg1; ++seg) {
+  ks = p.k_ranges[2 * seg];
+  ke = p.k_ranges[2 * seg + 1];
+  atype = p.attn_type_map ? p.attn_type_map[seg] : 0;
+  if (ke <= ks) continue;
+  seg_bounds();
+  if (k_lo >= k_hi) continue;
+  cur = 0;
+  stage_glds(0, k_lo);
+  if constexpr (NBUF == 3) stage_glds(1, k_lo + KITER);
+
+  for (int n0 = k_lo; n0 < k_hi; n0 += KITER) {
+    pipe_barrier<NBUF == 3 ? GOPS : 0>();
+    if constexpr (NBUF == 3) {
+      stage_glds(cur == 0 ? 2 : cur - 1, n0 + 2 * KITER);
+    } else {
+      if (n0 + KITER < k_hi) stage_glds(cur ^ 1, n0 + KITER);
+    }
+#pragma unroll
+    for (int sub = 0; sub < 2; ++sub) {
+    const int ns = n0 + sub * BWD_BN;
+    if (ns >= k_hi) break;
+    const __bf16* lkb = lds_k(cur) + sub * BWD_BN * D;
+    const __bf16* lvb = lds_v(cur) + sub * BWD_BN * D;
+
+    if (m0 < qe && ns + BWD_BN > wk_lo && ns < wk_hi) {
+      // ---- S^T / dP^T for BOTH halves off ONE kf/vf read pair ----
+      f32x16 sA0 = (f32x16)(0.f), dpA0 = (f32x16)(0.f);
+      f32x16 sA1 = (f32x16)(0.f), dpA1 = (f32x16)(0.f);
+#pragma unroll
+      for (int dd = 0; dd < DF; ++dd) {
+        const int off = swz(lo32, lo32 * ROWB + dd * 32 + hi * 16);
+        bf16x8 kf = *(const bf16x8*)((const char*)lkb + off);
+        bf16x8 vf = *(const bf16x8*)((const char*)lvb + off);
+        sA0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[0][dd], sA0, 0, 0, 0);
+        dpA0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dof[0][dd], dpA0, 0, 0, 0);
+        sA1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[1][dd], sA1, 0, 0, 0);
+        dpA1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dof[1][dd], dpA1, 0, 0, 0);
+      }
+
+      // ---- softmax recompute per half; `scale` folded into the FINAL dq
+      // store (dq is linear in dS), not the per-element dsv ----
+      float dsv[2][16];
+#pragma unroll
+      for (int hh = 0; hh < 2; ++hh) {
+        const f32x16& sA = hh ? sA1 : sA0;
+        const f32x16& dpA = hh ? dpA1 : dpA0;
+        const int m0h = m0 + 32 * hh;
+        const bool interior =
+            (qrow_h[hh] < qe) && row_live[hh] && (ns >= ks) &&
+            (ns + BWD_BN <= ke) &&
+            !((atype == 1 || atype == 3) &&
+              (ns + BWD_BN - 1 > m0h + (ke - qe))) &&
+            !((atype == 2 || atype == 3) && (ns < m0h + 31 + (ks - qs)));
+        if (__all(interior) && !HAS_SOFTCAP) {
+          const float lsc = lse_q[hh] * log2e;
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const float pij = fast_exp2(sA[r] * sl2 - lsc);
+            dsv[hh][r] = pij * (dpA[r] - dpsum_q[hh]);
+          }
+        } else {
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int kk = ns + crow(r, hi);
+            bool ok = row_live[hh] && kk < ke && kk >= ks;
+            if (atype == 1 || atype == 3)
+              ok = ok && (kk - qrow_h[hh] <= ke - qe);
+            if (atype == 2 || atype == 3)
+              ok = ok && (kk - qrow_h[hh] >= ks - qs);
+            float sv = sA[r];
+            float drel = 1.f;
+            float t;
+            if (HAS_SOFTCAP) {
+              const float th = tanhf(sv * cap_pre);
+              t = th * sl2;
+              drel = 1.f - th * th;
+            } else {
+              t = sv * sl2;
+            }
+            const float pij = ok ? fast_exp2(t - lse_q[hh] * log2e) : 0.f;
+            dsv[hh][r] = pij * (dpA[r] - dpsum_q[hh]) * drel;
+          }
+        }
+      }
+
+      // ---- dq += dS K: ONE tr16 B-frag feeds BOTH halves' MFMAs ----
+      bf16x8 a00 = cframe_to_afrag(dsv[0], 0);
+      bf16x8 a01 = cframe_to_afrag(dsv[0], 1);
+      bf16x8 a10 = cframe_to_afrag(dsv[1], 0);
+      bf16x8 a11 = cframe_to_afrag(dsv[1], 1);
+      {
+        const int qhalf2 = (lane >> 4) & 1;
+        const int jrow = (lane & 15) >> 2;
+        const int row0 = 8 * hi + jrow;
+        const int row1 = 8 * hi + 4 + jrow;
+        const int k_base = (int)(unsigned long long)(
+            (__attribute__((address_space(3))) const char*)lkb);
+        const int lane8 = (lane & 3) * 8;
+        const int sw0 = (row0 & SW32M) << 5;
+        const int sw1 = (row1 & SW32M) << 5;
+        const int rb0 = row0 * ROWB + lane8;
+        const int rb1 = row1 * ROWB + lane8;
+        const int sw2 = ((row0 + 16) & SW32M) << 5;
+        const int sw3 = ((row1 + 16) & SW32M) << 5;
+        const int rb2 = (row0 + 16) * ROWB + lane8;
+        const int rb3 = (row1 + 16) * ROWB + lane8;
+#pragma unroll
+        for (int dt = 0; dt < DT; ++dt) {
+          const int dcol = (dt * 32 + 16 * qhalf2) * 2;
+          bf16x8 b0 = tr16_frag(k_base + rb0 + (dcol ^ sw0),
+                                k_base + rb1 + (dcol ^ sw1));
+          acc_dq[0][dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              a00, b0, acc_dq[0][dt], 0, 0, 0);
+          acc_dq[1][dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              a10, b0, acc_dq[1][dt], 0, 0, 0);
+          bf16x8 b1 = tr16_frag(k_base + rb2 + (dcol ^ sw2),
+                                k_base + rb3 + (dcol ^ sw3));
+          acc_dq[0][dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              a01, b1, acc_dq[0][dt], 0, 0, 0);
+          acc_dq[1][dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              a11, b1, acc_dq[1][dt], 0, 0, 0);
+        }
+      }
+    }
+    }  // sub
+    cur = (NBUF == 3) ? (cur == 2 ? 0 : cur + 1) : (cur ^ 1);
+  }
+  __syncthreads();
+  }  // seg
+
+  if (skip_dq) return;
+#pragma unroll
+  for (int hh = 0; hh < 2; ++hh) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int qr = m0 + 32 * hh + crow(r, hi);
+      if (qr >= qe) continue;
+      float* dst = p.dq + (size_t)qr * q_pitch + (size_t)h * D;
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        const float val = acc_dq[hh][dt][r] * p.scale;
+        if (val != 0.f) unsafeAtomicAdd(dst + dt * 32 + lo32, val);
+      }
+    }
+  }
+}
+
 // ---------------- launchers ----------------
 extern "C" int magi_ffa_bwd_preprocess(const magi_ffa_bwd_args* a) {
   if (!a || !a->dout || !a->out || !a->dpsum) return -1;
@@ -949,27 +1225,50 @@ extern "C" int magi_ffa_bwd_dq(const magi_ffa_bwd_args* a) {
   const int dqw = a->max_seqlen_k >= 8192 ? 8 : 4;
   int nbuf = 3;
   { const char* e = getenv("MAGI_BWD_NBUF"); if (e) nbuf = atoi(e); }
+  int nt = 0;
+  { const char* e = getenv("MAGI_STAGE_NT"); if (e) nt = atoi(e); }
+  // measured 61.1 vs 54.8 ms at 64k: without a partner wave the exposed
+  // LDS/barrier latencies outweigh the halved memory ops — keep 2 waves/SIMD
+  int dq64 = 0;
+  { const char* e = getenv("MAGI_BWD_DQ64"); if (e) dq64 = atoi(e); }
+  if (a->n_ranges > 65535) return -5;
+  hipStream_t s = (hipStream_t)a->stream;
+  const bool sc = a->softcap > 0.f;
+  if (dqw == 8 && dq64) {
+    // 64-row q tiles, 4 waves, 1 wave/SIMD (same 256-row block span)
+    const int qblocks64 = (int)((a->total_q + 255) / 256);
+    dim3 grid64 = p.head_major
+                      ? dim3(a->hq, qblocks64, (unsigned)a->n_ranges)
+                      : dim3(qblocks64, a->hq, (unsigned)a->n_ranges);
+    dim3 block64(256);
+#define LAUNCH_DQ64(DD, SC) \
+  hipLaunchKernelGGL((ffa_bwd_dq64_kernel<DD, SC, 3>), grid64, block64, 0, s, p)
+    if (a->d == 64) { if (sc) LAUNCH_DQ64(64, true); else LAUNCH_DQ64(64, false); }
+    else            { if (sc) LAUNCH_DQ64(128, true); else LAUNCH_DQ64(128, false); }
+#undef LAUNCH_DQ64
+    return (int)hipGetLastError();
+  }
   const int qspan = BWD_BM * dqw;
   const int qblocks = (int)((a->total_q + qspan - 1) / qspan);
-  if (a->n_ranges > 65535) return -5;
   dim3 grid_q = p.head_major ? dim3(a->hq, qblocks, (unsigned)a->n_ranges)
                              : dim3(qblocks, a->hq, (unsigned)a->n_ranges);
   dim3 block(64 * dqw);
-  hipStream_t s = (hipStream_t)a->stream;
-  const bool sc = a->softcap > 0.f;
-#define LAUNCH_DQ(DD, SC, WW, NB) \
-  hipLaunchKernelGGL((ffa_bwd_dq_kernel<DD, SC, WW, NB>), grid_q, block, 0, s, p)
-#define PICK_DQ(DD) \
+#define LAUNCH_DQ(DD, SC, WW, NB, NTV) \
+  hipLaunchKernelGGL((ffa_bwd_dq_kernel<DD, SC, WW, NB, NTV>), grid_q, block, \
+                     0, s, p)
+#define PICK_DQ_NT(DD, NTV) \
   do { \
     if (dqw == 8) { \
-      if (nbuf == 3) { if (sc) LAUNCH_DQ(DD, true, 8, 3); else LAUNCH_DQ(DD, false, 8, 3); } \
-      else           { if (sc) LAUNCH_DQ(DD, true, 8, 2); else LAUNCH_DQ(DD, false, 8, 2); } \
+      if (nbuf == 3) { if (sc) LAUNCH_DQ(DD, true, 8, 3, NTV); else LAUNCH_DQ(DD, false, 8, 3, NTV); } \
+      else           { if (sc) LAUNCH_DQ(DD, true, 8, 2, NTV); else LAUNCH_DQ(DD, false, 8, 2, NTV); } \
     } else { \
-      if (sc) LAUNCH_DQ(DD, true, 4, 2); else LAUNCH_DQ(DD, false, 4, 2); \
+      if (sc) LAUNCH_DQ(DD, true, 4, 2, NTV); else LAUNCH_DQ(DD, false, 4, 2, NTV); \
     } \
   } while (0)
+#define PICK_DQ(DD) do { if (nt) PICK_DQ_NT(DD, 1); else PICK_DQ_NT(DD, 0); } while (0)
   if (a->d == 64) PICK_DQ(64); else PICK_DQ(128);
 #undef PICK_DQ
+#undef PICK_DQ_NT
 #undef LAUNCH_DQ
   return (int)hipGetLastError();
 }
@@ -985,10 +1284,14 @@ static int launch_bwd_dkv(const magi_ffa_bwd_args* a) {
   // staging. The fused mode keeps NBUF=2 (its V tiles + a 3-ring exceed the
   // 160 KB LDS); the split W8 modes default to the 3-slot ring.
   const bool big = a->max_seqlen_k >= 8192;
-  const int W = ((MODE == 2 || MODE == 0) && big) ? 8 : 4;
+  // W8 for ALL modes on long ranges: one 512-thread WG/CU halves the
+  // staging streams (PMC r2: dV at W4 fetched 174 GB/launch vs dK-W8's 71)
+  const int W = big ? 8 : 4;
   int nbuf = (MODE == 0 || W == 4) ? 2 : 3;
   { const char* e = getenv("MAGI_BWD_NBUF");
     if (e && nbuf == 3) nbuf = atoi(e); }
+  int nt = 0;
+  { const char* e = getenv("MAGI_STAGE_NT"); if (e) nt = atoi(e); }
   const int span = BWD_BN * W;
   const int nblocks = (a->max_seqlen_k + span - 1) / span;
   if (a->n_ranges > 65535) return -5;
@@ -998,28 +1301,30 @@ static int launch_bwd_dkv(const magi_ffa_bwd_args* a) {
   dim3 block(64 * W);
   hipStream_t s = (hipStream_t)a->stream;
   const bool sc = a->softcap > 0.f;
-#define LAUNCH_DKV(DD, SC, WW, NB) \
-  hipLaunchKernelGGL((ffa_bwd_dkv_kernel<DD, SC, MODE, WW, NB>), grid_kv, \
+#define LAUNCH_DKV(DD, SC, WW, NB, NTV) \
+  hipLaunchKernelGGL((ffa_bwd_dkv_kernel<DD, SC, MODE, WW, NB, NTV>), grid_kv, \
                      block, 0, s, p)
-#define PICK_DKV(DD) \
+#define PICK_DKV_NT(DD, NTV) \
   do { \
     if (W == 8) { \
       bool done = false; \
       if constexpr (MODE != 0) { /* MODE0 + 3-ring exceeds 160 KB LDS */ \
         if (nbuf == 3) { \
-          if (sc) LAUNCH_DKV(DD, true, 8, 3); else LAUNCH_DKV(DD, false, 8, 3); \
+          if (sc) LAUNCH_DKV(DD, true, 8, 3, NTV); else LAUNCH_DKV(DD, false, 8, 3, NTV); \
           done = true; \
         } \
       } \
       if (!done) { \
-        if (sc) LAUNCH_DKV(DD, true, 8, 2); else LAUNCH_DKV(DD, false, 8, 2); \
+        if (sc) LAUNCH_DKV(DD, true, 8, 2, NTV); else LAUNCH_DKV(DD, false, 8, 2, NTV); \
       } \
     } else { \
-      if (sc) LAUNCH_DKV(DD, true, 4, 2); else LAUNCH_DKV(DD, false, 4, 2); \
+      if (sc) LAUNCH_DKV(DD, true, 4, 2, NTV); else LAUNCH_DKV(DD, false, 4, 2, NTV); \
     } \
   } while (0)
+#define PICK_DKV(DD) do { if (nt) PICK_DKV_NT(DD, 1); else PICK_DKV_NT(DD, 0); } while (0)
   if (a->d == 64) PICK_DKV(64); else PICK_DKV(128);
 #undef PICK_DKV
+#undef PICK_DKV_NT
 #undef LAUNCH_DKV
   return (int)hipGetLastError();
 }

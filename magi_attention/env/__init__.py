@@ -57,12 +57,24 @@ def is_hierarchical_comm_enable() -> bool:
     return _get_bool("MAGI_ATTENTION_HIERARCHICAL_COMM")
 
 
-def is_bwd_split_dkv() -> bool:
-    """Default ON: dV and dK run as separate kernels (each 2 waves/SIMD,
-    S recomputed) — measured faster than the fused dK+dV kernel on both the
-    dense-64k (461 vs 444 TF) and varlen-16k (177 vs 169 TF) workloads.
-    MAGI_BWD_FUSED_DKV=1 opts back into the fused single kernel."""
-    return not _get_bool("MAGI_BWD_FUSED_DKV")
+def bwd_dkv_mode() -> str:
+    """"auto" (default): fused dK+dV kernel for long ranges (r2: 124.5 ms vs
+    the dv+dk split's 127.4 at 64k, one less launch + one less Q/dO staging
+    stream), split for short ranges (the fused W4 variant is occupancy-1).
+    MAGI_BWD_SPLIT_DKV=1 forces the split; MAGI_BWD_FUSED_DKV=1 forces fused.
+    """
+    if _get_bool("MAGI_BWD_SPLIT_DKV"):
+        return "split"
+    if _get_bool("MAGI_BWD_FUSED_DKV"):
+        return "fused"
+    return "auto"
+
+
+def is_bwd_split_dkv(max_seqlen_k: int = 0) -> bool:
+    mode = bwd_dkv_mode()
+    if mode == "auto":
+        return max_seqlen_k < 8192
+    return mode == "split"
 
 
 def ffa_forward_sm_margin() -> int:

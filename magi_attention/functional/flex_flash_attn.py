@@ -54,7 +54,7 @@ def run_bwd_deterministic(args, q_ranges, k_ranges, attn_type_map,
 
     launch(lib.magi_ffa_bwd_dq, q_groups, (q_ranges, k_ranges), "bwd_dq[det]")
     hs = gqa if gqa > 1 else 1
-    if env.is_bwd_split_dkv():
+    if env.is_bwd_split_dkv(args.max_seqlen_k):
         launch(lib.magi_ffa_bwd_dv, k_groups, (q_ranges, k_ranges),
                "bwd_dv[det]", head_splits=hs)
         launch(lib.magi_ffa_bwd_dk, k_groups, (q_ranges, k_ranges),
@@ -93,7 +93,7 @@ def run_bwd_passes(args, device, dq_tables=None, dkv_tables=None) -> None:
     check(lib.magi_ffa_bwd_dq(args), "magi_ffa_bwd_dq")
     args.stream = ctypes.c_void_p(main.cuda_stream)
     set_tables(dkv_tables, False)
-    if env.is_bwd_split_dkv():
+    if env.is_bwd_split_dkv(args.max_seqlen_k):
         check(lib.magi_ffa_bwd_dv(args), "magi_ffa_bwd_dv")
         check(lib.magi_ffa_bwd_dk(args), "magi_ffa_bwd_dk")
     else:

@@ -83,19 +83,16 @@ def main():
         print(f"{tag:28s} {ms:8.2f} ms   issued {tf:7.1f} TF/s ({eff:4.1f}% peak)")
 
     print(f"== 64k causal h{hq} d{d}: one matmul unit = {unit/1e12:.2f} TFLOP ==")
-    run("dq  NBUF=2 (r1)", "magi_ffa_bwd_dq", 3, {"MAGI_BWD_NBUF": "2"})
-    run("dq  NBUF=3 ring", "magi_ffa_bwd_dq", 3, {"MAGI_BWD_NBUF": "3"})
-    run("dv  NBUF=2 (r1)", "magi_ffa_bwd_dv", 2, {"MAGI_BWD_NBUF": "2"})
-    run("dv  NBUF=3 ring", "magi_ffa_bwd_dv", 2, {"MAGI_BWD_NBUF": "3"})
-    run("dk  NBUF=2 (r1)", "magi_ffa_bwd_dk", 3, {"MAGI_BWD_NBUF": "2"})
-    run("dk  NBUF=3 ring", "magi_ffa_bwd_dk", 3, {"MAGI_BWD_NBUF": "3"})
-    run("dkv fused r2", "magi_ffa_bwd_dkv", 4)
+    run("dq64 (1 wave/SIMD)", "magi_ffa_bwd_dq", 3)
+    run("dq32 (2 waves)", "magi_ffa_bwd_dq", 3, {"MAGI_BWD_DQ64": "0"})
+    run("dv  W8", "magi_ffa_bwd_dv", 2)
+    run("dk", "magi_ffa_bwd_dk", 3)
+    run("dkv fused", "magi_ffa_bwd_dkv", 4)
 
     # best-combo whole-backward estimate
-    best_dq = min(results["dq  NBUF=2 (r1)"][0], results["dq  NBUF=3 ring"][0])
-    split = (min(results["dv  NBUF=2 (r1)"][0], results["dv  NBUF=3 ring"][0])
-             + min(results["dk  NBUF=2 (r1)"][0], results["dk  NBUF=3 ring"][0]))
-    fused = results["dkv fused r2"][0]
+    best_dq = min(results["dq64 (1 wave/SIMD)"][0], results["dq32 (2 waves)"][0])
+    split = results["dv  W8"][0] + results["dk"][0]
+    fused = results["dkv fused"][0]
     bwd_flops = 2.5 * 2 * unit
     for tag, tot in (("split best", best_dq + split),
                      ("fused best", best_dq + fused)):
