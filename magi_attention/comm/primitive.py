@@ -73,14 +73,16 @@ def _rows_copy(src: torch.Tensor, dst: torch.Tensor, cmap: RowChunkMap,
 
 
 def group_cast(
-    kv_local: torch.Tensor,  # [2L, h, d] (k rows then v rows)
+    kv_local: torch.Tensor,  # [ncopies*L, h, d] (e.g. k rows then v rows)
     arg: GroupCastArg,
     group: dist.ProcessGroup,
     async_op: bool = True,
 ) -> WorkWithPostProcessFn:
-    """Multicast my hosted K/V token rows to the ranks that need them this
-    stage; returns a handle whose wait_post_process() yields the stage kv
-    buffer [2S, h, d] in globally-sorted order."""
+    """Multicast my hosted token rows to the ranks that need them this stage;
+    returns a handle whose wait_post_process() yields the stage buffer
+    [ncopies*S, h, d] in globally-sorted order. ncopies=2 packs (K,V) — or
+    any two row-shaped tensors, e.g. QO-comm's (q,do) / (lse,dpsum) — through
+    one wire; ncopies=1 is a single-tensor cast (QO-comm q)."""
     h_tail = kv_local.shape[1:]
     send = kv_local.new_empty((arg.send_pack.total_rows, *h_tail))
     _rows_copy(kv_local, send, arg.send_pack)
@@ -104,7 +106,7 @@ def group_cast(
             work = None
 
     def post() -> torch.Tensor:
-        stage = kv_local.new_zeros((2 * arg.stage_tokens, *h_tail))
+        stage = kv_local.new_zeros((arg.ncopies * arg.stage_tokens, *h_tail))
         _rows_copy(recv, stage, arg.recv_unpack)
         return stage
 
@@ -112,14 +114,19 @@ def group_cast(
 
 
 def group_reduce(
-    partial: torch.Tensor,   # [2S, h, d] partial dK/dV of the stage buffer
-    dst: torch.Tensor,       # [2L, h, d] local dK/dV accumulator (fp32)
+    partial: torch.Tensor,   # [n*S, h, d] partial dK/dV (or dq) stage rows
+    dst: torch.Tensor,       # [n*L, h, d] owner-local accumulator (fp32)
     arg: GroupReduceArg,
     group: dist.ProcessGroup,
     async_op: bool = True,
+    wire_dtype: Optional[torch.dtype] = None,
 ) -> WorkWithPostProcessFn:
-    """Return partial dK/dV rows to their owner ranks and sum-reduce them into
-    the owner-local accumulator."""
+    """Return partial rows to their owner ranks and sum-reduce them into the
+    owner-local accumulator. wire_dtype=bf16 (the reference's DEFAULT;
+    env/comm.py:107 high-precision-reduce flips to fp32) halves the wire:
+    the partial is downcast before packing and upcast before the sum."""
+    if wire_dtype is not None and wire_dtype != partial.dtype:
+        partial = partial.to(wire_dtype)
     h_tail = partial.shape[1:]
     send = partial.new_empty((arg.send_pack.total_rows, *h_tail))
     _rows_copy(partial, send, arg.send_pack)
@@ -143,8 +150,93 @@ def group_reduce(
             work = None
 
     def post() -> torch.Tensor:
-        _rows_copy(recv, dst, arg.recv_reduce, reduce="sum")
+        r = recv.float() if recv.dtype != dst.dtype else recv
+        _rows_copy(r, dst, arg.recv_reduce, reduce="sum")
         return dst
+
+    return WorkWithPostProcessFn(work, post)
+
+
+def group_reduce_out_lse(
+    out_part: torch.Tensor,  # [S, hq, d] fp32 partial out (stage q rows)
+    lse_part: torch.Tensor,  # [S, hq] fp32 partial lse
+    out_acc: torch.Tensor,   # [Lq, hq, d] fp32 owner accumulator
+    lse_acc: torch.Tensor,   # [Lq, hq] fp32 owner accumulator
+    arg: GroupReduceArg,     # ncopies=1 tables over q rows
+    group: dist.ProcessGroup,
+    high_precision: bool = False,
+    wire_dtype: Optional[torch.dtype] = None,
+) -> WorkWithPostProcessFn:
+    """QO-comm forward reduce: partial (out, lse) of remotely-computed q rows
+    returned to their owners and merged with the online-softmax correction
+    (reference _reduce_partial_out_lse dist_attn.py:1924; wire dtype = the
+    param dtype for out unless the fwd high-precision-reduce flag (then
+    fp32) — lse always fp32)."""
+    wire = (out_part if high_precision or wire_dtype is None
+            else out_part.to(wire_dtype))
+    h_tail = wire.shape[1:]
+    send_o = wire.new_empty((arg.send_pack.total_rows, *h_tail))
+    _rows_copy(wire, send_o, arg.send_pack)
+    send_l = lse_part.new_empty((arg.send_pack.total_rows, lse_part.shape[1]))
+    _rows_copy(lse_part, send_l, arg.send_pack)
+    recv_o = wire.new_empty((arg.total_recv, *h_tail))
+    recv_l = lse_part.new_empty((arg.total_recv, lse_part.shape[1]))
+    if dist.get_world_size(group) == 1:
+        work = None
+        recv_o, recv_l = send_o, send_l
+    else:
+        flat = 1
+        for t in h_tail:
+            flat *= t
+        work = dist.all_to_all_single(
+            recv_o.view(recv_o.shape[0], flat), send_o.view(send_o.shape[0], flat),
+            output_split_sizes=arg.output_split_sizes,
+            input_split_sizes=arg.input_split_sizes, group=group, async_op=True,
+        )
+        work2 = dist.all_to_all_single(
+            recv_l, send_l,
+            output_split_sizes=arg.output_split_sizes,
+            input_split_sizes=arg.input_split_sizes, group=group, async_op=True,
+        )
+        w1 = work
+
+        class _Both:
+            def wait(self):
+                w1.wait()
+                work2.wait()
+
+        work = _Both()
+
+    def post() -> torch.Tensor:
+        ro = recv_o.float() if recv_o.dtype != torch.float32 else recv_o
+        cmap = arg.recv_reduce
+        if out_acc.is_cuda:
+            from ..common.range_op import range_reduce
+
+            in_r, out_s = cmap.to_device(out_acc.device)
+            # pieces from different source ranks may target the SAME owner
+            # row: the lse merge is read-modify-write, so sequence per piece
+            for i in range(len(cmap.in_ranges)):
+                a, b = cmap.in_ranges[i]
+                range_reduce(ro, out_acc, in_r[i:i + 1], out_s[i:i + 1],
+                             op="lse", total_rows=b - a,
+                             in_lse=recv_l, out_lse=lse_acc)
+            return out_acc
+        for (a, b), o in zip(cmap.in_ranges, cmap.out_starts):
+            n = b - a
+            l_new = recv_l[a:b].double()
+            l_old = lse_acc[o:o + n].double()
+            l_m = torch.logaddexp(l_old, l_new)
+            w_old = torch.where(l_old == float("-inf"),
+                                torch.zeros_like(l_old), (l_old - l_m).exp())
+            w_new = torch.where(l_new == float("-inf"),
+                                torch.zeros_like(l_new), (l_new - l_m).exp())
+            out_acc[o:o + n] = (
+                out_acc[o:o + n].double() * w_old.unsqueeze(-1)
+                + ro[a:b].double() * w_new.unsqueeze(-1)
+            ).to(out_acc.dtype)
+            lse_acc[o:o + n] = l_m.to(lse_acc.dtype)
+        return out_acc
 
     return WorkWithPostProcessFn(work, post)
 

@@ -98,6 +98,13 @@ class DistAttnRuntimeMgr:
         self.solver, calc_meta, comm_meta = make_attn_meta_from_dispatch_meta(
             slices, self.dispatch_meta, dist_attn_config
         )
+        qo_meta = None
+        if env.is_qo_comm_enable():
+            from .meta import make_qo_meta_from_dispatch_meta
+
+            qo_meta = make_qo_meta_from_dispatch_meta(
+                slices, self.dispatch_meta, dist_attn_config
+            )
         intra_group = inter_group = None
         if mesh_groups is not None and env.is_hierarchical_comm_enable():
             intra_group, inter_group, wi, wn = mesh_groups
@@ -113,6 +120,7 @@ class DistAttnRuntimeMgr:
             total_local_q=total_padded // cp_size,
             intra_group=intra_group,
             inter_group=inter_group,
+            qo_meta=qo_meta,
         )
 
     # ---- ops ----

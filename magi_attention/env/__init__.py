@@ -97,6 +97,13 @@ def is_backward_high_precision_reduce_enable() -> bool:
     return _get_bool("MAGI_ATTENTION_BACKWARD_HIGH_PRECISION_REDUCE")
 
 
+def is_bwd_hide_tail_reduce() -> bool:
+    """MAGI_ATTENTION_BWD_HIDE_TAIL_REDUCE=1: overlap the tail-stage dKV
+    group-reduce with the local-grad dtype casts instead of waiting first
+    (reference env/general.py:233, dist_attn.py:2503)."""
+    return _get_bool("MAGI_ATTENTION_BWD_HIDE_TAIL_REDUCE")
+
+
 def is_native_grpcoll_enable() -> bool:
     return _get_bool("MAGI_ATTENTION_NATIVE_GRPCOLL")
 
@@ -109,6 +116,7 @@ def snapshot() -> tuple[tuple[str, Any], ...]:
         ("auto_range_merge", is_auto_range_merge_enable()),
         ("hierarchical_comm", is_hierarchical_comm_enable()),
         ("qo_comm", is_qo_comm_enable()),
+        ("bwd_hide_tail_reduce", is_bwd_hide_tail_reduce()),
         ("fwd_hp_reduce", is_forward_high_precision_reduce_enable()),
         ("bwd_hp_reduce", is_backward_high_precision_reduce_enable()),
         ("fwd_sm_margin", ffa_forward_sm_margin()),

@@ -22,10 +22,11 @@ from ..comm.primitive import (
     WorkWithPostProcessFn,
     group_cast,
     group_reduce,
+    group_reduce_out_lse,
     hier_group_cast,
     hier_group_reduce,
 )
-from ..meta.containers import AttnArg, CalcMeta, CommMeta
+from ..meta.containers import AttnArg, CalcMeta, CommMeta, QoCommMeta
 
 # test-only hook: lets CPU (gloo) tests run the runtime with an oracle
 # attention implementation. NEVER set in production; the product path on GPU
@@ -48,6 +49,9 @@ class DistAttnRuntime:
     # hierarchical 2D-mesh comm (set iff stages_cast_hier planned)
     intra_group: Optional[dist.ProcessGroup] = None
     inter_group: Optional[dist.ProcessGroup] = None
+    # QO-comm plan (MAGI_ATTENTION_QO_COMM=1): remote slices computed at the
+    # K-host; q (fwd) / q,do,lse,dpsum (bwd) travel instead of K/V
+    qo_meta: Optional[QoCommMeta] = None
 
     @property
     def overlap_degree(self) -> int:
@@ -86,12 +90,6 @@ class DistAttnRuntime:
     ) -> Tuple[torch.Tensor, torch.Tensor, Optional[torch.Tensor]]:
         tq, hq, d = q.shape
         scale = softmax_scale or self.softmax_scale or d ** (-0.5)
-        kv_local = torch.cat([k, v], dim=0)
-
-        # pre-issue ALL remote-stage group-casts (reference dist_attn.py:419-436)
-        works: List[WorkWithPostProcessFn] = [
-            self._cast(kv_local, s) for s in range(self.overlap_degree)
-        ]
 
         out_acc = torch.zeros(tq, hq, d, dtype=torch.float32, device=q.device)
         lse_acc = torch.full(
@@ -101,18 +99,57 @@ class DistAttnRuntime:
         if return_max_logits:
             max_logits = torch.full((hq,), float("-inf"), dtype=torch.float32,
                                     device=q.device)
-        self._fwd_partial(q, k, v, self.calc_meta.host_arg, out_acc, lse_acc,
-                          scale, softcap, max_logits)
-        for s in range(self.overlap_degree):
-            arg = self.calc_meta.stage_args[s]
-            stage_kv = works[s].wait_post_process()
-            S = self.comm_meta.stages_cast[s].stage_tokens
-            if S == 0 or arg.is_empty():
-                continue
-            self._fwd_partial(
-                q, stage_kv[:S], stage_kv[S:], arg, out_acc, lse_acc, scale,
-                softcap, max_logits,
-            )
+
+        if self.qo_meta is not None:
+            # QO-comm: cast q to the K-hosts, compute partial (out,lse) for
+            # remote q rows, lse-merge them back to owners (reference
+            # dist_attn.py:1570 _fetch_remote_q / :1924 _reduce_partial_out_lse)
+            qo = self.qo_meta
+            works = [
+                group_cast(q, qo.stages_cast1[s], self.cp_group)
+                for s in range(self.overlap_degree)
+            ]
+            self._fwd_partial(q, k, v, qo.calc.host_arg, out_acc, lse_acc,
+                              scale, softcap, max_logits)
+            rworks: List[WorkWithPostProcessFn] = []
+            hp = env.is_forward_high_precision_reduce_enable()
+            for s in range(self.overlap_degree):
+                arg = qo.calc.stage_args[s]
+                stage_q = works[s].wait_post_process()
+                S = qo.stages_cast1[s].stage_tokens
+                out_s = torch.zeros(S, hq, d, dtype=torch.float32,
+                                    device=q.device)
+                lse_s = torch.full((S, hq), float("-inf"),
+                                   dtype=torch.float32, device=q.device)
+                if S > 0 and not arg.is_empty():
+                    self._fwd_partial(stage_q, k, v, arg, out_s, lse_s,
+                                      scale, softcap, max_logits)
+                # collective: every rank joins every stage (it may RECEIVE
+                # partials for q rows it owns even with nothing to compute)
+                rworks.append(group_reduce_out_lse(
+                    out_s, lse_s, out_acc, lse_acc, qo.stages_reduce1[s],
+                    self.cp_group, high_precision=hp, wire_dtype=q.dtype,
+                ))
+            for w in rworks:
+                w.wait_post_process()
+        else:
+            kv_local = torch.cat([k, v], dim=0)
+            # pre-issue ALL remote-stage group-casts (reference dist_attn.py:419-436)
+            works = [
+                self._cast(kv_local, s) for s in range(self.overlap_degree)
+            ]
+            self._fwd_partial(q, k, v, self.calc_meta.host_arg, out_acc,
+                              lse_acc, scale, softcap, max_logits)
+            for s in range(self.overlap_degree):
+                arg = self.calc_meta.stage_args[s]
+                stage_kv = works[s].wait_post_process()
+                S = self.comm_meta.stages_cast[s].stage_tokens
+                if S == 0 or arg.is_empty():
+                    continue
+                self._fwd_partial(
+                    q, stage_kv[:S], stage_kv[S:], arg, out_acc, lse_acc,
+                    scale, softcap, max_logits,
+                )
         if sink is not None:
             # fold the replicated sink ONCE into this rank's (out, lse) — the
             # q rows are rank-disjoint, so per-rank postprocess is exact
@@ -184,10 +221,6 @@ class DistAttnRuntime:
         tq, hq, d = q.shape
         L = k.shape[0]
         scale = softmax_scale or self.softmax_scale or d ** (-0.5)
-        kv_local = torch.cat([k, v], dim=0)
-
-        # re-fetch remote KV (reference backward:3455 -> _fetch_remote_kv:1463)
-        works = [self._cast(kv_local, s) for s in range(self.overlap_degree)]
 
         dq_acc = torch.zeros(tq, hq, d, dtype=torch.float32, device=q.device)
         dkv_acc = torch.zeros(2 * L, *k.shape[1:], dtype=torch.float32,
@@ -196,33 +229,96 @@ class DistAttnRuntime:
         dsink = None
         if sink is not None:
             dsink = self._dsink(sink, lse, dpsum)
-        self._bwd_partial(
-            dout, q, k, v, out, lse, dpsum, self.calc_meta.host_arg,
-            dq_acc, dkv_acc[:L], dkv_acc[L:], scale, softcap,
-        )
+        hide_tail = env.is_bwd_hide_tail_reduce()
         rworks: List[WorkWithPostProcessFn] = []
-        for s in range(self.overlap_degree):
-            arg = self.calc_meta.stage_args[s]
-            stage_kv = works[s].wait_post_process()
-            S = self.comm_meta.stages_cast[s].stage_tokens
-            # group_reduce is a COLLECTIVE: every rank must join every stage
-            # even with an empty local partial (this rank may still RECEIVE
-            # dK/dV contributions for KV rows it hosts).
-            dkv_stage = torch.zeros(
-                2 * S, *k.shape[1:], dtype=torch.float32, device=q.device
+
+        if self.qo_meta is not None:
+            # QO-comm backward (reference dist_attn.py:1659
+            # _fetch_remote_qo_do_lse): cast (q,do) + (lse,dpsum) to the
+            # K-hosts; partial dK/dV accumulate LOCALLY (no dKV wire), only
+            # partial dq returns to q owners. We cast dpsum instead of the
+            # reference's o — the kernels only consume dpsum, which the
+            # owner has already computed for all its rows.
+            qo = self.qo_meta
+            qdo = torch.cat([q, dout], dim=0)
+            lsedp = torch.cat([lse, dpsum], dim=0)
+            works2 = [group_cast(qdo, qo.stages_cast2[s], self.cp_group)
+                      for s in range(self.overlap_degree)]
+            worksl = [group_cast(lsedp, qo.stages_cast2[s], self.cp_group)
+                      for s in range(self.overlap_degree)]
+            self._bwd_partial(
+                dout, q, k, v, out, lse, dpsum, qo.calc.host_arg,
+                dq_acc, dkv_acc[:L], dkv_acc[L:], scale, softcap,
             )
-            if S > 0 and not arg.is_empty():
-                self._bwd_partial(
-                    dout, q, stage_kv[:S], stage_kv[S:], out, lse, dpsum, arg,
-                    dq_acc, dkv_stage[:S], dkv_stage[S:], scale, softcap,
+            # low-precision (default) wire = the PARAM dtype (bf16 for bf16
+            # training; fp64 test runs stay exact), fp32 with the HP flag
+            wire = (None if env.is_backward_high_precision_reduce_enable()
+                    else k.dtype)
+            for s in range(self.overlap_degree):
+                arg = qo.calc.stage_args[s]
+                buf = works2[s].wait_post_process()
+                lbuf = worksl[s].wait_post_process()
+                S = qo.stages_cast2[s].stage_tokens
+                dq_stage = torch.zeros(S, hq, d, dtype=torch.float32,
+                                       device=q.device)
+                if S > 0 and not arg.is_empty():
+                    self._bwd_partial(
+                        buf[S:], buf[:S], k, v, out, lbuf[:S], lbuf[S:], arg,
+                        dq_stage, dkv_acc[:L], dkv_acc[L:], scale, softcap,
+                    )
+                rworks.append(group_reduce(
+                    dq_stage, dq_acc, qo.stages_reduce1[s], self.cp_group,
+                    wire_dtype=wire,
+                ))
+        else:
+            kv_local = torch.cat([k, v], dim=0)
+            # re-fetch remote KV (reference backward:3455 -> _fetch_remote_kv:1463)
+            works = [self._cast(kv_local, s)
+                     for s in range(self.overlap_degree)]
+            self._bwd_partial(
+                dout, q, k, v, out, lse, dpsum, self.calc_meta.host_arg,
+                dq_acc, dkv_acc[:L], dkv_acc[L:], scale, softcap,
+            )
+            # reference default: bf16 wire for partial dKV; the
+            # high-precision-reduce flag doubles the wire for fp32
+            # (env/comm.py:107)
+            # low-precision (default) wire = the PARAM dtype (bf16 for bf16
+            # training; fp64 test runs stay exact), fp32 with the HP flag
+            wire = (None if env.is_backward_high_precision_reduce_enable()
+                    else k.dtype)
+            for s in range(self.overlap_degree):
+                arg = self.calc_meta.stage_args[s]
+                stage_kv = works[s].wait_post_process()
+                S = self.comm_meta.stages_cast[s].stage_tokens
+                # group_reduce is a COLLECTIVE: every rank must join every
+                # stage even with an empty local partial (this rank may still
+                # RECEIVE dK/dV contributions for KV rows it hosts).
+                dkv_stage = torch.zeros(
+                    2 * S, *k.shape[1:], dtype=torch.float32, device=q.device
                 )
-            rworks.append(self._reduce(dkv_stage, dkv_acc, s))
-        for w in rworks:
-            w.wait_post_process()
+                if S > 0 and not arg.is_empty():
+                    self._bwd_partial(
+                        dout, q, stage_kv[:S], stage_kv[S:], out, lse, dpsum,
+                        arg, dq_acc, dkv_stage[:S], dkv_stage[S:], scale,
+                        softcap,
+                    )
+                rworks.append(self._reduce(dkv_stage, dkv_acc, s))
+        if hide_tail:
+            # MAGI_ATTENTION_BWD_HIDE_TAIL_REDUCE: run the local-grad dtype
+            # casts UNDER the tail-stage reduce instead of after it
+            # (reference _hide_tail_stage_reduce_backward dist_attn.py:2503)
+            dq = dq_acc.to(q.dtype) if self.qo_meta is None else None
+            for w in rworks:
+                w.wait_post_process()
+            if dq is None:
+                dq = dq_acc.to(q.dtype)
+        else:
+            for w in rworks:
+                w.wait_post_process()
+            dq = dq_acc.to(q.dtype)
         if dsink is not None:
             # "sh" sink is replicated; its gradient sums over ALL q rows
             dist.all_reduce(dsink, op=dist.ReduceOp.SUM, group=self.cp_group)
-        dq = dq_acc.to(q.dtype)
         dk = dkv_acc[:L].to(k.dtype)
         dv = dkv_acc[L:].to(v.dtype)
         return dq, dk, dv, dsink

@@ -113,3 +113,28 @@ def make_attn_meta_from_dispatch_meta(
     )
     rank = dispatch_meta.cp_rank
     return solver, solver.make_calc_meta(rank), solver.make_comm_meta(rank)
+
+
+def make_qo_meta_from_dispatch_meta(
+    slices: List[MaskSlice],
+    dispatch_meta: DispatchMeta,
+    dist_attn_config: DistAttnConfig,
+):
+    """QO-comm plan (MAGI_ATTENTION_QO_COMM=1): run the SAME solver on the
+    transposed mask — its remote-K machinery then produces remote-Q cast and
+    partial-(out,lse)/dq reduce tables (reference env/comm.py:72)."""
+    from .geometry import transpose_slice
+
+    oc = dist_attn_config.overlap_config
+    degree = (oc.degree or 1) if oc.enable else 1
+    slices_t = [t for sl in slices for t in transpose_slice(sl)]
+    solver_t = DistAttnSolver(
+        slices=slices_t,
+        partitions=dispatch_meta.partitions,
+        chunk_size=dispatch_meta.chunk_size,
+        total_seqlen=dispatch_meta.total_seqlen,
+        cp_size=dispatch_meta.cp_size,
+        overlap_degree=degree,
+        min_stage_tokens=oc.min_chunk_size,
+    )
+    return solver_t.make_qo_comm_meta(dispatch_meta.cp_rank)

@@ -20,7 +20,8 @@ from .containers import (
     HierGroupReduceArg,
     RowChunkMap,
 )
-from .geometry import MaskSlice, k_window, q_window
+from .containers import QoCommMeta
+from .geometry import MaskSlice, k_window, q_window, transpose_slice
 
 
 def _ranges_of(pieces: List[AttnRange]) -> AttnRanges:
@@ -172,7 +173,10 @@ class DistAttnSolver:
         ]
 
     # ---------------- comm meta ----------------
-    def make_comm_meta(self, rank: int) -> CommMeta:
+    def make_comm_meta(self, rank: int, ncopies: int = 2) -> CommMeta:
+        """ncopies: tensor copies packed through the same row tables — 2 for
+        (K,V) (also reused for QO-comm's (q,do) and (lse,dpsum) casts), 1
+        for single-tensor casts/reduces (QO-comm q / out / dq)."""
         cp = self.cp_size
         casts: List[GroupCastArg] = []
         reduces: List[GroupReduceArg] = []
@@ -189,20 +193,16 @@ class DistAttnSolver:
             for r in range(cp):
                 pieces = ov[rank][r]
                 tok = pieces.total_seqlen
-                # k rows then v rows within this dst segment
-                vcur = cursor + tok
-                for p in pieces:
-                    lp = my_hr.make_range_local(p, is_self_merged=True)
-                    send_in.append((lp.start, lp.end))
-                    send_out.append(cursor)
-                    cursor += p.seqlen
-                for p in pieces:
-                    lp = my_hr.make_range_local(p, is_self_merged=True)
-                    send_in.append((L + lp.start, L + lp.end))
-                    send_out.append(vcur)
-                    vcur += p.seqlen
-                cursor = vcur
-                in_splits.append(2 * tok)
+                # copy c's rows at source offset c*L within this dst segment
+                for c in range(ncopies):
+                    ccur = cursor + c * tok
+                    for p in pieces:
+                        lp = my_hr.make_range_local(p, is_self_merged=True)
+                        send_in.append((c * L + lp.start, c * L + lp.end))
+                        send_out.append(ccur)
+                        ccur += p.seqlen
+                cursor += ncopies * tok
+                in_splits.append(ncopies * tok)
             total_send = cursor
             # ---- cast: me as RECEIVER ----
             my_need = self.plans[rank].stages_need[s]
@@ -214,17 +214,13 @@ class DistAttnSolver:
             for o in range(cp):
                 pieces = ov[o][rank]
                 tok = pieces.total_seqlen
-                for p in pieces:  # k rows
-                    lp = my_need.make_range_local(p)
-                    recv_in.append((rcur, rcur + p.seqlen))
-                    recv_out.append(lp.start)
-                    rcur += p.seqlen
-                for p in pieces:  # v rows -> stage offset +S
-                    lp = my_need.make_range_local(p)
-                    recv_in.append((rcur, rcur + p.seqlen))
-                    recv_out.append(S + lp.start)
-                    rcur += p.seqlen
-                out_splits.append(2 * tok)
+                for c in range(ncopies):
+                    for p in pieces:
+                        lp = my_need.make_range_local(p)
+                        recv_in.append((rcur, rcur + p.seqlen))
+                        recv_out.append(c * S + lp.start)
+                        rcur += p.seqlen
+                out_splits.append(ncopies * tok)
             casts.append(
                 GroupCastArg(
                     send_pack=RowChunkMap(send_in, send_out, total_send),
@@ -232,10 +228,11 @@ class DistAttnSolver:
                     recv_unpack=RowChunkMap(recv_in, recv_out, rcur),
                     output_split_sizes=out_splits,
                     stage_tokens=S,
+                    ncopies=ncopies,
                 )
             )
             # ---- reduce (bwd): reverse tables ----
-            # me as SENDER of partial dkv (stage buffer rows -> owners)
+            # me as SENDER of partials (stage buffer rows -> owners)
             rs_in: List[Tuple[int, int]] = []
             rs_out: List[int] = []
             rs_splits: List[int] = []
@@ -243,20 +240,16 @@ class DistAttnSolver:
             for o in range(cp):
                 pieces = ov[o][rank]  # what I received from o = what I return
                 tok = pieces.total_seqlen
-                vcur = cursor + tok
-                for p in pieces:
-                    lp = my_need.make_range_local(p)
-                    rs_in.append((lp.start, lp.end))
-                    rs_out.append(cursor)
-                    cursor += p.seqlen
-                for p in pieces:
-                    lp = my_need.make_range_local(p)
-                    rs_in.append((S + lp.start, S + lp.end))
-                    rs_out.append(vcur)
-                    vcur += p.seqlen
-                cursor = vcur
-                rs_splits.append(2 * tok)
-            # me as RECEIVER of partial dkv for my hosted rows (sum-reduce)
+                for c in range(ncopies):
+                    ccur = cursor + c * tok
+                    for p in pieces:
+                        lp = my_need.make_range_local(p)
+                        rs_in.append((c * S + lp.start, c * S + lp.end))
+                        rs_out.append(ccur)
+                        ccur += p.seqlen
+                cursor += ncopies * tok
+                rs_splits.append(ncopies * tok)
+            # me as RECEIVER of partials for my hosted rows (sum-reduce)
             rr_in: List[Tuple[int, int]] = []
             rr_out: List[int] = []
             rr_splits: List[int] = []
@@ -264,17 +257,13 @@ class DistAttnSolver:
             for r in range(cp):
                 pieces = ov[rank][r]
                 tok = pieces.total_seqlen
-                for p in pieces:  # dk rows
-                    lp = my_hr.make_range_local(p, is_self_merged=True)
-                    rr_in.append((rcur, rcur + p.seqlen))
-                    rr_out.append(lp.start)
-                    rcur += p.seqlen
-                for p in pieces:  # dv rows
-                    lp = my_hr.make_range_local(p, is_self_merged=True)
-                    rr_in.append((rcur, rcur + p.seqlen))
-                    rr_out.append(L + lp.start)
-                    rcur += p.seqlen
-                rr_splits.append(2 * tok)
+                for c in range(ncopies):
+                    for p in pieces:
+                        lp = my_hr.make_range_local(p, is_self_merged=True)
+                        rr_in.append((rcur, rcur + p.seqlen))
+                        rr_out.append(c * L + lp.start)
+                        rcur += p.seqlen
+                rr_splits.append(ncopies * tok)
             reduces.append(
                 GroupReduceArg(
                     send_pack=RowChunkMap(rs_in, rs_out, cursor),
@@ -285,6 +274,56 @@ class DistAttnSolver:
                 )
             )
         return CommMeta(stages_cast=casts, stages_reduce=reduces)
+
+    # ---------------- QO-comm (solver runs on the TRANSPOSED mask) --------
+    def make_qo_calc_meta(self, rank: int) -> CalcMeta:
+        """Stage args with the ORIGINAL orientation restored: this solver was
+        built on transposed slices (q' = k), so host/remote slices transpose
+        back before local-coordinate mapping. Stage q coords live in the
+        stage-need buffer space; k coords in my hosted rows."""
+        plan = self.plans[rank]
+        hr = plan.host_ranges
+
+        def restored(slices_t: List[MaskSlice], qspace: AttnRanges,
+                     q_merged: bool) -> AttnArg:
+            qr, kr, tm = [], [], []
+            area = 0
+            mx = 0
+            for sl_t in slices_t:
+                for o in transpose_slice(sl_t):  # back to original orientation
+                    lq = (qspace.make_range_local(AttnRange(o.qs, o.qe),
+                                                  is_self_merged=True)
+                          if q_merged else
+                          qspace.make_range_local(AttnRange(o.qs, o.qe)))
+                    lk = hr.make_range_local(AttnRange(o.ks, o.ke),
+                                             is_self_merged=True)
+                    qr.append((lq.start, lq.end))
+                    kr.append((lk.start, lk.end))
+                    tm.append(o.t)
+                    area += o.area()
+                    mx = max(mx, lq.end - lq.start)
+            return AttnArg(qr, kr, tm, max_seqlen_q=mx, total_area=area)
+
+        host_arg = restored(plan.host_slices, hr, True)
+        stage_args = []
+        for s in range(self.overlap_degree):
+            st = plan.stages_need[s]
+            stage_slices: List[MaskSlice] = []
+            for sl in plan.remote_slices:
+                for piece in st:
+                    stage_slices.extend(k_window(sl, piece.start, piece.end))
+            stage_args.append(restored(stage_slices, st, False))
+        return CalcMeta(host_arg=host_arg, stage_args=stage_args)
+
+    def make_qo_comm_meta(self, rank: int) -> QoCommMeta:
+        m1 = self.make_comm_meta(rank, ncopies=1)
+        m2 = self.make_comm_meta(rank, ncopies=2)
+        return QoCommMeta(
+            calc=self.make_qo_calc_meta(rank),
+            stages_cast1=m1.stages_cast,
+            stages_cast2=m2.stages_cast,
+            stages_reduce1=m1.stages_reduce,
+        )
 
     # ---------------- hierarchical comm meta ----------------
     def make_hier_comm_meta(

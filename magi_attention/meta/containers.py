@@ -93,6 +93,9 @@ class GroupCastArg:
     recv_unpack: RowChunkMap          # recv buffer rows -> stage buffer
     output_split_sizes: List[int]     # per src rank (k rows)
     stage_tokens: int                 # stage buffer rows (k)
+    # tensor copies packed through the same row tables: 2 for (K,V) or
+    # (q,do) / (lse,dpsum); 1 for single-tensor casts (QO-comm q)
+    ncopies: int = 2
 
 
 @dataclass
@@ -159,6 +162,19 @@ class CommMeta:
     @property
     def overlap_degree(self) -> int:
         return len(self.stages_cast)
+
+
+@dataclass
+class QoCommMeta:
+    """QO-comm plan (MAGI_ATTENTION_QO_COMM=1; reference env/comm.py:72,
+    dist_attn.py:1659 _fetch_remote_qo_do_lse): remote slices are computed at
+    the rank hosting their K side. Built from the solver run on the
+    TRANSPOSED mask, so the remote-K machinery emits remote-Q tables."""
+
+    calc: "CalcMeta"                  # orientation-restored stage args
+    stages_cast1: List[GroupCastArg]  # single-tensor cast (fwd q)
+    stages_cast2: List[GroupCastArg]  # doubled cast (bwd q+do / lse+dpsum)
+    stages_reduce1: List[GroupReduceArg]  # partial (out,lse) fwd / dq bwd
 
 
 @dataclass

@@ -121,6 +121,17 @@ def normalize(
     return out
 
 
+def transpose_slice(sl: MaskSlice) -> List[MaskSlice]:
+    """The same mask region viewed with axes swapped: (q', k') = (k, q),
+    band lo <= k-q <= up  ->  -up <= k'-q' <= -lo. Used by QO-comm planning
+    (the solver runs on the transposed mask so remote-K machinery produces
+    remote-Q tables; reference env/comm.py:72 MAGI_ATTENTION_QO_COMM)."""
+    lo, up = sl.bounds()
+    lo2 = -up if up is not None else None
+    up2 = -lo if lo is not None else None
+    return normalize(sl.ks, sl.ke, sl.qs, sl.qe, lo2, up2)
+
+
 def q_window(sl: MaskSlice, a: int, b: int) -> List[MaskSlice]:
     """Sub-slices of sl restricted to q rows [a, b)."""
     lo, up = sl.bounds()
