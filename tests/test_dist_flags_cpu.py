@@ -151,3 +151,38 @@ def test_qo_ws4():
     port = _free_port()
     mp.spawn(_worker, args=(4, port, "varlen_mixed", 2, FLAG_SETS["qo"]),
              nprocs=4, join=True)
+
+
+def _worker_gatherv(rank, ws, port, _a, _b, _c):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=ws)
+    try:
+        from magi_attention.comm.primitive import all_gather_v, reduce_scatter_v
+
+        sizes = [3, 5, 0, 7][:ws]
+        g = torch.Generator().manual_seed(5 + rank)
+        local = torch.randn(sizes[rank], 4, generator=g)
+        full = all_gather_v(local, sizes, dist.group.WORLD)
+        # reference: gather via object exchange
+        obj = [None] * ws
+        dist.all_gather_object(obj, local)
+        torch.testing.assert_close(full, torch.cat(obj, dim=0))
+
+        contrib = torch.randn(sum(sizes), 4,
+                              generator=torch.Generator().manual_seed(50 + rank))
+        mine = reduce_scatter_v(contrib.clone(), sizes, dist.group.WORLD)
+        obj = [None] * ws
+        dist.all_gather_object(obj, contrib)
+        total = torch.stack(obj).sum(0)
+        start = sum(sizes[:rank])
+        torch.testing.assert_close(mine, total[start:start + sizes[rank]])
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("ws", [2, 4])
+def test_uneven_gather_reduce_scatter(ws):
+    port = _free_port()
+    mp.spawn(_worker_gatherv, args=(ws, port, None, None, None), nprocs=ws,
+             join=True)
