@@ -133,6 +133,7 @@ struct BwdParams {
   long long total_q, total_k;
   int debug_ablate;  // perf ablation only: 1=skip dq stores, 2=skip dkv stores
   int head_major;    // 1: blockIdx.x = head (XCD-affine); 0: work-major
+  int work_nx, work_ny, work_nz;  // flattened work space (strided-grid mode)
   int head_mult;     // deterministic GQA split: real head = off + idx*mult
   int head_off;
   int n_heads_launch;
@@ -206,13 +207,25 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   // Adaptive grid (see fwd kernel): head-major = XCD-affine atomics + K/V
   // L2 locality per head; work-major when one head's operands overflow an
   // XCD's L2.
-  const int ri = blockIdx.z;
-  const int hidx = p.head_major ? blockIdx.x : blockIdx.y;
+  // CU-margin strided-grid mode (see ffa_fwd.hip): margin>0 caps the grid
+  // below the resident-WG count; each WG walks the flattened work space.
+  const long long work_total =
+      (long long)p.work_nx * p.work_ny * p.work_nz;
+  const long long grid_span = (long long)gridDim.x * gridDim.y * gridDim.z;
+  for (long long w = blockIdx.x +
+           (long long)gridDim.x * (blockIdx.y + (long long)gridDim.y * blockIdx.z);
+       w < work_total; w += grid_span) {
+  const int wx = (int)(w % p.work_nx);
+  const long long w2 = w / p.work_nx;
+  const int wy = (int)(w2 % p.work_ny);
+  const int wz = (int)(w2 / p.work_ny);
+  const int ri = wz;
+  const int hidx = p.head_major ? wx : wy;
   const int h = p.head_off + hidx * p.head_mult;
-  const int wb = p.head_major ? blockIdx.y : blockIdx.x;
+  const int wb = p.head_major ? wy : wx;
   const int ks = p.k_ranges[2 * ri], ke = p.k_ranges[2 * ri + 1];
   const int nblk0 = ks + wb * (BWD_BN * WAVES);
-  if (nblk0 >= ke) return;
+  if (nblk0 >= ke) continue;
   const int seg0 = p.seg_starts ? p.seg_starts[ri] : ri;
   const int seg1 = p.seg_starts ? p.seg_starts[ri + 1] : ri + 1;
   int qs, qe, atype;
@@ -588,7 +601,7 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   }  // seg
 
   // ---- write dK/dV ----
-  if (!wave_live || (p.debug_ablate & 2)) return;
+  if (wave_live && !(p.debug_ablate & 2)) {
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     const int kr = n0 + crow(r, hi);
@@ -607,6 +620,8 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
       }
     }
   }
+  }  // wave_live store
+  }  // strided work loop
 }
 
 
@@ -628,12 +643,24 @@ void ffa_bwd_dq_kernel(BwdParams p) {
   auto swz = [](int row, int byte_off) {
     return byte_off ^ ((row & SW32M) << 5);
   };
-  const int ri = blockIdx.z;
-  const int h = p.head_major ? blockIdx.x : blockIdx.y;
-  const int wb = p.head_major ? blockIdx.y : blockIdx.x;
+  // CU-margin strided-grid mode (see ffa_fwd.hip): margin>0 caps the grid
+  // below the resident-WG count; each WG walks the flattened work space.
+  const long long work_total =
+      (long long)p.work_nx * p.work_ny * p.work_nz;
+  const long long grid_span = (long long)gridDim.x * gridDim.y * gridDim.z;
+  for (long long w = blockIdx.x +
+           (long long)gridDim.x * (blockIdx.y + (long long)gridDim.y * blockIdx.z);
+       w < work_total; w += grid_span) {
+  const int wx = (int)(w % p.work_nx);
+  const long long w2 = w / p.work_nx;
+  const int wy = (int)(w2 % p.work_ny);
+  const int wz = (int)(w2 / p.work_ny);
+  const int ri = wz;
+  const int h = p.head_major ? wx : wy;
+  const int wb = p.head_major ? wy : wx;
   const int qs = p.q_ranges[2 * ri], qe = p.q_ranges[2 * ri + 1];
   const int mblk0 = qs + wb * (BWD_BM * WAVES);
-  if (mblk0 >= qe) return;
+  if (mblk0 >= qe) continue;
   const int seg0 = p.seg_starts ? p.seg_starts[ri] : ri;
   const int seg1 = p.seg_starts ? p.seg_starts[ri + 1] : ri + 1;
   int ks, ke, atype;
@@ -860,7 +887,7 @@ void ffa_bwd_dq_kernel(BwdParams p) {
   }  // seg
 
   // ---- store dq once (atomicAdd: q_ranges of different slices may overlap) ----
-  if (skip_dq) return;
+  if (!skip_dq) {
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     const int qr = m0 + crow(r, hi);
@@ -872,6 +899,8 @@ void ffa_bwd_dq_kernel(BwdParams p) {
       if (val != 0.f) unsafeAtomicAdd(dst + dt * 32 + lo32, val);
     }
   }
+  }  // !skip_dq
+  }  // strided work loop
 }
 
 
@@ -1240,6 +1269,7 @@ extern "C" int magi_ffa_bwd_dq(const magi_ffa_bwd_args* a) {
     dim3 grid64 = p.head_major
                       ? dim3(a->hq, qblocks64, (unsigned)a->n_ranges)
                       : dim3(qblocks64, a->hq, (unsigned)a->n_ranges);
+    p.work_nx = grid64.x; p.work_ny = grid64.y; p.work_nz = grid64.z;
     dim3 block64(256);
 #define LAUNCH_DQ64(DD, SC) \
   hipLaunchKernelGGL((ffa_bwd_dq64_kernel<DD, SC, 3>), grid64, block64, 0, s, p)
@@ -1252,6 +1282,16 @@ extern "C" int magi_ffa_bwd_dq(const magi_ffa_bwd_args* a) {
   const int qblocks = (int)((a->total_q + qspan - 1) / qspan);
   dim3 grid_q = p.head_major ? dim3(a->hq, qblocks, (unsigned)a->n_ranges)
                              : dim3(qblocks, a->hq, (unsigned)a->n_ranges);
+  p.work_nx = grid_q.x; p.work_ny = grid_q.y; p.work_nz = grid_q.z;
+  const int margin = a->cu_margin & 0xFFFF;
+  if (margin > 0) {  // CU-margin: cap resident WGs (see ffa_fwd.hip)
+    const long long total = (long long)grid_q.x * grid_q.y * grid_q.z;
+    const int per_cu = (dqw == 4) ? 2 : 1;
+    long long cap = (long long)(256 - margin) * per_cu;
+    if (cap < 1) cap = 1;
+    if (cap > total) cap = total;
+    grid_q = dim3((unsigned)cap, 1, 1);
+  }
   dim3 block(64 * dqw);
 #define LAUNCH_DQ(DD, SC, WW, NB, NTV) \
   hipLaunchKernelGGL((ffa_bwd_dq_kernel<DD, SC, WW, NB, NTV>), grid_q, block, \
@@ -1298,6 +1338,16 @@ static int launch_bwd_dkv(const magi_ffa_bwd_args* a) {
   dim3 grid_kv = p.head_major
                      ? dim3(p.n_heads_launch, nblocks, (unsigned)a->n_ranges)
                      : dim3(nblocks, p.n_heads_launch, (unsigned)a->n_ranges);
+  p.work_nx = grid_kv.x; p.work_ny = grid_kv.y; p.work_nz = grid_kv.z;
+  const int margin = a->cu_margin & 0xFFFF;
+  if (margin > 0) {  // CU-margin: cap resident WGs (see ffa_fwd.hip)
+    const long long total = (long long)grid_kv.x * grid_kv.y * grid_kv.z;
+    const int per_cu = (W == 4 && MODE != 0) ? 2 : 1;
+    long long cap = (long long)(256 - margin) * per_cu;
+    if (cap < 1) cap = 1;
+    if (cap > total) cap = total;
+    grid_kv = dim3((unsigned)cap, 1, 1);
+  }
   dim3 block(64 * W);
   hipStream_t s = (hipStream_t)a->stream;
   const bool sc = a->softcap > 0.f;

@@ -93,6 +93,7 @@ struct FwdParams {
                          // per-segment attn types; NULL = one segment per ri
   int hq, hk, gqa;    // gqa = hq / hk
   int head_major;     // 1: grid.x = head (XCD-affine; per-head KV fits L2)
+  int work_nx, work_ny, work_nz;  // flattened work space (strided-grid mode)
   int n_lock_slots;
   float scale;        // softmax_scale
   float softcap;
@@ -109,12 +110,27 @@ __global__ __launch_bounds__(64 * WAVES, 8 / WAVES) void ffa_fwd_kernel(FwdParam
   // Adaptive grid: head-major (blockIdx.x = head -> one XCD per head: L2
   // locality for K/V and merge traffic) when one head's K/V fits an XCD's
   // 4 MB L2; m-block-major otherwise (all XCDs stream the same K window).
-  const int ri = p.head_major ? blockIdx.z : blockIdx.y;
-  const int h = p.head_major ? blockIdx.x : blockIdx.z;
-  const int mb = p.head_major ? blockIdx.y : blockIdx.x;
+  // CU-margin (reference env/comm.py:44 sm_margin): when cu_margin > 0 the
+  // launcher caps the grid below the chip's resident-WG count and each WG
+  // walks the flattened work space, leaving `margin` CUs permanently free
+  // for comm kernels. With margin 0 the grid covers the work exactly and
+  // this loop runs once per WG (identical to the r1 3D grid).
+  const long long work_total =
+      (long long)p.work_nx * p.work_ny * p.work_nz;
+  const long long grid_span = (long long)gridDim.x * gridDim.y * gridDim.z;
+  for (long long w = blockIdx.x +
+           (long long)gridDim.x * (blockIdx.y + (long long)gridDim.y * blockIdx.z);
+       w < work_total; w += grid_span) {
+  const int wx = (int)(w % p.work_nx);
+  const long long w2 = w / p.work_nx;
+  const int wy = (int)(w2 % p.work_ny);
+  const int wz = (int)(w2 / p.work_ny);
+  const int ri = p.head_major ? wz : wy;
+  const int h = p.head_major ? wx : wz;
+  const int mb = p.head_major ? wy : wx;
   const int qs = p.q_ranges[2 * ri], qe = p.q_ranges[2 * ri + 1];
   const int m0 = qs + mb * (32 * WAVES);
-  if (m0 >= qe) return;                       // uniform across block
+  if (m0 >= qe) continue;                     // uniform across block
   // auto_range_merge: iterate this unique q range's k segments in-kernel,
   // online softmax carried across them (reference merge_range.cu semantics)
   const int seg0 = p.qk_starts ? p.qk_starts[ri] : ri;
@@ -383,7 +399,7 @@ __global__ __launch_bounds__(64 * WAVES, 8 / WAVES) void ffa_fwd_kernel(FwdParam
     // LDS reads of this segment retired before the next segment's staging
     __syncthreads();
   }
-  if (!any_seg) return;
+  if (any_seg) {
 
   // ======================= epilogue =======================
   if (p.max_logits) {
@@ -424,8 +440,7 @@ __global__ __launch_bounds__(64 * WAVES, 8 / WAVES) void ffa_fwd_kernel(FwdParam
           p.out_f32[base + dt * 32 + lo32] = val;
       }
     }
-    return;
-  }
+  } else {
 
   // ---- lock-guarded read-merge-write (range locks; a 32*WAVES-row block
   // spans up to 32*WAVES/LOCK_GRAN+1 slots, acquired in ascending order —
@@ -487,6 +502,9 @@ __global__ __launch_bounds__(64 * WAVES, 8 / WAVES) void ffa_fwd_kernel(FwdParam
       __hip_atomic_store(p.locks + (size_t)sl * p.hq + h, 0, __ATOMIC_RELEASE,
                          __HIP_MEMORY_SCOPE_AGENT);
   }
+  }  // ATOMIC epilogue
+  }  // any_seg
+  }  // strided work loop
 }
 
 // ------------------------------------------------------------------
@@ -607,6 +625,19 @@ extern "C" int magi_ffa_fwd(const magi_ffa_fwd_args* a) {
   dim3 grid = p.head_major
                   ? dim3(a->hq, mblocks, (unsigned)a->n_ranges)
                   : dim3(mblocks, (unsigned)a->n_ranges, a->hq);
+  p.work_nx = grid.x;
+  p.work_ny = grid.y;
+  p.work_nz = grid.z;
+  const int margin = a->cu_margin & 0xFFFF;
+  if (margin > 0) {
+    // persistent-strided: cap resident WGs to (256 - margin) CUs
+    const long long total = (long long)grid.x * grid.y * grid.z;
+    const int per_cu = (fw == 4) ? 2 : 1;
+    long long cap = (long long)(256 - margin) * per_cu;
+    if (cap < 1) cap = 1;
+    if (cap > total) cap = total;
+    grid = dim3((unsigned)cap, 1, 1);
+  }
   dim3 block(64 * fw);
   hipStream_t stream = (hipStream_t)a->stream;
   if (a->d == 64)

@@ -200,7 +200,9 @@ class DistAttnRuntime:
             softmax_scale=scale, softcap=softcap, out_type=torch.float32,
             disable_fwd_atomic_reduction=False,
             deterministic=env.is_deterministic_mode_enable(),
-            sm_margin=env.ffa_forward_sm_margin(),
+            # CU reservation only pays when comm kernels need the CUs
+            sm_margin=(env.ffa_forward_sm_margin()
+                       if dist.get_world_size(self.cp_group) > 1 else 0),
             max_seqlen_q=arg.max_seqlen_q,
             max_logits=max_logits, qk_starts=qk_starts,
         )
@@ -391,7 +393,8 @@ class DistAttnRuntime:
             hq=hq, hk=hk, d=d, max_seqlen_k=max_k,
             out_is_fp32=int(out.dtype == torch.float32),
             softmax_scale=scale, softcap=softcap,
-            cu_margin=env.ffa_backward_sm_margin(),
+            cu_margin=(env.ffa_backward_sm_margin()
+                       if dist.get_world_size(self.cp_group) > 1 else 0),
             stream=current_stream_ptr(),
         )
         from .flex_flash_attn import run_bwd_deterministic, run_bwd_passes
