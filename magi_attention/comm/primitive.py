@@ -94,9 +94,14 @@ def group_cast(
         flat = 1
         for t in h_tail:
             flat *= t
+        rv = recv.view(recv.shape[0], flat)
+        sv = send.view(send.shape[0], flat)
+        if recv.dtype == torch.float8_e4m3fn:
+            # RCCL/gloo have no fp8 type: move the same bytes as uint8
+            rv = rv.view(torch.uint8)
+            sv = sv.view(torch.uint8)
         work = dist.all_to_all_single(
-            recv.view(recv.shape[0], flat),
-            send.view(send.shape[0], flat),
+            rv, sv,
             output_split_sizes=arg.output_split_sizes,
             input_split_sizes=arg.input_split_sizes,
             group=group,

@@ -89,6 +89,9 @@ class DistAttnRuntime:
         return_max_logits: bool = False,
     ) -> Tuple[torch.Tensor, torch.Tensor, Optional[torch.Tensor]]:
         tq, hq, d = q.shape
+        fp8 = q.dtype == torch.float8_e4m3fn
+        if fp8:
+            assert self.qo_meta is None, "fp8 + QO-comm lands later"
         scale = softmax_scale or self.softmax_scale or d ** (-0.5)
 
         out_acc = torch.zeros(tq, hq, d, dtype=torch.float32, device=q.device)
@@ -158,7 +161,9 @@ class DistAttnRuntime:
         if max_logits is not None:
             dist.all_reduce(max_logits, op=dist.ReduceOp.MAX,
                             group=self.cp_group)
-        out = out_acc.to(q.dtype)
+        # the fp8 extension returns bf16 out (same policy as the single-GPU
+        # wrapper: flex_flash_attn.py)
+        out = out_acc.to(torch.bfloat16 if fp8 else q.dtype)
         return out, lse_acc, max_logits
 
     def _sink_post(self, out_acc, lse_acc, sink):
@@ -223,6 +228,16 @@ class DistAttnRuntime:
         tq, hq, d = q.shape
         L = k.shape[0]
         scale = softmax_scale or self.softmax_scale or d ** (-0.5)
+        in_dtype = q.dtype
+        if in_dtype == torch.float8_e4m3fn:
+            # mixed-precision policy (single-GPU wrapper, flex_flash_attn.py
+            # :521): backward runs the bf16 kernels over upcast operands
+            q = q.bfloat16()
+            k = k.bfloat16()
+            v = v.bfloat16()
+            dout = dout.bfloat16()
+            if out.dtype == torch.float8_e4m3fn:
+                out = out.bfloat16()
 
         dq_acc = torch.zeros(tq, hq, d, dtype=torch.float32, device=q.device)
         dkv_acc = torch.zeros(2 * L, *k.shape[1:], dtype=torch.float32,
@@ -321,8 +336,10 @@ class DistAttnRuntime:
         if dsink is not None:
             # "sh" sink is replicated; its gradient sums over ALL q rows
             dist.all_reduce(dsink, op=dist.ReduceOp.SUM, group=self.cp_group)
-        dk = dkv_acc[:L].to(k.dtype)
-        dv = dkv_acc[L:].to(v.dtype)
+        if in_dtype == torch.float8_e4m3fn:
+            dq = dq_acc.to(in_dtype)
+        dk = dkv_acc[:L].to(in_dtype)
+        dv = dkv_acc[L:].to(in_dtype)
         return dq, dk, dv, dsink
 
     def _dsink(self, sink, lse, dpsum):

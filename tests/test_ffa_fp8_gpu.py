@@ -81,3 +81,109 @@ def test_fp8_fwd_bwd_autograd():
     torch.cuda.synchronize()
     assert q.grad is not None and q.grad.dtype == torch.float8_e4m3fn
     assert torch.isfinite(q.grad.float()).all()
+
+
+@requires_gpu
+def test_fp8_fwd_bwd_oracle_parity():
+    """Mixed-precision fp8 backward vs the fp64 oracle: ground truth uses the
+    UPCAST fp8 values as exact inputs (the quantisation is the input, not the
+    error); budget = the bf16-oracle error with the fp8-appropriate ratio
+    (SURVEY §8c fp8 caveat: calibrated, looser thresholds)."""
+    from oracle import make_attn_mask, ref_attn_with_grads
+    from tests.util import assert_close_to_ref
+    from magi_attention.functional import flex_flash_attn_func
+
+    tq = tk = 2048
+    hq, hk, d = 8, 2, 128
+    g = torch.Generator().manual_seed(21)
+    qf = (torch.randn(tq, hq, d, generator=g) * 0.5).to(torch.float8_e4m3fn)
+    kf = (torch.randn(tk, hk, d, generator=g) * 0.5).to(torch.float8_e4m3fn)
+    vf = (torch.randn(tk, hk, d, generator=g) * 0.5).to(torch.float8_e4m3fn)
+    dout = (torch.randn(tq, hq, d, generator=g) * 0.5).bfloat16()
+    qr_l = [[0, 1024], [1024, 2048]]
+    kr_l = [[0, 1024], [1024, 2048]]
+    ty_l = [1, 1]
+    q = qf.cuda().requires_grad_(True)
+    k = kf.cuda().requires_grad_(True)
+    v = vf.cuda().requires_grad_(True)
+    qr = torch.tensor(qr_l, dtype=torch.int32, device="cuda")
+    kr = torch.tensor(kr_l, dtype=torch.int32, device="cuda")
+    tm = torch.tensor(ty_l, dtype=torch.int32, device="cuda")
+    out, meta = flex_flash_attn_func(q, k, v, qr, kr, tm)
+    out.backward(dout.cuda())
+    torch.cuda.synchronize()
+
+    mask = make_attn_mask(tq, tk, qr_l, kr_l, ty_l)
+    qc, kc, vc = qf.float(), kf.float(), vf.float()
+    o_hi, _, dq_hi, dk_hi, dv_hi = ref_attn_with_grads(
+        qc, kc, vc, mask, dout.float()
+    )
+    o_lo, _, dq_lo, dk_lo, dv_lo = ref_attn_with_grads(
+        qc, kc, vc, mask, dout.float(), high_precision=False,
+        p_dtype=torch.bfloat16,
+    )
+    # fwd runs on fp8 MFMAs (bf16-converted operands inside): fp8-calibrated
+    assert_close_to_ref(out.detach().cpu().float(), o_hi.float(),
+                        o_lo.float(), "fp8:out", ratio=8.0, floor=2e-2)
+    # bwd runs the bf16 kernels over upcast operands: bf16-calibrated
+    assert_close_to_ref(q.grad.cpu().float(), dq_hi.float(), dq_lo.float(),
+                        "fp8:dq", ratio=4.5, floor=2.5e-2)
+    assert_close_to_ref(k.grad.cpu().float(), dk_hi.float(), dk_lo.float(),
+                        "fp8:dk", ratio=4.5, floor=2.5e-2)
+    assert_close_to_ref(v.grad.cpu().float(), dv_hi.float(), dv_lo.float(),
+                        "fp8:dv", ratio=4.5)
+
+
+@requires_gpu
+def test_fp8_distributed_cp1():
+    """Distributed fp8 (BASELINE config 5's CP form) through the product path
+    at cp=1: calc_attn with fp8 Q/K/V must match the single-GPU fp8 wrapper
+    bitwise (same kernels, same buffers; the wire short-circuits at cp1)."""
+    import os
+    import torch.distributed as dist
+    from magi_attention.api import calc_attn, dispatch, magi_attn_flex_key
+    from magi_attention.common.ranges import AttnRanges
+    from magi_attention.config import DispatchConfig, DistAttnConfig
+    from magi_attention.functional import flex_flash_attn_func
+
+    if not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29541")
+        dist.init_process_group("nccl", rank=0, world_size=1)
+    total, hq, hk, d = 4096, 16, 2, 128
+    g = torch.Generator().manual_seed(17)
+    q = (torch.randn(total, hq, d, generator=g) * 0.5).to(torch.float8_e4m3fn).cuda()
+    k = (torch.randn(total, hk, d, generator=g) * 0.5).to(torch.float8_e4m3fn).cuda()
+    v = (torch.randn(total, hk, d, generator=g) * 0.5).to(torch.float8_e4m3fn).cuda()
+    dout = torch.randn(total, hq, d, generator=g).bfloat16().cuda()
+    key = magi_attn_flex_key(
+        AttnRanges.from_ranges([[0, total]]),
+        AttnRanges.from_ranges([[0, total]]),
+        "causal", total, total, hq, hk, d,
+        cp_group_or_mesh=dist.group.WORLD,
+        dist_attn_config=DistAttnConfig(
+            dispatch_config=DispatchConfig(chunk_size=512)),
+    )
+    ql = dispatch(q, key).requires_grad_(True)
+    kl = dispatch(k, key).requires_grad_(True)
+    vl = dispatch(v, key).requires_grad_(True)
+    out, meta = calc_attn(ql, kl, vl, key)
+    assert out.dtype == torch.bfloat16
+    out.backward(dispatch(dout, key))
+    torch.cuda.synchronize()
+    assert ql.grad.dtype == torch.float8_e4m3fn
+
+    # single-GPU reference path on the SAME (dispatched) layout
+    from magi_attention.api import get_position_ids
+    pos = get_position_ids(key)
+    qr = torch.tensor([[0, total]], dtype=torch.int32, device="cuda")
+    tm = torch.tensor([1], dtype=torch.int32, device="cuda")
+    q2 = q.clone().requires_grad_(True)
+    k2 = k.clone().requires_grad_(True)
+    v2 = v.clone().requires_grad_(True)
+    o2, _ = flex_flash_attn_func(q2, k2, v2, qr, qr.clone(), tm)
+    o2.backward(dout)
+    torch.cuda.synchronize()
+    torch.testing.assert_close(out, o2[pos], atol=0, rtol=0)
+    torch.testing.assert_close(ql.grad.float(), q2.grad[pos].float(),
+                               atol=0.25, rtol=0.1)
