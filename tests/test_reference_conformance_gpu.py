@@ -109,14 +109,18 @@ def test_ffa_random(model, gen, attn_type):
 
 # named mask configs from the reference's test_pipeline.py:403-857 (scaled to
 # keep the fp64 CPU oracle fast; structure preserved)
+# NOTE: like the reference's configs these are AREA-disjoint — q ranges may
+# overlap (exercising the lock-merge epilogue) but no (q,k) pair is covered
+# twice (a doubly-covered pair legitimately counts twice in the softmax, in
+# this engine and in the reference alike, while a dense-mask oracle ORs it).
 PIPELINE_CONFIGS = {
-    # varlen_block_causal_12k_with_q_overlap (:520): causal blocks whose
-    # q ranges overlap previous blocks
+    # varlen_block_causal_12k_with_q_overlap (:520, scaled): causal blocks
+    # whose q ranges overlap previous blocks, disjoint k coverage
     "varlen_block_causal_with_q_overlap": dict(
         total=3072,
         q_ranges=[[0, 1024], [512, 2048], [1536, 3072], [2048, 3072]],
-        k_ranges=[[0, 1024], [0, 2048], [1024, 2560], [0, 512]],
-        types=[1, 1, 1, 0],
+        k_ranges=[[0, 1024], [1024, 2048], [2048, 2560], [0, 512]],
+        types=[1, 1, 0, 0],
     ),
     # full_mask_assembled_from_small_pieces_with_8k (:700, scaled)
     "full_from_small_pieces": dict(
@@ -127,11 +131,11 @@ PIPELINE_CONFIGS = {
                   [0, 1024], [1024, 2048]],
         types=[0, 0, 0, 0, 0, 0],
     ),
-    # bi_causal_12k_with_q_overlap (:610, scaled): bi-causal + overlap
+    # bi_causal_12k_with_q_overlap (:610, scaled): bi-causal bands + q overlap
     "bi_causal_with_q_overlap": dict(
         total=2560,
-        q_ranges=[[0, 1280], [640, 2560], [0, 2560]],
-        k_ranges=[[0, 1280], [1280, 2560], [0, 640]],
+        q_ranges=[[0, 1024], [512, 1536], [1536, 2560]],
+        k_ranges=[[0, 1280], [1280, 2304], [2304, 2560]],
         types=[3, 3, 0],
     ),
 }
