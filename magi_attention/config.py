@@ -46,7 +46,12 @@ class SortedSequentialSelectAlg(DispatchAlg):
 
 @dataclass(frozen=True)
 class ToppHeapDispatchAlg(DispatchAlg):
+    """MinHeap with IOU-affinity tie-breaks: the top max(1, ceil(cp*top_p))
+    least-loaded ranks are candidates, affinity picks among them (reference
+    dispatch_solver.py:207,990)."""
+
     type: DispatchAlgType = DispatchAlgType.TOPP_HEAP
+    top_p: float = 0.5
 
 
 # ---- overlap algorithms ----

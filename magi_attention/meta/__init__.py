@@ -79,8 +79,24 @@ def make_dispatch_meta_from_qk_ranges(
     for c in range(num_chunks):
         a, b = c * chunk_size, (c + 1) * chunk_size
         workloads.append(float(sum(area_in_rows(sl, a, b) for sl in slices)))
+    affinities = None
+    from ..common.enum import DispatchAlgType
+
+    if dist_attn_config.dispatch_config.alg.type == DispatchAlgType.TOPP_HEAP:
+        # per-chunk k coverage for the IOU-affinity tie-break
+        from .geometry import q_window
+
+        affinities = []
+        for c in range(num_chunks):
+            a, b = c * chunk_size, (c + 1) * chunk_size
+            rr = AttnRanges()
+            for sl in slices:
+                for sub in q_window(sl, a, b):
+                    rr.append(AttnRanges.from_ranges([[sub.ks, sub.ke]])[0])
+            affinities.append(rr.merge())
     sol = DispatchSolver(dist_attn_config.dispatch_config.alg).solve(
-        workloads, cp_size
+        workloads, cp_size, affinities=affinities,
+        uneven_shard=dist_attn_config.dispatch_config.uneven_shard,
     )
     return DispatchMeta(
         cp_size=cp_size,
@@ -102,6 +118,8 @@ def make_attn_meta_from_dispatch_meta(
     rank's calc/comm tables."""
     oc = dist_attn_config.overlap_config
     degree = (oc.degree or 1) if oc.enable else 1
+    from ..common.enum import AttnOverlapMode, OverlapAlgType
+
     solver = DistAttnSolver(
         slices=slices,
         partitions=dispatch_meta.partitions,
@@ -110,6 +128,11 @@ def make_attn_meta_from_dispatch_meta(
         cp_size=dispatch_meta.cp_size,
         overlap_degree=degree,
         min_stage_tokens=oc.min_chunk_size,
+        overlap_mode=("dynamic" if oc.enable
+                      and oc.mode == AttnOverlapMode.DYNAMIC else "static"),
+        overlap_alg=("greedy" if oc.alg.type == OverlapAlgType.GREEDY
+                     else "uniform"),
+        max_num_chunks=oc.max_num_chunks,
     )
     rank = dispatch_meta.cp_rank
     return solver, solver.make_calc_meta(rank), solver.make_comm_meta(rank)

@@ -62,6 +62,53 @@ def _split_stages(need: AttnRanges, degree: int, min_stage_tokens: int = 512
     return out
 
 
+def _split_stages_area(
+    need: AttnRanges, remote_slices: List[MaskSlice], degree: int,
+    min_stage_tokens: int,
+) -> List[AttnRanges]:
+    """Greedy area-balanced stage composition (reference GreedyOverlapAlg):
+    cut the remote-need ranges into ~min_stage_tokens pieces, compute the
+    mask AREA each unlocks, and min-heap the pieces into `degree` stages by
+    accumulated area — stages then carry comparable COMPUTE, not comparable
+    tokens (a token-balanced split under a causal mask gives the last stage
+    far more area than the first)."""
+    import heapq
+
+    pieces: List[Tuple[int, int]] = []
+    for r in need.merge():
+        a = r.start
+        while a < r.end:
+            b = min(a + min_stage_tokens, r.end)
+            pieces.append((a, b))
+            a = b
+    if not pieces:
+        return [AttnRanges() for _ in range(degree)]
+
+    def area_of(a: int, b: int) -> int:
+        tot = 0
+        for sl in remote_slices:
+            for sub in k_window(sl, a, b):
+                tot += sub.area()
+        return tot
+
+    degree = max(1, min(degree, len(pieces)))
+    scored = sorted(((area_of(a, b), a, b) for a, b in pieces), reverse=True)
+    heap = [(0.0, i) for i in range(degree)]
+    heapq.heapify(heap)
+    buckets: List[List[Tuple[int, int]]] = [[] for _ in range(degree)]
+    for ar, a, b in scored:
+        load, i = heapq.heappop(heap)
+        buckets[i].append((a, b))
+        heapq.heappush(heap, (load + ar, i))
+    out = []
+    for bk in buckets:
+        rr = AttnRanges()
+        for a, b in sorted(bk):
+            rr.append(AttnRange(a, b))
+        out.append(rr.merge())
+    return out
+
+
 @dataclass
 class RankAttnPlan:
     host_ranges: AttnRanges              # global rows hosted (merged)
@@ -83,7 +130,20 @@ class DistAttnSolver:
         cp_size: int,
         overlap_degree: int = 2,
         min_stage_tokens: int = 512,
+        overlap_mode: str = "static",
+        overlap_alg: str = "uniform",
+        max_num_chunks: int = 64,
     ):
+        """overlap_mode "dynamic" (reference AttnOverlapMode.DYNAMIC,
+        overlap_solver.py): the per-rank stage count follows the remote-need
+        size (one stage per ~4*min_stage_tokens, capped by max_num_chunks)
+        instead of a fixed degree; all ranks still pad to the global max so
+        every rank runs the same number of collective rounds (the reference
+        all-reduces the degree, dist_attn_solver.py:994 — here every rank
+        derives all plans deterministically, so it is a host max).
+        overlap_alg "greedy" (reference GreedyOverlapAlg, overlap_solver.py:
+        205): stages balance the CALC AREA they unlock, not the token count —
+        min-heap assignment of min_stage_tokens-sized pieces by area."""
         self.cp_size = cp_size
         self.chunk_size = chunk_size
         self.total_seqlen = total_seqlen
@@ -117,7 +177,18 @@ class DistAttnSolver:
                     remote_slices.extend(k_window(sl, p.start, p.end))
                     remote_need.append(p.clone())
             remote_need = remote_need.merge()
-            stages = _split_stages(remote_need, overlap_degree, min_stage_tokens)
+            deg = overlap_degree
+            if overlap_mode == "dynamic":
+                tok = remote_need.total_seqlen
+                deg = max(1, min((tok + 4 * min_stage_tokens - 1)
+                                 // (4 * min_stage_tokens),
+                                 max_num_chunks))
+            if overlap_alg == "greedy" and deg > 1:
+                stages = _split_stages_area(
+                    remote_need, remote_slices, deg, min_stage_tokens
+                )
+            else:
+                stages = _split_stages(remote_need, deg, min_stage_tokens)
             max_deg = max(max_deg, len(stages))
             self.plans.append(
                 RankAttnPlan(hr, host_slices, remote_slices, stages)

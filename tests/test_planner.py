@@ -323,3 +323,91 @@ def test_solver_coverage_fuzz():
                 add(arg, solver.plans[rank].stages_need[s])
         assert torch.equal(got.bool(), want), f"trial {trial}: coverage mismatch"
         assert (got <= 1).all(), f"trial {trial}: duplicated area"
+
+
+# ---- r2 solver depth: algs / overlap modes / uneven shards ----
+
+def _alg_by_name(name):
+    from magi_attention.config import (
+        LBDispatchAlg, MinHeapDispatchAlg, SequentialDispatchAlg,
+        SortedSequentialSelectAlg, ToppHeapDispatchAlg,
+    )
+    return {
+        "minheap": MinHeapDispatchAlg(), "seq": SequentialDispatchAlg(),
+        "sortedseq": SortedSequentialSelectAlg(),
+        "topp": ToppHeapDispatchAlg(), "lb": LBDispatchAlg(),
+    }[name]
+
+
+@pytest.mark.parametrize("alg", ["minheap", "seq", "sortedseq", "topp", "lb"])
+@pytest.mark.parametrize("mode_alg", [("static", "uniform"),
+                                      ("dynamic", "uniform"),
+                                      ("static", "greedy"),
+                                      ("dynamic", "greedy")])
+@pytest.mark.parametrize("seed", [1, 3])
+def test_planner_alg_mode_matrix(alg, mode_alg, seed):
+    """Every (dispatch alg x overlap mode x overlap alg) combination must
+    still satisfy exact coverage + zero redundancy (the invariant of
+    test_planner_covers_global_mask)."""
+    from magi_attention.common.enum import AttnOverlapMode
+    from magi_attention.config import GreedyOverlapAlg, UniformOverlapAlg
+
+    mode, oalg = mode_alg
+    total, cp, chunk = 1024, 4, 128
+    q_ranges, k_ranges, types, slices, _ = build_case(seed, total, cp, chunk)
+    cfg = DistAttnConfig(
+        dispatch_config=DispatchConfig(chunk_size=chunk, alg=_alg_by_name(alg)),
+        overlap_config=OverlapConfig(
+            degree=2, min_chunk_size=64,
+            mode=(AttnOverlapMode.DYNAMIC if mode == "dynamic"
+                  else AttnOverlapMode.STATIC),
+            alg=(GreedyOverlapAlg() if oalg == "greedy"
+                 else UniformOverlapAlg()),
+        ),
+    )
+    dm = make_dispatch_meta_from_qk_ranges(slices, total, cp, 0, cfg)
+    ref_mask = make_attn_mask(total, total, q_ranges, k_ranges, types)
+    cover = torch.zeros(total, total, dtype=torch.int32)
+    for r in range(cp):
+        dm.cp_rank = r
+        solver, calc, comm = make_attn_meta_from_dispatch_meta(slices, dm, cfg)
+        plan = solver.plans[r]
+        for sl in plan.host_slices + plan.remote_slices:
+            cover += to_dense([sl], total, total).int()
+        need = AttnRanges()
+        for sl in plan.remote_slices:
+            need.append(AttnRanges.from_ranges([(sl.ks, sl.ke)])[0])
+        st_tokens = sum(s.total_seqlen for s in plan.stages_need)
+        assert st_tokens == need.merge().total_seqlen, "zero redundancy"
+    assert torch.equal(cover.bool(), ref_mask)
+    assert int(cover.max()) <= 1
+
+
+def test_dispatch_solver_uneven_shards():
+    from magi_attention.meta.solver.dispatch_solver import DispatchSolver
+    from magi_attention.config import MinHeapDispatchAlg, SequentialDispatchAlg
+
+    w = [5.0, 1.0, 3.0, 2.0, 8.0, 1.0, 4.0]  # 7 chunks, cp=3
+    for alg in (MinHeapDispatchAlg(), SequentialDispatchAlg()):
+        sol = DispatchSolver(alg).solve(w, 3, uneven_shard=True)
+        counts = sorted(len(p) for p in sol.partitions)
+        assert counts == [2, 2, 3]
+        assert sorted(c for p in sol.partitions for c in p) == list(range(7))
+        for p, l in zip(sol.partitions, sol.loads):
+            assert abs(sum(w[c] for c in p) - l) < 1e-9
+
+
+def test_topp_heap_affinity_reduces_spread():
+    """IOU affinity must group chunks touching the same k rows onto the same
+    rank when loads are comparable."""
+    from magi_attention.meta.solver.dispatch_solver import DispatchSolver
+    from magi_attention.config import ToppHeapDispatchAlg
+
+    w = [1.0] * 8
+    aff = [AttnRanges.from_ranges([[0, 100]]) for _ in range(4)] + \
+          [AttnRanges.from_ranges([[100, 200]]) for _ in range(4)]
+    sol = DispatchSolver(ToppHeapDispatchAlg(top_p=1.0)).solve(
+        w, 2, affinities=aff)
+    for p in sol.partitions:
+        groups = {0 if c < 4 else 1 for c in p}
+        assert len(groups) == 1, f"affinity should not mix groups: {sol.partitions}"
