@@ -391,35 +391,47 @@ def measure_extra_configs(device):
 
 
 def measure_cpu_baseline():
-    """Oracle (CPU restatement, kind="port") timed on the host cores over a
-    BOUNDED sample: one causal 4096-token slice of the same workload,
-    fwd+bwd, scaled to the TFLOPS metric."""
+    """Oracle (CPU restatement, kind="port") timed on the host cores at the
+    REAL 64k workload shape, blockwise: four
+    1024-row q blocks spread across the 64k sequence, each attending its
+    full causal k span (up to 64k keys), fwd+bwd in fp32 — a bounded
+    (~15-25 s) sample of the exact per-row work of the metric's workload."""
     sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
-    from oracle import make_attn_mask, ref_attn_with_grads
 
-    n = 4096
     cores = os.cpu_count() or 1
     torch.set_num_threads(cores)
     g = torch.Generator().manual_seed(42)
-    q = torch.randn(n, HQ, D, generator=g) * 0.5
-    k = torch.randn(n, HKV, D, generator=g) * 0.5
-    v = torch.randn(n, HKV, D, generator=g) * 0.5
-    do = torch.randn(n, HQ, D, generator=g)
-    mask = make_attn_mask(n, n, [[0, n]], [[0, n]], [1])
-    # fp32 oracle pass for timing (the fp64 path is the parity oracle; fp32 is
-    # the fair CPU-throughput baseline)
+    k = torch.randn(SEQLEN, HKV, D, generator=g) * 0.5
+    v = torch.randn(SEQLEN, HKV, D, generator=g) * 0.5
+    qb = 1024
+    blocks = [16384, 49152]  # q block starts across the 64k seq
+    scale = D ** -0.5
     t0 = time.perf_counter()
-    ref_attn_with_grads(q, k, v, mask, do, high_precision=False)
+    area = 0
+    for b0 in blocks:
+        q = (torch.randn(qb, HQ, D, generator=g) * 0.5).requires_grad_(True)
+        kspan = b0 + qb  # causal: rows b0..b0+qb see keys [0, row]
+        ks = k[:kspan].clone().requires_grad_(True)
+        vs = v[:kspan].clone().requires_grad_(True)
+        s = torch.einsum("qhd,khd->hqk", q, ks) * scale
+        rows = torch.arange(b0, b0 + qb).unsqueeze(-1)
+        cols = torch.arange(kspan).unsqueeze(0)
+        s = s.masked_fill((cols > rows).unsqueeze(0), float("-inf"))
+        p = torch.softmax(s, dim=-1)
+        o = torch.einsum("hqk,khd->qhd", p, vs)
+        o.backward(torch.randn(qb, HQ, D, generator=g))
+        area += int((rows - cols.clamp(max=rows) >= 0).sum())
     dt = time.perf_counter() - t0
-    area = n * (n + 1) // 2
+    area = sum(b0 * qb + qb * (qb + 1) // 2 for b0 in blocks)
     flops = 4 * area * HQ * D * 3.5
     return {
         "value": flops / dt / 1e12,
         "unit": "TFLOPS/s",
         "cores": cores,
         "kind": "port",
-        "sample": f"causal {n}-token slice of the 64k workload, fwd+bwd, "
-                  f"fp32 torch oracle, {dt:.1f}s",
+        "sample": f"blockwise 64k shape: {len(blocks)}x{qb} q rows at offsets {blocks} "
+                  f"with full causal k spans (<=64k keys), fwd+bwd, fp32 "
+                  f"torch oracle, {dt:.1f}s",
     }
 
 
