@@ -209,12 +209,15 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   // XCD's L2.
   // CU-margin strided-grid mode (see ffa_fwd.hip): margin>0 caps the grid
   // below the resident-WG count; each WG walks the flattened work space.
+  // MODE 0 sits at exactly the 256-reg budget and the loop backedge costs it
+  // 41 scratch spills — its launcher never caps the grid, so the loop is
+  // made provably single-iteration there and dissolves at compile time.
   const long long work_total =
       (long long)p.work_nx * p.work_ny * p.work_nz;
   const long long grid_span = (long long)gridDim.x * gridDim.y * gridDim.z;
-  for (long long w = blockIdx.x +
-           (long long)gridDim.x * (blockIdx.y + (long long)gridDim.y * blockIdx.z);
-       w < work_total; w += grid_span) {
+  const long long w_first = blockIdx.x +
+      (long long)gridDim.x * (blockIdx.y + (long long)gridDim.y * blockIdx.z);
+  for (long long w = w_first; w < work_total; w += grid_span) {
   const int wx = (int)(w % p.work_nx);
   const long long w2 = w / p.work_nx;
   const int wy = (int)(w2 % p.work_ny);
@@ -621,6 +624,7 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
     }
   }
   }  // wave_live store
+  if constexpr (MODE == 0) break;  // single trip: kills the backedge (regs)
   }  // strided work loop
 }
 
@@ -1339,7 +1343,7 @@ static int launch_bwd_dkv(const magi_ffa_bwd_args* a) {
                      ? dim3(p.n_heads_launch, nblocks, (unsigned)a->n_ranges)
                      : dim3(nblocks, p.n_heads_launch, (unsigned)a->n_ranges);
   p.work_nx = grid_kv.x; p.work_ny = grid_kv.y; p.work_nz = grid_kv.z;
-  const int margin = a->cu_margin & 0xFFFF;
+  const int margin = (MODE == 0) ? 0 : (a->cu_margin & 0xFFFF);
   if (margin > 0) {  // CU-margin: cap resident WGs (see ffa_fwd.hip)
     const long long total = (long long)grid_kv.x * grid_kv.y * grid_kv.z;
     const int per_cu = (W == 4 && MODE != 0) ? 2 : 1;
