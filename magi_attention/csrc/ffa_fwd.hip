@@ -290,10 +290,29 @@ __global__ __launch_bounds__(64 * WAVES, 8 / WAVES) void ffa_fwd_kernel(FwdParam
     }
     mx = fmaxf(mx, warp_xor32(mx));
 
+    // ---- defer-max (guide RESCALE_THRESHOLD idiom): tolerate per-row max
+    // growth up to 2^8 without touching m_run — P stays bounded by 256 in
+    // the fp32 accumulators and the O-rescale (16 bpermute + 64 mults)
+    // vanishes on the vast majority of tiles. A lane whose first finite max
+    // arrives (m_run -inf -> finite) cannot defer. T13 hazard: the decision
+    // is made ONCE per tile before PV, and the same alpha feeds both l_run
+    // and the O rescale.
     const float m_new = fmaxf(m_run, mx);
-    const float m_use = (m_new == -INFINITY) ? 0.f : m_new;
-    const float alpha = (m_run == -INFINITY) ? 0.f : fast_exp2(m_run - m_use);
-    m_run = m_new;
+    // max_logits reads m_run as the TRUE row max in the epilogue — no slack
+    const float defer_thr = p.max_logits ? 0.f : 8.f;
+    const bool need_rescale =
+        (m_new > m_run + defer_thr) ||
+        (m_run == -INFINITY && m_new != -INFINITY);
+    float m_use;
+    float alpha;
+    if (!__any(need_rescale)) {
+      m_use = (m_run == -INFINITY) ? 0.f : m_run;  // keep the old reference
+      alpha = 1.f;
+    } else {
+      m_use = (m_new == -INFINITY) ? 0.f : m_new;
+      alpha = (m_run == -INFINITY) ? 0.f : fast_exp2(m_run - m_use);
+      m_run = m_new;
+    }
 
     float pr[16];
     float psum = 0.f;
@@ -611,11 +630,10 @@ extern "C" int magi_ffa_fwd(const magi_ffa_fwd_args* a) {
   p.total_q = a->total_q;
   p.total_k = a->total_k;
 
-  // Unlike the bwd dq/dk passes, the forward measured FASTER at 4 waves
-  // (58.1 vs 59.8 ms at 64k, same box A/B): its K/V staging is a smaller
-  // share of the loop than bwd's Q/dO staging, and the 256-row block widens
-  // the causal k-window per block. MAGI_FWD_WAVES=8 opts in for tuning.
-  int fw = 4;
+  // r2: 8 waves (one 512-thread WG/CU) now beat 4 on both 8k (505 vs 471
+  // TF) and 64k (742 vs 716) — the r1 preference for 4 predated the
+  // scheduler-visible tr16 reads, packed converts and raw-exp2 softmax.
+  int fw = 8;
   { const char* e = getenv("MAGI_FWD_WAVES"); if (e && atoi(e)) fw = atoi(e); }
   const int span = 32 * fw;
   const int mblocks = (a->max_seqlen_q + span - 1) / span;
