@@ -113,6 +113,24 @@ class MagiSinkArgs(ctypes.Structure):
     ]
 
 
+class MagiGrpCollPullArgs(ctypes.Structure):
+    """mirrors magi_grpcoll_pull_args in csrc/grpcoll.hip (native grpcoll)"""
+
+    _fields_ = [
+        ("pieces", ctypes.c_void_p),
+        ("n_pieces", ctypes.c_int32),
+        ("row_elems", ctypes.c_int32),
+        ("elem_size", ctypes.c_int32),
+        ("peer_ptrs", ctypes.c_void_p * 8),
+        ("peer_flags", ctypes.c_void_p * 8),
+        ("wait_value", ctypes.c_int32),
+        ("n_peers", ctypes.c_int32),
+        ("dst", ctypes.c_void_p),
+        ("reduce", ctypes.c_int32),
+        ("stream", ctypes.c_void_p),
+    ]
+
+
 class MagiCorrectArgs(ctypes.Structure):
     _fields_ = [
         ("out1", ctypes.c_void_p),
@@ -150,6 +168,20 @@ def _try_load() -> ctypes.CDLL | None:
                 "magi_probe_mfma",
                 [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p],
             ),
+            ("magi_ipc_get_handle", [ctypes.c_void_p, ctypes.c_void_p]),
+            ("magi_ipc_open",
+             [ctypes.c_void_p, ctypes.POINTER(ctypes.c_void_p)]),
+            ("magi_ipc_close", [ctypes.c_void_p]),
+            ("magi_grpcoll_signal",
+             [ctypes.c_void_p, ctypes.c_int32, ctypes.c_void_p]),
+            ("magi_grpcoll_wait",
+             [ctypes.c_void_p, ctypes.c_int32, ctypes.c_void_p]),
+            ("magi_grpcoll_ack",
+             [ctypes.c_void_p, ctypes.c_int32, ctypes.c_void_p]),
+            ("magi_grpcoll_pull", [ctypes.POINTER(MagiGrpCollPullArgs)]),
+            ("magi_ipc_base",
+             [ctypes.c_void_p, ctypes.POINTER(ctypes.c_void_p),
+              ctypes.POINTER(ctypes.c_uint64)]),
         ]:
             fn = getattr(lib, name)
             fn.argtypes = argtypes

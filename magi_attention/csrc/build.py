@@ -24,6 +24,7 @@ SOURCES = [
     CSRC / "range_ops.hip",
     CSRC / "ffa_fwd_fp8.hip",
     CSRC / "ext_utils.hip",
+    CSRC / "grpcoll.hip",
 ]
 
 HIPCC = "hipcc"

@@ -98,6 +98,10 @@ class DistAttnRuntimeMgr:
         self.solver, calc_meta, comm_meta = make_attn_meta_from_dispatch_meta(
             slices, self.dispatch_meta, dist_attn_config
         )
+        if env.is_native_grpcoll_enable():
+            natives, st_all = self.solver.make_native_comm_meta(cp_rank)
+            comm_meta.stages_native = natives
+            comm_meta.stage_tokens_all = st_all
         qo_meta = None
         if env.is_qo_comm_enable():
             from .meta import make_qo_meta_from_dispatch_meta
@@ -209,6 +213,9 @@ def check_flag_comb() -> None:
     if env.is_native_grpcoll_enable():
         assert not env.is_deterministic_mode_enable(), (
             "Native grpcoll is not compatible with deterministic mode for now"
+        )
+        assert not env.is_qo_comm_enable(), (
+            "Native grpcoll is not compatible with qo comm for now"
         )
 
 

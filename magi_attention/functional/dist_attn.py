@@ -52,6 +52,8 @@ class DistAttnRuntime:
     # QO-comm plan (MAGI_ATTENTION_QO_COMM=1): remote slices computed at the
     # K-host; q (fwd) / q,do,lse,dpsum (bwd) travel instead of K/V
     qo_meta: Optional[QoCommMeta] = None
+    # native HIP-IPC grpcoll transport (MAGI_ATTENTION_NATIVE_GRPCOLL=1)
+    _native: object = None
 
     @property
     def overlap_degree(self) -> int:
@@ -62,7 +64,32 @@ class DistAttnRuntime:
         return (self.comm_meta.stages_cast_hier is not None
                 and self.intra_group is not None)
 
+    def _native_grpcoll(self, kv_local: torch.Tensor):
+        """Lazy per-runtime IPC window setup (handles exchanged once)."""
+        if self._native is None:
+            from ..comm.native_grpcoll import NativeGrpColl
+
+            object.__setattr__(self, "_native", NativeGrpColl(
+                self.comm_meta.stages_native,
+                self.comm_meta.stage_tokens_all,
+                self.cp_group,
+                kv_rows=2 * (kv_local.shape[0] // 2),
+                h=kv_local.shape[1], d=kv_local.shape[2],
+                dtype=kv_local.dtype,
+            ))
+        return self._native
+
+    @property
+    def use_native(self) -> bool:
+        return (self.comm_meta.stages_native is not None
+                and env.is_native_grpcoll_enable())
+
     def _cast(self, kv_local: torch.Tensor, s: int) -> WorkWithPostProcessFn:
+        if self.use_native and kv_local.is_cuda:
+            n = self._native_grpcoll(kv_local)
+            if s == 0:
+                n.new_step()
+            return n.cast(kv_local, s)
         if self.use_hier:
             return hier_group_cast(
                 kv_local, self.comm_meta.stages_cast_hier[s],
@@ -72,6 +99,8 @@ class DistAttnRuntime:
                           self.cp_group)
 
     def _reduce(self, partial, dst, s: int) -> WorkWithPostProcessFn:
+        if self.use_native and dst.is_cuda and self._native is not None:
+            return self._native.reduce(partial, dst, s)
         if self.use_hier:
             return hier_group_reduce(
                 partial, dst, self.comm_meta.stages_reduce_hier[s],

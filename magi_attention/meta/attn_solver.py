@@ -346,6 +346,59 @@ class DistAttnSolver:
             )
         return CommMeta(stages_cast=casts, stages_reduce=reduces)
 
+    # ---------------- native grpcoll pull tables ----------------
+    def make_native_comm_meta(self, rank: int):
+        """Per-stage PULL plans for the HIP-IPC transport (csrc/grpcoll.hip):
+        cast: which rows of each peer's kv WINDOW land where in my stage
+        buffer; reduce: which rows of each peer's partial-dKV window sum into
+        my local dkv accumulator. All derived from the same overlap table as
+        the a2av path (every rank computes all plans deterministically)."""
+        from .containers import NativeStageMeta
+
+        cp = self.cp_size
+        my_hr = self.host_ranges_all[rank]
+        L = my_hr.total_seqlen
+        out = []
+        stage_tokens_all = []
+        for s in range(self.overlap_degree):
+            ov = self._overlap_table(s)
+            my_need = self.plans[rank].stages_need[s]
+            S = my_need.total_seqlen
+            cast = []
+            for o in range(cp):
+                L_o = self.host_ranges_all[o].total_seqlen
+                for p_ in ov[o][rank]:
+                    src = self.host_ranges_all[o].make_range_local(
+                        p_, is_self_merged=True)
+                    dst = my_need.make_range_local(p_)
+                    n = p_.seqlen
+                    cast.append((o, src.start, dst.start, n))          # k
+                    cast.append((o, L_o + src.start, S + dst.start, n))  # v
+            red = []
+            for r in range(cp):
+                S_r = self.plans[r].stages_need[s].total_seqlen
+                for p_ in ov[rank][r]:
+                    src = self.plans[r].stages_need[s].make_range_local(p_)
+                    dst = my_hr.make_range_local(p_, is_self_merged=True)
+                    n = p_.seqlen
+                    red.append((r, src.start, dst.start, n))            # dk
+                    red.append((r, S_r + src.start, L + dst.start, n))  # dv
+            cast_consumers = sum(
+                1 for r in range(cp) if ov[rank][r].total_seqlen > 0
+            )
+            reduce_consumers = sum(
+                1 for o in range(cp) if ov[o][rank].total_seqlen > 0
+            )
+            out.append(NativeStageMeta(
+                cast_pieces=cast, reduce_pieces=red, stage_tokens=S,
+                cast_consumers=cast_consumers,
+                reduce_consumers=reduce_consumers,
+            ))
+            stage_tokens_all.append(
+                [self.plans[r].stages_need[s].total_seqlen for r in range(cp)]
+            )
+        return out, stage_tokens_all
+
     # ---------------- QO-comm (solver runs on the TRANSPOSED mask) --------
     def make_qo_calc_meta(self, rank: int) -> CalcMeta:
         """Stage args with the ORIGINAL orientation restored: this solver was

@@ -158,10 +158,41 @@ class CommMeta:
     # hierarchical realisation of the same stages (set iff 2D mesh + env flag)
     stages_cast_hier: Optional[List[HierGroupCastArg]] = None
     stages_reduce_hier: Optional[List[HierGroupReduceArg]] = None
+    # native HIP-IPC pull plan (set iff MAGI_ATTENTION_NATIVE_GRPCOLL)
+    stages_native: Optional[List["NativeStageMeta"]] = None
+    stage_tokens_all: Optional[List[List[int]]] = None  # [stage][rank] S_r
 
     @property
     def overlap_degree(self) -> int:
         return len(self.stages_cast)
+
+
+@dataclass
+class NativeStageMeta:
+    """One overlap stage's PULL plan for the native (HIP-IPC over xGMI)
+    grpcoll transport (reference csrc/comm/grpcoll intranode kernels;
+    MI355X-first pull redesign — see csrc/grpcoll.hip header).
+    Pieces are [peer, src_row, dst_row, n_rows] int32 quadruples."""
+
+    cast_pieces: List[Tuple[int, int, int, int]]   # peer kv window -> stage
+    reduce_pieces: List[Tuple[int, int, int, int]]  # peer partial win -> dkv
+    stage_tokens: int            # S (stage buffer rows = 2*S)
+    cast_consumers: int          # ranks that pull from MY kv window
+    reduce_consumers: int        # ranks that pull from MY partial window
+    _cache: Dict[str, Tuple[torch.Tensor, torch.Tensor]] = field(
+        default_factory=dict, repr=False
+    )
+
+    def to_device(self, device):
+        key = str(device)
+        if key not in self._cache:
+            self._cache[key] = (
+                torch.tensor(self.cast_pieces or [[0, 0, 0, 0]],
+                             dtype=torch.int32, device=device).reshape(-1, 4),
+                torch.tensor(self.reduce_pieces or [[0, 0, 0, 0]],
+                             dtype=torch.int32, device=device).reshape(-1, 4),
+            )
+        return self._cache[key]
 
 
 @dataclass
