@@ -99,7 +99,11 @@ class NativeGrpColl:
         # flags: [deg cast_seq][deg cast_ack][deg red_seq][deg red_ack]
         self.flags = torch.zeros(4 * max(self.deg, 1), dtype=torch.int32,
                                  device=dev)
-        self.seq = 0
+        # separate sequence counters: cast rounds run every fwd AND bwd,
+        # reduce rounds only in bwd — sharing one counter deadlocks the
+        # reduce backpressure on its never-advanced ack counters
+        self.seq_cast = 0
+        self.seq_red = 0
 
         my = (_get_handle(self.kv_win), _get_handle(self.part_win),
               _get_handle(self.flags))
@@ -159,25 +163,23 @@ class NativeGrpColl:
             ctypes.cast(arr, ctypes.c_void_p), len(peers), _stream_ptr()),
             "grpcoll_ack")
 
-    def new_step(self):
-        self.seq += 1
-
     # ---- cast: my kv -> window; pull my stage buffer from peers ----
     def cast(self, kv_local: torch.Tensor, s: int) -> WorkWithPostProcessFn:
+        if s == 0:
+            self.seq_cast += 1
         meta = self.stages[s]
         slot_seq, slot_ack = s, self.deg + s
-        # backpressure: every consumer of my window acked the previous step
-        self._wait_acks(slot_ack, meta.cast_consumers * (self.seq - 1))
+        # backpressure: every consumer of my window acked the previous round
+        self._wait_acks(slot_ack, meta.cast_consumers * (self.seq_cast - 1))
         self.kv_win[: kv_local.shape[0]].copy_(kv_local)
-        self._signal(slot_seq, self.seq)
+        self._signal(slot_seq, self.seq_cast)
         S = meta.stage_tokens
         stage = kv_local.new_zeros((2 * S, self.h, self.d))
         cast_dev, _ = meta.to_device(kv_local.device)
         if meta.cast_pieces:
             self._pull(cast_dev, len(meta.cast_pieces), self.kv_ptrs,
-                       slot_seq, self.seq, stage, False,
+                       slot_seq, self.seq_cast, stage, False,
                        kv_local.element_size())
-            # ack every source peer's window (incl. self: uniform counters)
             srcs = sorted({p[0] for p in meta.cast_pieces})
             self._ack_peers(srcs, slot_ack)
         return WorkWithPostProcessFn(None, lambda: stage)
@@ -185,16 +187,18 @@ class NativeGrpColl:
     # ---- reduce: my partial -> window; owners pull-sum from peers ----
     def reduce(self, partial: torch.Tensor, dst: torch.Tensor, s: int
                ) -> WorkWithPostProcessFn:
+        if s == 0:
+            self.seq_red += 1
         meta = self.stages[s]
         slot_seq, slot_ack = 2 * self.deg + s, 3 * self.deg + s
-        self._wait_acks(slot_ack, meta.reduce_consumers * (self.seq - 1))
+        self._wait_acks(slot_ack, meta.reduce_consumers * (self.seq_red - 1))
         if partial.shape[0] > 0:
             self.part_win[: partial.shape[0]].copy_(partial)
-        self._signal(slot_seq, self.seq)
+        self._signal(slot_seq, self.seq_red)
         _, red_dev = meta.to_device(dst.device)
         if meta.reduce_pieces:
             self._pull(red_dev, len(meta.reduce_pieces), self.part_ptrs,
-                       slot_seq, self.seq, dst, True, 4)
+                       slot_seq, self.seq_red, dst, True, 4)
             srcs = sorted({p[0] for p in meta.reduce_pieces})
             self._ack_peers(srcs, slot_ack)
         return WorkWithPostProcessFn(None, lambda: dst)

@@ -86,10 +86,7 @@ class DistAttnRuntime:
 
     def _cast(self, kv_local: torch.Tensor, s: int) -> WorkWithPostProcessFn:
         if self.use_native and kv_local.is_cuda:
-            n = self._native_grpcoll(kv_local)
-            if s == 0:
-                n.new_step()
-            return n.cast(kv_local, s)
+            return self._native_grpcoll(kv_local).cast(kv_local, s)
         if self.use_hier:
             return hier_group_cast(
                 kv_local, self.comm_meta.stages_cast_hier[s],
