@@ -188,6 +188,29 @@ int magi_kernel_barrier_produce(int32_t* counter, void* stream);
 int magi_kernel_barrier_synchronize(const int32_t* counter, int32_t target,
                                     void* stream);
 
+/* native range utilities (reference csrc/extensions/
+   sort_and_reorder_ranges.cu, unique_consecutive_pairs.cu): single-WG HIP
+   kernels for the planner regime (N <= 8192); rc=2 => caller falls back */
+int magi_argsort_ranges(const void* ranges, void* out_idx, int n,
+                        void* stream);
+int magi_reorder_ranges(const void* q_ranges, const void* k_ranges,
+                        const void* attn_type_map, const void* order,
+                        void* q_out, void* k_out, void* t_out, int n,
+                        void* stream);
+int magi_unique_pairs(const void* sorted_ranges, void* uniq, void* inverse,
+                      void* count, int n, void* stream);
+
+/* native grpcoll transport (HIP-IPC pull over xGMI; csrc/grpcoll.hip) */
+int magi_ipc_get_handle(const void* dev_ptr, void* out_handle64);
+int magi_ipc_open(const void* handle64, void** out_ptr);
+int magi_ipc_close(void* ptr);
+int magi_ipc_base(const void* dev_ptr, void** base, unsigned long long* size);
+int magi_grpcoll_signal(void* flag_ptr, int value, void* stream);
+int magi_grpcoll_wait(const void* flag_ptr, int value, void* stream);
+int magi_grpcoll_ack(void* const* ack_ptrs, int n, void* stream);
+struct magi_grpcoll_pull_args;
+int magi_grpcoll_pull(const struct magi_grpcoll_pull_args* args);
+
 /* Version/identity probe so tests can verify the native library is loaded. */
 int magi_ffa_abi_version(void);
 

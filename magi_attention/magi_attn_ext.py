@@ -60,10 +60,22 @@ class KernelBarrier:
         return int(self._counter.item())
 
 
+_SORT_CAP = 8192  # LDS capacity of the native single-WG sort
+
+
 def argsort_ranges(ranges: torch.Tensor) -> torch.Tensor:
     """Indices sorting [N,2] int32 ranges lexicographically by (start, end)
-    (reference extensions/sort_and_reorder_ranges.cu) — identical pairs end
-    up adjacent, as unique_consecutive_pairs requires."""
+    (reference extensions/sort_and_reorder_ranges.cu). Native HIP bitonic
+    sort (csrc/ext_utils.hip magi_argsort_ranges) for CUDA tensors with
+    N <= 8192 (the planner regime); torch fallback above / on CPU."""
+    n = ranges.shape[0]
+    if ranges.is_cuda and 0 < n <= _SORT_CAP:
+        out = torch.empty(n, dtype=torch.int32, device=ranges.device)
+        rc = _ffa_lib.lib().magi_argsort_ranges(
+            _ffa_lib.ptr(ranges.contiguous()), _ffa_lib.ptr(out), n,
+            _ffa_lib.current_stream_ptr())
+        if rc == 0:
+            return out
     key = ranges[:, 0].long() * (1 << 31) + ranges[:, 1].long()
     return torch.argsort(key, stable=True).to(torch.int32)
 
@@ -74,6 +86,20 @@ def reorder_ranges_and_attn_type_maps(
     attn_type_map: torch.Tensor,
     order: torch.Tensor,
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    n = order.shape[0]
+    if q_ranges.is_cuda and n > 0:
+        qro = torch.empty_like(q_ranges)
+        kro = torch.empty_like(k_ranges)
+        tmo = torch.empty_like(attn_type_map)
+        rc = _ffa_lib.lib().magi_reorder_ranges(
+            _ffa_lib.ptr(q_ranges.contiguous()),
+            _ffa_lib.ptr(k_ranges.contiguous()),
+            _ffa_lib.ptr(attn_type_map.contiguous()),
+            _ffa_lib.ptr(order.contiguous()),
+            _ffa_lib.ptr(qro), _ffa_lib.ptr(kro), _ffa_lib.ptr(tmo), n,
+            _ffa_lib.current_stream_ptr())
+        if rc == 0:
+            return qro, kro, tmo
     idx = order.long()
     return (
         q_ranges.index_select(0, idx),
@@ -87,7 +113,20 @@ def unique_consecutive_pairs(
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     """Unique consecutive (start,end) pairs of sorted ranges + inverse map +
     count (reference extensions/unique_consecutive_pairs.cu; feeds
-    merge_ranges for auto_range_merge)."""
+    merge_ranges for auto_range_merge). Native HIP single-WG scan for CUDA
+    tensors with N <= 8192; torch fallback above / on CPU."""
+    n = ranges.shape[0]
+    if ranges.is_cuda and 0 < n <= _SORT_CAP:
+        uniq = torch.empty_like(ranges)
+        inverse = torch.empty(n, dtype=torch.int32, device=ranges.device)
+        count = torch.zeros(1, dtype=torch.int32, device=ranges.device)
+        rc = _ffa_lib.lib().magi_unique_pairs(
+            _ffa_lib.ptr(ranges.contiguous()), _ffa_lib.ptr(uniq),
+            _ffa_lib.ptr(inverse), _ffa_lib.ptr(count), n,
+            _ffa_lib.current_stream_ptr())
+        if rc == 0:
+            cnt = int(count.item())
+            return uniq[:cnt], inverse, count
     uniq, inverse = torch.unique_consecutive(
         ranges, dim=0, return_inverse=True
     )

@@ -475,3 +475,37 @@ def test_odd_head_dims(hd):
     ]:
         assert_close_to_ref(got.cpu().float(), ghi.float(), glo.float(),
                             f"hd{hd}:{name}")
+
+
+@requires_gpu
+def test_native_range_utils():
+    """Native argsort/reorder/unique kernels vs the torch reference
+    (reference extensions/sort_and_reorder_ranges.cu,
+    unique_consecutive_pairs.cu) — the magi_attn_ext boundary ops."""
+    import magi_attention.magi_attn_ext as ext
+
+    g = torch.Generator().manual_seed(9)
+    for n in (1, 7, 255, 1024, 4097):
+        starts = torch.randint(0, 500, (n,), generator=g, dtype=torch.int32)
+        ends = starts + torch.randint(1, 100, (n,), generator=g,
+                                      dtype=torch.int32)
+        ranges = torch.stack([starts, ends], 1).cuda()
+        idx = ext.argsort_ranges(ranges)
+        key = ranges[:, 0].long() * (1 << 31) + ranges[:, 1].long()
+        ref = torch.argsort(key.cpu(), stable=True).to(torch.int32)
+        assert torch.equal(idx.cpu(), ref), n
+
+        qro, kro, tmo = ext.reorder_ranges_and_attn_type_maps(
+            ranges, ranges + 1, torch.arange(n, dtype=torch.int32).cuda(),
+            idx,
+        )
+        assert torch.equal(qro.cpu(), ranges.cpu()[ref.long()])
+        assert torch.equal(tmo.cpu(), ref)
+
+        srt = ranges[idx.long()]
+        uniq, inv, cnt = ext.unique_consecutive_pairs(srt)
+        ru, rinv = torch.unique_consecutive(srt.cpu(), dim=0,
+                                            return_inverse=True)
+        assert int(cnt.item()) == ru.shape[0], n
+        assert torch.equal(uniq.cpu(), ru.to(torch.int32))
+        assert torch.equal(inv.cpu(), rinv.to(torch.int32))
