@@ -1255,7 +1255,11 @@ extern "C" int magi_ffa_bwd_dq(const magi_ffa_bwd_args* a) {
   // windows over-iterate masked edges on 2k varlen docs.
   // W8 runs the 3-slot LDS ring (constant-distance vmcnt barrier); W4 cannot
   // (3 buffers exceed the 80 KB/WG budget of 2 blocks/CU) and keeps 2.
-  const int dqw = a->max_seqlen_k >= 8192 ? 8 : 4;
+  // dq's q-outer blocks DO over-iterate causal k windows at W8 (r2 A/B on
+  // 2k varlen: 0.79 -> 1.61 ms), so it keeps the long-range threshold
+  int dq_bigseq = 8192;
+  { const char* e = getenv("MAGI_BWD_DQ_BIGSEQ"); if (e) dq_bigseq = atoi(e); }
+  const int dqw = a->max_seqlen_k >= dq_bigseq ? 8 : 4;
   int nbuf = 3;
   { const char* e = getenv("MAGI_BWD_NBUF"); if (e) nbuf = atoi(e); }
   int nt = 0;
@@ -1322,12 +1326,16 @@ static int launch_bwd_dkv(const magi_ffa_bwd_args* a) {
   BwdParams p;
   int rc = fill_bwd_params(a, &p);
   if (rc) return rc > 0 ? 0 : rc;
-  // dK (MODE 2) and fused (MODE 0) run 8 waves per WG for LONG ranges (one
-  // staged Q/dO image shared by all). dV (MODE 1) stays at 4 — its shorter
-  // MFMA chain loses more to the wider block windows than it saves in
-  // staging. The fused mode keeps NBUF=2 (its V tiles + a 3-ring exceed the
-  // 160 KB LDS); the split W8 modes default to the 3-slot ring.
-  const bool big = a->max_seqlen_k >= 8192;
+  // All dkv modes run 8 waves per WG (one 512-thread WG/CU, one shared
+  // staged Q/dO image) down to 1k ranges: these kernels' blocks span the
+  // K dim, so the wider workgroup does NOT over-iterate causal q windows,
+  // and halving the staging streams is a 50-60% win even on 2k varlen docs
+  // (r2 A/B: dv 0.71->0.48 ms, dk 0.90->0.56, fused 1.37->0.92). The fused
+  // mode keeps NBUF=2 (its V tiles + a 3-ring exceed the 160 KB LDS); the
+  // split W8 modes default to the 3-slot ring.
+  int bigseq = 1024;
+  { const char* e = getenv("MAGI_BWD_BIGSEQ"); if (e) bigseq = atoi(e); }
+  const bool big = a->max_seqlen_k >= bigseq;
   // W8 for ALL modes on long ranges: one 512-thread WG/CU halves the
   // staging streams (PMC r2: dV at W4 fetched 174 GB/launch vs dK-W8's 71)
   const int W = big ? 8 : 4;

@@ -73,7 +73,9 @@ def bwd_dkv_mode() -> str:
 def is_bwd_split_dkv(max_seqlen_k: int = 0) -> bool:
     mode = bwd_dkv_mode()
     if mode == "auto":
-        return max_seqlen_k < 8192
+        # r2: the fused W8 kernel wins from 1k ranges up (A/B on 2k varlen:
+        # fused 0.92 ms vs dv+dk 1.04); below that the W4 split is safer
+        return max_seqlen_k < 1024
     return mode == "split"
 
 
