@@ -193,7 +193,11 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   constexpr bool V_IN_LDS = MODE == 0;
   constexpr int DF = D / 16;
   constexpr int DT = D / 32;
-  constexpr int ROWB = D * 2;  // bytes per LDS tile row
+  // LDS rows padded to a power of two (D=192 -> 512-byte rows; see
+  // ffa_fwd.hip note on the swizzle bijection)
+  constexpr int ROWB = (D == 192) ? 512 : D * 2;
+  constexpr int ROWE = ROWB / 2;
+  constexpr int VSLOTS = D / 8;
   // XOR swizzle within one LDS row: spreads a b128 lane group over the row's
   // 16-B slots (T2/G4); mask keeps the XOR inside the row for D=64 too.
   // 32-B-granular XOR swizzle: spreads b128 lane groups over the row's 16-B
@@ -268,27 +272,27 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   // NBUF x 64-row Q/dO images: one barrier per 64 q rows — the doubled
   // compute phase covers the prefetch latency a 32-row phase could not.
   constexpr int QITER = 2 * BWD_BM;
-  constexpr int VLDS = V_IN_LDS ? WAVES * BWD_BN * D * 2 : 0;
-  static_assert(NBUF * 2 * QITER * (D * 2 + 4) + VLDS <= 163840,
+  constexpr int VLDS = V_IN_LDS ? WAVES * BWD_BN * ROWB : 0;
+  static_assert(NBUF * 2 * QITER * (ROWB + 4) + VLDS <= 163840,
                 "LDS budget");
   __shared__ __attribute__((aligned(16))) char smem[
-      NBUF * 2 * QITER * D * 2 + NBUF * 2 * QITER * 4 + (VLDS ? VLDS : 1)];
+      NBUF * 2 * QITER * ROWB + NBUF * 2 * QITER * 4 + (VLDS ? VLDS : 1)];
   auto lds_q = [&](int buf) -> __bf16* {
-    return (__bf16*)(smem + (2 * buf) * QITER * D * 2);
+    return (__bf16*)(smem + (2 * buf) * QITER * ROWB);
   };
   auto lds_do = [&](int buf) -> __bf16* {
-    return (__bf16*)(smem + (2 * buf + 1) * QITER * D * 2);
+    return (__bf16*)(smem + (2 * buf + 1) * QITER * ROWB);
   };
   // lse / dpsum, staged by LDS-DMA (one dword per lane: lanes 0-31 lse,
   // 32-63 dpsum), one call per 32-row group: layout per buffer is
   // [lse g0 (32) | dps g0 (32) | lse g1 (32) | dps g1 (32)]
   auto lds_lse = [&](int buf) -> float* {
-    return (float*)(smem + NBUF * 2 * QITER * D * 2 + buf * 2 * QITER * 4);
+    return (float*)(smem + NBUF * 2 * QITER * ROWB + buf * 2 * QITER * 4);
   };
   // MODE 0: per-wave V tile (BWD_BN rows x D, same 32-B XOR swizzle as the
   // Q/dO images so the dP B-fragment reads reuse the qf addressing)
-  __bf16* lds_vt = (__bf16*)(smem + NBUF * 2 * QITER * D * 2 +
-                             NBUF * 2 * QITER * 4) + wave * BWD_BN * D;
+  __bf16* lds_vt = (__bf16*)(smem + NBUF * 2 * QITER * ROWB +
+                             NBUF * 2 * QITER * 4) + wave * BWD_BN * ROWE;
 
   // K fragments (A-layout), loaded once per block; V fragments likewise in
   // the split dK mode — the fused mode stages V into LDS instead (register
@@ -318,11 +322,13 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
       const int r = r0v + lane / (ROWB / 16);
       const int c = lane % (ROWB / 16);
       const int kr = min(n0 + r, ke - 1);
-      const int csw = (c ^ ((r & SW32M) << 1)) * 8;
+      int cs = c ^ ((r & SW32M) << 1);
+      if (cs >= VSLOTS) cs = 0;
+      const int csw = cs * 8;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)(
               p.v + (size_t)kr * k_pitch + (size_t)kh * D + csw),
-          (__attribute__((address_space(3))) unsigned int*)&lds_vt[r0v * D],
+          (__attribute__((address_space(3))) unsigned int*)&lds_vt[r0v * ROWE],
           16, 0, STAGE_AUX);
     }
   }
@@ -359,16 +365,18 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
       const int r = r0 + lane / (ROWB / 16);
       const int c = lane % (ROWB / 16);
       const int qrow = min(m0x + r, qe - 1);
-      const int csw = (c ^ ((r & SW32M) << 1)) * 8;
+      int cs = c ^ ((r & SW32M) << 1);
+      if (cs >= VSLOTS) cs = 0;
+      const int csw = cs * 8;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)(
               p.q + (size_t)qrow * q_pitch + (size_t)h * D + csw),
-          (__attribute__((address_space(3))) unsigned int*)&lds_q(buf)[r0 * D],
+          (__attribute__((address_space(3))) unsigned int*)&lds_q(buf)[r0 * ROWE],
           16, 0, STAGE_AUX);
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)(
               p.dout + (size_t)qrow * q_pitch + (size_t)h * D + csw),
-          (__attribute__((address_space(3))) unsigned int*)&lds_do(buf)[r0 * D],
+          (__attribute__((address_space(3))) unsigned int*)&lds_do(buf)[r0 * ROWE],
           16, 0, STAGE_AUX);
     }
     // lse (lanes 0-31) / dpsum (lanes 32-63), one LDS-DMA per 32-row group
@@ -414,8 +422,8 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
     for (int sub = 0; sub < 2; ++sub) {
     const int ms = m0 + sub * BWD_BM;
     if (ms >= q_hi) break;  // q_hi is block-uniform
-    const __bf16* lqb = lds_q(cur) + sub * BWD_BM * D;
-    const __bf16* ldb = lds_do(cur) + sub * BWD_BM * D;
+    const __bf16* lqb = lds_q(cur) + sub * BWD_BM * ROWE;
+    const __bf16* ldb = lds_do(cur) + sub * BWD_BM * ROWE;
     const float* lse_t = lds_lse(cur) + sub * 2 * BWD_BM;
     const float* dps_t = lse_t + BWD_BM;
 
@@ -642,7 +650,9 @@ void ffa_bwd_dq_kernel(BwdParams p) {
   constexpr int STAGE_AUX = NT ? 2 : 0;  // see dkv kernel note
   constexpr int DF = D / 16;
   constexpr int DT = D / 32;
-  constexpr int ROWB = D * 2;
+  constexpr int ROWB = (D == 192) ? 512 : D * 2;  // padded pow2 LDS rows
+  constexpr int ROWE = ROWB / 2;
+  constexpr int VSLOTS = D / 8;
   constexpr int SW32M = ROWB / 32 - 1;  // 32-B-granular swizzle (tr16 reads)
   auto swz = [](int row, int byte_off) {
     return byte_off ^ ((row & SW32M) << 5);
@@ -700,13 +710,13 @@ void ffa_bwd_dq_kernel(BwdParams p) {
   // compute phase covers the prefetch latency a 32-row phase could not, with
   // HALF the barrier parking (PMC: SQ_WAIT_ANY 36% at 32-row iterations).
   constexpr int KITER = 2 * BWD_BN;
-  static_assert(NBUF * 2 * KITER * D * 2 <= 163840, "LDS budget");
-  __shared__ __attribute__((aligned(16))) char smem[NBUF * 2 * KITER * D * 2];
+  static_assert(NBUF * 2 * KITER * ROWB <= 163840, "LDS budget");
+  __shared__ __attribute__((aligned(16))) char smem[NBUF * 2 * KITER * ROWB];
   auto lds_k = [&](int buf) -> __bf16* {
-    return (__bf16*)(smem + (2 * buf) * KITER * D * 2);
+    return (__bf16*)(smem + (2 * buf) * KITER * ROWB);
   };
   auto lds_v = [&](int buf) -> __bf16* {
-    return (__bf16*)(smem + (2 * buf + 1) * KITER * D * 2);
+    return (__bf16*)(smem + (2 * buf + 1) * KITER * ROWB);
   };
 
   // persistent per-wave operands: Q and dO fragments (B-layout rows)
@@ -751,16 +761,18 @@ void ffa_bwd_dq_kernel(BwdParams p) {
       const int r = r0 + lane / (ROWB / 16);
       const int c = lane % (ROWB / 16);
       const int kr = min(n0x + r, ke - 1);
-      const int csw = (c ^ ((r & SW32M) << 1)) * 8;
+      int cs = c ^ ((r & SW32M) << 1);
+      if (cs >= VSLOTS) cs = 0;
+      const int csw = cs * 8;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)(
               p.k + (size_t)kr * k_pitch + (size_t)kh * D + csw),
-          (__attribute__((address_space(3))) unsigned int*)&lds_k(buf)[r0 * D],
+          (__attribute__((address_space(3))) unsigned int*)&lds_k(buf)[r0 * ROWE],
           16, 0, STAGE_AUX);
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)(
               p.v + (size_t)kr * k_pitch + (size_t)kh * D + csw),
-          (__attribute__((address_space(3))) unsigned int*)&lds_v(buf)[r0 * D],
+          (__attribute__((address_space(3))) unsigned int*)&lds_v(buf)[r0 * ROWE],
           16, 0, STAGE_AUX);
     }
   };
@@ -791,8 +803,8 @@ void ffa_bwd_dq_kernel(BwdParams p) {
     for (int sub = 0; sub < 2; ++sub) {
     const int ns = n0 + sub * BWD_BN;
     if (ns >= k_hi) break;  // k_hi is block-uniform
-    const __bf16* lkb = lds_k(cur) + sub * BWD_BN * D;
-    const __bf16* lvb = lds_v(cur) + sub * BWD_BN * D;
+    const __bf16* lkb = lds_k(cur) + sub * BWD_BN * ROWE;
+    const __bf16* lvb = lds_v(cur) + sub * BWD_BN * ROWE;
 
     if (wave_alive(m0, qe) && ns + BWD_BN > wk_lo && ns < wk_hi) {
       // ---- S^T = K Q^T ; dP^T = V dO^T (K/V A-frags from LDS rows) ----
@@ -1203,7 +1215,7 @@ extern "C" int magi_ffa_bwd_preprocess(const magi_ffa_bwd_args* a) {
 
 static int fill_bwd_params(const magi_ffa_bwd_args* a, BwdParams* p) {
   if (!a || !a->dout || !a->q || !a->k || !a->v || !a->lse) return -1;
-  if (a->d != 64 && a->d != 128) return -2;
+  if (a->d != 64 && a->d != 128 && a->d != 192) return -2;
   if (a->hq % a->hk != 0) return -3;
   if (a->n_ranges <= 0) return 1;  // caller treats >0 as "nothing to do"
   p->dout = (const bf16_t*)a->dout;
@@ -1259,7 +1271,8 @@ extern "C" int magi_ffa_bwd_dq(const magi_ffa_bwd_args* a) {
   // 2k varlen: 0.79 -> 1.61 ms), so it keeps the long-range threshold
   int dq_bigseq = 8192;
   { const char* e = getenv("MAGI_BWD_DQ_BIGSEQ"); if (e) dq_bigseq = atoi(e); }
-  const int dqw = a->max_seqlen_k >= dq_bigseq ? 8 : 4;
+  // D=192 W8 builds spill at the 256-reg cap: stay at 4 waves there
+  const int dqw = (a->d != 192 && a->max_seqlen_k >= dq_bigseq) ? 8 : 4;
   int nbuf = 3;
   { const char* e = getenv("MAGI_BWD_NBUF"); if (e) nbuf = atoi(e); }
   int nt = 0;
@@ -1271,7 +1284,7 @@ extern "C" int magi_ffa_bwd_dq(const magi_ffa_bwd_args* a) {
   if (a->n_ranges > 65535) return -5;
   hipStream_t s = (hipStream_t)a->stream;
   const bool sc = a->softcap > 0.f;
-  if (dqw == 8 && dq64) {
+  if (dqw == 8 && dq64 && a->d != 192) {
     // 64-row q tiles, 4 waves, 1 wave/SIMD (same 256-row block span)
     const int qblocks64 = (int)((a->total_q + 255) / 256);
     dim3 grid64 = p.head_major
@@ -1307,14 +1320,24 @@ extern "C" int magi_ffa_bwd_dq(const magi_ffa_bwd_args* a) {
 #define PICK_DQ_NT(DD, NTV) \
   do { \
     if (dqw == 8) { \
-      if (nbuf == 3) { if (sc) LAUNCH_DQ(DD, true, 8, 3, NTV); else LAUNCH_DQ(DD, false, 8, 3, NTV); } \
-      else           { if (sc) LAUNCH_DQ(DD, true, 8, 2, NTV); else LAUNCH_DQ(DD, false, 8, 2, NTV); } \
+      bool done3 = false; \
+      if constexpr (DD != 192) { /* a 3-ring of 512-B rows exceeds LDS */ \
+        if (nbuf == 3) { \
+          if (sc) LAUNCH_DQ(DD, true, 8, 3, NTV); else LAUNCH_DQ(DD, false, 8, 3, NTV); \
+          done3 = true; \
+        } \
+      } \
+      if (!done3) { \
+        if (sc) LAUNCH_DQ(DD, true, 8, 2, NTV); else LAUNCH_DQ(DD, false, 8, 2, NTV); \
+      } \
     } else { \
       if (sc) LAUNCH_DQ(DD, true, 4, 2, NTV); else LAUNCH_DQ(DD, false, 4, 2, NTV); \
     } \
   } while (0)
 #define PICK_DQ(DD) do { if (nt) PICK_DQ_NT(DD, 1); else PICK_DQ_NT(DD, 0); } while (0)
-  if (a->d == 64) PICK_DQ(64); else PICK_DQ(128);
+  if (a->d == 64) PICK_DQ(64);
+  else if (a->d == 192) PICK_DQ(192);
+  else PICK_DQ(128);
 #undef PICK_DQ
 #undef PICK_DQ_NT
 #undef LAUNCH_DQ
@@ -1335,7 +1358,7 @@ static int launch_bwd_dkv(const magi_ffa_bwd_args* a) {
   // split W8 modes default to the 3-slot ring.
   int bigseq = 1024;
   { const char* e = getenv("MAGI_BWD_BIGSEQ"); if (e) bigseq = atoi(e); }
-  const bool big = a->max_seqlen_k >= bigseq;
+  const bool big = a->d != 192 && a->max_seqlen_k >= bigseq;
   // W8 for ALL modes on long ranges: one 512-thread WG/CU halves the
   // staging streams (PMC r2: dV at W4 fetched 174 GB/launch vs dK-W8's 71)
   const int W = big ? 8 : 4;
@@ -1370,7 +1393,7 @@ static int launch_bwd_dkv(const magi_ffa_bwd_args* a) {
   do { \
     if (W == 8) { \
       bool done = false; \
-      if constexpr (MODE != 0) { /* MODE0 + 3-ring exceeds 160 KB LDS */ \
+      if constexpr (MODE != 0 && DD != 192) { /* 3-ring LDS limits */ \
         if (nbuf == 3) { \
           if (sc) LAUNCH_DKV(DD, true, 8, 3, NTV); else LAUNCH_DKV(DD, false, 8, 3, NTV); \
           done = true; \
@@ -1384,7 +1407,11 @@ static int launch_bwd_dkv(const magi_ffa_bwd_args* a) {
     } \
   } while (0)
 #define PICK_DKV(DD) do { if (nt) PICK_DKV_NT(DD, 1); else PICK_DKV_NT(DD, 0); } while (0)
-  if (a->d == 64) PICK_DKV(64); else PICK_DKV(128);
+  if (a->d == 64) PICK_DKV(64);
+  else if (a->d == 192) {
+    if constexpr (MODE != 0) PICK_DKV(192);
+    else return -6;  // fused + D=192 exceeds LDS; host routes to the split
+  } else PICK_DKV(128);
 #undef PICK_DKV
 #undef PICK_DKV_NT
 #undef LAUNCH_DKV

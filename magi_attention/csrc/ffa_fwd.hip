@@ -178,7 +178,12 @@ __global__ __launch_bounds__(64 * WAVES, 8 / WAVES) void ffa_fwd_kernel(FwdParam
       b_hi = b_lo;
     }
   };
-  constexpr int ROWB = D * 2;
+  // LDS rows are padded to a power of two so the XOR swizzle algebra holds
+  // for every D (D=192 pads 384-byte rows to 512; the slot permutation is a
+  // bijection, so padding slots only ever hold data no read requests).
+  constexpr int ROWB = (D == 192) ? 512 : D * 2;
+  constexpr int ROWE = ROWB / 2;       // LDS elements per padded row
+  constexpr int VSLOTS = D / 8;        // valid 16-B source slots per row
   // 32-B-granular swizzle: b128 lane groups spread (<=2-way) AND every 32-B
   // run stays physically contiguous for the tr16 V reads (see bwd kernels)
   constexpr int SW32M = ROWB / 32 - 1;
@@ -189,12 +194,12 @@ __global__ __launch_bounds__(64 * WAVES, 8 / WAVES) void ffa_fwd_kernel(FwdParam
   // backward kernels): one barrier per 64 k rows, the doubled compute phase
   // covers the prefetch latency the 32-row register pipeline could not
   constexpr int KITER = 2 * FFA_BN;
-  __shared__ __attribute__((aligned(16))) char fsmem[2 * 2 * KITER * D * 2];
+  __shared__ __attribute__((aligned(16))) char fsmem[2 * 2 * KITER * ROWB];
   auto lds_k = [&](int buf) -> __bf16* {
-    return (__bf16*)(fsmem + (2 * buf) * KITER * D * 2);
+    return (__bf16*)(fsmem + (2 * buf) * KITER * ROWB);
   };
   auto lds_v = [&](int buf) -> __bf16* {
-    return (__bf16*)(fsmem + (2 * buf + 1) * KITER * D * 2);
+    return (__bf16*)(fsmem + (2 * buf + 1) * KITER * ROWB);
   };
 
   // Q fragments in registers (8 x bf16x8 for D=128)
@@ -227,16 +232,18 @@ __global__ __launch_bounds__(64 * WAVES, 8 / WAVES) void ffa_fwd_kernel(FwdParam
       const int r = r0 + lane / (ROWB / 16);
       const int c = lane % (ROWB / 16);
       const int kr = min(n0x + r, ke - 1);
-      const int csw = (c ^ ((r & SW32M) << 1)) * 8;
+      int cs = c ^ ((r & SW32M) << 1);
+      if (cs >= VSLOTS) cs = 0;  // padding slot: any in-bounds source
+      const int csw = cs * 8;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)(
               kbase + (size_t)kr * k_pitch + csw),
-          (__attribute__((address_space(3))) unsigned int*)&lds_k(buf)[r0 * D],
+          (__attribute__((address_space(3))) unsigned int*)&lds_k(buf)[r0 * ROWE],
           16, 0, 0);
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)(
               vbase + (size_t)kr * k_pitch + csw),
-          (__attribute__((address_space(3))) unsigned int*)&lds_v(buf)[r0 * D],
+          (__attribute__((address_space(3))) unsigned int*)&lds_v(buf)[r0 * ROWE],
           16, 0, 0);
     }
   };
@@ -410,8 +417,8 @@ __global__ __launch_bounds__(64 * WAVES, 8 / WAVES) void ffa_fwd_kernel(FwdParam
       for (int sub = 0; sub < 2; ++sub) {
         const int ns = n0 + sub * FFA_BN;
         if (ns >= b_hi) break;  // b_hi is block-uniform
-        sub_body(ns, lds_k(cur) + sub * FFA_BN * D,
-                 lds_v(cur) + sub * FFA_BN * D);
+        sub_body(ns, lds_k(cur) + sub * FFA_BN * ROWE,
+                 lds_v(cur) + sub * FFA_BN * ROWE);
       }
       cur ^= 1;
     }
@@ -603,7 +610,7 @@ static int launch_fwd_d(const magi_ffa_fwd_args* a, const FwdParams& p,
 
 extern "C" int magi_ffa_fwd(const magi_ffa_fwd_args* a) {
   if (!a || !a->q || !a->k || !a->v || !a->out || !a->lse) return -1;
-  if (a->d != 64 && a->d != 128) return -2;
+  if (a->d != 64 && a->d != 128 && a->d != 192) return -2;
   if (a->hq % a->hk != 0) return -3;
   if (a->n_ranges <= 0) return 0;
   if (!a->disable_atomic_reduction && !a->locks) return -4;
@@ -635,6 +642,7 @@ extern "C" int magi_ffa_fwd(const magi_ffa_fwd_args* a) {
   // scheduler-visible tr16 reads, packed converts and raw-exp2 softmax.
   int fw = 8;
   { const char* e = getenv("MAGI_FWD_WAVES"); if (e && atoi(e)) fw = atoi(e); }
+  if (a->d == 192) fw = 4;  // the W8 D=192 build spills (256-reg cap)
   const int span = 32 * fw;
   const int mblocks = (a->max_seqlen_q + span - 1) / span;
   if (a->n_ranges > 65535) return -5;
@@ -661,6 +669,9 @@ extern "C" int magi_ffa_fwd(const magi_ffa_fwd_args* a) {
   if (a->d == 64)
     return fw == 8 ? launch_fwd_d<64, 8>(a, p, grid, block, stream)
                    : launch_fwd_d<64, 4>(a, p, grid, block, stream);
+  if (a->d == 192)
+    return fw == 8 ? launch_fwd_d<192, 8>(a, p, grid, block, stream)
+                   : launch_fwd_d<192, 4>(a, p, grid, block, stream);
   return fw == 8 ? launch_fwd_d<128, 8>(a, p, grid, block, stream)
                  : launch_fwd_d<128, 4>(a, p, grid, block, stream);
 }

@@ -70,8 +70,10 @@ def bwd_dkv_mode() -> str:
     return "auto"
 
 
-def is_bwd_split_dkv(max_seqlen_k: int = 0) -> bool:
+def is_bwd_split_dkv(max_seqlen_k: int = 0, head_dim: int = 128) -> bool:
     mode = bwd_dkv_mode()
+    if head_dim == 192:
+        return True  # fused + D=192 exceeds the 160 KB LDS (its V tiles)
     if mode == "auto":
         # r2: the fused W8 kernel wins from 1k ranges up (A/B on 2k varlen:
         # fused 0.92 ms vs dv+dk 1.04); below that the W4 split is safer

@@ -54,7 +54,7 @@ def run_bwd_deterministic(args, q_ranges, k_ranges, attn_type_map,
 
     launch(lib.magi_ffa_bwd_dq, q_groups, (q_ranges, k_ranges), "bwd_dq[det]")
     hs = gqa if gqa > 1 else 1
-    if env.is_bwd_split_dkv(args.max_seqlen_k):
+    if env.is_bwd_split_dkv(args.max_seqlen_k, args.d):
         launch(lib.magi_ffa_bwd_dv, k_groups, (q_ranges, k_ranges),
                "bwd_dv[det]", head_splits=hs)
         launch(lib.magi_ffa_bwd_dk, k_groups, (q_ranges, k_ranges),
@@ -93,7 +93,7 @@ def run_bwd_passes(args, device, dq_tables=None, dkv_tables=None) -> None:
     check(lib.magi_ffa_bwd_dq(args), "magi_ffa_bwd_dq")
     args.stream = ctypes.c_void_p(main.cuda_stream)
     set_tables(dkv_tables, False)
-    if env.is_bwd_split_dkv(args.max_seqlen_k):
+    if env.is_bwd_split_dkv(args.max_seqlen_k, args.d):
         check(lib.magi_ffa_bwd_dv(args), "magi_ffa_bwd_dv")
         check(lib.magi_ffa_bwd_dk(args), "magi_ffa_bwd_dk")
     else:
@@ -585,15 +585,15 @@ def flex_flash_attn_func(
         "index_attn lands in a later round"
     )
     d = q.shape[-1]
-    if d not in (64, 128):
+    if d not in (64, 128, 192):
         # arbitrary head dims run in the next bucket with zero feature
         # padding — exact for attention (padding contributes 0 to scores;
         # padded V columns are sliced off; gradients slice back through the
         # autograd pad). Reference buckets head_size <=64/<=128/<=192
         # (flash_api.cpp:322-334); the 192 bucket lands with a D=192 kernel.
-        assert d < 128, f"head_dim {d} > 128 lands in a later round"
+        assert d < 192, f"head_dim {d} > 192 is beyond the reference's buckets"
         assert q.dtype == torch.bfloat16, "padded head dims require bf16"
-        bucket = 64 if d <= 64 else 128
+        bucket = 64 if d <= 64 else (128 if d <= 128 else 192)
         if softmax_scale is None:
             softmax_scale = d ** (-0.5)  # scale from the REAL head dim
         pad = bucket - d

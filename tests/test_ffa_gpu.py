@@ -509,3 +509,43 @@ def test_native_range_utils():
         assert int(cnt.item()) == ru.shape[0], n
         assert torch.equal(uniq.cpu(), ru.to(torch.int32))
         assert torch.equal(inv.cpu(), rinv.to(torch.int32))
+
+
+@requires_gpu
+def test_head_dim_192_bucket():
+    """D=192 bucket (MLA head dims; reference flash_api.cpp:322 buckets
+    <=64/<=128/<=192, tile_size.h:42): exact D=192 plus a padded d=160 case,
+    fwd+bwd vs the fp64 oracle."""
+    from oracle import make_attn_mask, ref_attn_with_grads
+    from magi_attention.functional import flex_flash_attn_func
+
+    for d in (192, 160):
+        tq = tk = 1024
+        hq, hk = 4, 2
+        g = torch.Generator().manual_seed(33 + d)
+        q = (torch.randn(tq, hq, d, generator=g) * 0.5).bfloat16().cuda().requires_grad_(True)
+        k = (torch.randn(tk, hk, d, generator=g) * 0.5).bfloat16().cuda().requires_grad_(True)
+        v = (torch.randn(tk, hk, d, generator=g) * 0.5).bfloat16().cuda().requires_grad_(True)
+        dout = (torch.randn(tq, hq, d, generator=g) * 0.5).bfloat16().cuda()
+        qr_l = [[0, 512], [512, 1024]]
+        ty_l = [1, 0]
+        qr = torch.tensor(qr_l, dtype=torch.int32, device="cuda")
+        tm = torch.tensor(ty_l, dtype=torch.int32, device="cuda")
+        out, meta = flex_flash_attn_func(q, k, v, qr, qr.clone(), tm)
+        out.backward(dout)
+        torch.cuda.synchronize()
+
+        mask = make_attn_mask(tq, tk, qr_l, qr_l, ty_l)
+        qc, kc, vc, doc = [t.detach().cpu() for t in (q, k, v, dout)]
+        o_hi, _, dq_hi, dk_hi, dv_hi = ref_attn_with_grads(qc, kc, vc, mask, doc)
+        o_lo, _, dq_lo, dk_lo, dv_lo = ref_attn_with_grads(
+            qc, kc, vc, mask, doc, high_precision=False,
+            p_dtype=torch.bfloat16)
+        assert_close_to_ref(out.detach().cpu().float(), o_hi.float(),
+                            o_lo.float(), f"d{d}:out")
+        assert_close_to_ref(q.grad.cpu().float(), dq_hi.float(),
+                            dq_lo.float(), f"d{d}:dq", ratio=4.5)
+        assert_close_to_ref(k.grad.cpu().float(), dk_hi.float(),
+                            dk_lo.float(), f"d{d}:dk", ratio=4.5)
+        assert_close_to_ref(v.grad.cpu().float(), dv_hi.float(),
+                            dv_lo.float(), f"d{d}:dv", ratio=4.5)
