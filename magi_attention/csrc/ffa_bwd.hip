@@ -182,7 +182,7 @@ __global__ __launch_bounds__(256) void bwd_preprocess_kernel(BwdParams p, int d,
 // barrier: prefetch has TWO compute phases, and the barrier only waits for
 // the one stage it needs — attacks the 34-42% SQ_WAIT_ANY of the r1 PMC.
 template <int D, bool HAS_SOFTCAP, int MODE, int WAVES, int NBUF, int NT = 0>
-__global__ __launch_bounds__(64 * WAVES, 1)
+__global__ __launch_bounds__(64 * WAVES, WAVES == 8 ? 2 : 1)
 void ffa_bwd_dkv_kernel(BwdParams p) {
   // NT: stage with the non-temporal policy (aux=2) — staged rows are read
   // once per WG; keeping them out of L2 protects the resident K strips and
@@ -271,12 +271,23 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   // ".s-level traps" (a)).
   // NBUF x 64-row Q/dO images: one barrier per 64 q rows — the doubled
   // compute phase covers the prefetch latency a 32-row phase could not.
-  constexpr int QITER = 2 * BWD_BM;
-  constexpr int VLDS = V_IN_LDS ? WAVES * BWD_BN * ROWB : 0;
-  static_assert(NBUF * 2 * QITER * (ROWB + 4) + VLDS <= 163840,
+  // MODE 0 (fused, r2-v2): BOTH the K and V tiles of every wave live in LDS
+  // (2 x 64 KB) and the Q/dO ring shrinks to 32-row images (32 KB) — that
+  // is the LDS budget to the byte, so lse/dpsum move from LDS to lane
+  // registers + ds_bpermute. This removes the per-subtile K re-reads that
+  // made the previous fused kernel fabric-bound (PMC: 172 GB/launch of
+  // L2-miss fetch vs the split kernels' ~70) AND leaves ~36 registers of
+  // scheduler slack below the 256 cap.
+  // (the 32-row ring is only needed at D=128, where the K/V tiles take
+  // 128 KB; at D=64 everything fits beside a 64-row ring)
+  constexpr int QITER = (V_IN_LDS && D == 128) ? BWD_BM : 2 * BWD_BM;
+  constexpr int NSUB = QITER / BWD_BM;
+  constexpr int KVLDS = V_IN_LDS ? 2 * WAVES * BWD_BN * ROWB : 0;
+  constexpr int LSELDS = V_IN_LDS ? 0 : NBUF * 2 * QITER * 4;
+  static_assert(NBUF * 2 * QITER * ROWB + LSELDS + KVLDS <= 163840,
                 "LDS budget");
   __shared__ __attribute__((aligned(16))) char smem[
-      NBUF * 2 * QITER * ROWB + NBUF * 2 * QITER * 4 + (VLDS ? VLDS : 1)];
+      NBUF * 2 * QITER * ROWB + LSELDS + KVLDS];
   auto lds_q = [&](int buf) -> __bf16* {
     return (__bf16*)(smem + (2 * buf) * QITER * ROWB);
   };
@@ -289,10 +300,13 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   auto lds_lse = [&](int buf) -> float* {
     return (float*)(smem + NBUF * 2 * QITER * ROWB + buf * 2 * QITER * 4);
   };
-  // MODE 0: per-wave V tile (BWD_BN rows x D, same 32-B XOR swizzle as the
-  // Q/dO images so the dP B-fragment reads reuse the qf addressing)
-  __bf16* lds_vt = (__bf16*)(smem + NBUF * 2 * QITER * ROWB +
-                             NBUF * 2 * QITER * 4) + wave * BWD_BN * ROWE;
+  // MODE 0: per-wave K and V tiles (BWD_BN rows x D, same 32-B XOR swizzle
+  // as the Q/dO images so the A-fragment reads reuse the qf addressing)
+  __bf16* lds_kt = (__bf16*)(smem + NBUF * 2 * QITER * ROWB + LSELDS) +
+                   wave * BWD_BN * ROWE;
+  __bf16* lds_vt = (__bf16*)(smem + NBUF * 2 * QITER * ROWB + LSELDS +
+                             KVLDS / 2) +
+                   wave * BWD_BN * ROWE;
 
   // K fragments (A-layout), loaded once per block; V fragments likewise in
   // the split dK mode — the fused mode stages V into LDS instead (register
@@ -325,6 +339,11 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
       int cs = c ^ ((r & SW32M) << 1);
       if (cs >= VSLOTS) cs = 0;
       const int csw = cs * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)(
+              p.k + (size_t)kr * k_pitch + (size_t)kh * D + csw),
+          (__attribute__((address_space(3))) unsigned int*)&lds_kt[r0v * ROWE],
+          16, 0, STAGE_AUX);
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)(
               p.v + (size_t)kr * k_pitch + (size_t)kh * D + csw),
@@ -380,16 +399,20 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
           16, 0, STAGE_AUX);
     }
     // lse (lanes 0-31) / dpsum (lanes 32-63), one LDS-DMA per 32-row group
+    // (MODE 0 has no LDS lse region: compute reads them into lane registers
+    // and cross-lane-selects with ds_bpermute)
+    if constexpr (!V_IN_LDS) {
 #pragma unroll
-    for (int sg = 0; sg < 2; ++sg) {
-      const int qr = min(m0x + sg * BWD_BM + lo32, qe - 1);
-      const float* src = (lane < 32) ? p.lse + (size_t)qr * p.hq + h
-                                     : p.dpsum + (size_t)qr * p.hq + h;
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) unsigned int*)src,
-          (__attribute__((address_space(3))) unsigned int*)(
-              lds_lse(buf) + sg * 2 * BWD_BM),
-          4, 0, STAGE_AUX);
+      for (int sg = 0; sg < NSUB; ++sg) {
+        const int qr = min(m0x + sg * BWD_BM + lo32, qe - 1);
+        const float* src = (lane < 32) ? p.lse + (size_t)qr * p.hq + h
+                                       : p.dpsum + (size_t)qr * p.hq + h;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)src,
+            (__attribute__((address_space(3))) unsigned int*)(
+                lds_lse(buf) + sg * 2 * BWD_BM),
+            4, 0, STAGE_AUX);
+      }
     }
   };
 
@@ -419,13 +442,32 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
       if (m0 + QITER < q_hi) stage_glds(cur ^ 1, m0 + QITER);
     }
 #pragma unroll
-    for (int sub = 0; sub < 2; ++sub) {
+    for (int sub = 0; sub < NSUB; ++sub) {
     const int ms = m0 + sub * BWD_BM;
     if (ms >= q_hi) break;  // q_hi is block-uniform
     const __bf16* lqb = lds_q(cur) + sub * BWD_BM * ROWE;
     const __bf16* ldb = lds_do(cur) + sub * BWD_BM * ROWE;
-    const float* lse_t = lds_lse(cur) + sub * 2 * BWD_BM;
-    const float* dps_t = lse_t + BWD_BM;
+    const float* lse_t = nullptr;
+    const float* dps_t = nullptr;
+    float lsedp_reg = 0.f;  // MODE0: lanes 0-31 lse, 32-63 dpsum (bpermute)
+    if constexpr (!V_IN_LDS) {
+      lse_t = lds_lse(cur) + sub * 2 * BWD_BM;
+      dps_t = lse_t + BWD_BM;
+    } else {
+      const int qr = min(ms + lo32, qe - 1);
+      lsedp_reg = (lane < 32) ? p.lse[(size_t)qr * p.hq + h]
+                              : p.dpsum[(size_t)qr * p.hq + h];
+    }
+    auto lse_at = [&](int rl) -> float {
+      if constexpr (!V_IN_LDS) return lse_t[rl];
+      return __uint_as_float(__builtin_amdgcn_ds_bpermute(
+          rl << 2, __float_as_uint(lsedp_reg)));
+    };
+    auto dps_at = [&](int rl) -> float {
+      if constexpr (!V_IN_LDS) return dps_t[rl];
+      return __uint_as_float(__builtin_amdgcn_ds_bpermute(
+          (32 + rl) << 2, __float_as_uint(lsedp_reg)));
+    };
 
     if (wave_live && ms + BWD_BM > wq_lo && ms < wq_hi) {
       // ---- S = Q K^T ; dP = dO V^T, UN-swapped: C layout [q=crow][k=lo32],
@@ -433,25 +475,18 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
       // (cframe) instead of an LDS round-trip ----
       f32x16 s = (f32x16)(0.f), dp = (f32x16)(0.f);
       if constexpr (V_IN_LDS) {
-        // fused mode: the lane's K fragments re-load per subtile (keeping
-        // them loop-persistent overflows the 256-reg/2-wave budget). Issue
-        // the 8 loads FIRST, then run the INTERLEAVED dP/S chains — dP's
-        // LDS-only operands cover the K loads' latency, and the two
-        // independent MFMA streams cover each other's LDS reads.
-        bf16x8 kfs[DF];
-#pragma unroll
-        for (int dd = 0; dd < DF; ++dd)
-          kfs[dd] = *(const bf16x8*)(kp_row + dd * 16);
+        // fused mode: K and V fragments come straight off the wave's LDS
+        // tiles (staged once per block) — identical addressing to qf. The
+        // two independent MFMA streams cover each other's LDS read latency.
 #pragma unroll
         for (int dd = 0; dd < DF; ++dd) {
           const int off = swz(lo32, lo32 * ROWB + dd * 32 + hi * 16);
           bf16x8 dof = *(const bf16x8*)((const char*)ldb + off);
-          // lane holds V-tile row lo32 cols dd*16+hi*8 — identical
-          // addressing to qf, off the wave's swizzled V tile
           bf16x8 vf = *(const bf16x8*)((const char*)lds_vt + off);
           dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vf, dp, 0, 0, 0);
           bf16x8 qf = *(const bf16x8*)((const char*)lqb + off);
-          s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kfs[dd], s, 0, 0, 0);
+          bf16x8 kf = *(const bf16x8*)((const char*)lds_kt + off);
+          s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kf, s, 0, 0, 0);
         }
       } else {
 #pragma unroll
@@ -480,7 +515,7 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
       if (interior) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          const float lq = lse_t[crow(r, hi)];
+          const float lq = lse_at(crow(r, hi));
           all_live = all_live && (lq != INFINITY) && (lq != -INFINITY);
         }
       }
@@ -488,16 +523,16 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int rl = crow(r, hi);
-          const float pij = fast_exp2(s[r] * sl2 - lse_t[rl] * log2e);
+          const float pij = fast_exp2(s[r] * sl2 - lse_at(rl) * log2e);
           if constexpr (WANT_DV) pv[r] = pij;
           if constexpr (WANT_DK)
-            dsv[r] = pij * (dp[r] - dps_t[rl]) * p.scale;
+            dsv[r] = pij * (dp[r] - dps_at(rl)) * p.scale;
         }
       } else {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int qrow = ms + crow(r, hi);
-          const float lq = (qrow < qe) ? lse_t[crow(r, hi)] : INFINITY;
+          const float lq = (qrow < qe) ? lse_at(crow(r, hi)) : INFINITY;
           bool ok = (qrow < wq_hi) && (qrow >= qs) && lq != INFINITY &&
                     lq != -INFINITY && kk < ke;
           if (atype == 1 || atype == 3) ok = ok && (kk - qrow <= ke - qe);
@@ -515,7 +550,7 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
           const float pij = ok ? fast_exp2(t - lq * log2e) : 0.f;
           if constexpr (WANT_DV) pv[r] = pij;
           if constexpr (WANT_DK)
-            dsv[r] = pij * (dp[r] - dps_t[crow(r, hi)]) * dscale;
+            dsv[r] = pij * (dp[r] - dps_at(crow(r, hi))) * dscale;
         }
       }
 
