@@ -57,6 +57,14 @@ def is_hierarchical_comm_enable() -> bool:
     return _get_bool("MAGI_ATTENTION_HIERARCHICAL_COMM")
 
 
+def is_bwd_cosched() -> bool:
+    """Default ON: the dq pass runs on a side stream concurrent with the
+    dkv pass (A/B at 64k: 209.6 vs 218.3 ms/step serial, despite the fused
+    kernel's full-LDS workgroups partitioning CUs). MAGI_BWD_COSCHED=0
+    serializes the passes."""
+    return _get("MAGI_BWD_COSCHED", "1") != "0"
+
+
 def bwd_dkv_mode() -> str:
     """"auto" (default): fused dK+dV kernel for long ranges (r2: 124.5 ms vs
     the dv+dk split's 127.4 at 64k, one less launch + one less Q/dO staging

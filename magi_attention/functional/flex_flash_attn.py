@@ -72,10 +72,12 @@ def run_bwd_passes(args, device, dq_tables=None, dkv_tables=None) -> None:
     attn_type_map, seg_starts) — outer = the pass's OWN loop dim."""
     lib = _ffa_lib.lib()
     main = torch.cuda.current_stream(device)
-    side = _get_side_stream(device)
-    ev = torch.cuda.Event()
-    ev.record(main)          # dpsum + inputs ready
-    side.wait_event(ev)
+    cosched = env.is_bwd_cosched()
+    side = _get_side_stream(device) if cosched else main
+    if cosched:
+        ev = torch.cuda.Event()
+        ev.record(main)          # dpsum + inputs ready
+        side.wait_event(ev)
     keep = (dq_tables, dkv_tables)  # outlive the async launches
 
     def set_tables(t, outer_is_q):
@@ -98,9 +100,10 @@ def run_bwd_passes(args, device, dq_tables=None, dkv_tables=None) -> None:
         check(lib.magi_ffa_bwd_dk(args), "magi_ffa_bwd_dk")
     else:
         check(lib.magi_ffa_bwd_dkv(args), "magi_ffa_bwd_dkv")
-    ev2 = torch.cuda.Event()
-    ev2.record(side)
-    main.wait_event(ev2)
+    if cosched:
+        ev2 = torch.cuda.Event()
+        ev2.record(side)
+        main.wait_event(ev2)
     del keep
 
 from .. import _ffa_lib, env
