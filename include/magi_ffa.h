@@ -211,6 +211,32 @@ int magi_grpcoll_ack(void* const* ack_ptrs, int n, void* stream);
 struct magi_grpcoll_pull_args;
 int magi_grpcoll_pull(const struct magi_grpcoll_pull_args* args);
 
+/* Index-attention (token-gather) forward — the reference's
+ * index_attn_indices direct-to-kernel path (flex_flash_attn.py:1358-1390).
+ * indices_2d[i] lists the GLOBAL K rows attended by q token row i, shared
+ * by all hq query heads of that row; -1 = contiguous tail padding. KV heads
+ * are folded into the K row dimension (hk must be 1). Forward only. */
+typedef struct magi_ffa_index_args {
+  const void* q;              /* bf16 [total_q, hq, d] */
+  const void* k;              /* bf16 [total_k, 1, d] */
+  const void* v;              /* bf16 [total_k, 1, d] */
+  void* out;                  /* bf16 or f32 [total_q, hq, d], zero-init */
+  float* lse;                 /* f32 [total_q, hq], caller-initialised -inf */
+  const int32_t* indices_2d;  /* [total_q, max_topk] global K row ids, -1 pad */
+  int64_t total_q;
+  int64_t total_k;
+  int32_t max_topk;           /* multiple of 64 */
+  int32_t hq;
+  int32_t hk;                 /* must be 1 */
+  int32_t d;                  /* 64 or 128 */
+  float softmax_scale;
+  float softcap;
+  int32_t out_is_fp32;
+  void* stream;               /* hipStream_t */
+} magi_ffa_index_args;
+
+int magi_ffa_fwd_index(const magi_ffa_index_args* args);
+
 /* Version/identity probe so tests can verify the native library is loaded. */
 int magi_ffa_abi_version(void);
 

@@ -131,6 +131,29 @@ class MagiGrpCollPullArgs(ctypes.Structure):
     ]
 
 
+class MagiFfaIndexArgs(ctypes.Structure):
+    """mirrors magi_ffa_index_args (index-attention token-gather forward)"""
+
+    _fields_ = [
+        ("q", ctypes.c_void_p),
+        ("k", ctypes.c_void_p),
+        ("v", ctypes.c_void_p),
+        ("out", ctypes.c_void_p),
+        ("lse", ctypes.c_void_p),
+        ("indices_2d", ctypes.c_void_p),
+        ("total_q", ctypes.c_int64),
+        ("total_k", ctypes.c_int64),
+        ("max_topk", ctypes.c_int32),
+        ("hq", ctypes.c_int32),
+        ("hk", ctypes.c_int32),
+        ("d", ctypes.c_int32),
+        ("softmax_scale", ctypes.c_float),
+        ("softcap", ctypes.c_float),
+        ("out_is_fp32", ctypes.c_int32),
+        ("stream", ctypes.c_void_p),
+    ]
+
+
 class MagiCorrectArgs(ctypes.Structure):
     _fields_ = [
         ("out1", ctypes.c_void_p),
@@ -153,6 +176,7 @@ def _try_load() -> ctypes.CDLL | None:
         for name, argtypes in [
             ("magi_ffa_fwd", [ctypes.POINTER(MagiFfaFwdArgs)]),
             ("magi_ffa_fwd_fp8", [ctypes.POINTER(MagiFfaFwdArgs)]),
+            ("magi_ffa_fwd_index", [ctypes.POINTER(MagiFfaIndexArgs)]),
             ("magi_ffa_bwd", [ctypes.POINTER(MagiFfaBwdArgs)]),
             ("magi_ffa_bwd_dq", [ctypes.POINTER(MagiFfaBwdArgs)]),
             ("magi_ffa_bwd_dkv", [ctypes.POINTER(MagiFfaBwdArgs)]),

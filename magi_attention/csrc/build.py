@@ -20,6 +20,7 @@ STAMP = LIBDIR / ".build_stamp"
 
 SOURCES = [
     CSRC / "ffa_fwd.hip",
+    CSRC / "ffa_index.hip",
     CSRC / "ffa_bwd.hip",
     CSRC / "range_ops.hip",
     CSRC / "ffa_fwd_fp8.hip",

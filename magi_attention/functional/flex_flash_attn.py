@@ -350,6 +350,50 @@ def _apply_sink_postprocess(out, lse, sink, sink_layout, tq, hq, d):
     check(_ffa_lib.lib().magi_ffa_sink_postprocess(args), "sink_postprocess")
 
 
+def _flex_flash_attn_forward_index(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    indices_2d: torch.Tensor,
+    softmax_scale: float,
+    softcap: float,
+) -> tuple[torch.Tensor, AttnForwardMeta]:
+    """Index-attention (token-gather) forward: the reference's
+    index_attn_indices direct-to-kernel path (flex_flash_attn.py:1358-1390).
+    indices_2d [total_q, max_topk] int32 lists the GLOBAL K rows each q token
+    row attends, shared by all hq query heads; -1 = contiguous tail padding.
+    Forward only (the reference backward receives None for index_attn)."""
+    from .._ffa_lib import MagiFfaIndexArgs
+
+    assert q.dtype == torch.bfloat16, "index_attn requires bf16 inputs"
+    q, k, v, indices_2d = [maybe_contiguous(x) for x in (q, k, v, indices_2d)]
+    tq, hq, d = q.shape
+    tk, hk, _ = k.shape
+    assert hk == 1, (
+        "index_attn expects KV heads folded into the K row dimension "
+        "(k of shape [total_k, 1, d]; global id = (b*S_kv + t)*NHK + h)"
+    )
+    assert indices_2d.dtype == torch.int32 and indices_2d.dim() == 2
+    assert indices_2d.shape[0] == tq, (
+        f"indices_2d rows ({indices_2d.shape[0]}) must match q token rows ({tq})"
+    )
+    max_topk = indices_2d.shape[1]
+
+    out = torch.zeros(tq, hq, d, dtype=q.dtype, device=q.device)
+    lse = torch.full((tq, hq), float("-inf"), dtype=torch.float32, device=q.device)
+
+    args = MagiFfaIndexArgs(
+        q=ptr(q), k=ptr(k), v=ptr(v), out=ptr(out), lse=ptr(lse),
+        indices_2d=ptr(indices_2d),
+        total_q=tq, total_k=tk, max_topk=max_topk,
+        hq=hq, hk=hk, d=d,
+        softmax_scale=softmax_scale, softcap=softcap,
+        out_is_fp32=0, stream=current_stream_ptr(),
+    )
+    check(_ffa_lib.lib().magi_ffa_fwd_index(args), "magi_ffa_fwd_index")
+    return out, AttnForwardMeta(lse=lse, max_logits=None)
+
+
 def _flex_flash_attn_backward(
     dout: torch.Tensor,
     q: torch.Tensor,
@@ -583,10 +627,62 @@ def flex_flash_attn_func(
 ) -> tuple[torch.Tensor, AttnForwardMeta]:
     """Single-GPU flex-flash-attention (drop-in for the reference
     flex_flash_attn_func, flex_flash_attn.py:1066; full mask semantics in the
-    reference docstring :1247-1341). Returns (out, AttnForwardMeta(lse=...))."""
-    assert index_attn_indices is None and not index_attn, (
-        "index_attn lands in a later round"
+    reference docstring :1247-1341). Returns (out, AttnForwardMeta(lse=...)).
+
+    index_attn_indices (reference :1358-1390): (total_q, num_kv_heads,
+    max_topk) int32 GLOBAL K row ids, -1 tail padding, mutually exclusive
+    with q_ranges/k_ranges; max_topk must be a multiple of 64 (the MI355X
+    staged-tile size; the reference requires 128, so any reference-legal
+    input is accepted). Forward only."""
+    # ── sparse-input validation (mirrors reference :1344-1384) ──
+    _has_ranges = q_ranges is not None
+    _has_index = index_attn_indices is not None
+    assert int(_has_ranges) + int(_has_index) == 1, (
+        "Exactly one of (q_ranges + k_ranges) or index_attn_indices must be "
+        "provided"
     )
+    assert not (sparse_load and _has_index), (
+        "sparse_load and index_attn_indices are mutually exclusive"
+    )
+    if _has_index:
+        assert index_attn_indices.dim() == 3, (
+            f"index_attn_indices must be 3D (total_q, num_kv_heads, max_topk), "
+            f"got shape {tuple(index_attn_indices.shape)}"
+        )
+        assert q_block_size == 1 and k_block_size == 1, (
+            "only q_block_size=1 / k_block_size=1 (token granularity) is "
+            "supported for index_attn"
+        )
+        max_topk = index_attn_indices.shape[2]
+        assert max_topk % 64 == 0, (
+            f"index_attn max_topk={max_topk} must be a multiple of 64 "
+            f"(pad with -1)"
+        )
+        assert not (
+            torch.is_grad_enabled()
+            and any(t.requires_grad for t in (q, k, v))
+        ), "index_attn is forward-only (matches the reference)"
+        if softmax_scale is None:
+            softmax_scale = q.shape[-1] ** (-0.5)
+        d = q.shape[-1]
+        if d not in (64, 128):
+            # zero-pad odd head dims into the next bucket (same scheme as the
+            # dense path below; exact for attention)
+            assert d < 128, f"index_attn head_dim {d} > 128 unsupported"
+            bucket = 64 if d <= 64 else 128
+            pad = bucket - d
+            out, meta = _flex_flash_attn_forward_index(
+                torch.nn.functional.pad(q, (0, pad)),
+                torch.nn.functional.pad(k, (0, pad)),
+                torch.nn.functional.pad(v, (0, pad)),
+                index_attn_indices.reshape(-1, max_topk),
+                softmax_scale, softcap,
+            )
+            return out[..., :d], meta
+        return _flex_flash_attn_forward_index(
+            q, k, v, index_attn_indices.reshape(-1, max_topk),
+            softmax_scale, softcap,
+        )
     d = q.shape[-1]
     if d not in (64, 128, 192):
         # arbitrary head dims run in the next bucket with zero feature
