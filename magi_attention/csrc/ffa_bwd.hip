@@ -182,7 +182,7 @@ __global__ __launch_bounds__(256) void bwd_preprocess_kernel(BwdParams p, int d,
 // barrier: prefetch has TWO compute phases, and the barrier only waits for
 // the one stage it needs — attacks the 34-42% SQ_WAIT_ANY of the r1 PMC.
 template <int D, bool HAS_SOFTCAP, int MODE, int WAVES, int NBUF, int NT = 0>
-__global__ __launch_bounds__(64 * WAVES, WAVES == 8 ? 2 : 1)
+__global__ __launch_bounds__(64 * WAVES, WAVES >= 6 ? 2 : 1)
 void ffa_bwd_dkv_kernel(BwdParams p) {
   // NT: stage with the non-temporal policy (aux=2) — staged rows are read
   // once per WG; keeping them out of L2 protects the resident K strips and
@@ -190,7 +190,13 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   constexpr int STAGE_AUX = NT ? 2 : 0;
   constexpr bool WANT_DV = MODE != 2;
   constexpr bool WANT_DK = MODE != 1;
-  constexpr bool V_IN_LDS = MODE == 0;
+  // MODE 3 = fused "fat wave" (r2-v3): 4 waves x 64 k rows (2 column tiles)
+  // at 1 wave/SIMD — the unified 512-reg file holds BOTH columns' dK/dV
+  // accumulators (256 regs -> AGPRs), 64 MFMAs per wave between barriers
+  // instead of 32, and half the waves to sync per barrier. Same 256-row
+  // k span per WG as MODE 0 at 8 waves, identical grid.
+  constexpr bool V_IN_LDS = (MODE == 0 || MODE == 3);
+  constexpr int NCOL = (MODE == 3) ? 2 : 1;
   constexpr int DF = D / 16;
   constexpr int DT = D / 32;
   // LDS rows padded to a power of two (D=192 -> 512-byte rows; see
@@ -231,7 +237,7 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   const int h = p.head_off + hidx * p.head_mult;
   const int wb = p.head_major ? wy : wx;
   const int ks = p.k_ranges[2 * ri], ke = p.k_ranges[2 * ri + 1];
-  const int nblk0 = ks + wb * (BWD_BN * WAVES);
+  const int nblk0 = ks + wb * (BWD_BN * NCOL * WAVES);
   if (nblk0 >= ke) continue;
   const int seg0 = p.seg_starts ? p.seg_starts[ri] : ri;
   const int seg1 = p.seg_starts ? p.seg_starts[ri + 1] : ri + 1;
@@ -248,7 +254,7 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   const int lo32 = lane & 31;
   const int hi = lane >> 5;
 
-  const int n0 = nblk0 + wave * BWD_BN;  // this wave's k tile
+  const int n0 = nblk0 + wave * (BWD_BN * NCOL);  // this wave's k tile(s)
   const bool wave_live = n0 < ke;
   const bool skip_dq = (p.debug_ablate & 1) != 0;
 
@@ -280,9 +286,13 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   // scheduler slack below the 256 cap.
   // (the 32-row ring is only needed at D=128, where the K/V tiles take
   // 128 KB; at D=64 everything fits beside a 64-row ring)
-  constexpr int QITER = (V_IN_LDS && D == 128) ? BWD_BM : 2 * BWD_BM;
+  // Ring depth: the fused modes' K/V tiles crowd the ring out of LDS at
+  // D=128 when the WG spans 256 k rows (W8, or W4x2col); the W6 fused
+  // variant (192 k rows -> 96 KB of tiles) fits a 64-row ring exactly.
+  constexpr int QITER =
+      (V_IN_LDS && D == 128 && WAVES * NCOL >= 8) ? BWD_BM : 2 * BWD_BM;
   constexpr int NSUB = QITER / BWD_BM;
-  constexpr int KVLDS = V_IN_LDS ? 2 * WAVES * BWD_BN * ROWB : 0;
+  constexpr int KVLDS = V_IN_LDS ? 2 * WAVES * NCOL * BWD_BN * ROWB : 0;
   constexpr int LSELDS = V_IN_LDS ? 0 : NBUF * 2 * QITER * 4;
   static_assert(NBUF * 2 * QITER * ROWB + LSELDS + KVLDS <= 163840,
                 "LDS budget");
@@ -303,10 +313,10 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   // MODE 0: per-wave K and V tiles (BWD_BN rows x D, same 32-B XOR swizzle
   // as the Q/dO images so the A-fragment reads reuse the qf addressing)
   __bf16* lds_kt = (__bf16*)(smem + NBUF * 2 * QITER * ROWB + LSELDS) +
-                   wave * BWD_BN * ROWE;
+                   wave * NCOL * BWD_BN * ROWE;
   __bf16* lds_vt = (__bf16*)(smem + NBUF * 2 * QITER * ROWB + LSELDS +
                              KVLDS / 2) +
-                   wave * BWD_BN * ROWE;
+                   wave * NCOL * BWD_BN * ROWE;
 
   // K fragments (A-layout), loaded once per block; V fragments likewise in
   // the split dK mode — the fused mode stages V into LDS instead (register
@@ -331,7 +341,7 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   constexpr int ROWS_PER_GLDS_V = 1024 / ROWB;
   if constexpr (V_IN_LDS) {
 #pragma unroll
-    for (int gi = 0; gi < BWD_BN / ROWS_PER_GLDS_V; ++gi) {
+    for (int gi = 0; gi < (NCOL * BWD_BN) / ROWS_PER_GLDS_V; ++gi) {
       const int r0v = ROWS_PER_GLDS_V * gi;
       const int r = r0v + lane / (ROWB / 16);
       const int c = lane % (ROWB / 16);
@@ -353,7 +363,7 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   }
 
   // block + wave q loop bounds (recomputed per q segment)
-  const int nlast = min(nblk0 + BWD_BN * WAVES, ke) - 1;
+  const int nlast = min(nblk0 + BWD_BN * NCOL * WAVES, ke) - 1;
   int q_lo = 0, q_hi = 0, wq_lo = 0, wq_hi = 0;
   auto seg_bounds = [&]() {
     q_lo = qs; q_hi = qe;
@@ -362,25 +372,31 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
     wq_lo = qs; wq_hi = qe;
     if (atype == 1 || atype == 3) wq_lo = max(wq_lo, n0 - (ke - qe));
     if (atype == 2 || atype == 3)
-      wq_hi = min(wq_hi, (n0 + BWD_BN - 1) - (ks - qs) + 1);
+      wq_hi = min(wq_hi, (n0 + NCOL * BWD_BN - 1) - (ks - qs) + 1);
   };
 
-  f32x16 acc_dk[WANT_DK ? DT : 1], acc_dv[WANT_DV ? DT : 1];
+  f32x16 acc_dk[WANT_DK ? NCOL * DT : 1], acc_dv[WANT_DV ? NCOL * DT : 1];
 #pragma unroll
-  for (int dt = 0; dt < DT; ++dt) {
+  for (int dt = 0; dt < NCOL * DT; ++dt) {
     if constexpr (WANT_DK) acc_dk[dt] = (f32x16)(0.f);
     if constexpr (WANT_DV) acc_dv[dt] = (f32x16)(0.f);
   }
 
   // each wave's glds covers 4 rows (64 lanes x 16B = 1 KiB = 4 rows at D=128);
-  // wave w owns rows [8w, 8w+8) via 2 issues per tensor
+  // glds issues are dealt round-robin over the waves (wave-strided), so any
+  // WAVES count works; the NBUF=3 constant-distance barrier needs a UNIFORM
+  // per-wave issue count, so it requires divisibility.
   constexpr int ROWS_PER_GLDS = 1024 / ROWB;
-  static_assert(QITER / WAVES >= ROWS_PER_GLDS, "stage rows per wave");
-  constexpr int GLDS_PER_WAVE = (QITER / WAVES) / ROWS_PER_GLDS;
+  constexpr int GLDS_TOTAL = QITER / ROWS_PER_GLDS;
+  static_assert(NBUF == 2 || GLDS_TOTAL % WAVES == 0, "uniform stage count");
+  constexpr int GLDS_PER_WAVE =
+      (GLDS_TOTAL % WAVES == 0) ? GLDS_TOTAL / WAVES : 0;  // 0 = non-uniform
   auto stage_glds = [&](int buf, int m0x) {
 #pragma unroll
-    for (int gi = 0; gi < GLDS_PER_WAVE; ++gi) {
-      const int r0 = (QITER / WAVES) * wave + ROWS_PER_GLDS * gi;
+    for (int gi0 = 0; gi0 < (GLDS_TOTAL + WAVES - 1) / WAVES; ++gi0) {
+      const int gi = gi0 * WAVES + wave;
+      if (GLDS_TOTAL % WAVES != 0 && gi >= GLDS_TOTAL) break;
+      const int r0 = ROWS_PER_GLDS * gi;
       const int r = r0 + lane / (ROWB / 16);
       const int c = lane % (ROWB / 16);
       const int qrow = min(m0x + r, qe - 1);
@@ -470,6 +486,13 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
     };
 
     if (wave_live && ms + BWD_BM > wq_lo && ms < wq_hi) {
+      // MODE 3: the wave's two 32-column tiles in sequence — each column has
+      // its own C tiles and accumulator slice; Q/dO fragments and the lse /
+      // dpsum selects are re-read per column (LDS/lane-local, cheap), the
+      // barrier/staging cost is paid once for both.
+#pragma unroll
+      for (int ch = 0; ch < NCOL; ++ch) {
+      const int nch = n0 + ch * BWD_BN;
       // ---- S = Q K^T ; dP = dO V^T, UN-swapped: C layout [q=crow][k=lo32],
       // so the dV/dK A-fragments come from the in-register permlane transform
       // (cframe) instead of an LDS round-trip ----
@@ -478,14 +501,16 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
         // fused mode: K and V fragments come straight off the wave's LDS
         // tiles (staged once per block) — identical addressing to qf. The
         // two independent MFMA streams cover each other's LDS read latency.
+        const char* kt_c = (const char*)(lds_kt + ch * BWD_BN * ROWE);
+        const char* vt_c = (const char*)(lds_vt + ch * BWD_BN * ROWE);
 #pragma unroll
         for (int dd = 0; dd < DF; ++dd) {
           const int off = swz(lo32, lo32 * ROWB + dd * 32 + hi * 16);
           bf16x8 dof = *(const bf16x8*)((const char*)ldb + off);
-          bf16x8 vf = *(const bf16x8*)((const char*)lds_vt + off);
+          bf16x8 vf = *(const bf16x8*)(vt_c + off);
           dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vf, dp, 0, 0, 0);
           bf16x8 qf = *(const bf16x8*)((const char*)lqb + off);
-          bf16x8 kf = *(const bf16x8*)((const char*)lds_kt + off);
+          bf16x8 kf = *(const bf16x8*)(kt_c + off);
           s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kf, s, 0, 0, 0);
         }
       } else {
@@ -501,16 +526,16 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
       }
       }
 
-      const int kk = n0 + lo32;  // this lane's k column
+      const int kk = nch + lo32;  // this lane's k column
       // S is dead once P is computed, dP once dS is — reuse their registers
       // (all indices are compile-time constants after unrolling, so the
       // pointer cast stays in VGPRs; checked: ScratchSize 0)
       float* pv = (float*)&s;
       float* dsv = (float*)&dp;
       const bool interior =
-          (ms + BWD_BM <= wq_hi) && (ms >= qs) && (n0 + BWD_BN <= ke) &&
-          !((atype == 1 || atype == 3) && (n0 + BWD_BN - 1 > ms + (ke - qe))) &&
-          !((atype == 2 || atype == 3) && (n0 < ms + BWD_BM - 1 + (ks - qs)));
+          (ms + BWD_BM <= wq_hi) && (ms >= qs) && (nch + BWD_BN <= ke) &&
+          !((atype == 1 || atype == 3) && (nch + BWD_BN - 1 > ms + (ke - qe))) &&
+          !((atype == 2 || atype == 3) && (nch < ms + BWD_BM - 1 + (ks - qs)));
       bool all_live = interior;
       if (interior) {
 #pragma unroll
@@ -603,10 +628,10 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
         };
         if constexpr (MODE == 1) {
           pipe_mm(do_base, cframe_to_afrag(pv, 0), cframe_to_afrag(pv, 1),
-                  acc_dv);
+                  acc_dv + ch * DT);
         } else if constexpr (MODE == 2) {
           pipe_mm(q_base, cframe_to_afrag(dsv, 0), cframe_to_afrag(dsv, 1),
-                  acc_dk);
+                  acc_dk + ch * DT);
         } else {
           // MODE 0 (fused dK+dV): INTERLEAVE the two independent MFMA
           // streams — each fragment load's LDS latency is covered by the
@@ -619,26 +644,28 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
           bf16x8 da1 = cframe_to_afrag(dsv, 1);
 #pragma unroll
           for (int dt = 0; dt < DT; ++dt) {
+            const int at = ch * DT + dt;
             const int dcol = (dt * 32 + 16 * qhalf) * 2;
             bf16x8 bdo = tr16_frag(do_base + rb0 + (dcol ^ sw0),
                                    do_base + rb1 + (dcol ^ sw1));
-            acc_dv[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                pa0, bdo, acc_dv[dt], 0, 0, 0);
+            acc_dv[at] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                pa0, bdo, acc_dv[at], 0, 0, 0);
             bf16x8 bq = tr16_frag(q_base + rb0 + (dcol ^ sw0),
                                   q_base + rb1 + (dcol ^ sw1));
-            acc_dk[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                da0, bq, acc_dk[dt], 0, 0, 0);
+            acc_dk[at] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                da0, bq, acc_dk[at], 0, 0, 0);
             bf16x8 bdo1 = tr16_frag(do_base + rb2 + (dcol ^ sw2),
                                     do_base + rb3 + (dcol ^ sw3));
-            acc_dv[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                pa1, bdo1, acc_dv[dt], 0, 0, 0);
+            acc_dv[at] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                pa1, bdo1, acc_dv[at], 0, 0, 0);
             bf16x8 bq1 = tr16_frag(q_base + rb2 + (dcol ^ sw2),
                                    q_base + rb3 + (dcol ^ sw3));
-            acc_dk[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                da1, bq1, acc_dk[dt], 0, 0, 0);
+            acc_dk[at] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                da1, bq1, acc_dk[at], 0, 0, 0);
           }
         }
       }
+      }  // ch (column tile)
     }
     }  // sub
     cur = (NBUF == 3) ? (cur == 2 ? 0 : cur + 1) : (cur ^ 1);
@@ -649,25 +676,28 @@ void ffa_bwd_dkv_kernel(BwdParams p) {
   // ---- write dK/dV ----
   if (wave_live && !(p.debug_ablate & 2)) {
 #pragma unroll
+  for (int ch = 0; ch < NCOL; ++ch) {
+#pragma unroll
   for (int r = 0; r < 16; ++r) {
-    const int kr = n0 + crow(r, hi);
+    const int kr = n0 + ch * BWD_BN + crow(r, hi);
     if (kr >= ke) continue;
     float* dkp = p.dk + (size_t)kr * k_pitch + (size_t)kh * D;
     float* dvp = p.dv + (size_t)kr * k_pitch + (size_t)kh * D;
 #pragma unroll
     for (int dt = 0; dt < DT; ++dt) {
       if constexpr (WANT_DK) {
-        if (acc_dk[dt][r] != 0.f)
-          unsafeAtomicAdd(dkp + dt * 32 + lo32, acc_dk[dt][r]);
+        if (acc_dk[ch * DT + dt][r] != 0.f)
+          unsafeAtomicAdd(dkp + dt * 32 + lo32, acc_dk[ch * DT + dt][r]);
       }
       if constexpr (WANT_DV) {
-        if (acc_dv[dt][r] != 0.f)
-          unsafeAtomicAdd(dvp + dt * 32 + lo32, acc_dv[dt][r]);
+        if (acc_dv[ch * DT + dt][r] != 0.f)
+          unsafeAtomicAdd(dvp + dt * 32 + lo32, acc_dv[ch * DT + dt][r]);
       }
     }
   }
+  }  // ch
   }  // wave_live store
-  if constexpr (MODE == 0) break;  // single trip: kills the backedge (regs)
+  if constexpr (MODE == 0 || MODE == 3) break;  // single trip: kills the backedge (regs)
   }  // strided work loop
 }
 
@@ -1397,12 +1427,23 @@ static int launch_bwd_dkv(const magi_ffa_bwd_args* a) {
   // W8 for ALL modes on long ranges: one 512-thread WG/CU halves the
   // staging streams (PMC r2: dV at W4 fetched 174 GB/launch vs dK-W8's 71)
   const int W = big ? 8 : 4;
+  // MODE 3 "fat" fused: 4 waves x 2 column tiles at 1 wave/SIMD (same
+  // 256-row span/grid as MODE 0 at W8) — A/B gate, see kernel note
+  int fat = 0;
+  { const char* e = getenv("MAGI_BWD_FAT"); if (e) fat = atoi(e); }
+  const bool use_fat = (MODE == 0) && big && fat != 0;
+  // W6 fused: 192-row k span, 64-row Q/dO ring (2x the MFMAs per barrier of
+  // the W8 32-row ring), 1.5 waves/SIMD — A/B gate
+  int w6 = 0;
+  { const char* e = getenv("MAGI_BWD_W6"); if (e) w6 = atoi(e); }
+  const bool use_w6 = (MODE == 0) && big && !use_fat && w6 != 0;
   int nbuf = (MODE == 0 || W == 4) ? 2 : 3;
   { const char* e = getenv("MAGI_BWD_NBUF");
     if (e && nbuf == 3) nbuf = atoi(e); }
   int nt = 0;
   { const char* e = getenv("MAGI_STAGE_NT"); if (e) nt = atoi(e); }
-  const int span = BWD_BN * W;
+  const int span =
+      use_fat ? (2 * 4 * BWD_BN) : (use_w6 ? 6 * BWD_BN : BWD_BN * W);
   const int nblocks = (a->max_seqlen_k + span - 1) / span;
   if (a->n_ranges > 65535) return -5;
   dim3 grid_kv = p.head_major
@@ -1418,14 +1459,32 @@ static int launch_bwd_dkv(const magi_ffa_bwd_args* a) {
     if (cap > total) cap = total;
     grid_kv = dim3((unsigned)cap, 1, 1);
   }
-  dim3 block(64 * W);
+  dim3 block(use_fat ? 256 : (use_w6 ? 384 : 64 * W));
   hipStream_t s = (hipStream_t)a->stream;
   const bool sc = a->softcap > 0.f;
 #define LAUNCH_DKV(DD, SC, WW, NB, NTV) \
   hipLaunchKernelGGL((ffa_bwd_dkv_kernel<DD, SC, MODE, WW, NB, NTV>), grid_kv, \
                      block, 0, s, p)
+#define LAUNCH_FAT(DD, SC, NTV) \
+  hipLaunchKernelGGL((ffa_bwd_dkv_kernel<DD, SC, 3, 4, 2, NTV>), grid_kv, \
+                     block, 0, s, p)
+#define LAUNCH_W6(DD, SC, NTV) \
+  hipLaunchKernelGGL((ffa_bwd_dkv_kernel<DD, SC, 0, 6, 2, NTV>), grid_kv, \
+                     block, 0, s, p)
 #define PICK_DKV_NT(DD, NTV) \
   do { \
+    if (use_fat) { \
+      if constexpr (MODE == 0 && DD != 192) { \
+        if (sc) LAUNCH_FAT(DD, true, NTV); else LAUNCH_FAT(DD, false, NTV); \
+        break; \
+      } \
+    } \
+    if (use_w6) { \
+      if constexpr (MODE == 0 && DD != 192) { \
+        if (sc) LAUNCH_W6(DD, true, NTV); else LAUNCH_W6(DD, false, NTV); \
+        break; \
+      } \
+    } \
     if (W == 8) { \
       bool done = false; \
       if constexpr (MODE != 0 && DD != 192) { /* 3-ring LDS limits */ \
@@ -1449,6 +1508,8 @@ static int launch_bwd_dkv(const magi_ffa_bwd_args* a) {
   } else PICK_DKV(128);
 #undef PICK_DKV
 #undef PICK_DKV_NT
+#undef LAUNCH_W6
+#undef LAUNCH_FAT
 #undef LAUNCH_DKV
   return (int)hipGetLastError();
 }
