@@ -255,6 +255,29 @@ def measure_extra_configs(device):
     }
     del q, k, v
 
+    # ---- index_attn: DiT ratio-128 token-gather forward (ffa_index.hip) ----
+    from magi_attention.utils import build_index_attn_indices
+
+    toks, ihq, skv, topk = 4096, 128, 8192, 2048
+    q = (torch.randn(toks, ihq, D, generator=g) * 0.5).bfloat16().to(device)
+    k = (torch.randn(skv, 1, D, generator=g) * 0.5).bfloat16().to(device)
+    v = (torch.randn(skv, 1, D, generator=g) * 0.5).bfloat16().to(device)
+    idx = build_index_attn_indices(1, 1, toks, skv, topk, topk, device=device)
+
+    def idx_fwd():
+        with torch.no_grad():
+            flex_flash_attn_func(q, k, v, index_attn_indices=idx)
+
+    dt = time_fn(idx_fwd, steps=10, warm=3)
+    fl = 4.0 * toks * ihq * topk * D
+    out["index_attn_dit_fwd"] = {
+        "tflops": fl / dt / 1e12, "ms": dt * 1e3,
+        "frac_of_mfma_peak": fl / dt / MFMA_PEAK_BF16,
+        "config": "index_attn 4096 tokens x 128 qheads, topk 2048 of 8192 kv,"
+                  " d128 bf16, fwd only (forward-only feature)",
+    }
+    del q, k, v, idx
+
     # ---- config 2: varlen packed 16k (8 causal docs of 2k), hq16 d128 ----
     n, hq, hkv = 16384, 16, 16
     q = (torch.randn(n, hq, D, generator=g) * 0.5).bfloat16().to(device)
