@@ -29,7 +29,6 @@
 #include "../../include/magi_ffa.h"
 
 #define IDX_BN 32  // k rows per inner MFMA tile
-#define IDX_KITER 64  // k rows per staged LDS buffer
 
 using f32x16 = __attribute__((ext_vector_type(16))) float;
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
@@ -128,6 +127,12 @@ __global__ __launch_bounds__(64 * WAVES, 8 / WAVES) void ffa_index_fwd_kernel(
     return byte_off ^ ((row & SW32M) << 5);
   };
 
+  // Staged-image depth scales with the wave count: small-ratio workgroups
+  // (1-2 waves, hq<=64) keep 32-row buffers so 4-5 WGs co-reside per CU
+  // (the 64-row image at 64 KB capped LDS occupancy at 2 WGs -> 0.5-1
+  // waves/SIMD, measured 88 TF at hq=32); 4-wave WGs keep the 64-row image
+  // (2 WGs/CU, 2 waves/SIMD, one barrier per 64 gathered rows).
+  constexpr int IDX_KITER = (WAVES >= 4) ? 64 : 32;
   __shared__ __attribute__((aligned(16))) char ismem[2 * 2 * IDX_KITER * ROWB];
   auto lds_k = [&](int buf) -> __bf16* {
     return (__bf16*)(ismem + (2 * buf) * IDX_KITER * ROWB);
@@ -312,7 +317,7 @@ __global__ __launch_bounds__(64 * WAVES, 8 / WAVES) void ffa_index_fwd_kernel(
     __syncthreads();  // buf[cur] glds drained
     if (n0 + IDX_KITER < count) stage_glds(cur ^ 1, n0 + IDX_KITER);
 #pragma unroll
-    for (int sub = 0; sub < 2; ++sub) {
+    for (int sub = 0; sub < IDX_KITER / IDX_BN; ++sub) {
       const int ns = n0 + sub * IDX_BN;
       if (ns >= count) break;  // count is block-uniform
       sub_body(ns, lds_k(cur) + sub * IDX_BN * ROWE,
