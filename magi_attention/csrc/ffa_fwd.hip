@@ -37,6 +37,16 @@ using bf16_t = __bf16;
 
 #define DEV_INLINE __device__ __forceinline__
 
+// Software-pipelined LDS-DMA barrier (see ffa_bwd.hip): wait until at most
+// VM vector-memory ops are outstanding (the one prefetch stage this wave
+// still has in flight), instead of the full vmcnt(0) drain __syncthreads()
+// would insert. Valid because the k loop issues NO other vector-memory ops.
+template <int VM>
+DEV_INLINE void fwd_pipe_barrier() {
+  asm volatile("s_waitcnt vmcnt(%0)\n\ts_waitcnt lgkmcnt(0)\n\ts_barrier"
+               ::"i"(VM) : "memory");
+}
+
 DEV_INLINE float warp_xor32(float v) { return __shfl_xor(v, 32, 64); }
 
 // one packed convert (RNE, same as the scalar bf16 cast) — the C version
@@ -103,7 +113,8 @@ struct FwdParams {
 // WAVES: q-tiles per workgroup sharing ONE staged K/V image (see the bwd
 // kernels): 8 waves (one 512-thread WG/CU = 2 waves/SIMD) halve the staging
 // and barrier cost per MFMA for long ranges; 4 for short ranges.
-template <int D, bool HAS_SOFTCAP, bool ATOMIC, bool OUT_BF16, int WAVES>
+template <int D, bool HAS_SOFTCAP, bool ATOMIC, bool OUT_BF16, int WAVES,
+          int NBUF = 2>
 __global__ __launch_bounds__(64 * WAVES, 8 / WAVES) void ffa_fwd_kernel(FwdParams p) {
   constexpr int DF = D / 16;    // # of 16-wide d fragments
   constexpr int DT = D / 32;    // # of 32-wide output d tiles
@@ -190,11 +201,15 @@ __global__ __launch_bounds__(64 * WAVES, 8 / WAVES) void ffa_fwd_kernel(FwdParam
   auto swz = [](int row, int byte_off) {
     return byte_off ^ ((row & SW32M) << 5);
   };
-  // 2 buffers x 64-row K/V images staged by LDS-DMA (same scheme as the
+  // NBUF x 64-row K/V images staged by LDS-DMA (same scheme as the
   // backward kernels): one barrier per 64 k rows, the doubled compute phase
-  // covers the prefetch latency the 32-row register pipeline could not
+  // covers the prefetch latency the 32-row register pipeline could not.
+  // NBUF=3 (r2, ported from the bwd 3-ring): constant-distance vmcnt wait —
+  // the barrier waits only for the one stage it needs and the prefetch gets
+  // two compute phases to land, removing the full vmcnt(0) drain per
+  // iteration (the dominant stall on short ranges / 8k).
   constexpr int KITER = 2 * FFA_BN;
-  __shared__ __attribute__((aligned(16))) char fsmem[2 * 2 * KITER * ROWB];
+  __shared__ __attribute__((aligned(16))) char fsmem[NBUF * 2 * KITER * ROWB];
   auto lds_k = [&](int buf) -> __bf16* {
     return (__bf16*)(fsmem + (2 * buf) * KITER * ROWB);
   };
@@ -225,6 +240,7 @@ __global__ __launch_bounds__(64 * WAVES, 8 / WAVES) void ffa_fwd_kernel(FwdParam
   constexpr int ROWS_PER_GLDS = 1024 / ROWB;
   static_assert(KITER / WAVES >= ROWS_PER_GLDS, "stage rows per wave");
   constexpr int GLDS_PER_WAVE = (KITER / WAVES) / ROWS_PER_GLDS;
+  constexpr int GOPS = 2 * GLDS_PER_WAVE;  // vm ops per stage call per wave
   auto stage_glds = [&](int buf, int n0x) {
 #pragma unroll
     for (int gi = 0; gi < GLDS_PER_WAVE; ++gi) {
@@ -410,9 +426,17 @@ __global__ __launch_bounds__(64 * WAVES, 8 / WAVES) void ffa_fwd_kernel(FwdParam
     any_seg = true;
     cur = 0;
     stage_glds(0, b_lo);
+    if constexpr (NBUF == 3) stage_glds(1, b_lo + KITER);  // rows clamp
     for (int n0 = b_lo; n0 < b_hi; n0 += KITER) {
-      __syncthreads();  // buf[cur] glds drained (full 64-row phase of cover)
-      if (n0 + KITER < b_hi) stage_glds(cur ^ 1, n0 + KITER);
+      // NBUF=3: wait ONLY for buf[cur] (leave the next stage in flight),
+      // then prefetch two stages ahead — always, with clamped rows, so the
+      // vmcnt distance stays constant. NBUF=2: full drain.
+      fwd_pipe_barrier<NBUF == 3 ? GOPS : 0>();
+      if constexpr (NBUF == 3) {
+        stage_glds(cur == 0 ? 2 : cur - 1, n0 + 2 * KITER);
+      } else {
+        if (n0 + KITER < b_hi) stage_glds(cur ^ 1, n0 + KITER);
+      }
 #pragma unroll
       for (int sub = 0; sub < 2; ++sub) {
         const int ns = n0 + sub * FFA_BN;
@@ -420,9 +444,10 @@ __global__ __launch_bounds__(64 * WAVES, 8 / WAVES) void ffa_fwd_kernel(FwdParam
         sub_body(ns, lds_k(cur) + sub * FFA_BN * ROWE,
                  lds_v(cur) + sub * FFA_BN * ROWE);
       }
-      cur ^= 1;
+      cur = (NBUF == 3) ? (cur == 2 ? 0 : cur + 1) : (cur ^ 1);
     }
     // LDS reads of this segment retired before the next segment's staging
+    // (implicit vmcnt(0) drains the in-flight prefetches too)
     __syncthreads();
   }
   if (any_seg) {
@@ -584,15 +609,15 @@ extern "C" int magi_probe_mfma(const void* a, const void* b, void* d,
 // ------------------------------------------------------------------
 // launcher
 // ------------------------------------------------------------------
-template <int D, int W>
+template <int D, int W, int NB>
 static int launch_fwd_d(const magi_ffa_fwd_args* a, const FwdParams& p,
                         dim3 grid, dim3 block, hipStream_t stream) {
   const bool sc = a->softcap > 0.f;
   const bool atomic = !a->disable_atomic_reduction;
   const bool obf16 = !a->out_is_fp32;
   if (atomic && obf16) return -10;  // atomic merge requires fp32 out
-#define LAUNCH(SC, AT, OB)                                              \
-  hipLaunchKernelGGL((ffa_fwd_kernel<D, SC, AT, OB, W>), grid, block,   \
+#define LAUNCH(SC, AT, OB)                                                 \
+  hipLaunchKernelGGL((ffa_fwd_kernel<D, SC, AT, OB, W, NB>), grid, block,  \
                      0, stream, p)
   if (atomic) {
     if (sc) LAUNCH(true, true, false);
@@ -666,14 +691,23 @@ extern "C" int magi_ffa_fwd(const magi_ffa_fwd_args* a) {
   }
   dim3 block(64 * fw);
   hipStream_t stream = (hipStream_t)a->stream;
+  // 3-slot staging ring (constant-distance vmcnt barrier) for the W8 path;
+  // the W4/D=192 builds keep NBUF=2 (a 3-ring of 512-B rows exceeds LDS,
+  // and W4 runs 2 WGs/CU which the 96 KB ring would halve)
+  int nbuf = 3;
+  { const char* e = getenv("MAGI_FWD_NBUF"); if (e) nbuf = atoi(e); }
   if (a->d == 64)
-    return fw == 8 ? launch_fwd_d<64, 8>(a, p, grid, block, stream)
-                   : launch_fwd_d<64, 4>(a, p, grid, block, stream);
+    return fw == 8 ? (nbuf == 3
+                          ? launch_fwd_d<64, 8, 3>(a, p, grid, block, stream)
+                          : launch_fwd_d<64, 8, 2>(a, p, grid, block, stream))
+                   : launch_fwd_d<64, 4, 2>(a, p, grid, block, stream);
   if (a->d == 192)
-    return fw == 8 ? launch_fwd_d<192, 8>(a, p, grid, block, stream)
-                   : launch_fwd_d<192, 4>(a, p, grid, block, stream);
-  return fw == 8 ? launch_fwd_d<128, 8>(a, p, grid, block, stream)
-                 : launch_fwd_d<128, 4>(a, p, grid, block, stream);
+    return fw == 8 ? launch_fwd_d<192, 8, 2>(a, p, grid, block, stream)
+                   : launch_fwd_d<192, 4, 2>(a, p, grid, block, stream);
+  return fw == 8 ? (nbuf == 3
+                        ? launch_fwd_d<128, 8, 3>(a, p, grid, block, stream)
+                        : launch_fwd_d<128, 8, 2>(a, p, grid, block, stream))
+               : launch_fwd_d<128, 4, 2>(a, p, grid, block, stream);
 }
 
 extern "C" int magi_ffa_abi_version(void) { return 1; }
