@@ -549,3 +549,75 @@ def test_head_dim_192_bucket():
                             dk_lo.float(), f"d{d}:dk", ratio=4.5)
         assert_close_to_ref(v.grad.cpu().float(), dv_hi.float(),
                             dv_lo.float(), f"d{d}:dv", ratio=4.5)
+
+
+@requires_gpu
+def test_fwd_staging_ring_equivalence():
+    """The 3-slot constant-distance staging ring must be bitwise-identical to
+    the 2-slot full-drain ring (same MFMA order; only the barrier/prefetch
+    schedule differs). Regression for the r2 fwd pipeline port."""
+    import os
+
+    from magi_attention.functional import flex_flash_attn_func
+
+    torch.manual_seed(5)
+    n, hq, d = 2048 + 192, 4, 128
+    q = (torch.randn(n, hq, d) * 0.5).bfloat16().cuda()
+    k = (torch.randn(n, hq, d) * 0.5).bfloat16().cuda()
+    v = (torch.randn(n, hq, d) * 0.5).bfloat16().cuda()
+    qr = torch.tensor([[0, n]], dtype=torch.int32, device="cuda")
+    tm = torch.tensor([1], dtype=torch.int32, device="cuda")
+
+    outs = {}
+    try:
+        for nbuf in ("2", "3"):
+            os.environ["MAGI_FWD_NBUF"] = nbuf
+            with torch.no_grad():
+                o, meta = flex_flash_attn_func(q, k, v, qr, qr.clone(), tm,
+                                               max_seqlen_q=n)
+            torch.cuda.synchronize()
+            outs[nbuf] = (o, meta.lse)
+    finally:
+        os.environ.pop("MAGI_FWD_NBUF", None)
+    assert torch.equal(outs["2"][0], outs["3"][0])
+    assert torch.equal(outs["2"][1], outs["3"][1])
+
+
+@requires_gpu
+def test_bwd_fused_variant_equivalence():
+    """The gated fused-dkv cadence variants (MODE3 fat-wave, W6 64-row ring)
+    must produce bit-identical dk/dv to the default v2 kernel (documented
+    measurement points; profiles/r2_final_pmc.md)."""
+    import os
+
+    from magi_attention.functional import flex_flash_attn_func
+
+    torch.manual_seed(6)
+    n, hq, d = 2048 + 320, 2, 128
+    qr = torch.tensor([[0, n]], dtype=torch.int32, device="cuda")
+    tm = torch.tensor([1], dtype=torch.int32, device="cuda")
+
+    def grads(env):
+        for kk, vv in env.items():
+            os.environ[kk] = vv
+        try:
+            torch.manual_seed(6)
+            q = (torch.randn(n, hq, d) * 0.5).bfloat16().cuda().requires_grad_(True)
+            k = (torch.randn(n, hq, d) * 0.5).bfloat16().cuda().requires_grad_(True)
+            v = (torch.randn(n, hq, d) * 0.5).bfloat16().cuda().requires_grad_(True)
+            o, _ = flex_flash_attn_func(q, k, v, qr, qr.clone(), tm,
+                                        max_seqlen_q=n)
+            torch.manual_seed(7)
+            o.backward(torch.randn_like(o))
+            torch.cuda.synchronize()
+            return q.grad.clone(), k.grad.clone(), v.grad.clone()
+        finally:
+            for kk in env:
+                os.environ.pop(kk, None)
+
+    g0 = grads({})
+    for tag, env in (("fat", {"MAGI_BWD_FAT": "1"}),
+                     ("w6", {"MAGI_BWD_W6": "1"})):
+        g = grads(env)
+        for name, x, y in zip("dq dk dv".split(), g, g0):
+            assert torch.equal(x, y), (tag, name)
