@@ -6,6 +6,8 @@ from .forward_meta import AttnForwardMeta  # noqa: F401
 from .mask import AttnMask  # noqa: F401
 from .range import AttnRange, RangeError  # noqa: F401
 from .ranges import AttnRanges  # noqa: F401
+from .rectangle import AttnRectangle  # noqa: F401
+from .rectangles import AttnRectangles  # noqa: F401
 
 # The reference swaps its Python range/geometry types for a pybind C++
 # backend under MAGI_ATTENTION_CPP_BACKEND (common/__init__.py:36-68). The
@@ -22,6 +24,8 @@ __all__ = [
     "AttnRange",
     "RangeError",
     "AttnRanges",
+    "AttnRectangle",
+    "AttnRectangles",
     "range_op",
     "USE_CPP_BACKEND",
     "is_cpp_backend_enable",

@@ -17,20 +17,25 @@ class AttnRange:
     __slots__ = ("_start", "_end")
 
     def __init__(self, start: int, end: int) -> None:
-        if start < 0 or end < start:
-            raise RangeError(f"invalid range [{start}, {end})")
+        # reference semantics (common/range.py:61): start <= end; negative
+        # indices are legal (AttnRectangle d_ranges are k-q diagonals)
+        if end < start:
+            raise RangeError(
+                f"The attn_range {(start, end)} is invalid against the rule: "
+                f"'start <= end'"
+            )
         self._start = int(start)
         self._end = int(end)
 
     # -- properties -------------------------------------------------------
+    # setters are UNCHECKED like the reference's: rectangle shrinking sets
+    # temporarily-inverted bounds and re-validates afterwards
     @property
     def start(self) -> int:
         return self._start
 
     @start.setter
     def start(self, value: int) -> None:
-        if value < 0 or value > self._end:
-            raise RangeError(f"invalid start {value} for end {self._end}")
         self._start = int(value)
 
     @property
@@ -39,9 +44,25 @@ class AttnRange:
 
     @end.setter
     def end(self, value: int) -> None:
-        if value < self._start:
-            raise RangeError(f"invalid end {value} for start {self._start}")
         self._end = int(value)
+
+    # -- validity (reference common/range.py:238-281) ----------------------
+    def is_valid_close(self, start: int | None = None, end: int | None = None) -> bool:
+        start = self._start if start is None else start
+        end = self._end if end is None else end
+        return start <= end
+
+    def is_valid_open(self, start: int | None = None, end: int | None = None) -> bool:
+        start = self._start if start is None else start
+        end = self._end if end is None else end
+        return start < end
+
+    def check_valid(self, start: int | None = None, end: int | None = None) -> None:
+        if not self.is_valid_close(start, end):
+            raise RangeError(
+                f"The attn_range {(start, end)} is invalid against the rule: "
+                f"'start <= end'"
+            )
 
     @property
     def seqlen(self) -> int:
