@@ -1,6 +1,6 @@
 # Common types for the MI355X-native magi_attention rebuild.
 from ..env import is_cpp_backend_enable  # noqa: F401
-from . import enum, range_op  # noqa: F401
+from . import enum, jit, range_op  # noqa: F401
 from .enum import AttnMaskType  # noqa: F401
 from .forward_meta import AttnForwardMeta  # noqa: F401
 from .mask import AttnMask  # noqa: F401
@@ -18,6 +18,7 @@ USE_CPP_BACKEND = False
 
 __all__ = [
     "enum",
+    "jit",
     "AttnMask",
     "AttnMaskType",
     "AttnForwardMeta",
