@@ -181,6 +181,10 @@ class AttnRanges:
         total = sum(r.seqlen for r in self._ranges)
         return total - self.merge().total_seqlen
 
+    def union_size(self) -> int:
+        """Total token count covered by the union of self's own ranges."""
+        return self.merge().total_seqlen
+
     def intersect_size_with(self, other: "AttnRanges") -> int:
         return sum(r.seqlen for r in self.find_overlap_ranges(other))
 

@@ -5,6 +5,7 @@ from __future__ import annotations
 
 from dataclasses import dataclass, field
 
+from . import env  # noqa: F401  (reference config.py:17 re-export)
 from .common.enum import AttnOverlapMode, DispatchAlgType, OverlapAlgType
 
 

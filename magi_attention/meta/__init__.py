@@ -161,3 +161,12 @@ def make_qo_meta_from_dispatch_meta(
         min_stage_tokens=oc.min_chunk_size,
     )
     return solver_t.make_qo_comm_meta(dispatch_meta.cp_rank)
+
+
+# reference-surface submodules + bucket factories (imported at the bottom:
+# collection re-imports DispatchMeta from this partially-initialized module)
+from . import collection, container  # noqa: E402,F401
+from ._buckets import (  # noqa: E402,F401
+    make_bucket_per_rank_from_qk_ranges,
+    make_global_bucket_from_qk_ranges,
+)
