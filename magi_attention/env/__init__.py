@@ -136,3 +136,8 @@ def snapshot() -> tuple[tuple[str, Any], ...]:
         ("native_grpcoll", is_native_grpcoll_enable()),
         ("min_chunks_per_rank", min_chunks_per_rank()),
     )
+
+
+# reference-named submodule views (env/general.py, env/comm.py, env/build.py)
+# — imported at the bottom: they re-import this module's flag functions
+from . import build, comm, general  # noqa: E402,F401
