@@ -16,12 +16,15 @@ from typing import Dict, List, Tuple
 import torch
 
 from . import _ffa_lib
+from .common.enum import AttnMaskType  # noqa: F401  (C++ type alias)
 from .common.range import AttnRange, RangeError  # noqa: F401  (C++ ranges alias)
 from .common.ranges import (  # noqa: F401
     AttnRanges,
     check_valid_cu_seqlens,
     is_valid_cu_seqlens,
 )
+from .common.rectangle import AttnRectangle  # noqa: F401
+from .common.rectangles import AttnRectangles  # noqa: F401
 
 
 class KernelBarrier:
@@ -167,6 +170,31 @@ def elapsed_ms(name: str) -> float:
         return (b - a) * 1e3
     torch.cuda.synchronize()
     return a.elapsed_time(b)
+
+
+def elapsed_ms_event(name: str) -> float:
+    """Reference magi_attn_ext.elapsed_ms_event: ms between a named event's
+    start/stop records."""
+    return elapsed_ms(name)
+
+
+def produce(kernel_barrier: "KernelBarrier | None") -> None:
+    """Reference magi_attn_ext.produce: launch a producer kernel that
+    increments the barrier count by 1 (no-op for None)."""
+    if kernel_barrier is not None:
+        kernel_barrier.produce()
+
+
+def expand_attn_ranges(ranges: AttnRanges, stride: int, num_heads_group: int) -> AttnRanges:
+    """Reference magi_attn_ext.expand_attn_ranges (the DynamicAttnSolver's
+    range expansion): each range [s, e) expands to the num_heads_group
+    per-head-group row blocks [h*stride + s, h*stride + e)."""
+    out = AttnRanges()
+    for h in range(int(num_heads_group)):
+        off = h * int(stride)
+        for r in ranges:
+            out.append(AttnRange(r.start + off, r.end + off))
+    return out
 
 
 def destroy_event(name: str) -> None:

@@ -115,3 +115,23 @@ def test_singleton_meta():
         pass
 
     assert A() is A() and B() is B() and A() is not B()
+
+
+def test_magi_attn_ext_surface_aliases():
+    """The ext module exposes the reference pyi's type aliases + helpers
+    (magi_attn_ext.pyi: AttnMaskType/AttnRectangle(s)/expand_attn_ranges/
+    elapsed_ms_event/produce)."""
+    import magi_attention.magi_attn_ext as ext
+    from magi_attention.common import AttnRanges
+
+    assert ext.AttnMaskType is not None and ext.AttnRectangles is not None
+    rr = AttnRanges.from_ranges([(0, 3), (5, 8)])
+    out = ext.expand_attn_ranges(rr, stride=10, num_heads_group=2)
+    assert [(r.start, r.end) for r in out] == [
+        (0, 3), (5, 8), (10, 13), (15, 18)
+    ]
+    ext.start_event("t")
+    ext.stop_event("t")
+    assert ext.elapsed_ms_event("t") >= 0.0
+    ext.destroy_event("t")
+    ext.produce(None)  # no-op contract
