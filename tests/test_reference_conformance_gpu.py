@@ -57,7 +57,7 @@ GEN_CONFIGS = [
 
 
 def _run_case(tq, tk, hq, hk, d, qr_l, kr_l, ty_l, seed, tag,
-              deterministic=False):
+              deterministic=False, dense_override=None):
     from magi_attention.functional import flex_flash_attn_func
 
     q, k, v, dout, qr, kr, tm = make_flex_case(
@@ -70,7 +70,8 @@ def _run_case(tq, tk, hq, hk, d, qr_l, kr_l, ty_l, seed, tag,
                                      deterministic=deterministic)
     out.backward(dout)
     torch.cuda.synchronize()
-    mask = make_attn_mask(tq, tk, qr_l, kr_l, ty_l)
+    mask = (dense_override if dense_override is not None
+            else make_attn_mask(tq, tk, qr_l, kr_l, ty_l))
     qc, kc, vc, doc = [t.detach().cpu() for t in (q, k, v, dout)]
     o_hi, lse_hi, dq_hi, dk_hi, dv_hi = ref_attn_with_grads(qc, kc, vc, mask, doc)
     o_lo, _, dq_lo, dk_lo, dv_lo = ref_attn_with_grads(
@@ -158,3 +159,36 @@ def test_ffa_random_deterministic():
     ty_l = [rng.randint(0, 3) for _ in qr_l]
     _run_case(2048, 2048, 8, 8, 128, qr_l, kr_l, ty_l, seed=3,
               tag="det_random", deterministic=True)
+
+
+@requires_gpu
+def test_from_mask_inferred_ranges_drive_kernel():
+    """End-to-end over the AttnMask.from_mask INFERENCE path (reference
+    common/mask.py:165): build a random row-contiguous dense mask, infer
+    canonical (q_ranges, k_ranges, types), run the HIP kernel on the
+    inferred triples and compare against the fp64 oracle on the ORIGINAL
+    dense mask — proving the inference and the kernel agree cell-for-cell."""
+    import numpy as np
+
+    from magi_attention.common import AttnMask
+
+    rng = np.random.default_rng(23)
+    for trial in range(3):
+        n = 512
+        # random staircase mask: contiguous k-span per row, slowly drifting
+        start = np.zeros(n, dtype=int)
+        end = np.zeros(n, dtype=int)
+        s, e = 0, int(rng.integers(1, 64))
+        for r in range(n):
+            s = min(max(0, s + int(rng.integers(-1, 2))), n - 1)
+            e = min(max(s + 1, e + int(rng.integers(0, 2))), n)
+            start[r], end[r] = s, e
+        dense = torch.zeros(n, n, dtype=torch.int32)
+        for r in range(n):
+            dense[r, start[r]:end[r]] = 1
+        m = AttnMask.from_mask(dense)
+        qr = [[r.start, r.end] for r in m.q_ranges]
+        kr = [[r.start, r.end] for r in m.k_ranges]
+        ty = [t.to_int_type() for t in m.attn_mask_type]
+        _run_case(n, n, 2, 2, 128, qr, kr, ty, seed=100 + trial,
+                  tag=f"from_mask/{trial}", dense_override=dense.bool())
