@@ -23,9 +23,17 @@ def run(tokens=4096, hq=128, d=128, s_kv=8192, topk=2048, iters=20):
     q = torch.randn(tokens, hq, d, dtype=torch.bfloat16, device=dev)
     k = torch.randn(s_kv, 1, d, dtype=torch.bfloat16, device=dev)
     v = torch.randn_like(k)
-    idx = build_index_attn_indices(
-        1, 1, tokens, s_kv, topk, topk, device=dev
-    ).view(tokens, topk)
+    if s_kv <= 65536:
+        idx = build_index_attn_indices(
+            1, 1, tokens, s_kv, topk, topk, device=dev
+        ).view(tokens, topk)
+    else:
+        # HBM-resident regime probe: sample WITH replacement (perf-only —
+        # the without-replacement builder would need tokens x s_kv scores)
+        idx = (
+            torch.randint(0, s_kv, (tokens, topk), device=dev)
+            .sort(dim=-1).values.int()
+        )
     scale = d ** (-0.5)
 
     for _ in range(3):
@@ -54,3 +62,6 @@ if __name__ == "__main__":
         run(16384, 128, 128, 16384, 2048)   # bigger grid
         run(8192, 32, 128, 16384, 1024)     # ratio-32 shape (W1 head path)
         run(8192, 64, 64, 16384, 2048)      # d=64
+        # HBM-resident gathers (KV pool 512 MB >> 32 MB aggregate L2)
+        run(8192, 128, 128, 1048576, 2048)  # ratio-128, cold gather
+        run(8192, 32, 128, 1048576, 1024)   # ratio-32, cold gather
