@@ -1306,7 +1306,11 @@ static int fill_bwd_params(const magi_ffa_bwd_args* a, BwdParams* p) {
   p->seg_starts = a->seg_starts;
   { const char* e = getenv("MAGI_BWD_ABLATE"); p->debug_ablate = e ? atoi(e) : 0; }
   // head-major pays when one head's K+V (bf16) fits a 4 MB XCD L2
-  p->head_major = ((long long)a->total_k * a->d * 4 <= (4 << 20)) ? 1 : 0;
+  // late-r2 A/B (profiles/r2_headmajor_ab.md): dq wins head-major up to
+  // ~16 MB of per-head K+V (1.20x at 16k) and loses beyond (0.93x at 64k,
+  // 0.91x at 128k) — same shape as the fwd kernel; the dkv launcher
+  // OVERRIDES this to always-head-major below.
+  p->head_major = ((long long)a->total_k * a->d * 4 <= (16 << 20)) ? 1 : 0;
   { const char* e = getenv("MAGI_BWD_HEAD_MAJOR"); if (e) p->head_major = atoi(e); }
   // deterministic-mode GQA split: cu_margin's top bits carry (mult<<8)|off;
   // 0 = all heads in one launch (default)
@@ -1414,6 +1418,10 @@ static int launch_bwd_dkv(const magi_ffa_bwd_args* a) {
   BwdParams p;
   int rc = fill_bwd_params(a, &p);
   if (rc) return rc > 0 ? 0 : rc;
+  // dkv passes win head-major at EVERY measured size (their dk/dv atomics
+  // cluster per head in one XCD's L2): 1.54x at 16k, 1.10x at 64k, 1.07x
+  // at 128k (profiles/r2_headmajor_ab.md) — unconditional unless A/B'd
+  if (!getenv("MAGI_BWD_HEAD_MAJOR")) p.head_major = 1;
   // All dkv modes run 8 waves per WG (one 512-thread WG/CU, one shared
   // staged Q/dO image) down to 1k ranges: these kernels' blocks span the
   // K dim, so the wider workgroup does NOT over-iterate causal q windows,

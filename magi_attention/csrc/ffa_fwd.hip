@@ -671,8 +671,14 @@ extern "C" int magi_ffa_fwd(const magi_ffa_fwd_args* a) {
   const int span = 32 * fw;
   const int mblocks = (a->max_seqlen_q + span - 1) / span;
   if (a->n_ranges > 65535) return -5;
-  // head-major XCD affinity pays when one head's K+V fits a 4 MB XCD L2
-  p.head_major = ((long long)a->total_k * a->d * 4 <= (4 << 20)) ? 1 : 0;
+  // head-major XCD affinity (grid.x = head): late-r2 A/B moved the
+  // threshold from 4 MB to 16 MB of per-head K+V — head-major wins 1.26x
+  // at 4k, 1.45x at 8k, 1.29x at 16k (8.4 MB), is neutral at 32k
+  // (16.8 MB) and loses 0.93x at 64k where the streamed K window
+  // L3-shares better work-major (profiles/r2_headmajor_ab.md)
+  p.head_major = ((long long)a->total_k * a->d * 4 <= (16 << 20)) ? 1 : 0;
+  { const char* e = getenv("MAGI_FWD_HEADMAJOR");  // A/B override
+    if (e) p.head_major = atoi(e) ? 1 : 0; }
   dim3 grid = p.head_major
                   ? dim3(a->hq, mblocks, (unsigned)a->n_ranges)
                   : dim3(mblocks, (unsigned)a->n_ranges, a->hq);
