@@ -58,11 +58,13 @@ def is_hierarchical_comm_enable() -> bool:
 
 
 def is_bwd_cosched() -> bool:
-    """Default ON: the dq pass runs on a side stream concurrent with the
-    dkv pass (A/B at 64k: 209.6 vs 218.3 ms/step serial, despite the fused
-    kernel's full-LDS workgroups partitioning CUs). MAGI_BWD_COSCHED=0
-    serializes the passes."""
-    return _get("MAGI_BWD_COSCHED", "1") != "0"
+    """Default OFF since the late-r2 head-major dkv gate: the fused dkv's
+    per-head XCD-L2 affinity (99.1 vs 109.4 ms solo) is diluted when dq
+    waves co-reside, so serializing the passes now wins the step A/B
+    (205.7 vs 208.9 ms at 64k; earlier in r2, pre-gate, co-scheduling had
+    won 209.6 vs 218.3). MAGI_BWD_COSCHED=1 re-enables the side-stream
+    co-schedule."""
+    return _get("MAGI_BWD_COSCHED", "0") == "1"
 
 
 def bwd_dkv_mode() -> str:
