@@ -181,7 +181,39 @@ def measure_roofline(state, device):
     bwd_ms = ev0.elapsed_time(ev1) / reps
     bwd_flops = host_fwd_flops * 2.5
 
-    achieved = bwd_flops / (bwd_ms * 1e-3)
+    # dominant-kernel-only: the fused dkv launch (4 of the 5 issued
+    # matmul-areas of the serial bwd pipeline), timed solo with HIP events
+    # on the launch stream
+    import ctypes
+
+    from magi_attention import _ffa_lib
+    from magi_attention._ffa_lib import MagiFfaBwdArgs, check, ptr
+
+    ha = rt.calc_meta.host_arg
+    qr_t, kr_t, tm_t = ha.to_device(device)
+    dkv_args = MagiFfaBwdArgs(
+        dout=ptr(dout), q=ptr(ql.detach()), k=ptr(kl.detach()),
+        v=ptr(vl.detach()), out=ptr(out), lse=ptr(lse),
+        dq=ptr(dq), dk=ptr(dkv[: kl.shape[0]]), dv=ptr(dkv[kl.shape[0]:]),
+        dpsum=ptr(dpsum),
+        q_ranges=ptr(qr_t), k_ranges=ptr(kr_t),
+        attn_type_map=ptr(tm_t),
+        n_ranges=qr_t.shape[0],
+        total_q=ql.shape[0], total_k=kl.shape[0],
+        hq=HQ, hk=HKV, d=D, max_seqlen_k=kl.shape[0],
+        out_is_fp32=int(out.dtype == torch.float32),
+        softmax_scale=scale, softcap=0.0, cu_margin=0,
+        stream=ctypes.c_void_p(torch.cuda.current_stream().cuda_stream),
+    )
+    ev0.record()
+    for _ in range(reps):
+        check(_ffa_lib.lib().magi_ffa_bwd_dkv(dkv_args), "roofline dkv")
+    ev1.record()
+    torch.cuda.synchronize()
+    dkv_ms = ev0.elapsed_time(ev1) / reps
+    dkv_flops = host_fwd_flops * 2.0  # S, dP, dV, dK over the host area
+
+    achieved = dkv_flops / (dkv_ms * 1e-3)
     # traffic: per-launch FETCH+WRITE bytes of the dominant kernel (fused
     # dkv), from the committed rocprofv3 PMC passes on this same workload
     # (profiles/r2_pmc_traffic.json; FETCH corrected 2x per the gfx950
@@ -205,8 +237,11 @@ def measure_roofline(state, device):
         "frac": achieved / MFMA_PEAK_BF16,
         "traffic": traffic,
         "detail": {
-            "kernel": "ffa_bwd_dkv_kernel<fused> (dominant; dq pass co-scheduled on a side stream)",
-            "bwd_ms_per_launch": bwd_ms,
+            "kernel": "ffa_bwd_dkv_kernel<fused> (dominant; head-major; "
+                      "serial bwd schedule)",
+            "dkv_ms_per_launch": dkv_ms,
+            "bwd_pipeline_ms": bwd_ms,
+            "bwd_pipeline_flops_per_s": bwd_flops / (bwd_ms * 1e-3),
             "fwd_ms_per_launch": fwd_ms,
             "fwd_achieved_flops_per_s": host_fwd_flops / (fwd_ms * 1e-3),
             "traffic_note": traffic_note,
