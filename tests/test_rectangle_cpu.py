@@ -174,3 +174,67 @@ def test_rectangles_indexing_and_eq():
         [(0, 4), (4, 8)], [(0, 4), (0, 8)], [1, 1]
     )
     assert clone == rects and hash(clone) == hash(rects)
+
+
+def test_reference_golden_cases():
+    """Concrete expectations ported from the reference's own
+    tests/test_common/test_rectangle.py (init shrinks, mask-type d-band
+    seeding, boundary intersections, validity flips)."""
+    r = AttnRectangle(AttnRange(0, 10), AttnRange(0, 20), AttnRange(-5, 5))
+    assert (r.q_range.start, r.q_range.end) == (0, 10)
+    assert (r.k_range.start, r.k_range.end) == (0, 15)  # k shrunk by d_end
+    assert (r.d_range.start, r.d_range.end) == (-5, 5)
+    assert r.intersection_q_id_on_left_boundary() == 5
+    assert r.intersection_q_id_on_right_boundary() == 9
+
+    r = AttnRectangle(AttnRange(0, 10), AttnRange(0, 20), AttnRange(-100, 100))
+    assert (r.d_range.start, r.d_range.end) == (-9, 19)
+
+    r = AttnRectangle(AttnRange(0, 10), AttnRange(0, 20),
+                      mask_type=AttnMaskType.CAUSAL)
+    assert r.d_range.end == 10  # k_end - q_end
+    r = AttnRectangle(AttnRange(0, 10), AttnRange(0, 20),
+                      mask_type=AttnMaskType.BICAUSAL)
+    assert (r.d_range.start, r.d_range.end) == (0, 10)
+    r = AttnRectangle(AttnRange(0, 10), AttnRange(0, 20),
+                      mask_type=AttnMaskType.INVCAUSAL)
+    assert r.d_range.start == 0
+
+    # unchecked setters allow a temporarily-invalid state; is_valid flags it
+    bad = AttnRectangle(AttnRange(0, 10), AttnRange(0, 20), AttnRange(-5, 5))
+    bad.q_range.start = 10
+    bad.q_range.end = 0
+    assert not bad.is_valid() and bad.get_valid_or_none() is None
+
+    # segment clips (reference expectations)
+    r = AttnRectangle(AttnRange(0, 20), AttnRange(0, 20), AttnRange(-5, 5))
+    seg = r.get_rect_within_q_segment(5, 15)
+    assert (seg.q_range.start, seg.q_range.end) == (5, 15)
+    assert r.get_rect_within_q_segment(25, 35) is None
+    seg = r.get_rect_within_k_segment(5, 15)
+    assert (seg.k_range.start, seg.k_range.end) == (5, 15)
+    assert r.get_rect_within_k_segment(25, 35) is None
+
+
+def test_decomposition_reference_band_regime():
+    """The reference's to_qk_range_mask_type contract on its own sampling
+    regime (d band spanning both corner diagonals): at most 3 parts, and
+    the parts RECONSTRUCTED from (q, k, mask_type) alone reproduce the
+    area (round-trips the mask-type constructor)."""
+    rng = np.random.default_rng(9)
+    for _ in range(100):
+        qs = int(rng.integers(0, 99)); qe = int(rng.integers(qs + 1, 101))
+        ks = int(rng.integers(0, 99)); ke = int(rng.integers(ks + 1, 101))
+        d_min = ks - (qe - 1)
+        lo_c, hi_c = sorted((ks - qs, ke - qe))
+        d_max = ke - 1 - qs
+        ds = int(rng.integers(d_min, lo_c + 1))
+        de = int(rng.integers(max(hi_c, ds), d_max + 1))
+        rect = AttnRectangle(AttnRange(qs, qe), AttnRange(ks, ke),
+                             AttnRange(ds, de))
+        parts = rect.to_qk_range_mask_type()
+        assert len(parts) <= 3, rect
+        rebuilt = sum(
+            AttnRectangle(qr, kr, mask_type=t).area() for qr, kr, t in parts
+        )
+        assert rebuilt == rect.area(), rect
