@@ -172,3 +172,54 @@ def test_from_mask_roundtrip_random():
         except AssertionError:
             continue  # overlapping causal overwrite can break row contiguity
         assert np.equal(m2.mask_flag_array, m.mask_flag_array).all()
+
+
+def test_sub_mask_reference_golden():
+    """The reference's own test_make_sub_mask_with_calc_sub_area expectations
+    (tests/test_common/test_attn_mask.py:132): exact sub-areas AND the exact
+    canonical tuples from_mask infers on each sub-window."""
+    m = AttnMask.from_ranges(
+        AttnRanges.from_ranges([(0, 6), (6, 9), (9, 12), (12, 16)]),
+        AttnRanges.from_ranges([(0, 4), (4, 12), (12, 15), (1, 13)]),
+        [AttnMaskType.CAUSAL, AttnMaskType.FULL,
+         AttnMaskType.CAUSAL, AttnMaskType.CAUSAL],
+        16, 16,
+    )
+    assert m.area == 82
+
+    s1 = m.make_sub_mask(AttnRange(4, 13), AttnRange(1, 13))
+    assert m.calc_sub_area(AttnRange(4, 13), AttnRange(1, 13)) == 41
+    assert s1.area == 41
+    assert s1.q_ranges == AttnRanges.from_ranges([[0, 2], [2, 5], [5, 8], [8, 9]])
+    assert s1.k_ranges == AttnRanges.from_ranges([[0, 3], [3, 11], [11, 12], [0, 9]])
+    assert s1.attn_mask_type == [
+        AttnMaskType.CAUSAL, AttnMaskType.FULL, AttnMaskType.FULL,
+        AttnMaskType.FULL,
+    ]
+
+    s2 = m.make_sub_mask(AttnRange(0, 14), AttnRange(0, 7))
+    assert m.calc_sub_area(AttnRange(0, 14), AttnRange(0, 7)) == 31
+    assert s2.area == 31
+    assert s2.q_ranges == AttnRanges.from_ranges([[0, 6], [6, 9], [9, 12], [12, 14]])
+    assert s2.k_ranges == AttnRanges.from_ranges([[0, 4], [4, 7], [9, 9], [1, 7]])
+    assert s2.attn_mask_type == [
+        AttnMaskType.CAUSAL, AttnMaskType.FULL, AttnMaskType.CAUSAL,
+        AttnMaskType.FULL,
+    ]
+
+    s3 = m.make_sub_mask(AttnRange(5, 16), AttnRange(3, 11))
+    assert m.calc_sub_area(AttnRange(5, 16), AttnRange(3, 11)) == 53
+    assert s3.area == 53
+    assert s3.q_ranges == AttnRanges.from_ranges(
+        [[0, 1], [1, 4], [4, 7], [7, 9], [9, 11]]
+    )
+    assert s3.k_ranges == AttnRanges.from_ranges(
+        [[0, 1], [1, 8], [4, 4], [0, 8], [0, 8]]
+    )
+    assert s3.attn_mask_type == [
+        AttnMaskType.FULL, AttnMaskType.FULL, AttnMaskType.CAUSAL,
+        AttnMaskType.CAUSAL, AttnMaskType.FULL,
+    ]
+
+    assert m.calc_sub_area(AttnRange(4, 12), AttnRange(4, 12)) == 24
+    assert m.make_sub_mask(AttnRange(4, 12), AttnRange(4, 12)).area == 24
