@@ -99,6 +99,21 @@ class AttnRange:
     def intersect_size(self, other: "AttnRange") -> int:
         return max(0, min(self._end, other._end) - max(self._start, other._start))
 
+    def union(self, other: "AttnRange") -> list["AttnRange"]:
+        """One merged range when the two overlap or nest, else both
+        (reference common/range.py union; adjacency without overlap keeps
+        both, like the reference)."""
+        if self.is_empty() or other.is_empty():
+            return [self, other]
+        if self.is_subrange_of(other):
+            return [other]
+        if other.is_subrange_of(self):
+            return [self]
+        if self.is_overlap_with(other):
+            return [AttnRange(min(self._start, other._start),
+                              max(self._end, other._end))]
+        return [self, other]
+
     def union_size(self, other: "AttnRange") -> int:
         return self.seqlen + other.seqlen - self.intersect_size(other)
 

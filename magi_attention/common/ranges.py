@@ -105,6 +105,17 @@ class AttnRanges:
         s = self.sort()
         return all(a.end <= b.start for a, b in zip(s._ranges, s._ranges[1:]))
 
+    def is_valid(self) -> bool:
+        """All member ranges satisfy start <= end (reference ranges.py:158)."""
+        return all(r.is_valid_close() for r in self._ranges)
+
+    def check_valid(self) -> None:
+        if not self.is_valid():
+            raise ValueError(
+                f"Some of the {self._ranges=} is invalid against the rule: "
+                f"'start <= end'"
+            )
+
     def is_cu_seqlens(self, seqlen: int) -> bool:
         if not self._ranges:
             return False
