@@ -186,3 +186,63 @@ def test_uneven_gather_reduce_scatter(ws):
     port = _free_port()
     mp.spawn(_worker_gatherv, args=(ws, port, None, None, None), nprocs=ws,
              join=True)
+
+
+def _worker_functional(rank, ws, port, _a, _b, _c):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=ws)
+    try:
+        from magi_attention.comm.functional import (
+            all_gather_fwd_scatter_bwd,
+            scatter_fwd_all_gather_bwd,
+        )
+
+        group = dist.group.WORLD
+        sizes = [3, 5, 2, 7][:ws]
+        g = torch.Generator().manual_seed(9 + rank)
+        local = torch.randn(sizes[rank], 4, generator=g).requires_grad_(True)
+
+        # gather-fwd: output is the concatenation; backward scatters the
+        # grad back to this rank's slice
+        full = all_gather_fwd_scatter_bwd(local, group, dim=0,
+                                          split_sizes=sizes)
+        obj = [None] * ws
+        dist.all_gather_object(obj, local.detach())
+        torch.testing.assert_close(full.detach(), torch.cat(obj, dim=0))
+        gout = torch.randn(full.shape, generator=torch.Generator().manual_seed(77))
+        full.backward(gout)
+        start = sum(sizes[:rank])
+        torch.testing.assert_close(local.grad,
+                                   gout[start:start + sizes[rank]])
+
+        # scatter-fwd: output is this rank's slice of the input; backward
+        # all-gathers the slice grads
+        big = torch.randn(sum(sizes), 4,
+                          generator=torch.Generator().manual_seed(31)
+                          ).requires_grad_(True)
+        mine = scatter_fwd_all_gather_bwd(big, group, dim=0, split_sizes=sizes)
+        torch.testing.assert_close(mine.detach(),
+                                   big.detach()[start:start + sizes[rank]])
+        gmine = torch.randn(mine.shape,
+                            generator=torch.Generator().manual_seed(100 + rank))
+        mine.backward(gmine)
+        obj = [None] * ws
+        dist.all_gather_object(obj, gmine)
+        torch.testing.assert_close(big.grad, torch.cat(obj, dim=0))
+
+        # even-split default (split_sizes=None), non-zero dim
+        x = torch.randn(4, 2 * ws, generator=g).requires_grad_(True)
+        piece = scatter_fwd_all_gather_bwd(x, group, dim=1, split_sizes=None)
+        torch.testing.assert_close(
+            piece.detach(), x.detach()[:, 2 * rank : 2 * rank + 2]
+        )
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("ws", [2, 4])
+def test_comm_functional_gather_scatter(ws):
+    port = _free_port()
+    mp.spawn(_worker_functional, args=(ws, port, None, None, None), nprocs=ws,
+             join=True)
