@@ -112,3 +112,40 @@ def test_chunk_iou_and_slices():
     assert abs(c.iou - 4 / 12) < 1e-9
     c2 = AttnChunk(chunk_id=1, q_slices=[s2])
     assert abs(c.iou_with(c2) - 8 / 12) < 1e-9
+
+
+def test_gt_dispatcher_fuzz_vs_factory():
+    """Randomized agreement between the dense-mask GroundTruthDispatcher and
+    the arithmetic bucket factory (FULL/CAUSAL blocks — AttnMask.from_ranges'
+    domain), per-chunk."""
+    from magi_attention.config import DispatchAlg
+    from magi_attention.testing import GroundTruthDispatcher
+
+    rng = np.random.default_rng(12)
+    for _ in range(15):
+        chunk = int(rng.choice([2, 4, 8]))
+        n_chunks = int(rng.integers(2, 7))
+        n = chunk * n_chunks
+        cuts = sorted(set(int(c) for c in rng.integers(1, n, size=2)))
+        bounds = [0] + cuts + [n]
+        qrs, krs, tts = [], [], []
+        for a, b in zip(bounds, bounds[1:]):
+            if a == b:
+                continue
+            ks = int(rng.integers(0, n))
+            ke = int(rng.integers(ks + 1, n + 1))
+            qrs.append((a, b))
+            krs.append((ks, ke))
+            tts.append(AttnMaskType.CAUSAL if rng.integers(0, 2) else
+                       AttnMaskType.FULL)
+        gt = GroundTruthDispatcher(alg=DispatchAlg())
+        b_gt = gt._compute_self_attn_areas(
+            AttnRanges.from_ranges(qrs), AttnRanges.from_ranges(krs), tts,
+            chunk_size=chunk,
+        )
+        b_ar = make_global_bucket_from_qk_ranges(
+            AttnRanges.from_ranges(qrs), AttnRanges.from_ranges(krs), tts,
+            num_chunks=n_chunks, chunk_size=chunk,
+        )
+        assert b_gt.areas == b_ar.areas, (qrs, krs, tts, chunk)
+        assert b_gt.area == b_ar.area
