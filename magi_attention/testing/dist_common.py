@@ -92,24 +92,25 @@ def should_run_attn_config(name: str) -> bool:
     return _match_patterns(name, raw)
 
 
+def _case_value_str(value: object) -> str:
+    """Canonical string of one parametrize value for filtering: named
+    dicts match by their NAME entry, tuples join with underscores."""
+    if isinstance(value, dict) and NAME in value:
+        return str(value[NAME])
+    if isinstance(value, tuple):
+        return "_".join(str(v) for v in value)
+    return str(value)
+
+
 def should_run_test_case(**parametrize_args: object) -> bool:
     """AND-combined fnmatch filters per parametrize dimension, driven by
-    MAGI_ATTENTION_TEST_<DIMENSION> env vars; dicts match by their NAME key,
-    tuples join with underscores."""
+    MAGI_ATTENTION_TEST_<DIMENSION> env vars."""
     for dim_name, value in parametrize_args.items():
         envvar = _TEST_FILTER_ENVVARS.get(
             dim_name, _TEST_FILTER_ENV_PREFIX + dim_name.upper()
         )
         raw = os.environ.get(envvar, "").strip()
-        if not raw:
-            continue
-        if isinstance(value, dict) and NAME in value:
-            value_str = str(value[NAME])
-        elif isinstance(value, tuple):
-            value_str = "_".join(str(v) for v in value)
-        else:
-            value_str = str(value)
-        if not _match_patterns(value_str, raw):
+        if raw and not _match_patterns(_case_value_str(value), raw):
             return False
     return True
 
@@ -123,35 +124,35 @@ NUM_DEVICES = 4
 RUN_IN_MP = "MAGI_ATTENTION_PARAMETERIZE_RUN_IN_MP"
 
 
+_SUPPORTED_BACKENDS = ("nccl", "gloo", "mpi", "cpu:gloo,cuda:nccl")
+
+
 class DistTestBase(MultiProcessTestCase):
-    @property
-    def seed(self) -> int:
-        return 42
+    """Multi-process test base: spawns world_size workers, each of which
+    calls init_pg() (via with_comms) to join a file-rendezvous process group
+    with a rank-offset seed. On ROCm the "nccl" backend is RCCL."""
 
-    @property
-    def world_size(self) -> int:
-        return NUM_DEVICES
-
-    @property
-    def backend(self) -> str:
-        return PG_DEFAULT_BACKEND
+    seed = property(lambda self: 42)
+    world_size = property(lambda self: NUM_DEVICES)
+    backend = property(lambda self: PG_DEFAULT_BACKEND)
 
     def init_pg(self) -> None:
-        if "nccl" in self.backend and torch.cuda.device_count() < self.world_size:
+        be = self.backend
+        if "nccl" in be and torch.cuda.device_count() < self.world_size:
             raise RuntimeError(
                 f"nccl backend requires {self.world_size} GPUs, but only "
                 f"{torch.cuda.device_count()} are available"
             )
-        if self.backend not in ["nccl", "gloo", "mpi", "cpu:gloo,cuda:nccl"]:
-            raise RuntimeError(f"Backend {self.backend} not supported!")
+        if be not in _SUPPORTED_BACKENDS:
+            raise RuntimeError(f"Backend {be} not supported!")
         dist.init_process_group(
-            backend=self.backend,
+            backend=be,
             world_size=self.world_size,
             rank=self.rank,
             init_method=f"file://{self.file_name}",
             timeout=datetime.timedelta(minutes=30),
         )
-        if "nccl" in self.backend:
+        if "nccl" in be:
             torch.cuda.set_device(self.rank)
         self._set_random_seed()
 
