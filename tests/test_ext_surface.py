@@ -122,3 +122,54 @@ def test_correct_attn_helpers_match_oracle_merge():
     o_f, l_f = correct_attn_out_lse_with_sink(o_ns.float(), l_ns, sink)
     torch.testing.assert_close(o_f.double(), o_s, atol=1e-5, rtol=1e-5)
     torch.testing.assert_close(l_f, l_s, atol=1e-5, rtol=1e-5)
+
+
+def test_cabi_validation_codes_cpu():
+    """The launchers' argument-validation paths return their documented
+    negative codes BEFORE any GPU work — checkable in the CPU container
+    (include/magi_ffa.h conventions; no kernel is launched)."""
+    import ctypes
+
+    import pytest
+
+    from magi_attention import _ffa_lib
+    from magi_attention._ffa_lib import MagiFfaFwdArgs, MagiFfaIndexArgs
+
+    if not _ffa_lib.is_available():
+        pytest.skip("native library not built")
+    lib = _ffa_lib.lib()
+
+    # index launcher
+    a = MagiFfaIndexArgs(
+        q=1, k=1, v=1, out=1, lse=1, indices_2d=1,
+        total_q=8, total_k=8, max_topk=64, hq=4, hk=2, d=128,
+        softmax_scale=1.0, softcap=0.0, out_is_fp32=0, stream=None,
+    )
+    assert lib.magi_ffa_fwd_index(ctypes.byref(a)) == -3  # hk must be 1
+    a.hk = 1
+    a.d = 96
+    assert lib.magi_ffa_fwd_index(ctypes.byref(a)) == -2  # d bucket
+    a.d = 128
+    a.max_topk = 50
+    assert lib.magi_ffa_fwd_index(ctypes.byref(a)) == -4  # 64 | max_topk
+    a.max_topk = 64
+    a.indices_2d = None
+    assert lib.magi_ffa_fwd_index(ctypes.byref(a)) == -1  # null pointer
+
+    # fwd launcher
+    f = MagiFfaFwdArgs(
+        q=1, k=1, v=1, out=1, lse=1, q_ranges=1, k_ranges=1,
+        attn_type_map=None, locks=None, max_logits=None, qk_starts=None,
+        n_ranges=1, total_q=8, total_k=8, hq=4, hk=2, d=96, max_seqlen_q=8,
+        softmax_scale=1.0, softcap=0.0, out_is_fp32=1,
+        disable_atomic_reduction=1, cu_margin=0, stream=None,
+    )
+    assert lib.magi_ffa_fwd(ctypes.byref(f)) == -2  # d bucket
+    f.d = 128
+    f.hq = 3  # not divisible by hk=2
+    assert lib.magi_ffa_fwd(ctypes.byref(f)) == -3
+    f.hq = 4
+    f.disable_atomic_reduction = 0  # atomic merge without locks
+    assert lib.magi_ffa_fwd(ctypes.byref(f)) == -4
+    f.q = None
+    assert lib.magi_ffa_fwd(ctypes.byref(f)) == -1
